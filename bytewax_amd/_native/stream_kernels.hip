@@ -99,41 +99,51 @@ __global__ void k_window_agg_insert(
     int64_t len_ms,
     unsigned long long* __restrict__ max_ts,  // device scalar (atomicMax)
     int* __restrict__ error_flag) {
-  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int lane = threadIdx.x & (WAVE - 1);
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   int64_t local_max = 0;
-  for (; i < n; i += stride) {
-    int64_t t = ts[i];
+  // Wave-uniform iteration: every lane of a wave runs every loop
+  // iteration (masked by `valid`) so cross-lane ops in the DEDUP path
+  // never read from exited lanes.
+  int64_t first = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  for (int64_t i = first; i - lane < n; i += stride) {
+    bool valid = i < n;
+    int64_t t = valid ? ts[i] : 0;
     if (t > local_max) local_max = t;
-    int64_t win = (t - align_ms) / len_ms;
-    uint64_t packed =
-        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
-    unsigned long long inc =
-        (MODE == AGG_COUNT) ? 1ULL : (unsigned long long)vals[i];
+    uint64_t packed = 0;
+    unsigned long long inc = 0;
+    if (valid) {
+      int64_t win = (t - align_ms) / len_ms;
+      packed = ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+      inc = (MODE == AGG_COUNT) ? 1ULL : (unsigned long long)vals[i];
+    }
 
     bool ok = true;
     if (DEDUP) {
       // Wave-aggregate duplicate (key, window) pairs: one atomic per
       // distinct pair per wave.  Worth it only for low-cardinality
       // keys; high-cardinality batches use the plain path.
-      unsigned long long remaining = __ballot(1);
-      int lane = threadIdx.x & (WAVE - 1);
+      unsigned long long remaining = __ballot(valid);
       bool leader = false;
       unsigned long long agg = 0;
       while (remaining) {
         int l = __ffsll((unsigned long long)remaining) - 1;
-        uint64_t lk = __shfl((long long)packed, l);
-        unsigned long long match = __ballot(packed == (uint64_t)lk);
+        uint64_t lk = (uint64_t)__shfl((long long)packed, l);
+        unsigned long long match =
+            __ballot(valid && packed == lk);
         if (MODE == AGG_COUNT) {
           if (lane == l) {
             leader = true;
             agg = __popcll(match);
           }
         } else {
-          // Masked segmented sum of `inc` over matching lanes.
-          unsigned long long contrib = (packed == (uint64_t)lk) ? inc : 0;
+          // Segmented sum over matching lanes: XOR butterfly gives
+          // every lane (incl. the leader) the full total.
+          unsigned long long contrib =
+              (valid && packed == lk) ? inc : 0ULL;
           for (int off = WAVE / 2; off > 0; off >>= 1) {
-            contrib += __shfl_down((long long)contrib, off);
+            contrib += (unsigned long long)__shfl_xor(
+                (long long)contrib, off);
           }
           if (lane == l) {
             leader = true;
@@ -143,7 +153,7 @@ __global__ void k_window_agg_insert(
         remaining &= ~match;
       }
       if (leader) ok = hash_add(tkeys, tvals, mask, packed, agg);
-    } else {
+    } else if (valid) {
       ok = hash_add(tkeys, tvals, mask, packed, inc);
     }
     if (!ok) atomicExch(error_flag, 1);
