@@ -78,6 +78,7 @@ class _LocalCluster:
         self.eof_votes = [False] * worker_count
         self.close = False
         self.all_eof = False
+        self.abort_vote = False
         self.abort: Optional[BaseException] = None
         self.epoch_deadline = 0.0
 
@@ -89,6 +90,14 @@ class _LocalCluster:
 
 
 class _WorkerCtx:
+    """Thread-transport worker context.
+
+    The transport interface (`barrier`, `exchange_round`, `vote_close`)
+    is shared with the multi-process ``torch.distributed`` transport in
+    :mod:`bytewax_amd._engine.dist` (gloo for CPU processes, RCCL for
+    GPU workers).
+    """
+
     def __init__(self, cluster: _LocalCluster, worker_index: int):
         self.cluster = cluster
         self.worker_index = worker_index
@@ -106,30 +115,49 @@ class _WorkerCtx:
         if self.cluster.abort is not None:
             raise _Interrupted()
 
-    def send(self, dst: int, step_idx: int, input_idx: int, items: List[Any]) -> None:
-        with self.cluster.lock:
-            self.cluster.inboxes[dst].append((step_idx, input_idx, items))
+    def fail(self, ex: BaseException) -> None:
+        self.cluster.fail(ex)
 
-    def deliver(self) -> List[Tuple[int, int, List[Any]]]:
-        with self.cluster.lock:
-            msgs = self.cluster.inboxes[self.worker_index]
-            self.cluster.inboxes[self.worker_index] = []
+    def exchange_round(
+        self, outbox: Dict[int, List[Tuple[int, int, List[Any]]]]
+    ) -> List[Tuple[int, int, List[Any]]]:
+        """Deliver per-destination buffered messages; returns messages
+        addressed to this worker.  Collective: all workers call this
+        the same number of times per epoch."""
+        c = self.cluster
+        with c.lock:
+            for dst, msgs in outbox.items():
+                c.inboxes[dst].extend(msgs)
+        self.barrier()
+        with c.lock:
+            msgs = c.inboxes[self.worker_index]
+            c.inboxes[self.worker_index] = []
         return msgs
 
-    def vote_close(self, local_eof: bool, deadline: float) -> Tuple[bool, bool]:
-        """Returns (close, all_eof); identical on every worker."""
+    def vote_close(
+        self, local_eof: bool, deadline: float, aborting: bool
+    ) -> Tuple[bool, bool, bool]:
+        """Returns (close, all_eof, abort); identical on every worker."""
         c = self.cluster
         if self.worker_count == 1:
             all_eof = local_eof
-            return (all_eof or time.monotonic() >= deadline, all_eof)
+            return (
+                all_eof or aborting or time.monotonic() >= deadline,
+                all_eof,
+                aborting,
+            )
         with c.lock:
             c.eof_votes[self.worker_index] = local_eof
+            if aborting:
+                c.abort_vote = True
         self.barrier()
         if self.worker_index == 0:
             c.all_eof = all(c.eof_votes)
-            c.close = c.all_eof or time.monotonic() >= deadline
+            c.close = (
+                c.all_eof or c.abort_vote or time.monotonic() >= deadline
+            )
         self.barrier()
-        return (c.close, c.all_eof)
+        return (c.close, c.all_eof, c.abort_vote)
 
 
 # ---------------------------------------------------------------------------
@@ -496,6 +524,7 @@ class _Worker:
         self.epoch = resume_epoch
         self.epoch_interval = epoch_interval
         self.fifo: deque = deque()
+        self.outbox: Dict[int, List[Tuple[int, int, List[Any]]]] = {}
         self.execs: List[_Exec] = []
         w, n = ctx.worker_index, ctx.worker_count
 
@@ -620,7 +649,9 @@ class _Worker:
             if dst == self.ctx.worker_index:
                 self.fifo.append((step_idx, input_idx, chunk))
             else:
-                self.ctx.send(dst, step_idx, input_idx, chunk)
+                self.outbox.setdefault(dst, []).append(
+                    (step_idx, input_idx, chunk)
+                )
 
     def _drain(self) -> None:
         while self.fifo:
@@ -635,8 +666,8 @@ class _Worker:
         if self.ctx.worker_count == 1:
             return
         for _ in range(self.graph.n_exchange_rounds):
-            self.ctx.barrier()
-            for msg in self.ctx.deliver():
+            outbox, self.outbox = self.outbox, {}
+            for msg in self.ctx.exchange_round(outbox):
                 self.fifo.append(msg)
             self._drain()
 
@@ -648,25 +679,30 @@ class _Worker:
             self.ctx.barrier()  # align epoch start
             deadline = time.monotonic() + interval_s
             all_eof = False
-            aborted = False
+            abort_exc: Optional[BaseException] = None
             while True:
                 now = _now()
                 progressed = False
-                try:
-                    for src in self.input_execs:
-                        progressed |= src.poll(now, self._emit)
-                except AbortExecution as ex:
-                    self.ctx.cluster.fail(ex)
-                    aborted = True
-                if aborted:
-                    raise _Interrupted()
-                for sf in self.stateful_execs:
-                    for stream_id, out_items in sf.fire_timers(now):
-                        progressed = True
-                        self._emit(stream_id, out_items)
+                if abort_exc is None:
+                    try:
+                        for src in self.input_execs:
+                            progressed |= src.poll(now, self._emit)
+                    except AbortExecution as ex:
+                        abort_exc = ex
+                    for sf in self.stateful_execs:
+                        for stream_id, out_items in sf.fire_timers(now):
+                            progressed = True
+                            self._emit(stream_id, out_items)
                 self._exchange_rounds()
                 local_eof = all(e.eof_all() for e in self.input_execs)
-                close, all_eof = self.ctx.vote_close(local_eof, deadline)
+                close, all_eof, abort = self.ctx.vote_close(
+                    local_eof, deadline, abort_exc is not None
+                )
+                if abort:
+                    # Abort the whole execution: no snapshot for the
+                    # open epoch; resume replays it.
+                    self.ctx.fail(abort_exc or AbortExecution())
+                    raise _Interrupted()
                 if close:
                     break
                 if not progressed:
