@@ -60,9 +60,42 @@ class RecordBatch:
         return int(self.keys.numel())
 
 
+def _mix64_torch(x):
+    """splitmix64 finalizer on int64 tensors, bit-identical to the
+    device `mix64` (logical shifts emulated by masking the arithmetic
+    shift's sign extension; multiplication wraps in both)."""
+
+    def shr(v, s):
+        return (v >> s) & ((1 << (64 - s)) - 1)
+
+    x = x ^ shr(x, 33)
+    x = x * -48064316817838067  # 0xff51afd7ed558ccd as int64
+    x = x ^ shr(x, 33)
+    x = x * -4265267296055464877  # 0xc4ceb9fe1a85ec53 as int64
+    x = x ^ shr(x, 33)
+    return x
+
+
+def _bucket_cpu(batch: RecordBatch, world: int):
+    """Host twin of the bucketing kernels (power-of-two worlds match
+    the device hash exactly)."""
+    import torch
+
+    dst = torch.remainder(
+        _mix64_torch(batch.keys.to(torch.int64)), world
+    )
+    order = torch.argsort(dst, stable=True)
+    counts = torch.bincount(dst, minlength=world).to(torch.int32)
+    send_keys = batch.keys[order]
+    send_ts = batch.ts[order]
+    send_vals = batch.vals[order] if batch.vals is not None else None
+    return counts, send_keys, send_ts, send_vals
+
+
 def exchange_by_key(batch: RecordBatch, group=None) -> RecordBatch:
     """Exchange a batch across all workers so each key lands on its
-    owning GPU: bucketing kernel → RCCL all-to-allv over xGMI.
+    owning worker: bucketing kernel → RCCL all-to-allv over xGMI (gloo
+    when the batch lives on CPU).
 
     Collective: every rank must call this once per scheduling step.
     """
@@ -72,29 +105,32 @@ def exchange_by_key(batch: RecordBatch, group=None) -> RecordBatch:
     world = dist.get_world_size(group)
     if world == 1:
         return batch
-    k = ext()
     dev = batch.keys.device
     n = len(batch)
-    counts = torch.zeros(world, dtype=torch.int32, device=dev)
-    k.bucket_hist(batch.keys, world, counts)
-    offsets = torch.cumsum(counts, 0, dtype=torch.int32) - counts
-    cursors = offsets.clone()
-    send_keys = torch.empty(n, dtype=torch.int32, device=dev)
-    send_ts = torch.empty(n, dtype=torch.int64, device=dev)
     has_vals = batch.vals is not None
-    send_vals = torch.empty(
-        n if has_vals else 0, dtype=torch.int64, device=dev
-    )
-    k.bucket_scatter(
-        batch.keys,
-        batch.ts,
-        batch.vals if has_vals else None,
-        world,
-        cursors,
-        send_keys,
-        send_ts,
-        send_vals,
-    )
+    if dev.type == "cpu":
+        counts, send_keys, send_ts, send_vals = _bucket_cpu(batch, world)
+    else:
+        k = ext()
+        counts = torch.zeros(world, dtype=torch.int32, device=dev)
+        k.bucket_hist(batch.keys, world, counts)
+        offsets = torch.cumsum(counts, 0, dtype=torch.int32) - counts
+        cursors = offsets.clone()
+        send_keys = torch.empty(n, dtype=torch.int32, device=dev)
+        send_ts = torch.empty(n, dtype=torch.int64, device=dev)
+        send_vals = torch.empty(
+            n if has_vals else 0, dtype=torch.int64, device=dev
+        )
+        k.bucket_scatter(
+            batch.keys,
+            batch.ts,
+            batch.vals if has_vals else None,
+            world,
+            cursors,
+            send_keys,
+            send_ts,
+            send_vals,
+        )
     # Exchange split sizes (host sync of 2*world ints — control plane).
     recv_counts = torch.empty_like(counts)
     dist.all_to_all_single(recv_counts, counts, group=group)
@@ -137,12 +173,22 @@ class WindowAggState:
     ):
         import torch
 
-        self.k = ext()
         self.device = device
         self.align_ms = align_ms
         self.len_ms = len_ms
         self.mode = mode
         self.dedup = dedup
+        self.max_ts_host = 0  # watermark if batches carry max_ts
+        self.closed_horizon = -(1 << 62)  # window ids below are closed
+        self.cpu = device.type == "cpu"
+        if self.cpu:
+            # Host twin of the device table: used for CPU-only test
+            # runs of the columnar path (gloo, no GPU).  Not a
+            # production fallback — `ext()` still hard-fails on GPU
+            # machines if the extension is missing.
+            self._table: Dict[Tuple[int, int], int] = {}
+            return
+        self.k = ext()
         self.nslots = 1 << slots_pow
         self.tkeys = torch.full(
             (self.nslots,), -1, dtype=torch.int64, device=device
@@ -157,10 +203,32 @@ class WindowAggState:
         self.out_n = torch.zeros(1, dtype=torch.int32, device=device)
         # Pinned staging for snapshot spill to host DRAM.
         self._pin_keys = None
-        self.max_ts_host = 0  # watermark if batches carry max_ts
-        self.closed_horizon = -(1 << 62)  # window ids below are closed
+
+    def _insert_cpu(self, batch: RecordBatch) -> None:
+        import numpy as np
+
+        keys = batch.keys.numpy()
+        wins = ((batch.ts.numpy() - self.align_ms) // self.len_ms).astype(
+            "int64"
+        )
+        if self.mode == AGG_COUNT:
+            vals = np.ones(len(keys), dtype="int64")
+        else:
+            vals = batch.vals.numpy()
+        packed = (wins << 32) | keys.astype("uint32")
+        uniq, inv = np.unique(packed, return_inverse=True)
+        sums = np.bincount(inv, weights=vals).astype("int64")
+        for p, s in zip(uniq.tolist(), sums.tolist()):
+            kw = (int(np.uint32(p & 0xFFFFFFFF)), int(p >> 32))
+            self._table[kw] = self._table.get(kw, 0) + s
+        mx = int(batch.ts.max().item()) if len(batch) else 0
+        if mx > self.max_ts_host:
+            self.max_ts_host = mx
 
     def insert(self, batch: RecordBatch) -> None:
+        if self.cpu:
+            self._insert_cpu(batch)
+            return
         self.k.window_agg_insert(
             batch.keys,
             batch.ts,
@@ -178,13 +246,32 @@ class WindowAggState:
             self.max_ts_host = batch.max_ts
 
     def watermark_ms(self, sync: bool = False) -> int:
-        if sync:
+        if sync and not self.cpu:
             dev = int(self.max_ts_dev.item())
             if dev > self.max_ts_host:
                 self.max_ts_host = dev
         return self.max_ts_host
 
+    def _extract_cpu(self, horizon: int, clear: bool) -> Optional[RecordBatch]:
+        import torch
+
+        hit = [(k, w, v) for (k, w), v in self._table.items() if w < horizon]
+        if not hit:
+            return None
+        if clear:
+            for k, w, _v in hit:
+                del self._table[(k, w)]
+        keys = torch.tensor([k for k, _w, _v in hit], dtype=torch.int32)
+        ts = torch.tensor(
+            [w * self.len_ms + self.align_ms for _k, w, _v in hit],
+            dtype=torch.int64,
+        )
+        vals = torch.tensor([v for _k, _w, v in hit], dtype=torch.int64)
+        return RecordBatch(keys, ts, vals)
+
     def _extract(self, horizon: int, clear: bool) -> Optional[RecordBatch]:
+        if self.cpu:
+            return self._extract_cpu(horizon, clear)
         import torch
 
         self.out_n.zero_()
@@ -234,6 +321,17 @@ class WindowAggState:
         """Spill the live table to pinned host memory (recovery)."""
         import torch
 
+        if self.cpu:
+            import numpy as np
+
+            items = list(self._table.items())
+            return {
+                "keys": np.array([k for (k, _w), _v in items], dtype="int32"),
+                "wins": np.array([w for (_k, w), _v in items], dtype="int32"),
+                "vals": np.array([v for _kw, v in items], dtype="int64"),
+                "max_ts": self.watermark_ms(),
+                "closed_horizon": self.closed_horizon,
+            }
         self.out_n.zero_()
         self.k.close_extract(
             self.tkeys,
@@ -269,6 +367,17 @@ class WindowAggState:
         import torch
 
         n = len(snap["keys"])
+        if self.cpu:
+            for k, w, v in zip(
+                snap["keys"].tolist(),
+                snap["wins"].tolist(),
+                snap["vals"].tolist(),
+            ):
+                kw = (int(k), int(w))
+                self._table[kw] = self._table.get(kw, 0) + int(v)
+            self.max_ts_host = snap["max_ts"]
+            self.closed_horizon = snap["closed_horizon"]
+            return
         if n:
             keys = torch.as_tensor(snap["keys"]).to(self.device)
             wins = torch.as_tensor(snap["wins"]).to(torch.int64)

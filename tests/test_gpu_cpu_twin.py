@@ -1,0 +1,159 @@
+"""CPU-twin tests of the columnar path (no GPU required).
+
+The same `WindowAggState` / `exchange_by_key` / `keyed_window_agg`
+APIs run against a host table, letting the full bench pipeline —
+including the multi-rank gloo collectives — be validated on CPU.
+The device implementations are checked against host references in
+``tests/test_gpu_kernels.py`` (marked gpu).
+"""
+
+import json
+import os
+import subprocess
+import sys
+from collections import Counter
+from datetime import datetime, timedelta, timezone
+from pathlib import Path
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+from bytewax_amd.gpu import (  # noqa: E402
+    AGG_COUNT,
+    AGG_SUM,
+    RecordBatch,
+    WindowAggState,
+    _ms,
+)
+
+ALIGN = datetime(2024, 1, 1, tzinfo=timezone.utc)
+
+
+def test_cpu_window_count():
+    align_ms = _ms(ALIGN)
+    state = WindowAggState(torch.device("cpu"), align_ms, 60_000, AGG_COUNT)
+    keys = torch.tensor([1, 1, 2, 1], dtype=torch.int32)
+    ts = torch.tensor(
+        [align_ms, align_ms + 1, align_ms + 61_000, align_ms + 60_000],
+        dtype=torch.int64,
+    )
+    state.insert(RecordBatch(keys, ts, max_ts=align_ms + 61_000))
+    closed = state.close_due()
+    assert closed is not None
+    got = Counter(
+        zip(closed.keys.tolist(), closed.ts.tolist(), closed.vals.tolist())
+    )
+    assert got == Counter({(1, align_ms, 2): 1})
+    rest = state.close_all()
+    got2 = sorted(zip(rest.keys.tolist(), rest.vals.tolist()))
+    assert got2 == [(1, 1), (2, 1)]
+
+
+def test_cpu_snapshot_roundtrip():
+    align_ms = _ms(ALIGN)
+    a = WindowAggState(torch.device("cpu"), align_ms, 60_000, AGG_SUM)
+    keys = torch.tensor([5, 5, 9], dtype=torch.int32)
+    ts = torch.full((3,), align_ms + 10, dtype=torch.int64)
+    vals = torch.tensor([10, 20, 30], dtype=torch.int64)
+    a.insert(RecordBatch(keys, ts, vals, max_ts=align_ms + 10))
+    snap = a.snapshot_to_host()
+
+    b = WindowAggState(torch.device("cpu"), align_ms, 60_000, AGG_SUM)
+    b.restore_from_host(snap)
+    out = b.close_all()
+    got = sorted(zip(out.keys.tolist(), out.vals.tolist()))
+    assert got == [(5, 30), (9, 30)]
+
+
+def test_keyed_window_agg_pipeline_cpu():
+    import bytewax_amd.operators as op
+    from bytewax_amd.dataflow import Dataflow
+    from bytewax_amd.gpu.operators import (
+        CollectCountsSink,
+        SyntheticEventSource,
+        keyed_window_agg,
+    )
+    from bytewax_amd.testing import run_main
+
+    out = []
+    flow = Dataflow("cpu_twin")
+    s = op.input(
+        "inp",
+        flow,
+        SyntheticEventSource(
+            events_per_batch=10_000,
+            n_batches=4,
+            vocab=100,
+            align_to=ALIGN,
+            sim_ms_per_batch=30_000,
+            device="cpu",
+        ),
+    )
+    agg = keyed_window_agg(
+        "agg",
+        s,
+        align_to=ALIGN,
+        length=timedelta(minutes=1),
+        mode="count",
+        device="cpu",
+        exchange=False,
+    )
+    op.output("out", agg, CollectCountsSink(out))
+    run_main(flow)
+    total = sum(int(b.vals.sum().item()) for b in out)
+    assert total == 40_000
+
+
+@pytest.mark.timeout(300)
+def test_bench_two_ranks_gloo(tmp_path: Path):
+    """The exact bench.py path on 2 CPU ranks over gloo: validates the
+    collective alignment (exchange per step, split-size exchange,
+    vote/EOF) that the driver's 8-GPU scale run depends on."""
+    repo = Path(__file__).resolve().parent.parent
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(repo)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["MASTER_PORT"] = str(29650 + os.getpid() % 200)
+    procs = []
+    for rank in range(2):
+        e = dict(env)
+        e["RANK"] = str(rank)
+        e["LOCAL_RANK"] = str(rank)
+        e["WORLD_SIZE"] = "2"
+        procs.append(
+            subprocess.Popen(
+                [
+                    sys.executable,
+                    str(repo / "bench.py"),
+                    "--gpus",
+                    "2",
+                    "--steps",
+                    "3",
+                    "--warmup",
+                    "1",
+                    "--events-per-batch",
+                    "20000",
+                    "--vocab",
+                    "1000",
+                    "--device",
+                    "cpu",
+                ],
+                env=e,
+                stdout=subprocess.PIPE,
+                stderr=subprocess.PIPE,
+                cwd=str(repo),
+            )
+        )
+    outs = [p.communicate(timeout=240) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se.decode()[-2000:]
+    line = [
+        ln
+        for ln in outs[0][0].decode().splitlines()
+        if ln.startswith("{")
+    ][-1]
+    res = json.loads(line)
+    assert res["n_gpus"] == 2
+    assert res["value"] > 0
+    assert res["config"]["parallelism"] == "key-hash all-to-allv dp2"
