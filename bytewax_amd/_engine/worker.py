@@ -654,12 +654,33 @@ class _Worker:
                 )
 
     def _drain(self) -> None:
+        from .._metrics import metrics_enabled, observe_batch
+
+        instrumented = metrics_enabled()
         while self.fifo:
             step_idx, input_idx, items = self.fifo.popleft()
             ex = self.execs[step_idx]
-            for stream_id, out_items in ex.process(input_idx, items, self.epoch):
-                if out_items:
-                    self._emit(stream_id, out_items)
+            if instrumented:
+                t0 = time.perf_counter()
+                outs = ex.process(input_idx, items, self.epoch)
+                n_out = 0
+                for stream_id, out_items in outs:
+                    n_out += len(out_items)
+                    if out_items:
+                        self._emit(stream_id, out_items)
+                observe_batch(
+                    ex.step.step_id,
+                    self.ctx.worker_index,
+                    len(items),
+                    n_out,
+                    time.perf_counter() - t0,
+                )
+            else:
+                for stream_id, out_items in ex.process(
+                    input_idx, items, self.epoch
+                ):
+                    if out_items:
+                        self._emit(stream_id, out_items)
 
     def _exchange_rounds(self) -> None:
         self._drain()
@@ -798,6 +819,10 @@ def _run_cluster(
         msg = "epoch interval must be non-negative"
         raise ValueError(msg)
     graph = compile_graph(flow)
+
+    from .._webserver import maybe_start_webserver
+
+    maybe_start_webserver(flow)
 
     store = None
     ex_num, resume_epoch = 0, 1
