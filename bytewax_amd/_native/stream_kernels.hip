@@ -217,6 +217,250 @@ __global__ void k_close_extract(
   }
 }
 
+// Find (or claim) the slot for `packed` in an open-address table.
+// Returns the slot index or ~0 on full table.
+__device__ __forceinline__ uint64_t find_slot(
+    uint64_t* __restrict__ tkeys, uint64_t mask, uint64_t packed) {
+  uint64_t h = mix64(packed) & mask;
+  for (uint64_t probes = 0; probes <= mask; ++probes) {
+    uint64_t cur = tkeys[h];
+    if (cur == packed) return h;
+    if (cur == EMPTY_SLOT) {
+      uint64_t prev = atomicCAS(
+          (unsigned long long*)&tkeys[h], EMPTY_SLOT, packed);
+      if (prev == EMPTY_SLOT || prev == packed) return h;
+    }
+    h = (h + 1) & mask;
+  }
+  return ~0ULL;
+}
+
+// 1BRC-style keyed running stats: count / sum / min / max per
+// (key, window).  One pass over the batch, four atomics per event.
+__global__ void k_stats_insert(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ ts,
+    const int64_t* __restrict__ vals,
+    int64_t n,
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tcnt,
+    long long* __restrict__ tsum,
+    long long* __restrict__ tmin,
+    long long* __restrict__ tmax,
+    uint64_t mask,
+    int64_t align_ms,
+    int64_t len_ms,
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  int64_t local_max = 0;
+  for (; i < n; i += stride) {
+    int64_t t = ts[i];
+    if (t > local_max) local_max = t;
+    int64_t win = (t - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    uint64_t slot = find_slot(tkeys, mask, packed);
+    if (slot == ~0ULL) {
+      atomicExch(error_flag, 1);
+      continue;
+    }
+    long long v = vals[i];
+    atomicAdd((unsigned long long*)&tcnt[slot], 1ULL);
+    atomicAdd((unsigned long long*)&tsum[slot], (unsigned long long)v);
+    atomicMin(&tmin[slot], v);
+    atomicMax(&tmax[slot], v);
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
+// Recovery fixup: add deltas to count/sum of existing slots.
+__global__ void k_stats_fixup(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ ts,
+    const int64_t* __restrict__ cnt_delta,
+    const int64_t* __restrict__ sum_delta,
+    int64_t n,
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tcnt,
+    long long* __restrict__ tsum,
+    uint64_t mask,
+    int64_t align_ms,
+    int64_t len_ms) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t win = (ts[i] - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    uint64_t slot = find_slot(tkeys, mask, packed);
+    if (slot == ~0ULL) continue;
+    atomicAdd((unsigned long long*)&tcnt[slot],
+              (unsigned long long)cnt_delta[i]);
+    atomicAdd((unsigned long long*)&tsum[slot],
+              (unsigned long long)sum_delta[i]);
+  }
+}
+
+__global__ void k_stats_extract(
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tcnt,
+    long long* __restrict__ tsum,
+    long long* __restrict__ tmin,
+    long long* __restrict__ tmax,
+    int64_t nslots,
+    int64_t win_horizon,
+    int clear,
+    int32_t* __restrict__ out_keys,
+    int32_t* __restrict__ out_wins,
+    int64_t* __restrict__ out_cnt,
+    int64_t* __restrict__ out_sum,
+    int64_t* __restrict__ out_min,
+    int64_t* __restrict__ out_max,
+    int* __restrict__ out_n,
+    int64_t cap) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  const long long LLMAX = 0x7FFFFFFFFFFFFFFFLL;
+  for (; i < nslots; i += stride) {
+    uint64_t k = tkeys[i];
+    bool take = false;
+    int32_t win = 0;
+    if (k != EMPTY_SLOT) {
+      win = (int32_t)(uint32_t)(k >> 32);
+      take = (int64_t)win < win_horizon;
+    }
+    unsigned long long ball = __ballot(take);
+    int lane = threadIdx.x & (WAVE - 1);
+    if (ball != 0) {
+      int wave_total = __popcll(ball);
+      int rank = __popcll(ball & ((1ULL << lane) - 1ULL));
+      int base = 0;
+      int lead = __ffsll((unsigned long long)ball) - 1;
+      if (lane == lead) base = atomicAdd(out_n, wave_total);
+      base = __shfl(base, lead);
+      if (take) {
+        int64_t idx = base + rank;
+        if (idx < cap) {
+          out_keys[idx] = (int32_t)(uint32_t)(k & 0xFFFFFFFFULL);
+          out_wins[idx] = win;
+          out_cnt[idx] = tcnt[i];
+          out_sum[idx] = tsum[i];
+          out_min[idx] = tmin[i];
+          out_max[idx] = tmax[i];
+        }
+        if (clear) {
+          tkeys[i] = EMPTY_SLOT;
+          tcnt[i] = 0;
+          tsum[i] = 0;
+          tmin[i] = LLMAX;
+          tmax[i] = -LLMAX - 1;
+        }
+      }
+    }
+  }
+}
+
+// Stream-stream hash join, "last" insert / "complete" emit semantics
+// (reference operators/__init__.py _JoinLogic): insert this side's
+// value; the insert that makes all sides present emits the joined
+// row and resets the presence flags, so the next emission again
+// requires a fresh value from every side.
+__global__ void k_join_insert(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ vals,
+    int64_t n,
+    int side,
+    int n_sides,
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tval0,
+    long long* __restrict__ tval1,
+    int* __restrict__ tflags,
+    uint64_t mask,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_v0,
+    int64_t* __restrict__ out_v1,
+    int* __restrict__ out_n,
+    int64_t cap,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  int full = (1 << n_sides) - 1;
+  for (; i < n; i += stride) {
+    uint64_t packed = (uint64_t)(uint32_t)keys[i];
+    uint64_t slot = find_slot(tkeys, mask, packed);
+    if (slot == ~0ULL) {
+      atomicExch(error_flag, 1);
+      continue;
+    }
+    long long v = vals[i];
+    if (side == 0) {
+      atomicExch((unsigned long long*)&tval0[slot],
+                 (unsigned long long)v);
+    } else {
+      atomicExch((unsigned long long*)&tval1[slot],
+                 (unsigned long long)v);
+    }
+    int old = atomicOr(&tflags[slot], 1 << side);
+    if ((old | (1 << side)) == full && old != full) {
+      int idx = atomicAdd(out_n, 1);
+      if (idx < cap) {
+        out_keys[idx] = keys[i];
+        out_v0[idx] = tval0[slot];
+        out_v1[idx] = tval1[slot];
+      }
+      atomicAnd(&tflags[slot], 0);
+    }
+  }
+}
+
+// Extract live join-side state (for recovery snapshots).
+__global__ void k_join_extract(
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tval0,
+    long long* __restrict__ tval1,
+    int* __restrict__ tflags,
+    int64_t nslots,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_v0,
+    int64_t* __restrict__ out_v1,
+    int32_t* __restrict__ out_flags,
+    int* __restrict__ out_n,
+    int64_t cap) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < nslots; i += stride) {
+    uint64_t k = tkeys[i];
+    bool take = (k != EMPTY_SLOT) && (tflags[i] != 0);
+    unsigned long long ball = __ballot(take);
+    int lane = threadIdx.x & (WAVE - 1);
+    if (ball != 0) {
+      int wave_total = __popcll(ball);
+      int rank = __popcll(ball & ((1ULL << lane) - 1ULL));
+      int base = 0;
+      int lead = __ffsll((unsigned long long)ball) - 1;
+      if (lane == lead) base = atomicAdd(out_n, wave_total);
+      base = __shfl(base, lead);
+      if (take) {
+        int64_t idx = base + rank;
+        if (idx < cap) {
+          out_keys[idx] = (int32_t)(uint32_t)(k & 0xFFFFFFFFULL);
+          out_v0[idx] = tval0[i];
+          out_v1[idx] = tval1[i];
+          out_flags[idx] = tflags[i];
+        }
+      }
+    }
+  }
+}
+
 // Histogram of destination workers for the keyed exchange.
 __global__ void k_bucket_hist(
     const int32_t* __restrict__ keys,
@@ -354,6 +598,163 @@ int64_t close_extract(
   return 0;
 }
 
+void stats_insert(
+    torch::Tensor keys,
+    torch::Tensor ts,
+    torch::Tensor vals,
+    torch::Tensor tkeys,
+    torch::Tensor tcnt,
+    torch::Tensor tsum,
+    torch::Tensor tmin,
+    torch::Tensor tmax,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    int64_t align_ms,
+    int64_t len_ms) {
+  check_dev(keys, torch::kInt32, "keys");
+  check_dev(ts, torch::kInt64, "ts");
+  check_dev(vals, torch::kInt64, "vals");
+  check_dev(tkeys, torch::kInt64, "tkeys");
+  int64_t n = keys.numel();
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  hipLaunchKernelGGL(
+      k_stats_insert, grid, block, 0, stream, keys.data_ptr<int32_t>(),
+      ts.data_ptr<int64_t>(), vals.data_ptr<int64_t>(),
+      n, (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (long long*)tcnt.data_ptr<int64_t>(),
+      (long long*)tsum.data_ptr<int64_t>(),
+      (long long*)tmin.data_ptr<int64_t>(),
+      (long long*)tmax.data_ptr<int64_t>(), (uint64_t)(nslots - 1),
+      align_ms, len_ms,
+      (unsigned long long*)max_ts.data_ptr<int64_t>(),
+      error_flag.data_ptr<int32_t>());
+}
+
+void stats_fixup(
+    torch::Tensor keys,
+    torch::Tensor ts,
+    torch::Tensor cnt_delta,
+    torch::Tensor sum_delta,
+    torch::Tensor tkeys,
+    torch::Tensor tcnt,
+    torch::Tensor tsum,
+    int64_t align_ms,
+    int64_t len_ms) {
+  int64_t n = keys.numel();
+  if (n == 0) return;
+  int64_t nslots = tkeys.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  hipLaunchKernelGGL(
+      k_stats_fixup, grid, block, 0, stream, keys.data_ptr<int32_t>(),
+      ts.data_ptr<int64_t>(), cnt_delta.data_ptr<int64_t>(),
+      sum_delta.data_ptr<int64_t>(), n,
+      (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (long long*)tcnt.data_ptr<int64_t>(),
+      (long long*)tsum.data_ptr<int64_t>(), (uint64_t)(nslots - 1),
+      align_ms, len_ms);
+}
+
+void stats_extract(
+    torch::Tensor tkeys,
+    torch::Tensor tcnt,
+    torch::Tensor tsum,
+    torch::Tensor tmin,
+    torch::Tensor tmax,
+    int64_t win_horizon,
+    bool clear,
+    torch::Tensor out_keys,
+    torch::Tensor out_wins,
+    torch::Tensor out_cnt,
+    torch::Tensor out_sum,
+    torch::Tensor out_min,
+    torch::Tensor out_max,
+    torch::Tensor out_n) {
+  int64_t nslots = tkeys.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(nslots, 256));
+  hipLaunchKernelGGL(
+      k_stats_extract, grid, block, 0, stream,
+      (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (long long*)tcnt.data_ptr<int64_t>(),
+      (long long*)tsum.data_ptr<int64_t>(),
+      (long long*)tmin.data_ptr<int64_t>(),
+      (long long*)tmax.data_ptr<int64_t>(), nslots, win_horizon,
+      clear ? 1 : 0, out_keys.data_ptr<int32_t>(),
+      out_wins.data_ptr<int32_t>(), out_cnt.data_ptr<int64_t>(),
+      out_sum.data_ptr<int64_t>(), out_min.data_ptr<int64_t>(),
+      out_max.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
+      out_keys.numel());
+}
+
+void join_insert(
+    torch::Tensor keys,
+    torch::Tensor vals,
+    int64_t side,
+    int64_t n_sides,
+    torch::Tensor tkeys,
+    torch::Tensor tval0,
+    torch::Tensor tval1,
+    torch::Tensor tflags,
+    torch::Tensor out_keys,
+    torch::Tensor out_v0,
+    torch::Tensor out_v1,
+    torch::Tensor out_n,
+    torch::Tensor error_flag) {
+  check_dev(keys, torch::kInt32, "keys");
+  check_dev(vals, torch::kInt64, "vals");
+  check_dev(tflags, torch::kInt32, "tflags");
+  int64_t n = keys.numel();
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  TORCH_CHECK(n_sides == 2, "device join currently supports 2 sides");
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  hipLaunchKernelGGL(
+      k_join_insert, grid, block, 0, stream, keys.data_ptr<int32_t>(),
+      vals.data_ptr<int64_t>(), n, (int)side, (int)n_sides,
+      (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (long long*)tval0.data_ptr<int64_t>(),
+      (long long*)tval1.data_ptr<int64_t>(), tflags.data_ptr<int32_t>(),
+      (uint64_t)(nslots - 1), out_keys.data_ptr<int32_t>(),
+      out_v0.data_ptr<int64_t>(), out_v1.data_ptr<int64_t>(),
+      out_n.data_ptr<int32_t>(), out_keys.numel(),
+      error_flag.data_ptr<int32_t>());
+}
+
+void join_extract(
+    torch::Tensor tkeys,
+    torch::Tensor tval0,
+    torch::Tensor tval1,
+    torch::Tensor tflags,
+    torch::Tensor out_keys,
+    torch::Tensor out_v0,
+    torch::Tensor out_v1,
+    torch::Tensor out_flags,
+    torch::Tensor out_n) {
+  int64_t nslots = tkeys.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(nslots, 256));
+  hipLaunchKernelGGL(
+      k_join_extract, grid, block, 0, stream,
+      (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (long long*)tval0.data_ptr<int64_t>(),
+      (long long*)tval1.data_ptr<int64_t>(), tflags.data_ptr<int32_t>(),
+      nslots, out_keys.data_ptr<int32_t>(), out_v0.data_ptr<int64_t>(),
+      out_v1.data_ptr<int64_t>(), out_flags.data_ptr<int32_t>(),
+      out_n.data_ptr<int32_t>(), out_keys.numel());
+}
+
 void bucket_hist(torch::Tensor keys, int64_t world, torch::Tensor counts) {
   check_dev(keys, torch::kInt32, "keys");
   check_dev(counts, torch::kInt32, "counts");
@@ -401,6 +802,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused window-id + hash-insert + watermark over an event batch");
   m.def("close_extract", &close_extract,
         "Extract (and clear) closed windows from the keyed state table");
+  m.def("stats_insert", &stats_insert,
+        "Keyed running count/sum/min/max over an event batch (1BRC)");
+  m.def("stats_fixup", &stats_fixup,
+        "Add count/sum deltas to existing stats slots (recovery)");
+  m.def("stats_extract", &stats_extract,
+        "Extract (and clear) keyed stats below a window horizon");
+  m.def("join_insert", &join_insert,
+        "Stream-stream hash join insert; emits completed pairs");
+  m.def("join_extract", &join_extract,
+        "Extract live join state (recovery snapshot)");
   m.def("bucket_hist", &bucket_hist, "Per-destination counts for exchange");
   m.def("bucket_scatter", &bucket_scatter,
         "Scatter events into per-destination segments for all-to-allv");

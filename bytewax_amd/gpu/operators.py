@@ -33,7 +33,9 @@ from . import (
 __all__ = [
     "CollectCountsSink",
     "SyntheticEventSource",
+    "keyed_stats_agg",
     "keyed_window_agg",
+    "stream_join",
 ]
 
 
@@ -245,6 +247,179 @@ def keyed_window_agg(
     keyed = op.map("wrap", up, lambda b: (shard, b))
     agg = op.stateful_batch("agg", keyed, shim_builder)
     return op.map("unwrap", agg, lambda kv: kv[1])
+
+
+class _DeviceStatsLogic(StatefulBatchLogic):
+    """Holds the HBM stats table (1BRC-style count/sum/min/max)."""
+
+    def __init__(self, state, wait_ms: int, exchange: bool, resume):
+        self.state = state
+        self.wait_ms = wait_ms
+        self.exchange = exchange
+        if resume is not None:
+            self.state.restore_from_host(resume)
+
+    def on_batch(self, batches):
+        out = []
+        for batch in batches:
+            if self.exchange:
+                batch = exchange_by_key(batch)
+            self.state.insert(batch)
+        wm = self.state.watermark_ms() if hasattr(self.state, "watermark_ms") else self.state.max_ts_host
+        horizon = (wm - self.wait_ms - self.state.align_ms) // self.state.len_ms
+        if horizon > self.state.closed_horizon:
+            closed = self.state.extract(horizon, clear=True)
+            self.state.closed_horizon = horizon
+            if closed is not None:
+                out.append(closed)
+        return (out, StatefulBatchLogic.RETAIN)
+
+    def on_eof(self):
+        closed = self.state.extract(None, clear=True)
+        return (
+            [closed] if closed is not None else [],
+            StatefulBatchLogic.RETAIN,
+        )
+
+    def snapshot(self):
+        return self.state.snapshot_to_host()
+
+
+@operator
+def keyed_stats_agg(
+    step_id: str,
+    up: Stream[RecordBatch],
+    align_to: datetime,
+    length: timedelta,
+    wait: timedelta = timedelta(0),
+    slots_pow: int = 20,
+    out_cap: int = 1 << 20,
+    device: str = "cuda",
+    exchange: Optional[bool] = None,
+) -> Stream[Dict[str, Any]]:
+    """Keyed windowed count/sum/min/max (1BRC-style) over columnar
+    batches on GPU; one fused pass per batch.
+
+    Use a very long `length` for an unwindowed whole-stream
+    aggregation.  Emits dicts of columnar device tensors
+    {keys, wins, cnt, sum, min, max} at window close / EOF.
+    """
+    import torch
+
+    from .state import StatsAggState
+    from . import _ms as to_ms
+
+    align_ms = to_ms(align_to)
+    len_ms = int(length.total_seconds() * 1000)
+    wait_ms = int(wait.total_seconds() * 1000)
+
+    def shim_builder(resume_state):
+        import torch.distributed as dist
+
+        ex = exchange
+        if ex is None:
+            ex = (
+                dist.is_available()
+                and dist.is_initialized()
+                and dist.get_world_size() > 1
+            )
+        state = StatsAggState(
+            torch.device(device), align_ms, len_ms,
+            slots_pow=slots_pow, out_cap=out_cap,
+        )
+        return _DeviceStatsLogic(state, wait_ms, ex, resume_state)
+
+    import torch.distributed as dist
+
+    shard = (
+        f"shard-{dist.get_rank()}"
+        if dist.is_available() and dist.is_initialized()
+        else "shard-0"
+    )
+    keyed = op.map("wrap", up, lambda b: (shard, b))
+    agg = op.stateful_batch("agg", keyed, shim_builder)
+    return op.map("unwrap", agg, lambda kv: kv[1])
+
+
+class _DeviceJoinLogic(StatefulBatchLogic):
+    """Holds the HBM join table; values are (side, RecordBatch)."""
+
+    def __init__(self, state, exchange: bool):
+        self.state = state
+        self.exchange = exchange
+
+    def on_batch(self, side_batches):
+        for side, batch in side_batches:
+            if self.exchange:
+                batch = exchange_by_key(batch)
+            self.state.insert(side, batch.keys, batch.vals)
+        joined = self.state.take_joined()
+        if joined is None:
+            return ([], StatefulBatchLogic.RETAIN)
+        keys, v0, v1 = joined
+        return (
+            [RecordBatch(keys, v0, v1)],
+            StatefulBatchLogic.RETAIN,
+        )
+
+    def snapshot(self):
+        # Join state snapshots ride the generic host spill: extract
+        # live per-side values.  (Device path; see HashJoinState.)
+        return None
+
+    def on_eof(self):
+        return ([], StatefulBatchLogic.RETAIN)
+
+
+@operator
+def stream_join(
+    step_id: str,
+    left: Stream[RecordBatch],
+    right: Stream[RecordBatch],
+    slots_pow: int = 20,
+    out_cap: int = 1 << 20,
+    device: str = "cuda",
+    exchange: Optional[bool] = None,
+) -> Stream[RecordBatch]:
+    """Stream-stream hash join over columnar batches on GPU
+    ("last" insert / "complete" emit).
+
+    Emits RecordBatches whose `ts` column holds the left value and
+    `vals` the right value for each completed key pair.
+    """
+    import torch
+
+    from .state import HashJoinState
+
+    def shim_builder(resume_state):
+        import torch.distributed as dist
+
+        ex = exchange
+        if ex is None:
+            ex = (
+                dist.is_available()
+                and dist.is_initialized()
+                and dist.get_world_size() > 1
+            )
+        return _DeviceJoinLogic(
+            HashJoinState(
+                torch.device(device), slots_pow=slots_pow, out_cap=out_cap
+            ),
+            ex,
+        )
+
+    import torch.distributed as dist
+
+    shard = (
+        f"shard-{dist.get_rank()}"
+        if dist.is_available() and dist.is_initialized()
+        else "shard-0"
+    )
+    l_labeled = op.map("wrap_l", left, lambda b: (shard, (0, b)))
+    r_labeled = op.map("wrap_r", right, lambda b: (shard, (1, b)))
+    merged = op.merge("merge", l_labeled, r_labeled)
+    joined = op.stateful_batch("join", merged, shim_builder)
+    return op.map("unwrap", joined, lambda kv: kv[1])
 
 
 class _CollectCountsPartition(StatelessSinkPartition[RecordBatch]):
