@@ -33,6 +33,14 @@ def parse_args():
     p.add_argument("--window-sec", type=int, default=60)
     p.add_argument("--sim-ms-per-batch", type=int, default=5000)
     p.add_argument("--device", type=str, default="cuda")
+    p.add_argument(
+        "--engine",
+        choices=["auto", "python", "native"],
+        default="auto",
+        help="'native' = C++ step loop (single-GPU); 'python' = the "
+        "per-step engine loop (required for multi-GPU exchange); "
+        "'auto' picks native for world==1 on GPU",
+    )
     return p.parse_args()
 
 
@@ -113,23 +121,76 @@ def main():
                 seed=42 + rank * 7919,
             )
 
-    out: List = []
-    flow = Dataflow("bench_wordcount")
-    s = op.input("inp", flow, BenchSource())
-    agg = keyed_window_agg(
-        "window_count",
-        s,
-        align_to=align,
-        length=timedelta(seconds=args.window_sec),
-        mode="count",
-        slots_pow=max(14, (args.vocab * 4).bit_length()),
-        out_cap=max(1 << 20, args.vocab * 2),
-        device=device,
-        exchange=(world > 1),
-    )
-    op.output("out", agg, CollectCountsSink(out))
+    engine = args.engine
+    if engine == "auto":
+        engine = "native" if (world == 1 and on_gpu) else "python"
 
-    run_main(flow, epoch_interval=timedelta(days=365))
+    out: List = []
+    closed_rows = 0
+    lat: List[float] = []
+    if engine == "native":
+        # C++ step loop: the host thread is a pure kernel-launch
+        # engine; no Python between steps.
+        from bytewax_amd.gpu import WindowAggState, _ms
+
+        part = _SyntheticPartition(
+            torch.device(device),
+            E,
+            None,
+            args.vocab,
+            args.sim_ms_per_batch,
+            _ms(align),
+            seed=42 + rank * 7919,
+        )
+        # Zero-based ts templates; the kernel applies the per-step base.
+        ts_pool = [part.ts_template] * len(part.key_pool)
+        state = WindowAggState(
+            torch.device(device),
+            _ms(align),
+            args.window_sec * 1000,
+            slots_pow=max(14, (args.vocab * 4).bit_length()),
+            out_cap=max(1 << 20, args.vocab * 2),
+        )
+        # Warmup.
+        r, _ = state.native_run(
+            part.key_pool, ts_pool, 0, W, args.sim_ms_per_batch
+        )
+        closed_rows += r
+        barrier_sync()
+        t0 = time.perf_counter()
+        r, step_ns = state.native_run(
+            part.key_pool, ts_pool, W, K, args.sim_ms_per_batch
+        )
+        closed_rows += r
+        barrier_sync()
+        t1 = time.perf_counter()
+        timings["t0"], timings["t1"] = t0, t1
+        ns = step_ns.tolist()
+        lat = sorted(
+            (b - a) / 1e6 for a, b in zip(ns[:-1], ns[1:])
+        )
+    else:
+        flow = Dataflow("bench_wordcount")
+        s = op.input("inp", flow, BenchSource())
+        agg = keyed_window_agg(
+            "window_count",
+            s,
+            align_to=align,
+            length=timedelta(seconds=args.window_sec),
+            mode="count",
+            slots_pow=max(14, (args.vocab * 4).bit_length()),
+            out_cap=max(1 << 20, args.vocab * 2),
+            device=device,
+            exchange=(world > 1),
+        )
+        op.output("out", agg, CollectCountsSink(out))
+
+        run_main(flow, epoch_interval=timedelta(days=365))
+        steps = timings["step_starts"]
+        lat = sorted(
+            (b - a) * 1000.0 for a, b in zip(steps[:-1], steps[1:])
+        )
+        closed_rows = sum(len(b) for b in out)
 
     elapsed = timings["t1"] - timings["t0"]
     # Max over ranks.
@@ -144,13 +205,7 @@ def main():
     events_per_sec = total_events / elapsed
     ms_per_step = elapsed / K * 1000.0
 
-    steps = timings["step_starts"]
-    lat = sorted(
-        (b - a) * 1000.0 for a, b in zip(steps[:-1], steps[1:])
-    )
-    p99_ms = lat[int(len(lat) * 0.99) - 1] if lat else None
-
-    closed_rows = sum(len(b) for b in out)
+    p99_ms = lat[max(0, int(len(lat) * 0.99) - 1)] if lat else None
 
     if rank == 0:
         print(
@@ -175,6 +230,7 @@ def main():
                         "window_sec": args.window_sec,
                         "sim_ms_per_batch": args.sim_ms_per_batch,
                         "parallelism": f"key-hash all-to-allv dp{world}",
+                        "engine": engine,
                         "p99_step_ms": p99_ms,
                         "closed_window_rows": closed_rows,
                     },

@@ -29,10 +29,18 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <chrono>
 #include <cstdint>
+#include <vector>
 
 #define WAVE 64
 #define EMPTY_SLOT 0xFFFFFFFFFFFFFFFFULL
+
+#define HIP_CHECK(expr)                                               \
+  do {                                                                \
+    hipError_t _e = (expr);                                           \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e)); \
+  } while (0)
 
 static inline int n_blocks(int64_t n, int block) {
   int64_t b = (n + block - 1) / block;
@@ -97,6 +105,8 @@ __global__ void k_window_agg_insert(
     uint64_t mask,
     int64_t align_ms,
     int64_t len_ms,
+    int64_t ts_base,  // added to every timestamp (columnar sources can
+                      // reuse one template batch across steps)
     unsigned long long* __restrict__ max_ts,  // device scalar (atomicMax)
     int* __restrict__ error_flag) {
   int lane = threadIdx.x & (WAVE - 1);
@@ -108,7 +118,7 @@ __global__ void k_window_agg_insert(
   int64_t first = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   for (int64_t i = first; i - lane < n; i += stride) {
     bool valid = i < n;
-    int64_t t = valid ? ts[i] : 0;
+    int64_t t = valid ? (ts[i] + ts_base) : 0;
     if (t > local_max) local_max = t;
     uint64_t packed = 0;
     unsigned long long inc = 0;
@@ -530,7 +540,8 @@ void window_agg_insert(
     int64_t align_ms,
     int64_t len_ms,
     int64_t mode,
-    bool dedup) {
+    bool dedup,
+    int64_t ts_base) {
   check_dev(keys, torch::kInt32, "keys");
   check_dev(ts, torch::kInt64, "ts");
   check_dev(tkeys, torch::kInt64, "tkeys");
@@ -559,7 +570,7 @@ void window_agg_insert(
         keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), vptr, n,
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(),
-        (uint64_t)(nslots - 1), align_ms, len_ms,
+        (uint64_t)(nslots - 1), align_ms, len_ms, ts_base,
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>());
   };
@@ -797,6 +808,112 @@ void bucket_scatter(
       out_ts.data_ptr<int64_t>(), out_vals.data_ptr<int64_t>());
 }
 
+// ---------------------------------------------------------------------------
+// Native step-loop executor: runs the steady-state columnar window
+// pipeline (source batch -> fused insert -> watermark close) for a
+// whole run of steps with NO Python between steps.  This is the role
+// the reference's Rust `step_or_park` worker loop plays
+// (reference src/worker.rs:68-83), specialized for the device path:
+// the host thread is just a kernel-launch engine on one HIP stream.
+// Returns the total number of closed-window rows; per-step host
+// launch timestamps (ns) are written into `step_ns_out` for latency
+// percentiles.  The GIL is released for the duration.
+// ---------------------------------------------------------------------------
+
+int64_t native_run_window_steps(
+    std::vector<torch::Tensor> key_pool,
+    std::vector<torch::Tensor> ts_pool,
+    int64_t start_step,
+    int64_t n_steps,
+    int64_t sim_ms_per_batch,
+    torch::Tensor tkeys,
+    torch::Tensor tvals,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    torch::Tensor out_keys,
+    torch::Tensor out_wins,
+    torch::Tensor out_vals,
+    torch::Tensor out_n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t wait_ms,
+    int64_t mode,
+    bool dedup,
+    int64_t closed_horizon_in,
+    torch::Tensor step_ns_out,  // int64 CPU tensor [n_steps]
+    torch::Tensor state_out  // int64 CPU tensor [2]: closed_horizon, rows
+) {
+  TORCH_CHECK(!key_pool.empty(), "empty key pool");
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  int64_t n = key_pool[0].numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  int64_t* step_ns =
+      step_ns_out.numel() >= n_steps ? step_ns_out.data_ptr<int64_t>()
+                                     : nullptr;
+
+  int64_t closed_horizon = closed_horizon_in;
+  int64_t total_rows = 0;
+  int pool = (int)key_pool.size();
+
+  // Pinned host scalar for the close-count readback.
+  int* h_n = nullptr;
+  HIP_CHECK(hipHostMalloc((void**)&h_n, sizeof(int), hipHostMallocDefault));
+
+  {
+    pybind11::gil_scoped_release release;
+    for (int64_t s = 0; s < n_steps; ++s) {
+      if (step_ns != nullptr) {
+        step_ns[s] = std::chrono::duration_cast<std::chrono::nanoseconds>(
+                         std::chrono::steady_clock::now().time_since_epoch())
+                         .count();
+      }
+      int64_t step = start_step + s;
+      auto& keys = key_pool[step % pool];
+      auto& ts = ts_pool[step % pool];
+      int64_t base = align_ms + step * sim_ms_per_batch;
+      auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(
+            kern, grid, block, 0, stream, keys.data_ptr<int32_t>(),
+            ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
+            (uint64_t*)tkeys.data_ptr<int64_t>(),
+            (unsigned long long*)tvals.data_ptr<int64_t>(),
+            (uint64_t)(nslots - 1), align_ms, len_ms, base,
+            (unsigned long long*)max_ts.data_ptr<int64_t>(),
+            error_flag.data_ptr<int32_t>());
+      };
+      if (dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
+      else launch(k_window_agg_insert<AGG_COUNT, false>);
+
+      int64_t wm = base + sim_ms_per_batch - 1;
+      int64_t horizon = (wm - wait_ms - align_ms) / len_ms;
+      if (horizon > closed_horizon) {
+        HIP_CHECK(hipMemsetAsync(out_n.data_ptr<int32_t>(), 0, sizeof(int), stream));
+        hipLaunchKernelGGL(
+            k_close_extract, dim3(n_blocks(nslots, 256)), block, 0, stream,
+            (uint64_t*)tkeys.data_ptr<int64_t>(),
+            (unsigned long long*)tvals.data_ptr<int64_t>(), nslots, horizon,
+            1, out_keys.data_ptr<int32_t>(), out_wins.data_ptr<int32_t>(),
+            out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
+            out_keys.numel());
+        HIP_CHECK(hipMemcpyAsync(h_n, out_n.data_ptr<int32_t>(), sizeof(int),
+                       hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        total_rows += *h_n;
+        closed_horizon = horizon;
+      }
+    }
+    HIP_CHECK(hipStreamSynchronize(stream));
+  }
+  HIP_CHECK(hipHostFree(h_n));
+  auto* st = state_out.data_ptr<int64_t>();
+  st[0] = closed_horizon;
+  st[1] = total_rows;
+  return total_rows;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("window_agg_insert", &window_agg_insert,
         "Fused window-id + hash-insert + watermark over an event batch");
@@ -815,4 +932,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bucket_hist", &bucket_hist, "Per-destination counts for exchange");
   m.def("bucket_scatter", &bucket_scatter,
         "Scatter events into per-destination segments for all-to-allv");
+  m.def("native_run_window_steps", &native_run_window_steps,
+        "Run N steps of the columnar window pipeline with no Python "
+        "between steps (native step loop)");
 }

@@ -55,6 +55,9 @@ class RecordBatch:
     ts: Any  # torch.Tensor int64
     vals: Optional[Any] = None  # torch.Tensor int64
     max_ts: Optional[int] = None
+    #: Scalar added to every `ts` on the fly by consuming kernels —
+    #: lets sources reuse one timestamp-template tensor across steps.
+    ts_base: int = 0
 
     def __len__(self) -> int:
         return int(self.keys.numel())
@@ -105,6 +108,12 @@ def exchange_by_key(batch: RecordBatch, group=None) -> RecordBatch:
     world = dist.get_world_size(group)
     if world == 1:
         return batch
+    if batch.ts_base != 0:
+        msg = (
+            "exchange_by_key requires ts_base == 0 (bases are "
+            "per-producer; bake them into `ts` before exchanging)"
+        )
+        raise ValueError(msg)
     dev = batch.keys.device
     n = len(batch)
     has_vals = batch.vals is not None
@@ -208,9 +217,9 @@ class WindowAggState:
         import numpy as np
 
         keys = batch.keys.numpy()
-        wins = ((batch.ts.numpy() - self.align_ms) // self.len_ms).astype(
-            "int64"
-        )
+        wins = (
+            (batch.ts.numpy() + batch.ts_base - self.align_ms) // self.len_ms
+        ).astype("int64")
         if self.mode == AGG_COUNT:
             vals = np.ones(len(keys), dtype="int64")
         else:
@@ -221,7 +230,7 @@ class WindowAggState:
         for p, s in zip(uniq.tolist(), sums.tolist()):
             kw = (int(np.uint32(p & 0xFFFFFFFF)), int(p >> 32))
             self._table[kw] = self._table.get(kw, 0) + s
-        mx = int(batch.ts.max().item()) if len(batch) else 0
+        mx = int(batch.ts.max().item()) + batch.ts_base if len(batch) else 0
         if mx > self.max_ts_host:
             self.max_ts_host = mx
 
@@ -241,6 +250,7 @@ class WindowAggState:
             self.len_ms,
             self.mode,
             self.dedup,
+            batch.ts_base,
         )
         if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
             self.max_ts_host = batch.max_ts
@@ -363,6 +373,61 @@ class WindowAggState:
             "closed_horizon": self.closed_horizon,
         }
 
+    def native_run(
+        self,
+        key_pool,
+        ts_pool,
+        start_step: int,
+        n_steps: int,
+        sim_ms_per_batch: int,
+        wait_ms: int = 0,
+    ):
+        """Run `n_steps` of the window pipeline through the native C++
+        step loop (no Python between steps).  Single-worker path; the
+        multi-GPU exchange path uses the per-step Python loop.
+
+        Returns (closed_rows, step_launch_ns: torch int64 tensor).
+        """
+        import torch
+
+        if self.cpu:
+            msg = "native_run requires a device table"
+            raise RuntimeError(msg)
+        step_ns = torch.zeros(n_steps, dtype=torch.int64)
+        state_out = torch.zeros(2, dtype=torch.int64)
+        rows = self.k.native_run_window_steps(
+            list(key_pool),
+            list(ts_pool),
+            start_step,
+            n_steps,
+            sim_ms_per_batch,
+            self.tkeys,
+            self.tvals,
+            self.max_ts_dev,
+            self.error_flag,
+            self.out_keys,
+            self.out_wins,
+            self.out_vals,
+            self.out_n,
+            self.align_ms,
+            self.len_ms,
+            wait_ms,
+            self.mode,
+            self.dedup,
+            self.closed_horizon,
+            step_ns,
+            state_out,
+        )
+        self.closed_horizon = int(state_out[0].item())
+        self.max_ts_host = max(
+            self.max_ts_host,
+            self.align_ms + (start_step + n_steps) * sim_ms_per_batch - 1,
+        )
+        if int(self.error_flag.item()) != 0:
+            msg = "keyed window state table overflowed; increase slots_pow"
+            raise RuntimeError(msg)
+        return rows, step_ns
+
     def restore_from_host(self, snap: Dict[str, Any]) -> None:
         import torch
 
@@ -396,6 +461,7 @@ class WindowAggState:
                 self.len_ms,
                 AGG_SUM,  # re-add saved accumulators regardless of mode
                 False,
+                0,
             )
         self.max_ts_host = snap["max_ts"]
         self.closed_horizon = snap["closed_horizon"]
