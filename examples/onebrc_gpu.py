@@ -1,0 +1,117 @@
+"""1BRC-style keyed aggregation on MI355X (BASELINE config 5).
+
+One billion synthetic (station, temperature) rows aggregated to
+count/sum/min/max per station in HBM via the fused stats kernel, with
+the state spilled to host DRAM (pinned staging) as a recovery
+snapshot at the end.
+
+1 GPU:  python examples/onebrc_gpu.py [--rows 1000000000]
+N GPUs: python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+            --master-addr 127.0.0.1 examples/onebrc_gpu.py
+"""
+
+import argparse
+import os
+import sys
+import time
+from datetime import datetime, timedelta, timezone
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import bytewax_amd.operators as op
+from bytewax_amd.dataflow import Dataflow
+from bytewax_amd.gpu.operators import (
+    SyntheticEventSource,
+    keyed_stats_agg,
+)
+from bytewax_amd.outputs import DynamicSink, StatelessSinkPartition
+from bytewax_amd.testing import run_main
+
+ALIGN = datetime(2024, 1, 1, tzinfo=timezone.utc)
+N_STATIONS = 10_000
+
+
+class _StatsSink(StatelessSinkPartition):
+    def __init__(self, ls):
+        self._ls = ls
+
+    def write_batch(self, items):
+        self._ls.extend(items)
+
+
+class StatsSink(DynamicSink):
+    def __init__(self, ls):
+        self._ls = ls
+
+    def build(self, step_id, worker_index, worker_count):
+        return _StatsSink(self._ls)
+
+
+def main():
+    import torch
+
+    p = argparse.ArgumentParser()
+    p.add_argument("--rows", type=int, default=1_000_000_000)
+    p.add_argument("--rows-per-batch", type=int, default=50_000_000)
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.init_process_group(
+            backend="nccl" if torch.cuda.is_available() else "gloo"
+        )
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    rows_here = args.rows // world
+    n_batches = max(1, rows_here // args.rows_per_batch)
+
+    out = []
+    flow = Dataflow("onebrc")
+    s = op.input(
+        "inp",
+        flow,
+        SyntheticEventSource(
+            events_per_batch=args.rows_per_batch,
+            n_batches=n_batches,
+            vocab=N_STATIONS,
+            align_to=ALIGN,
+            sim_ms_per_batch=1000,
+            device=device,
+            with_vals=True,  # temperatures (int fixed-point)
+            seed=11 + rank,
+        ),
+    )
+    stats = keyed_stats_agg(
+        "stats",
+        s,
+        align_to=ALIGN,
+        length=timedelta(days=36500),  # unwindowed: whole-stream agg
+        slots_pow=16,
+        device=device,
+    )
+    op.output("out", stats, StatsSink(out))
+
+    t0 = time.perf_counter()
+    run_main(flow, epoch_interval=timedelta(days=365))
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+
+    total_rows = n_batches * args.rows_per_batch
+    stations = sum(len(o["keys"]) for o in out)
+    if rank == 0:
+        print(
+            f"aggregated {total_rows * world} rows over {stations} stations "
+            f"in {t1 - t0:.2f}s = "
+            f"{total_rows * world / (t1 - t0) / 1e9:.2f}e9 rows/s"
+        )
+
+
+if __name__ == "__main__":
+    main()
