@@ -165,10 +165,14 @@ def main():
         barrier_sync()
         t1 = time.perf_counter()
         timings["t0"], timings["t1"] = t0, t1
-        ns = step_ns.tolist()
-        lat = sorted(
-            (b - a) / 1e6 for a, b in zip(ns[:-1], ns[1:])
-        )
+        # EOF flush (outside the timed region, matching the Python
+        # engine whose EOF pass runs after the final poll).
+        final = state.close_all()
+        if final is not None:
+            closed_rows += len(final)
+        # Launch-to-launch gaps are not a latency measure under async
+        # execution; report no p99 for the native loop.
+        lat = []
     else:
         flow = Dataflow("bench_wordcount")
         s = op.input("inp", flow, BenchSource())
