@@ -34,6 +34,11 @@ def parse_args():
     p.add_argument("--sim-ms-per-batch", type=int, default=5000)
     p.add_argument("--device", type=str, default="cuda")
     p.add_argument(
+        "--radix",
+        action="store_true",
+        help="use the radix-partitioned LDS-staged insert path",
+    )
+    p.add_argument(
         "--engine",
         choices=["auto", "python", "native"],
         default="auto",
@@ -150,6 +155,8 @@ def main():
             args.window_sec * 1000,
             slots_pow=max(14, (args.vocab * 4).bit_length()),
             out_cap=max(1 << 20, args.vocab * 2),
+            radix=args.radix,
+            max_batch=E,
         )
         # Warmup.
         r, _ = state.native_run(
@@ -235,6 +242,7 @@ def main():
                         "sim_ms_per_batch": args.sim_ms_per_batch,
                         "parallelism": f"key-hash all-to-allv dp{world}",
                         "engine": engine,
+                        "radix": args.radix,
                         "p99_step_ms": p99_ms,
                         "closed_window_rows": closed_rows,
                     },

@@ -59,34 +59,58 @@ __device__ __forceinline__ uint64_t mix64(uint64_t x) {
   return x;
 }
 
+// Table slot addressing.  Two layouts share one code path:
+//   region_bits == 0: classic whole-table linear probing.
+//   region_bits  > 0: the table is partitioned into `nslots >>
+//     region_bits` contiguous *regions*; a key's region comes from the
+//     low hash bits and probing stays inside the region.  This is the
+//     layout the radix-partitioned LDS aggregation path flushes into
+//     (one workgroup owns one region → L2-local updates).
+__device__ __forceinline__ uint64_t region_of(
+    uint64_t h64, uint64_t mask, int region_bits) {
+  return (h64 & mask) >> region_bits;
+}
+
 // Insert `inc` into the open-address table for `packed`, creating the
 // slot if needed.  Table size is a power of two (`mask = nslots-1`).
-// Returns false if the table is full (caller-side growth handles it).
+// Returns false if the table (region) is full.
 __device__ __forceinline__ bool hash_add(
     uint64_t* __restrict__ tkeys,
     unsigned long long* __restrict__ tvals,
     uint64_t mask,
+    int region_bits,
     uint64_t packed,
     unsigned long long inc) {
-  uint64_t h = mix64(packed) & mask;
-  for (uint64_t probes = 0; probes <= mask; ++probes) {
-    uint64_t cur = tkeys[h];
+  uint64_t h64 = mix64(packed);
+  uint64_t h, probe_mask, base;
+  if (region_bits == 0) {
+    base = 0;
+    probe_mask = mask;
+    h = h64 & mask;
+  } else {
+    base = region_of(h64, mask, region_bits) << region_bits;
+    probe_mask = (1ULL << region_bits) - 1;
+    h = (h64 >> 32) & probe_mask;
+  }
+  for (uint64_t probes = 0; probes <= probe_mask; ++probes) {
+    uint64_t slot = base | h;
+    uint64_t cur = tkeys[slot];
     if (cur == packed) {
-      atomicAdd(&tvals[h], inc);
+      atomicAdd(&tvals[slot], inc);
       return true;
     }
     if (cur == EMPTY_SLOT) {
       // Plain load may be stale across XCDs; the CAS is the truth.
       uint64_t prev = atomicCAS(
-          (unsigned long long*)&tkeys[h], EMPTY_SLOT, packed);
+          (unsigned long long*)&tkeys[slot], EMPTY_SLOT, packed);
       if (prev == EMPTY_SLOT || prev == packed) {
-        atomicAdd(&tvals[h], inc);
+        atomicAdd(&tvals[slot], inc);
         return true;
       }
       // Someone else claimed the slot with a different key; keep
       // probing.
     }
-    h = (h + 1) & mask;
+    h = (h + 1) & probe_mask;
   }
   return false;
 }
@@ -107,6 +131,7 @@ __global__ void k_window_agg_insert(
     int64_t len_ms,
     int64_t ts_base,  // added to every timestamp (columnar sources can
                       // reuse one template batch across steps)
+    int region_bits,  // table layout (see hash_add)
     unsigned long long* __restrict__ max_ts,  // device scalar (atomicMax)
     int* __restrict__ error_flag) {
   int lane = threadIdx.x & (WAVE - 1);
@@ -162,9 +187,9 @@ __global__ void k_window_agg_insert(
         }
         remaining &= ~match;
       }
-      if (leader) ok = hash_add(tkeys, tvals, mask, packed, agg);
+      if (leader) ok = hash_add(tkeys, tvals, mask, region_bits, packed, agg);
     } else if (valid) {
-      ok = hash_add(tkeys, tvals, mask, packed, inc);
+      ok = hash_add(tkeys, tvals, mask, region_bits, packed, inc);
     }
     if (!ok) atomicExch(error_flag, 1);
   }
@@ -175,6 +200,174 @@ __global__ void k_window_agg_insert(
   }
   if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
     atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Radix-partitioned LDS-staged aggregation (the high-cardinality fast
+// path).  Instead of one random device-scope atomic per event into an
+// L2-missing table, events are first partitioned into contiguous
+// per-region segments (two streaming passes), then one workgroup
+// aggregates each region's events in an LDS-resident open-address
+// table (LDS atomics) and flushes distinct keys once into the region's
+// contiguous slice of the HBM table — turning ~N random global
+// atomics into ~distinct-keys L2-local ones.
+// ---------------------------------------------------------------------------
+
+template <int MODE>
+__global__ void k_radix_hist(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ ts,
+    int64_t n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    uint64_t mask,
+    int region_bits,
+    int* __restrict__ counts,
+    unsigned long long* __restrict__ max_ts) {
+  extern __shared__ int lhist[];
+  int nb = (int)(((mask + 1) >> region_bits));
+  for (int b = threadIdx.x; b < nb; b += blockDim.x) lhist[b] = 0;
+  __syncthreads();
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  int64_t local_max = 0;
+  for (; i < n; i += stride) {
+    int64_t t = ts[i] + ts_base;
+    if (t > local_max) local_max = t;
+    int64_t win = (t - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    int b = (int)region_of(mix64(packed), mask, region_bits);
+    atomicAdd(&lhist[b], 1);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < nb; b += blockDim.x) {
+    if (lhist[b] > 0) atomicAdd(&counts[b], lhist[b]);
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
+template <int MODE>
+__global__ void k_radix_scatter(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ ts,
+    const int64_t* __restrict__ vals,
+    int64_t n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    uint64_t mask,
+    int region_bits,
+    int* __restrict__ cursors,
+    uint64_t* __restrict__ ev_packed,
+    int64_t* __restrict__ ev_vals) {
+  // Two sweeps over this block's (deterministic) grid-stride slice:
+  // count per region into LDS, reserve global ranges, then write.
+  extern __shared__ int lmem[];
+  int nb = (int)(((mask + 1) >> region_bits));
+  int* lhist = lmem;
+  int* lbase = lmem + nb;
+  for (int b = threadIdx.x; b < nb; b += blockDim.x) lhist[b] = 0;
+  __syncthreads();
+  int64_t start = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (int64_t i = start; i < n; i += stride) {
+    int64_t win = (ts[i] + ts_base - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    atomicAdd(&lhist[(int)region_of(mix64(packed), mask, region_bits)], 1);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < nb; b += blockDim.x) {
+    int c = lhist[b];
+    lbase[b] = c > 0 ? atomicAdd(&cursors[b], c) : 0;
+    lhist[b] = 0;  // reused as the block-local write cursor
+  }
+  __syncthreads();
+  for (int64_t i = start; i < n; i += stride) {
+    int64_t win = (ts[i] + ts_base - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    int b = (int)region_of(mix64(packed), mask, region_bits);
+    int pos = lbase[b] + atomicAdd(&lhist[b], 1);
+    ev_packed[pos] = packed;
+    if (MODE == AGG_SUM) ev_vals[pos] = vals[i];
+  }
+}
+
+template <int MODE>
+__global__ __launch_bounds__(256) void k_radix_agg(
+    const uint64_t* __restrict__ ev_packed,
+    const int64_t* __restrict__ ev_vals,
+    const int* __restrict__ offsets,
+    const int* __restrict__ counts,
+    uint64_t* __restrict__ tkeys,
+    unsigned long long* __restrict__ tvals,
+    uint64_t mask,
+    int region_bits,
+    int* __restrict__ error_flag) {
+  extern __shared__ char smem[];
+  int region = 1 << region_bits;
+  uint64_t* lkeys = (uint64_t*)smem;
+  unsigned long long* lvals =
+      (unsigned long long*)(smem + (size_t)region * sizeof(uint64_t));
+  for (int s = threadIdx.x; s < region; s += blockDim.x) {
+    lkeys[s] = EMPTY_SLOT;
+    lvals[s] = 0;
+  }
+  __syncthreads();
+  int b = blockIdx.x;
+  int cnt = counts[b];
+  int start = offsets[b];
+  for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
+    uint64_t packed = ev_packed[start + j];
+    unsigned long long inc =
+        (MODE == AGG_SUM) ? (unsigned long long)ev_vals[start + j] : 1ULL;
+    uint64_t h64 = mix64(packed);
+    int lh = (int)((h64 >> 32) & (region - 1));
+    bool done = false;
+    for (int p = 0; p < region; ++p) {
+      uint64_t cur = lkeys[lh];
+      if (cur == packed) {
+        atomicAdd(&lvals[lh], inc);
+        done = true;
+        break;
+      }
+      if (cur == EMPTY_SLOT) {
+        uint64_t prev = atomicCAS(
+            (unsigned long long*)&lkeys[lh], EMPTY_SLOT, packed);
+        if (prev == EMPTY_SLOT || prev == packed) {
+          atomicAdd(&lvals[lh], inc);
+          done = true;
+          break;
+        }
+      }
+      lh = (lh + 1) & (region - 1);
+    }
+    if (!done) {
+      // LDS region full (only possible when the global region is at
+      // least as full): flush straight to the global region.
+      if (!hash_add(tkeys, tvals, mask, region_bits, packed, inc)) {
+        atomicExch(error_flag, 1);
+      }
+    }
+  }
+  __syncthreads();
+  // Flush distinct keys once into the region's contiguous HBM slice.
+  for (int s = threadIdx.x; s < region; s += blockDim.x) {
+    if (lkeys[s] != EMPTY_SLOT) {
+      if (!hash_add(tkeys, tvals, mask, region_bits, lkeys[s], lvals[s])) {
+        atomicExch(error_flag, 1);
+      }
+    }
   }
 }
 
@@ -541,7 +734,8 @@ void window_agg_insert(
     int64_t len_ms,
     int64_t mode,
     bool dedup,
-    int64_t ts_base) {
+    int64_t ts_base,
+    int64_t region_bits) {
   check_dev(keys, torch::kInt32, "keys");
   check_dev(ts, torch::kInt64, "ts");
   check_dev(tkeys, torch::kInt64, "tkeys");
@@ -571,6 +765,7 @@ void window_agg_insert(
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(),
         (uint64_t)(nslots - 1), align_ms, len_ms, ts_base,
+        (int)region_bits,
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>());
   };
@@ -578,6 +773,92 @@ void window_agg_insert(
   else if (mode == AGG_COUNT && dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
   else if (mode == AGG_SUM && !dedup) launch(k_window_agg_insert<AGG_SUM, false>);
   else launch(k_window_agg_insert<AGG_SUM, true>);
+}
+
+void radix_window_insert(
+    torch::Tensor keys,
+    torch::Tensor ts,
+    c10::optional<torch::Tensor> vals,
+    torch::Tensor tkeys,
+    torch::Tensor tvals,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    torch::Tensor counts,     // int32 [n_regions], zeroed here
+    torch::Tensor cursors,    // int32 [n_regions]
+    torch::Tensor ev_packed,  // int64 [>= n]
+    torch::Tensor ev_vals,    // int64 [>= n] (sum mode)
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t mode,
+    int64_t ts_base,
+    int64_t region_bits) {
+  check_dev(keys, torch::kInt32, "keys");
+  check_dev(ts, torch::kInt64, "ts");
+  int64_t n = keys.numel();
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  TORCH_CHECK(region_bits > 0, "radix path requires region_bits > 0");
+  TORCH_CHECK(region_bits <= 11, "region_bits > 11 exceeds the 64 KiB "
+              "dynamic-LDS-per-workgroup limit");
+  int64_t nb = nslots >> region_bits;
+  TORCH_CHECK(nb >= 1 && nb <= 8192, "region count out of range");
+  TORCH_CHECK(counts.numel() >= nb, "counts too small");
+  TORCH_CHECK(ev_packed.numel() >= n, "ev_packed too small");
+  const int64_t* vptr = nullptr;
+  if (mode == AGG_SUM) {
+    TORCH_CHECK(vals.has_value(), "sum mode requires vals");
+    TORCH_CHECK(ev_vals.numel() >= n, "ev_vals too small");
+    vptr = vals->data_ptr<int64_t>();
+  }
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  uint64_t mask = (uint64_t)(nslots - 1);
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  counts.zero_();
+  size_t hist_lds = (size_t)nb * sizeof(int);
+
+  auto hist = [&](auto kern) {
+    hipLaunchKernelGGL(
+        kern, grid, block, hist_lds, stream, keys.data_ptr<int32_t>(),
+        ts.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
+        (int)region_bits, counts.data_ptr<int32_t>(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>());
+  };
+  if (mode == AGG_COUNT) hist(k_radix_hist<AGG_COUNT>);
+  else hist(k_radix_hist<AGG_SUM>);
+
+  // Exclusive prefix sums (device-side, stays on the stream).
+  auto counts64 = counts.narrow(0, 0, nb);
+  auto offsets = at::cumsum(counts64, 0, at::kInt) - counts64;
+  cursors.narrow(0, 0, nb).copy_(offsets);
+
+  auto scat = [&](auto kern) {
+    hipLaunchKernelGGL(
+        kern, grid, block, 2 * hist_lds, stream, keys.data_ptr<int32_t>(),
+        ts.data_ptr<int64_t>(), vptr, n, align_ms, len_ms, ts_base, mask,
+        (int)region_bits, cursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr);
+  };
+  if (mode == AGG_COUNT) scat(k_radix_scatter<AGG_COUNT>);
+  else scat(k_radix_scatter<AGG_SUM>);
+
+  int region = 1 << region_bits;
+  size_t agg_lds = (size_t)region * 16;
+  auto offsets_i32 = offsets.contiguous();
+  auto agg = [&](auto kern) {
+    hipLaunchKernelGGL(
+        kern, dim3((unsigned)nb), block, agg_lds, stream,
+        (const uint64_t*)ev_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
+        offsets_i32.data_ptr<int32_t>(), counts.data_ptr<int32_t>(),
+        (uint64_t*)tkeys.data_ptr<int64_t>(),
+        (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
+        (int)region_bits, error_flag.data_ptr<int32_t>());
+  };
+  if (mode == AGG_COUNT) agg(k_radix_agg<AGG_COUNT>);
+  else agg(k_radix_agg<AGG_SUM>);
 }
 
 int64_t close_extract(
@@ -841,8 +1122,13 @@ int64_t native_run_window_steps(
     bool dedup,
     int64_t closed_horizon_in,
     torch::Tensor step_ns_out,  // int64 CPU tensor [n_steps]
-    torch::Tensor state_out  // int64 CPU tensor [2]: closed_horizon, rows
-) {
+    torch::Tensor state_out,  // int64 CPU tensor [2]: closed_horizon, rows
+    int64_t region_bits,
+    bool use_radix,
+    c10::optional<torch::Tensor> counts,
+    c10::optional<torch::Tensor> cursors,
+    c10::optional<torch::Tensor> ev_packed,
+    c10::optional<torch::Tensor> ev_vals) {
   TORCH_CHECK(!key_pool.empty(), "empty key pool");
   int64_t nslots = tkeys.numel();
   TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
@@ -874,18 +1160,26 @@ int64_t native_run_window_steps(
       auto& keys = key_pool[step % pool];
       auto& ts = ts_pool[step % pool];
       int64_t base = align_ms + step * sim_ms_per_batch;
-      auto launch = [&](auto kern) {
-        hipLaunchKernelGGL(
-            kern, grid, block, 0, stream, keys.data_ptr<int32_t>(),
-            ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
-            (uint64_t*)tkeys.data_ptr<int64_t>(),
-            (unsigned long long*)tvals.data_ptr<int64_t>(),
-            (uint64_t)(nslots - 1), align_ms, len_ms, base,
-            (unsigned long long*)max_ts.data_ptr<int64_t>(),
-            error_flag.data_ptr<int32_t>());
-      };
-      if (dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
-      else launch(k_window_agg_insert<AGG_COUNT, false>);
+      if (use_radix) {
+        radix_window_insert(
+            keys, ts, c10::nullopt, tkeys, tvals, max_ts, error_flag,
+            *counts, *cursors, *ev_packed, *ev_vals, align_ms, len_ms,
+            AGG_COUNT, base, region_bits);
+      } else {
+        auto launch = [&](auto kern) {
+          hipLaunchKernelGGL(
+              kern, grid, block, 0, stream, keys.data_ptr<int32_t>(),
+              ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
+              (uint64_t*)tkeys.data_ptr<int64_t>(),
+              (unsigned long long*)tvals.data_ptr<int64_t>(),
+              (uint64_t)(nslots - 1), align_ms, len_ms, base,
+              (int)region_bits,
+              (unsigned long long*)max_ts.data_ptr<int64_t>(),
+              error_flag.data_ptr<int32_t>());
+        };
+        if (dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
+        else launch(k_window_agg_insert<AGG_COUNT, false>);
+      }
 
       int64_t wm = base + sim_ms_per_batch - 1;
       int64_t horizon = (wm - wait_ms - align_ms) / len_ms;
@@ -917,6 +1211,8 @@ int64_t native_run_window_steps(
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("window_agg_insert", &window_agg_insert,
         "Fused window-id + hash-insert + watermark over an event batch");
+  m.def("radix_window_insert", &radix_window_insert,
+        "Radix-partitioned LDS-staged keyed window aggregation");
   m.def("close_extract", &close_extract,
         "Extract (and clear) closed windows from the keyed state table");
   m.def("stats_insert", &stats_insert,

@@ -179,7 +179,15 @@ class WindowAggState:
         slots_pow: int = 20,
         dedup: bool = False,
         out_cap: int = 1 << 20,
+        radix: bool = False,
+        region_bits: int = 11,
+        max_batch: int = 0,
     ):
+        """:arg radix: Use the radix-partitioned LDS-staged insert path
+        (fastest for high-cardinality keys; requires `max_batch`, the
+        largest batch size that will be inserted).
+        :arg region_bits: log2 of the table-region size when `radix`.
+        """
         import torch
 
         self.device = device
@@ -189,7 +197,31 @@ class WindowAggState:
         self.dedup = dedup
         self.max_ts_host = 0  # watermark if batches carry max_ts
         self.closed_horizon = -(1 << 62)  # window ids below are closed
+        self.radix = radix
+        self.region_bits = region_bits if radix else 0
         self.cpu = device.type == "cpu"
+        if radix and not self.cpu:
+            if max_batch <= 0:
+                msg = "radix path requires max_batch"
+                raise ValueError(msg)
+            n_regions = (1 << slots_pow) >> region_bits
+            if n_regions < 1:
+                msg = "slots_pow must exceed region_bits"
+                raise ValueError(msg)
+            self.rx_counts = torch.zeros(
+                n_regions, dtype=torch.int32, device=device
+            )
+            self.rx_cursors = torch.zeros(
+                n_regions, dtype=torch.int32, device=device
+            )
+            self.rx_packed = torch.empty(
+                max_batch, dtype=torch.int64, device=device
+            )
+            self.rx_vals = torch.empty(
+                max_batch if mode == AGG_SUM else 0,
+                dtype=torch.int64,
+                device=device,
+            )
         if self.cpu:
             # Host twin of the device table: used for CPU-only test
             # runs of the columnar path (gloo, no GPU).  Not a
@@ -238,20 +270,41 @@ class WindowAggState:
         if self.cpu:
             self._insert_cpu(batch)
             return
-        self.k.window_agg_insert(
-            batch.keys,
-            batch.ts,
-            batch.vals,
-            self.tkeys,
-            self.tvals,
-            self.max_ts_dev,
-            self.error_flag,
-            self.align_ms,
-            self.len_ms,
-            self.mode,
-            self.dedup,
-            batch.ts_base,
-        )
+        if self.radix:
+            self.k.radix_window_insert(
+                batch.keys,
+                batch.ts,
+                batch.vals,
+                self.tkeys,
+                self.tvals,
+                self.max_ts_dev,
+                self.error_flag,
+                self.rx_counts,
+                self.rx_cursors,
+                self.rx_packed,
+                self.rx_vals,
+                self.align_ms,
+                self.len_ms,
+                self.mode,
+                batch.ts_base,
+                self.region_bits,
+            )
+        else:
+            self.k.window_agg_insert(
+                batch.keys,
+                batch.ts,
+                batch.vals,
+                self.tkeys,
+                self.tvals,
+                self.max_ts_dev,
+                self.error_flag,
+                self.align_ms,
+                self.len_ms,
+                self.mode,
+                self.dedup,
+                batch.ts_base,
+                self.region_bits,
+            )
         if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
             self.max_ts_host = batch.max_ts
 
@@ -417,6 +470,12 @@ class WindowAggState:
             self.closed_horizon,
             step_ns,
             state_out,
+            self.region_bits,
+            self.radix,
+            self.rx_counts if self.radix else None,
+            self.rx_cursors if self.radix else None,
+            self.rx_packed if self.radix else None,
+            self.rx_vals if self.radix else None,
         )
         self.closed_horizon = int(state_out[0].item())
         self.max_ts_host = max(
@@ -462,6 +521,7 @@ class WindowAggState:
                 AGG_SUM,  # re-add saved accumulators regardless of mode
                 False,
                 0,
+                self.region_bits,
             )
         self.max_ts_host = snap["max_ts"]
         self.closed_horizon = snap["closed_horizon"]

@@ -69,6 +69,44 @@ def test_window_count_matches_reference(dedup):
     assert got == ref
 
 
+@pytest.mark.parametrize("mode_name", ["count", "sum"])
+def test_window_radix_matches_reference(mode_name):
+    """The radix-partitioned LDS-staged path must agree exactly with
+    the host reference (and hence with the single-pass path)."""
+    _skip_no_gpu()
+    from bytewax_amd.gpu import AGG_COUNT, AGG_SUM, RecordBatch, WindowAggState, _ms
+
+    torch.manual_seed(9)
+    n = 500_000
+    align_ms = _ms(ALIGN)
+    len_ms = 60_000
+    keys = torch.randint(0, 20_000, (n,), dtype=torch.int32)
+    ts = align_ms + torch.randint(0, 300_000, (n,), dtype=torch.int64)
+    vals = torch.randint(0, 100, (n,), dtype=torch.int64)
+    mode = AGG_COUNT if mode_name == "count" else AGG_SUM
+    if mode == AGG_COUNT:
+        ref = _ref_counts(keys, ts, align_ms, len_ms)
+    else:
+        ref = Counter(
+            {
+                k: v
+                for k, v in _ref_sums(
+                    keys, ts, vals, align_ms, len_ms
+                ).items()
+            }
+        )
+
+    state = WindowAggState(
+        torch.device("cuda:0"), align_ms, len_ms, mode,
+        slots_pow=18, radix=True, region_bits=11, max_batch=n,
+    )
+    state.insert(
+        RecordBatch(keys.cuda(), ts.cuda(), vals.cuda())
+    )
+    got = _extract_to_counter(state)
+    assert got == ref
+
+
 def test_window_sum_matches_reference():
     _skip_no_gpu()
     from bytewax_amd.gpu import AGG_SUM, RecordBatch, WindowAggState, _ms
