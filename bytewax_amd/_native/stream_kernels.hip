@@ -371,15 +371,19 @@ __global__ __launch_bounds__(256) void k_radix_agg(
   }
 }
 
-// Extract (and optionally clear) all slots whose window id is below
-// `win_horizon`.  Output compaction: wave ballot + one atomic per
-// wave, lanes write at their popcount rank.
+// Extract all slots whose window id is in [win_lo, win_hi) WITHOUT
+// mutating the table.  Deleting slots in place is forbidden: it breaks
+// open-addressing probe chains for keys displaced past the cleared
+// slot (a later re-insert of such a key would claim its home slot and
+// duplicate the cell).  Reclamation happens via k_close_migrate.
+// Output compaction: wave ballot + one atomic per wave, lanes write at
+// their popcount rank.
 __global__ void k_close_extract(
-    uint64_t* __restrict__ tkeys,
-    unsigned long long* __restrict__ tvals,
+    const uint64_t* __restrict__ tkeys,
+    const unsigned long long* __restrict__ tvals,
     int64_t nslots,
-    int64_t win_horizon,
-    int clear,
+    int64_t win_lo,
+    int64_t win_hi,
     int32_t* __restrict__ out_keys,
     int32_t* __restrict__ out_wins,
     int64_t* __restrict__ out_vals,
@@ -393,7 +397,7 @@ __global__ void k_close_extract(
     int32_t win = 0;
     if (k != EMPTY_SLOT) {
       win = (int32_t)(uint32_t)(k >> 32);
-      take = (int64_t)win < win_horizon;
+      take = (int64_t)win >= win_lo && (int64_t)win < win_hi;
     }
     unsigned long long ball = __ballot(take);
     int lane = threadIdx.x & (WAVE - 1);
@@ -411,9 +415,62 @@ __global__ void k_close_extract(
           out_wins[idx] = win;
           out_vals[idx] = (int64_t)tvals[i];
         }
-        if (clear) {
-          tkeys[i] = EMPTY_SLOT;
-          tvals[i] = 0;
+      }
+    }
+  }
+}
+
+// Window close with reclamation: emit cells below `win_hi` and
+// migrate the still-open cells into a fresh (pre-reset) table.  The
+// caller swaps tables afterwards and resets the old one
+// asynchronously.  Migrated cells are unique by construction so the
+// rebuild cannot create duplicate chains.
+__global__ void k_close_migrate(
+    const uint64_t* __restrict__ tkeys,
+    const unsigned long long* __restrict__ tvals,
+    int64_t nslots,
+    int64_t win_hi,
+    uint64_t* __restrict__ nkeys,
+    unsigned long long* __restrict__ nvals,
+    uint64_t mask,
+    int region_bits,
+    int32_t* __restrict__ out_keys,
+    int32_t* __restrict__ out_wins,
+    int64_t* __restrict__ out_vals,
+    int* __restrict__ out_n,
+    int64_t cap,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < nslots; i += stride) {
+    uint64_t k = tkeys[i];
+    bool occupied = (k != EMPTY_SLOT);
+    int32_t win = 0;
+    bool take = false;
+    if (occupied) {
+      win = (int32_t)(uint32_t)(k >> 32);
+      take = (int64_t)win < win_hi;
+      if (!take) {
+        if (!hash_add(nkeys, nvals, mask, region_bits, k, tvals[i])) {
+          atomicExch(error_flag, 1);
+        }
+      }
+    }
+    unsigned long long ball = __ballot(take);
+    int lane = threadIdx.x & (WAVE - 1);
+    if (ball != 0) {
+      int wave_total = __popcll(ball);
+      int rank = __popcll(ball & ((1ULL << lane) - 1ULL));
+      int base = 0;
+      int lead = __ffsll((unsigned long long)ball) - 1;
+      if (lane == lead) base = atomicAdd(out_n, wave_total);
+      base = __shfl(base, lead);
+      if (take) {
+        int64_t idx = base + rank;
+        if (idx < cap) {
+          out_keys[idx] = (int32_t)(uint32_t)(k & 0xFFFFFFFFULL);
+          out_wins[idx] = win;
+          out_vals[idx] = (int64_t)tvals[i];
         }
       }
     }
@@ -513,14 +570,14 @@ __global__ void k_stats_fixup(
 }
 
 __global__ void k_stats_extract(
-    uint64_t* __restrict__ tkeys,
-    long long* __restrict__ tcnt,
-    long long* __restrict__ tsum,
-    long long* __restrict__ tmin,
-    long long* __restrict__ tmax,
+    const uint64_t* __restrict__ tkeys,
+    const long long* __restrict__ tcnt,
+    const long long* __restrict__ tsum,
+    const long long* __restrict__ tmin,
+    const long long* __restrict__ tmax,
     int64_t nslots,
-    int64_t win_horizon,
-    int clear,
+    int64_t win_lo,
+    int64_t win_hi,
     int32_t* __restrict__ out_keys,
     int32_t* __restrict__ out_wins,
     int64_t* __restrict__ out_cnt,
@@ -531,14 +588,13 @@ __global__ void k_stats_extract(
     int64_t cap) {
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
-  const long long LLMAX = 0x7FFFFFFFFFFFFFFFLL;
   for (; i < nslots; i += stride) {
     uint64_t k = tkeys[i];
     bool take = false;
     int32_t win = 0;
     if (k != EMPTY_SLOT) {
       win = (int32_t)(uint32_t)(k >> 32);
-      take = (int64_t)win < win_horizon;
+      take = (int64_t)win >= win_lo && (int64_t)win < win_hi;
     }
     unsigned long long ball = __ballot(take);
     int lane = threadIdx.x & (WAVE - 1);
@@ -558,13 +614,6 @@ __global__ void k_stats_extract(
           out_sum[idx] = tsum[i];
           out_min[idx] = tmin[i];
           out_max[idx] = tmax[i];
-        }
-        if (clear) {
-          tkeys[i] = EMPTY_SLOT;
-          tcnt[i] = 0;
-          tsum[i] = 0;
-          tmin[i] = LLMAX;
-          tmax[i] = -LLMAX - 1;
         }
       }
     }
@@ -864,8 +913,8 @@ void radix_window_insert(
 int64_t close_extract(
     torch::Tensor tkeys,
     torch::Tensor tvals,
-    int64_t win_horizon,
-    bool clear,
+    int64_t win_lo,
+    int64_t win_hi,
     torch::Tensor out_keys,
     torch::Tensor out_wins,
     torch::Tensor out_vals,
@@ -882,12 +931,43 @@ int64_t close_extract(
   dim3 grid(n_blocks(nslots, 256));
   hipLaunchKernelGGL(
       k_close_extract, grid, block, 0, stream,
-      (uint64_t*)tkeys.data_ptr<int64_t>(),
-      (unsigned long long*)tvals.data_ptr<int64_t>(), nslots, win_horizon,
-      clear ? 1 : 0, out_keys.data_ptr<int32_t>(),
+      (const uint64_t*)tkeys.data_ptr<int64_t>(),
+      (const unsigned long long*)tvals.data_ptr<int64_t>(), nslots, win_lo,
+      win_hi, out_keys.data_ptr<int32_t>(),
       out_wins.data_ptr<int32_t>(), out_vals.data_ptr<int64_t>(),
       out_n.data_ptr<int32_t>(), out_keys.numel());
   return 0;
+}
+
+void close_migrate(
+    torch::Tensor tkeys,
+    torch::Tensor tvals,
+    torch::Tensor nkeys,
+    torch::Tensor nvals,
+    int64_t win_hi,
+    int64_t region_bits,
+    torch::Tensor out_keys,
+    torch::Tensor out_wins,
+    torch::Tensor out_vals,
+    torch::Tensor out_n,
+    torch::Tensor error_flag) {
+  check_dev(tkeys, torch::kInt64, "tkeys");
+  check_dev(nkeys, torch::kInt64, "nkeys");
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK(nkeys.numel() == nslots, "table size mismatch");
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(nslots, 256));
+  hipLaunchKernelGGL(
+      k_close_migrate, grid, block, 0, stream,
+      (const uint64_t*)tkeys.data_ptr<int64_t>(),
+      (const unsigned long long*)tvals.data_ptr<int64_t>(), nslots, win_hi,
+      (uint64_t*)nkeys.data_ptr<int64_t>(),
+      (unsigned long long*)nvals.data_ptr<int64_t>(),
+      (uint64_t)(nslots - 1), (int)region_bits,
+      out_keys.data_ptr<int32_t>(), out_wins.data_ptr<int32_t>(),
+      out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
+      out_keys.numel(), error_flag.data_ptr<int32_t>());
 }
 
 void stats_insert(
@@ -959,8 +1039,8 @@ void stats_extract(
     torch::Tensor tsum,
     torch::Tensor tmin,
     torch::Tensor tmax,
-    int64_t win_horizon,
-    bool clear,
+    int64_t win_lo,
+    int64_t win_hi,
     torch::Tensor out_keys,
     torch::Tensor out_wins,
     torch::Tensor out_cnt,
@@ -974,12 +1054,12 @@ void stats_extract(
   dim3 grid(n_blocks(nslots, 256));
   hipLaunchKernelGGL(
       k_stats_extract, grid, block, 0, stream,
-      (uint64_t*)tkeys.data_ptr<int64_t>(),
-      (long long*)tcnt.data_ptr<int64_t>(),
-      (long long*)tsum.data_ptr<int64_t>(),
-      (long long*)tmin.data_ptr<int64_t>(),
-      (long long*)tmax.data_ptr<int64_t>(), nslots, win_horizon,
-      clear ? 1 : 0, out_keys.data_ptr<int32_t>(),
+      (const uint64_t*)tkeys.data_ptr<int64_t>(),
+      (const long long*)tcnt.data_ptr<int64_t>(),
+      (const long long*)tsum.data_ptr<int64_t>(),
+      (const long long*)tmin.data_ptr<int64_t>(),
+      (const long long*)tmax.data_ptr<int64_t>(), nslots, win_lo,
+      win_hi, out_keys.data_ptr<int32_t>(),
       out_wins.data_ptr<int32_t>(), out_cnt.data_ptr<int64_t>(),
       out_sum.data_ptr<int64_t>(), out_min.data_ptr<int64_t>(),
       out_max.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
@@ -1122,13 +1202,16 @@ int64_t native_run_window_steps(
     bool dedup,
     int64_t closed_horizon_in,
     torch::Tensor step_ns_out,  // int64 CPU tensor [n_steps]
-    torch::Tensor state_out,  // int64 CPU tensor [2]: closed_horizon, rows
+    // int64 CPU tensor [3]: closed_horizon, rows, table-swap parity
+    torch::Tensor state_out,
     int64_t region_bits,
     bool use_radix,
     c10::optional<torch::Tensor> counts,
     c10::optional<torch::Tensor> cursors,
     c10::optional<torch::Tensor> ev_packed,
-    c10::optional<torch::Tensor> ev_vals) {
+    c10::optional<torch::Tensor> ev_vals,
+    torch::Tensor alt_tkeys,
+    torch::Tensor alt_tvals) {
   TORCH_CHECK(!key_pool.empty(), "empty key pool");
   int64_t nslots = tkeys.numel();
   TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
@@ -1143,6 +1226,11 @@ int64_t native_run_window_steps(
   int64_t closed_horizon = closed_horizon_in;
   int64_t total_rows = 0;
   int pool = (int)key_pool.size();
+  // Double-buffered tables: close = emit + migrate live cells into the
+  // (pre-reset) alternate table, then swap and reset the old one.
+  torch::Tensor cur_k = tkeys, cur_v = tvals;
+  torch::Tensor alt_k = alt_tkeys, alt_v = alt_tvals;
+  int swap_parity = 0;
 
   // Pinned host scalar for the close-count readback.
   int* h_n = nullptr;
@@ -1162,7 +1250,7 @@ int64_t native_run_window_steps(
       int64_t base = align_ms + step * sim_ms_per_batch;
       if (use_radix) {
         radix_window_insert(
-            keys, ts, c10::nullopt, tkeys, tvals, max_ts, error_flag,
+            keys, ts, c10::nullopt, cur_k, cur_v, max_ts, error_flag,
             *counts, *cursors, *ev_packed, *ev_vals, align_ms, len_ms,
             AGG_COUNT, base, region_bits);
       } else {
@@ -1170,8 +1258,8 @@ int64_t native_run_window_steps(
           hipLaunchKernelGGL(
               kern, grid, block, 0, stream, keys.data_ptr<int32_t>(),
               ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
-              (uint64_t*)tkeys.data_ptr<int64_t>(),
-              (unsigned long long*)tvals.data_ptr<int64_t>(),
+              (uint64_t*)cur_k.data_ptr<int64_t>(),
+              (unsigned long long*)cur_v.data_ptr<int64_t>(),
               (uint64_t)(nslots - 1), align_ms, len_ms, base,
               (int)region_bits,
               (unsigned long long*)max_ts.data_ptr<int64_t>(),
@@ -1186,14 +1274,25 @@ int64_t native_run_window_steps(
       if (horizon > closed_horizon) {
         HIP_CHECK(hipMemsetAsync(out_n.data_ptr<int32_t>(), 0, sizeof(int), stream));
         hipLaunchKernelGGL(
-            k_close_extract, dim3(n_blocks(nslots, 256)), block, 0, stream,
-            (uint64_t*)tkeys.data_ptr<int64_t>(),
-            (unsigned long long*)tvals.data_ptr<int64_t>(), nslots, horizon,
-            1, out_keys.data_ptr<int32_t>(), out_wins.data_ptr<int32_t>(),
+            k_close_migrate, dim3(n_blocks(nslots, 256)), block, 0, stream,
+            (const uint64_t*)cur_k.data_ptr<int64_t>(),
+            (const unsigned long long*)cur_v.data_ptr<int64_t>(), nslots,
+            horizon, (uint64_t*)alt_k.data_ptr<int64_t>(),
+            (unsigned long long*)alt_v.data_ptr<int64_t>(),
+            (uint64_t)(nslots - 1), (int)region_bits,
+            out_keys.data_ptr<int32_t>(), out_wins.data_ptr<int32_t>(),
             out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
-            out_keys.numel());
+            out_keys.numel(), error_flag.data_ptr<int32_t>());
         HIP_CHECK(hipMemcpyAsync(h_n, out_n.data_ptr<int32_t>(), sizeof(int),
                        hipMemcpyDeviceToHost, stream));
+        std::swap(cur_k, alt_k);
+        std::swap(cur_v, alt_v);
+        swap_parity ^= 1;
+        // Reset the retired table asynchronously for the next close.
+        HIP_CHECK(hipMemsetAsync(alt_k.data_ptr<int64_t>(), 0xFF,
+                                 (size_t)nslots * 8, stream));
+        HIP_CHECK(hipMemsetAsync(alt_v.data_ptr<int64_t>(), 0,
+                                 (size_t)nslots * 8, stream));
         HIP_CHECK(hipStreamSynchronize(stream));
         total_rows += *h_n;
         closed_horizon = horizon;
@@ -1205,6 +1304,7 @@ int64_t native_run_window_steps(
   auto* st = state_out.data_ptr<int64_t>();
   st[0] = closed_horizon;
   st[1] = total_rows;
+  st[2] = swap_parity;
   return total_rows;
 }
 
@@ -1213,6 +1313,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused window-id + hash-insert + watermark over an event batch");
   m.def("radix_window_insert", &radix_window_insert,
         "Radix-partitioned LDS-staged keyed window aggregation");
+  m.def("close_migrate", &close_migrate,
+        "Window close with reclamation: emit closed cells and migrate "
+        "live cells into a fresh table");
   m.def("close_extract", &close_extract,
         "Extract (and clear) closed windows from the keyed state table");
   m.def("stats_insert", &stats_insert,

@@ -235,6 +235,15 @@ class WindowAggState:
             (self.nslots,), -1, dtype=torch.int64, device=device
         )
         self.tvals = torch.zeros(self.nslots, dtype=torch.int64, device=device)
+        # Double-buffered: window close migrates live cells into the
+        # alternate table (in-place slot deletion would break probe
+        # chains) and swaps.
+        self.tkeys_alt = torch.full(
+            (self.nslots,), -1, dtype=torch.int64, device=device
+        )
+        self.tvals_alt = torch.zeros(
+            self.nslots, dtype=torch.int64, device=device
+        )
         self.max_ts_dev = torch.zeros(1, dtype=torch.int64, device=device)
         self.error_flag = torch.zeros(1, dtype=torch.int32, device=device)
         self.out_cap = out_cap
@@ -315,13 +324,19 @@ class WindowAggState:
                 self.max_ts_host = dev
         return self.max_ts_host
 
-    def _extract_cpu(self, horizon: int, clear: bool) -> Optional[RecordBatch]:
+    def _extract_cpu(
+        self, win_lo: int, win_hi: int, delete: bool
+    ) -> Optional[RecordBatch]:
         import torch
 
-        hit = [(k, w, v) for (k, w), v in self._table.items() if w < horizon]
+        hit = [
+            (k, w, v)
+            for (k, w), v in self._table.items()
+            if win_lo <= w < win_hi
+        ]
         if not hit:
             return None
-        if clear:
+        if delete:
             for k, w, _v in hit:
                 del self._table[(k, w)]
         keys = torch.tensor([k for k, _w, _v in hit], dtype=torch.int32)
@@ -332,25 +347,10 @@ class WindowAggState:
         vals = torch.tensor([v for _k, _w, v in hit], dtype=torch.int64)
         return RecordBatch(keys, ts, vals)
 
-    def _extract(self, horizon: int, clear: bool) -> Optional[RecordBatch]:
-        if self.cpu:
-            return self._extract_cpu(horizon, clear)
+    def _read_out(self) -> Optional[RecordBatch]:
         import torch
 
-        self.out_n.zero_()
-        self.k.close_extract(
-            self.tkeys,
-            self.tvals,
-            horizon,
-            clear,
-            self.out_keys,
-            self.out_wins,
-            self.out_vals,
-            self.out_n,
-        )
         n = int(self.out_n.item())  # syncs; amortized over window period
-        if n == 0:
-            return None
         if n > self.out_cap:
             msg = (
                 f"window close produced {n} rows > out_cap {self.out_cap}; "
@@ -360,25 +360,72 @@ class WindowAggState:
         if int(self.error_flag.item()) != 0:
             msg = "keyed window state table overflowed; increase slots_pow"
             raise RuntimeError(msg)
+        if n == 0:
+            return None
         return RecordBatch(
             self.out_keys[:n].clone(),
             self.out_wins[:n].to(torch.int64) * self.len_ms + self.align_ms,
             self.out_vals[:n].clone(),
         )
 
+    def _extract_range(self, win_lo: int, win_hi: int) -> Optional[RecordBatch]:
+        """Read cells in [win_lo, win_hi) without mutating the table."""
+        if self.cpu:
+            return self._extract_cpu(win_lo, win_hi, delete=False)
+        self.out_n.zero_()
+        self.k.close_extract(
+            self.tkeys,
+            self.tvals,
+            win_lo,
+            win_hi,
+            self.out_keys,
+            self.out_wins,
+            self.out_vals,
+            self.out_n,
+        )
+        return self._read_out()
+
     def close_due(self, wait_ms: int = 0) -> Optional[RecordBatch]:
-        """Extract all windows fully below the current watermark."""
+        """Extract all windows fully below the current watermark and
+        reclaim their space (migrate live cells to a fresh table)."""
         wm = self.watermark_ms()
         horizon = (wm - wait_ms - self.align_ms) // self.len_ms
         if horizon <= self.closed_horizon:
             return None
-        out = self._extract(horizon, clear=True)
+        if self.cpu:
+            out = self._extract_cpu(-(1 << 40), horizon, delete=True)
+            self.closed_horizon = horizon
+            return out
+        self.out_n.zero_()
+        self.k.close_migrate(
+            self.tkeys,
+            self.tvals,
+            self.tkeys_alt,
+            self.tvals_alt,
+            horizon,
+            self.region_bits,
+            self.out_keys,
+            self.out_wins,
+            self.out_vals,
+            self.out_n,
+            self.error_flag,
+        )
+        self.tkeys, self.tkeys_alt = self.tkeys_alt, self.tkeys
+        self.tvals, self.tvals_alt = self.tvals_alt, self.tvals
+        # Reset the retired table asynchronously for the next close.
+        self.tkeys_alt.fill_(-1)
+        self.tvals_alt.zero_()
+        out = self._read_out()
         self.closed_horizon = horizon
         return out
 
     def close_all(self) -> Optional[RecordBatch]:
-        """EOF: close every open window."""
-        return self._extract(1 << 40, clear=True)
+        """EOF: close every window at or above the closed horizon."""
+        if self.cpu:
+            return self._extract_cpu(self.closed_horizon, 1 << 40, delete=True)
+        out = self._extract_range(self.closed_horizon, 1 << 40)
+        self.closed_horizon = 1 << 40
+        return out
 
     def snapshot_to_host(self) -> Dict[str, Any]:
         """Spill the live table to pinned host memory (recovery)."""
@@ -399,8 +446,8 @@ class WindowAggState:
         self.k.close_extract(
             self.tkeys,
             self.tvals,
+            self.closed_horizon,
             1 << 40,
-            False,
             self.out_keys,
             self.out_wins,
             self.out_vals,
@@ -447,7 +494,7 @@ class WindowAggState:
             msg = "native_run requires a device table"
             raise RuntimeError(msg)
         step_ns = torch.zeros(n_steps, dtype=torch.int64)
-        state_out = torch.zeros(2, dtype=torch.int64)
+        state_out = torch.zeros(3, dtype=torch.int64)
         rows = self.k.native_run_window_steps(
             list(key_pool),
             list(ts_pool),
@@ -476,7 +523,12 @@ class WindowAggState:
             self.rx_cursors if self.radix else None,
             self.rx_packed if self.radix else None,
             self.rx_vals if self.radix else None,
+            self.tkeys_alt,
+            self.tvals_alt,
         )
+        if int(state_out[2].item()) == 1:
+            self.tkeys, self.tkeys_alt = self.tkeys_alt, self.tkeys
+            self.tvals, self.tvals_alt = self.tvals_alt, self.tvals
         self.closed_horizon = int(state_out[0].item())
         self.max_ts_host = max(
             self.max_ts_host,

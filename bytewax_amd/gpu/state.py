@@ -110,17 +110,26 @@ class StatsAggState:
     def extract(
         self, horizon: Optional[int] = None, clear: bool = True
     ) -> Optional[Dict[str, Any]]:
-        """Extract all (key, window) stats below `horizon` (all if
-        None).  Returns columnar host-free dict of device tensors."""
+        """Extract (key, window) stats in [closed_horizon, horizon)
+        (up to +inf if None).  On device the table is never mutated —
+        already-emitted cells are excluded by the horizon range (no
+        in-place deletion: it would corrupt probe chains); callers
+        advance `closed_horizon` after a closing extract.  Size
+        `slots_pow` for the stream's total distinct (key, window)
+        cells (1BRC-style whole-stream aggregation has exactly one
+        window)."""
         import torch
 
         if horizon is None:
             horizon = 1 << 40
+        win_lo = self.closed_horizon
+        if win_lo < -(1 << 39):
+            win_lo = -(1 << 39)
         if self.cpu:
             hit = [
                 (k, w, v)
                 for (k, w), v in self._table.items()
-                if w < horizon
+                if win_lo <= w < horizon
             ]
             if not hit:
                 return None
@@ -142,8 +151,8 @@ class StatsAggState:
             self.tsum,
             self.tmin,
             self.tmax,
+            win_lo,
             horizon,
-            clear,
             self.out_keys,
             self.out_wins,
             self.out["cnt"],
