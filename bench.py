@@ -35,8 +35,17 @@ def parse_args():
     p.add_argument("--device", type=str, default="cuda")
     p.add_argument(
         "--radix",
-        action="store_true",
-        help="use the radix-partitioned LDS-staged insert path",
+        action=argparse.BooleanOptionalAction,
+        default=True,
+        help="use the radix-partitioned LDS-staged insert path "
+        "(--no-radix for the single-pass path)",
+    )
+    p.add_argument(
+        "--latency-probes",
+        type=int,
+        default=20,
+        help="extra synced single-step runs after the timed region to "
+        "measure per-step latency (native engine)",
     )
     p.add_argument(
         "--engine",
@@ -172,14 +181,23 @@ def main():
         barrier_sync()
         t1 = time.perf_counter()
         timings["t0"], timings["t1"] = t0, t1
+        # Latency probes: single steps with a sync each, after the
+        # throughput region (launch-to-launch gaps are not a latency
+        # measure under async execution).
+        for i in range(args.latency_probes):
+            lt0 = time.perf_counter()
+            r, _ = state.native_run(
+                part.key_pool, ts_pool, W + K + i, 1, args.sim_ms_per_batch
+            )
+            closed_rows += r
+            torch.cuda.synchronize()
+            lat.append((time.perf_counter() - lt0) * 1000.0)
+        lat.sort()
         # EOF flush (outside the timed region, matching the Python
         # engine whose EOF pass runs after the final poll).
         final = state.close_all()
         if final is not None:
             closed_rows += len(final)
-        # Launch-to-launch gaps are not a latency measure under async
-        # execution; report no p99 for the native loop.
-        lat = []
     else:
         flow = Dataflow("bench_wordcount")
         s = op.input("inp", flow, BenchSource())
@@ -193,6 +211,7 @@ def main():
             out_cap=max(1 << 20, args.vocab * 2),
             device=device,
             exchange=(world > 1),
+            radix=args.radix and on_gpu,
         )
         op.output("out", agg, CollectCountsSink(out))
 

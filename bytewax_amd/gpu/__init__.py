@@ -201,9 +201,6 @@ class WindowAggState:
         self.region_bits = region_bits if radix else 0
         self.cpu = device.type == "cpu"
         if radix and not self.cpu:
-            if max_batch <= 0:
-                msg = "radix path requires max_batch"
-                raise ValueError(msg)
             n_regions = (1 << slots_pow) >> region_bits
             if n_regions < 1:
                 msg = "slots_pow must exceed region_bits"
@@ -214,14 +211,7 @@ class WindowAggState:
             self.rx_cursors = torch.zeros(
                 n_regions, dtype=torch.int32, device=device
             )
-            self.rx_packed = torch.empty(
-                max_batch, dtype=torch.int64, device=device
-            )
-            self.rx_vals = torch.empty(
-                max_batch if mode == AGG_SUM else 0,
-                dtype=torch.int64,
-                device=device,
-            )
+            self._alloc_rx(max(max_batch, 1))
         if self.cpu:
             # Host twin of the device table: used for CPU-only test
             # runs of the columnar path (gloo, no GPU).  Not a
@@ -275,10 +265,26 @@ class WindowAggState:
         if mx > self.max_ts_host:
             self.max_ts_host = mx
 
+    def _alloc_rx(self, cap: int) -> None:
+        import torch
+
+        self.rx_packed = torch.empty(
+            cap, dtype=torch.int64, device=self.device
+        )
+        self.rx_vals = torch.empty(
+            cap if self.mode == AGG_SUM else 0,
+            dtype=torch.int64,
+            device=self.device,
+        )
+
     def insert(self, batch: RecordBatch) -> None:
         if self.cpu:
             self._insert_cpu(batch)
             return
+        if self.radix and len(batch) > self.rx_packed.numel():
+            # Scatter buffers grow to fit the largest batch seen
+            # (exchange-received batches vary in size).
+            self._alloc_rx(int(len(batch) * 5 // 4))
         if self.radix:
             self.k.radix_window_insert(
                 batch.keys,

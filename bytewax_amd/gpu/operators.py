@@ -195,6 +195,8 @@ def keyed_window_agg(
     out_cap: int = 1 << 20,
     device: str = "cuda",
     exchange: Optional[bool] = None,
+    radix: bool = False,
+    region_bits: int = 11,
 ) -> Stream[RecordBatch]:
     """Keyed tumbling-window aggregation over columnar batches on GPU.
 
@@ -226,14 +228,17 @@ def keyed_window_agg(
 
     def shim_builder(resume_state):
         ex = exchange if exchange is not None else make_exchange_flag()
+        dev = torch.device(device)
         state = WindowAggState(
-            torch.device(device),
+            dev,
             align_ms,
             len_ms,
             agg_mode,
             slots_pow=slots_pow,
             dedup=dedup,
             out_cap=out_cap,
+            radix=radix and dev.type != "cpu",
+            region_bits=region_bits,
         )
         return _DeviceWindowLogic(state, wait_ms, ex, resume_state)
 
