@@ -95,19 +95,27 @@ def _bucket_cpu(batch: RecordBatch, world: int):
     return counts, send_keys, send_ts, send_vals
 
 
-def exchange_by_key(batch: RecordBatch, group=None) -> RecordBatch:
+def exchange_by_key(
+    batch: RecordBatch, group=None, async_op: bool = False
+):
     """Exchange a batch across all workers so each key lands on its
     owning worker: bucketing kernel → RCCL all-to-allv over xGMI (gloo
     when the batch lives on CPU).
 
     Collective: every rank must call this once per scheduling step.
+
+    With ``async_op`` the value collectives are issued asynchronously
+    (the split-size exchange still synchronizes) and the result is
+    ``(works, batch)`` — call ``w.wait()`` on each work handle before
+    consuming the batch.  This lets the exchange of step N overlap the
+    aggregation compute of step N-1 on the compute stream.
     """
     import torch
     import torch.distributed as dist
 
     world = dist.get_world_size(group)
     if world == 1:
-        return batch
+        return ([], batch) if async_op else batch
     if batch.ts_base != 0:
         msg = (
             "exchange_by_key requires ts_base == 0 (bases are "
@@ -148,19 +156,32 @@ def exchange_by_key(batch: RecordBatch, group=None) -> RecordBatch:
     m = int(sum(out_splits))
     recv_keys = torch.empty(m, dtype=torch.int32, device=dev)
     recv_ts = torch.empty(m, dtype=torch.int64, device=dev)
-    dist.all_to_all_single(
-        recv_keys, send_keys, out_splits, in_splits, group=group
+    works = []
+    w = dist.all_to_all_single(
+        recv_keys, send_keys, out_splits, in_splits, group=group,
+        async_op=async_op,
     )
-    dist.all_to_all_single(
-        recv_ts, send_ts, out_splits, in_splits, group=group
+    if w is not None:
+        works.append(w)
+    w = dist.all_to_all_single(
+        recv_ts, send_ts, out_splits, in_splits, group=group,
+        async_op=async_op,
     )
+    if w is not None:
+        works.append(w)
     recv_vals = None
     if has_vals:
         recv_vals = torch.empty(m, dtype=torch.int64, device=dev)
-        dist.all_to_all_single(
-            recv_vals, send_vals, out_splits, in_splits, group=group
+        w = dist.all_to_all_single(
+            recv_vals, send_vals, out_splits, in_splits, group=group,
+            async_op=async_op,
         )
-    return RecordBatch(recv_keys, recv_ts, recv_vals, max_ts=batch.max_ts)
+        if w is not None:
+            works.append(w)
+    out = RecordBatch(recv_keys, recv_ts, recv_vals, max_ts=batch.max_ts)
+    if async_op:
+        return (works, out)
+    return out
 
 
 class WindowAggState:

@@ -145,7 +145,13 @@ class SyntheticEventSource(DynamicSource):
 
 
 class _DeviceWindowLogic(StatefulBatchLogic):
-    """Holds the HBM window table; exchanges, inserts, closes."""
+    """Holds the HBM window table; exchanges, inserts, closes.
+
+    With the exchange enabled, the RCCL all-to-allv of step N is
+    issued asynchronously and overlaps the insert of step N-1's
+    received batch on the compute stream (one-step pipeline; the
+    pending batch is flushed at EOF and before snapshots).
+    """
 
     def __init__(
         self,
@@ -157,21 +163,35 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         self.state = state
         self.wait_ms = wait_ms
         self.exchange = exchange
+        self.pending: Optional[tuple] = None  # (works, RecordBatch)
         if resume is not None:
             self.state.restore_from_host(resume)
+
+    def _flush_pending(self) -> None:
+        if self.pending is None:
+            return
+        works, batch = self.pending
+        self.pending = None
+        for w in works:
+            w.wait()
+        self.state.insert(batch)
 
     def on_batch(self, batches: List[RecordBatch]):
         out = []
         for batch in batches:
             if self.exchange:
-                batch = exchange_by_key(batch)
-            self.state.insert(batch)
+                works, exchanged = exchange_by_key(batch, async_op=True)
+                self._flush_pending()
+                self.pending = (works, exchanged)
+            else:
+                self.state.insert(batch)
         closed = self.state.close_due(self.wait_ms)
         if closed is not None:
             out.append(closed)
         return (out, StatefulBatchLogic.RETAIN)
 
     def on_eof(self):
+        self._flush_pending()
         closed = self.state.close_all()
         return (
             [closed] if closed is not None else [],
@@ -179,6 +199,7 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         )
 
     def snapshot(self) -> Dict[str, Any]:
+        self._flush_pending()
         return self.state.snapshot_to_host()
 
 
