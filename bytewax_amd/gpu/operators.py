@@ -66,23 +66,35 @@ class _SyntheticPartition(StatelessSourcePartition[RecordBatch]):
         self.emitted = 0
         self.sim_ms_per_batch = sim_ms_per_batch
         self.align_ms = align_ms
-        g = torch.Generator(device="cpu").manual_seed(seed)
+        # Generate pools directly on the target device (CPU-side
+        # generation + H2D copies would dominate short runs).
+        g = torch.Generator(device=device).manual_seed(seed)
         self.key_pool = [
             torch.randint(
-                0, vocab, (events_per_batch,), dtype=torch.int32, generator=g
-            ).to(device)
+                0,
+                vocab,
+                (events_per_batch,),
+                dtype=torch.int32,
+                generator=g,
+                device=device,
+            )
             for _ in range(pool)
         ]
-        base = torch.arange(events_per_batch, dtype=torch.int64)
+        base = torch.arange(events_per_batch, dtype=torch.int64, device=device)
         self.ts_template = (
-            (base * sim_ms_per_batch) // max(events_per_batch, 1)
-        ).to(device)
+            base * sim_ms_per_batch
+        ) // max(events_per_batch, 1)
         self.val_pool = None
         if vals:
             self.val_pool = [
                 torch.randint(
-                    0, 100, (events_per_batch,), dtype=torch.int64, generator=g
-                ).to(device)
+                    0,
+                    100,
+                    (events_per_batch,),
+                    dtype=torch.int64,
+                    generator=g,
+                    device=device,
+                )
                 for _ in range(pool)
             ]
 
