@@ -118,6 +118,9 @@ __device__ __forceinline__ bool hash_add(
 // Aggregation modes baked at compile time per kernel instantiation.
 enum AggMode { AGG_COUNT = 0, AGG_SUM = 1 };
 
+__device__ __forceinline__ uint64_t find_slot(
+    uint64_t* __restrict__ tkeys, uint64_t mask, uint64_t packed);
+
 template <int MODE, bool DEDUP>
 __global__ void k_window_agg_insert(
     const int32_t* __restrict__ keys,
@@ -367,6 +370,100 @@ __global__ __launch_bounds__(256) void k_radix_agg(
       if (!hash_add(tkeys, tvals, mask, region_bits, lkeys[s], lvals[s])) {
         atomicExch(error_flag, 1);
       }
+    }
+  }
+}
+
+// Radix aggregation for the 4-accumulator stats table (count / sum /
+// min / max): one workgroup per region, LDS-resident accumulators,
+// one flush per distinct cell.  Events must already be partitioned
+// (k_radix_scatter with values).
+__global__ __launch_bounds__(256) void k_radix_agg_stats(
+    const uint64_t* __restrict__ ev_packed,
+    const int64_t* __restrict__ ev_vals,
+    const int* __restrict__ offsets,
+    const int* __restrict__ counts,
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tcnt,
+    long long* __restrict__ tsum,
+    long long* __restrict__ tmin,
+    long long* __restrict__ tmax,
+    uint64_t mask,
+    int region_bits,
+    int* __restrict__ error_flag) {
+  extern __shared__ char smem[];
+  int region = 1 << region_bits;
+  uint64_t* lkeys = (uint64_t*)smem;
+  long long* lcnt = (long long*)(smem + (size_t)region * 8);
+  long long* lsum = (long long*)(smem + (size_t)region * 16);
+  long long* lmin = (long long*)(smem + (size_t)region * 24);
+  long long* lmax = (long long*)(smem + (size_t)region * 32);
+  const long long LLMAX = 0x7FFFFFFFFFFFFFFFLL;
+  for (int s = threadIdx.x; s < region; s += blockDim.x) {
+    lkeys[s] = EMPTY_SLOT;
+    lcnt[s] = 0;
+    lsum[s] = 0;
+    lmin[s] = LLMAX;
+    lmax[s] = -LLMAX - 1;
+  }
+  __syncthreads();
+  int b = blockIdx.x;
+  int cnt = counts[b];
+  int start = offsets[b];
+  for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
+    uint64_t packed = ev_packed[start + j];
+    long long v = ev_vals[start + j];
+    uint64_t h64 = mix64(packed);
+    int lh = (int)((h64 >> 32) & (region - 1));
+    int slot = -1;
+    for (int p = 0; p < region; ++p) {
+      uint64_t cur = lkeys[lh];
+      if (cur == packed) {
+        slot = lh;
+        break;
+      }
+      if (cur == EMPTY_SLOT) {
+        uint64_t prev = atomicCAS(
+            (unsigned long long*)&lkeys[lh], EMPTY_SLOT, packed);
+        if (prev == EMPTY_SLOT || prev == packed) {
+          slot = lh;
+          break;
+        }
+      }
+      lh = (lh + 1) & (region - 1);
+    }
+    if (slot >= 0) {
+      atomicAdd((unsigned long long*)&lcnt[slot], 1ULL);
+      atomicAdd((unsigned long long*)&lsum[slot], (unsigned long long)v);
+      atomicMin(&lmin[slot], v);
+      atomicMax(&lmax[slot], v);
+    } else {
+      // LDS region full: fall through to the global region directly.
+      uint64_t gslot = find_slot(tkeys, mask, packed);
+      if (gslot == ~0ULL) {
+        atomicExch(error_flag, 1);
+      } else {
+        atomicAdd((unsigned long long*)&tcnt[gslot], 1ULL);
+        atomicAdd((unsigned long long*)&tsum[gslot], (unsigned long long)v);
+        atomicMin(&tmin[gslot], v);
+        atomicMax(&tmax[gslot], v);
+      }
+    }
+  }
+  __syncthreads();
+  for (int s = threadIdx.x; s < region; s += blockDim.x) {
+    if (lkeys[s] != EMPTY_SLOT) {
+      uint64_t gslot = find_slot(tkeys, mask, lkeys[s]);
+      if (gslot == ~0ULL) {
+        atomicExch(error_flag, 1);
+        continue;
+      }
+      atomicAdd((unsigned long long*)&tcnt[gslot],
+                (unsigned long long)lcnt[s]);
+      atomicAdd((unsigned long long*)&tsum[gslot],
+                (unsigned long long)lsum[s]);
+      atomicMin(&tmin[gslot], lmin[s]);
+      atomicMax(&tmax[gslot], lmax[s]);
     }
   }
 }
@@ -1007,6 +1104,74 @@ void stats_insert(
       error_flag.data_ptr<int32_t>());
 }
 
+void radix_stats_insert(
+    torch::Tensor keys,
+    torch::Tensor ts,
+    torch::Tensor vals,
+    torch::Tensor tkeys,
+    torch::Tensor tcnt,
+    torch::Tensor tsum,
+    torch::Tensor tmin,
+    torch::Tensor tmax,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    torch::Tensor counts,
+    torch::Tensor cursors,
+    torch::Tensor ev_packed,
+    torch::Tensor ev_vals,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    int64_t region_bits) {
+  check_dev(keys, torch::kInt32, "keys");
+  check_dev(ts, torch::kInt64, "ts");
+  check_dev(vals, torch::kInt64, "vals");
+  int64_t n = keys.numel();
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  TORCH_CHECK(region_bits > 0 && region_bits <= 10,
+              "stats radix needs 0 < region_bits <= 10 (40 B/slot LDS)");
+  int64_t nb = nslots >> region_bits;
+  TORCH_CHECK(nb >= 1 && nb <= 8192, "region count out of range");
+  TORCH_CHECK(ev_packed.numel() >= n && ev_vals.numel() >= n,
+              "scatter buffers too small");
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  uint64_t mask = (uint64_t)(nslots - 1);
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  counts.zero_();
+  size_t hist_lds = (size_t)nb * sizeof(int);
+  hipLaunchKernelGGL(
+      k_radix_hist<AGG_SUM>, grid, block, hist_lds, stream,
+      keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), n, align_ms,
+      len_ms, ts_base, mask, (int)region_bits, counts.data_ptr<int32_t>(),
+      (unsigned long long*)max_ts.data_ptr<int64_t>());
+  auto counts64 = counts.narrow(0, 0, nb);
+  auto offsets = at::cumsum(counts64, 0, at::kInt) - counts64;
+  cursors.narrow(0, 0, nb).copy_(offsets);
+  hipLaunchKernelGGL(
+      k_radix_scatter<AGG_SUM>, grid, block, 2 * hist_lds, stream,
+      keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
+      vals.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
+      (int)region_bits, cursors.data_ptr<int32_t>(),
+      (uint64_t*)ev_packed.data_ptr<int64_t>(),
+      ev_vals.data_ptr<int64_t>());
+  int region = 1 << region_bits;
+  size_t agg_lds = (size_t)region * 40;
+  auto offsets_i32 = offsets.contiguous();
+  hipLaunchKernelGGL(
+      k_radix_agg_stats, dim3((unsigned)nb), block, agg_lds, stream,
+      (const uint64_t*)ev_packed.data_ptr<int64_t>(),
+      ev_vals.data_ptr<int64_t>(), offsets_i32.data_ptr<int32_t>(),
+      counts.data_ptr<int32_t>(), (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (long long*)tcnt.data_ptr<int64_t>(),
+      (long long*)tsum.data_ptr<int64_t>(),
+      (long long*)tmin.data_ptr<int64_t>(),
+      (long long*)tmax.data_ptr<int64_t>(), mask, (int)region_bits,
+      error_flag.data_ptr<int32_t>());
+}
+
 void stats_fixup(
     torch::Tensor keys,
     torch::Tensor ts,
@@ -1320,6 +1485,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Extract (and clear) closed windows from the keyed state table");
   m.def("stats_insert", &stats_insert,
         "Keyed running count/sum/min/max over an event batch (1BRC)");
+  m.def("radix_stats_insert", &radix_stats_insert,
+        "Radix-partitioned LDS-staged keyed stats aggregation");
   m.def("stats_fixup", &stats_fixup,
         "Add count/sum deltas to existing stats slots (recovery)");
   m.def("stats_extract", &stats_extract,

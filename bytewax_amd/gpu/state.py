@@ -31,6 +31,8 @@ class StatsAggState:
         len_ms: int,
         slots_pow: int = 20,
         out_cap: int = 1 << 20,
+        radix: bool = True,
+        region_bits: int = 10,
     ):
         import torch
 
@@ -44,7 +46,23 @@ class StatsAggState:
             self._table: Dict[Tuple[int, int], Tuple[int, int, int, int]] = {}
             return
         self.k = ext()
+        # The radix path needs enough regions for parallelism; keep at
+        # least 1024 workgroups' worth.
+        if radix:
+            slots_pow = max(slots_pow, region_bits + 10)
+        self.radix = radix
+        self.region_bits = region_bits
         self.nslots = 1 << slots_pow
+        if radix:
+            n_regions = self.nslots >> region_bits
+            self.rx_counts = torch.zeros(
+                n_regions, dtype=torch.int32, device=device
+            )
+            self.rx_cursors = torch.zeros(
+                n_regions, dtype=torch.int32, device=device
+            )
+            self.rx_packed = torch.empty(1, dtype=torch.int64, device=device)
+            self.rx_vals = torch.empty(1, dtype=torch.int64, device=device)
         self.tkeys = torch.full(
             (self.nslots,), -1, dtype=torch.int64, device=device
         )
@@ -73,6 +91,37 @@ class StatsAggState:
             raise ValueError(msg)
         if self.cpu:
             self._insert_cpu(batch)
+        elif self.radix:
+            import torch
+
+            if len(batch) > self.rx_packed.numel():
+                cap = int(len(batch) * 5 // 4)
+                self.rx_packed = torch.empty(
+                    cap, dtype=torch.int64, device=self.device
+                )
+                self.rx_vals = torch.empty(
+                    cap, dtype=torch.int64, device=self.device
+                )
+            self.k.radix_stats_insert(
+                batch.keys,
+                batch.ts,
+                batch.vals,
+                self.tkeys,
+                self.tcnt,
+                self.tsum,
+                self.tmin,
+                self.tmax,
+                self.max_ts_dev,
+                self.error_flag,
+                self.rx_counts,
+                self.rx_cursors,
+                self.rx_packed,
+                self.rx_vals,
+                self.align_ms,
+                self.len_ms,
+                batch.ts_base,
+                self.region_bits,
+            )
         else:
             self.k.stats_insert(
                 batch.keys,

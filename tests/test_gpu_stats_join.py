@@ -93,14 +93,15 @@ def test_join_cpu_twin():
 
 
 @pytest.mark.gpu
-def test_stats_gpu_matches_reference():
+@pytest.mark.parametrize("radix", [False, True])
+def test_stats_gpu_matches_reference(radix):
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
     align_ms = _ms(ALIGN)
     keys, ts, vals = _mk_events(300_000, 500, 180_000, seed=3)
     ref = _ref_stats(keys, ts, vals, align_ms, 60_000)
     st = StatsAggState(
-        torch.device("cuda:0"), align_ms, 60_000, slots_pow=13
+        torch.device("cuda:0"), align_ms, 60_000, slots_pow=13, radix=radix
     )
     st.insert(RecordBatch(keys.cuda(), ts.cuda(), vals.cuda()))
     out = st.extract()
