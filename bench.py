@@ -20,7 +20,7 @@ import json
 import os
 import time
 from datetime import datetime, timedelta, timezone
-from typing import List, Optional
+from typing import List
 
 
 def parse_args():

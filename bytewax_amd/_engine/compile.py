@@ -8,7 +8,7 @@ exchange metadata for the BSP scheduler.
 """
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Tuple
 
 from ..dataflow import Dataflow, Operator, Stream
 

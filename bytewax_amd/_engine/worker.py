@@ -25,8 +25,6 @@ Epoch semantics preserved from the reference:
   no snapshot for the open epoch.
 """
 
-import copy
-import pickle
 import random
 import threading
 import time
@@ -40,7 +38,6 @@ from ..inputs import (
     AbortExecution,
     DynamicSource,
     FixedPartitionedSource,
-    StatefulSourcePartition,
 )
 from ..outputs import DynamicSink, FixedPartitionedSink
 from ..recovery import RecoveryConfig, RecoveryStore, de_state, ser_state

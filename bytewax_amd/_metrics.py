@@ -11,8 +11,6 @@ enabled by the webserver (``BYTEWAX_DATAFLOW_API_ENABLED``) or by
 calling :func:`enable_metrics`.
 """
 
-from typing import Optional
-
 __all__ = ["enable_metrics", "generate_python_metrics", "metrics_enabled"]
 
 _BUCKETS = (

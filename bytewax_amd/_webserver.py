@@ -7,7 +7,6 @@ serves the rendered graph JSON cached at startup, ``/metrics`` serves
 the prometheus exposition.
 """
 
-import json
 import os
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer

@@ -15,7 +15,7 @@ are importable and testable without a broker.
 """
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, Generic, Iterable, List, Optional, Tuple, TypeVar
+from typing import Any, Dict, Generic, List, Optional, Tuple, TypeVar
 
 from ...inputs import FixedPartitionedSource, StatefulSourcePartition
 from ...outputs import DynamicSink, StatelessSinkPartition

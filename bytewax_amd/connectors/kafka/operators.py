@@ -6,7 +6,7 @@ kafka/operators.py:49-434): ``kop.input`` splitting oks/errs,
 """
 
 from dataclasses import dataclass
-from typing import Any, Dict, List, Optional, TypeVar
+from typing import Dict, List, Optional, TypeVar
 
 import bytewax_amd.operators as op
 from ...dataflow import Dataflow, Stream, operator

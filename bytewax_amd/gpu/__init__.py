@@ -22,8 +22,8 @@ stays on the host path.
 """
 
 from dataclasses import dataclass
-from datetime import datetime, timedelta, timezone
-from typing import Any, Dict, List, Optional, Tuple
+from datetime import datetime, timezone
+from typing import Any, Dict, Optional, Tuple
 
 from ._ext import build as build_ext  # noqa: F401
 from ._ext import ext

@@ -14,10 +14,10 @@ keyed exchange is an RCCL collective.
 
 from dataclasses import dataclass
 from datetime import datetime, timedelta
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional
 
 import bytewax_amd.operators as op
-from ..dataflow import Dataflow, Stream, operator
+from ..dataflow import Stream, operator
 from ..inputs import DynamicSource, StatelessSourcePartition
 from ..outputs import DynamicSink, StatelessSinkPartition
 from ..operators import StatefulBatchLogic

@@ -37,7 +37,6 @@ from typing import (
     Set,
     Tuple,
     TypeVar,
-    Union,
 )
 
 import bytewax_amd.operators as op

@@ -13,7 +13,7 @@ import os
 import sys
 from datetime import timedelta
 from pathlib import Path
-from typing import Any, Optional
+from typing import Optional
 
 from ._engine import cluster_main, run_main
 from .dataflow import Dataflow

@@ -6,7 +6,7 @@ Mermaid diagram; ``python -m bytewax_amd.visualize`` CLI.
 """
 
 import json
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List
 
 from .dataflow import Dataflow, Operator
