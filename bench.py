@@ -28,7 +28,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--events-per-batch", type=int, default=32_000_000)
+    p.add_argument("--events-per-batch", type=int, default=64_000_000)
     p.add_argument("--vocab", type=int, default=1_000_000)
     p.add_argument("--window-sec", type=int, default=60)
     p.add_argument("--sim-ms-per-batch", type=int, default=5000)

@@ -424,6 +424,13 @@ def stream_join(
 
     Emits RecordBatches whose `ts` column holds the left value and
     `vals` the right value for each completed key pair.
+
+    Concurrency semantic: within one batch, duplicate keys on the same
+    side race — which duplicate's value lands (and whether a duplicate
+    arriving during a completion's reset survives as a fresh presence)
+    is unspecified, mirroring the reference's unspecified cross-worker
+    arrival order.  Pre-deduplicate per batch if you need the serial
+    "last" semantics exactly.
     """
     import torch
 
