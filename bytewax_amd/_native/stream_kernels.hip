@@ -810,6 +810,40 @@ __global__ void k_join_extract(
   }
 }
 
+// Stream compaction: keep events where mask != 0, preserving relative
+// order per wave (wave-ballot ranks + one atomic per wave).
+__global__ void k_filter_compact(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ ts,
+    const int64_t* __restrict__ vals,  // may be nullptr
+    const uint8_t* __restrict__ mask,
+    int64_t n,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_ts,
+    int64_t* __restrict__ out_vals,
+    int* __restrict__ out_n) {
+  int lane = threadIdx.x & (WAVE - 1);
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  int64_t first = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  for (int64_t i = first; i - lane < n; i += stride) {
+    bool keep = (i < n) && mask[i] != 0;
+    unsigned long long ball = __ballot(keep);
+    if (ball == 0) continue;
+    int wave_total = __popcll(ball);
+    int rank = __popcll(ball & ((1ULL << lane) - 1ULL));
+    int base = 0;
+    int lead = __ffsll(ball) - 1;
+    if (lane == lead) base = atomicAdd(out_n, wave_total);
+    base = __shfl(base, lead);
+    if (keep) {
+      int64_t o = base + rank;
+      out_keys[o] = keys[i];
+      out_ts[o] = ts[i];
+      if (vals != nullptr) out_vals[o] = vals[i];
+    }
+  }
+}
+
 // Histogram of destination workers for the keyed exchange.
 __global__ void k_bucket_hist(
     const int32_t* __restrict__ keys,
@@ -1292,6 +1326,35 @@ void join_extract(
       out_n.data_ptr<int32_t>(), out_keys.numel());
 }
 
+int64_t filter_compact(
+    torch::Tensor keys,
+    torch::Tensor ts,
+    c10::optional<torch::Tensor> vals,
+    torch::Tensor mask,
+    torch::Tensor out_keys,
+    torch::Tensor out_ts,
+    torch::Tensor out_vals,
+    torch::Tensor out_n) {
+  check_dev(keys, torch::kInt32, "keys");
+  check_dev(ts, torch::kInt64, "ts");
+  TORCH_CHECK(mask.scalar_type() == torch::kUInt8 ||
+                  mask.scalar_type() == torch::kBool,
+              "mask must be bool/uint8");
+  int64_t n = keys.numel();
+  if (n == 0) return 0;
+  const int64_t* vptr = nullptr;
+  if (vals.has_value()) vptr = vals->data_ptr<int64_t>();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  hipLaunchKernelGGL(
+      k_filter_compact, grid, block, 0, stream, keys.data_ptr<int32_t>(),
+      ts.data_ptr<int64_t>(), vptr, (const uint8_t*)mask.data_ptr(), n,
+      out_keys.data_ptr<int32_t>(), out_ts.data_ptr<int64_t>(),
+      out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>());
+  return 0;
+}
+
 void bucket_hist(torch::Tensor keys, int64_t world, torch::Tensor counts) {
   check_dev(keys, torch::kInt32, "keys");
   check_dev(counts, torch::kInt32, "counts");
@@ -1495,6 +1558,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Stream-stream hash join insert; emits completed pairs");
   m.def("join_extract", &join_extract,
         "Extract live join state (recovery snapshot)");
+  m.def("filter_compact", &filter_compact,
+        "Stream compaction of a RecordBatch by a boolean mask");
   m.def("bucket_hist", &bucket_hist, "Per-destination counts for exchange");
   m.def("bucket_scatter", &bucket_scatter,
         "Scatter events into per-destination segments for all-to-allv");

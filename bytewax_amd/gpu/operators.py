@@ -33,6 +33,8 @@ from . import (
 __all__ = [
     "CollectCountsSink",
     "SyntheticEventSource",
+    "filter_batch",
+    "map_batch",
     "keyed_stats_agg",
     "keyed_window_agg",
     "stream_join",
@@ -465,6 +467,83 @@ def stream_join(
     merged = op.merge("merge", l_labeled, r_labeled)
     joined = op.stateful_batch("join", merged, shim_builder)
     return op.map("unwrap", joined, lambda kv: kv[1])
+
+
+@operator
+def map_batch(
+    step_id: str,
+    up: Stream[RecordBatch],
+    mapper,
+) -> Stream[RecordBatch]:
+    """Transform RecordBatches one-to-one with device tensor ops.
+
+    ``mapper(batch) -> RecordBatch`` runs on the GPU via torch ops (or
+    custom kernels); this is the columnar twin of ``op.map``.
+    """
+    return op.flat_map_batch(
+        "flat_map_batch", up, lambda bs: [mapper(b) for b in bs]
+    )
+
+
+@operator
+def filter_batch(
+    step_id: str,
+    up: Stream[RecordBatch],
+    predicate,
+) -> Stream[RecordBatch]:
+    """Keep only events where ``predicate(batch)`` (a device bool mask
+    over the batch) is true; compaction runs as a wave-ballot HIP
+    kernel — the columnar twin of ``op.filter``.
+    """
+
+    def shim(batches):
+        import torch
+
+        from ._ext import ext
+
+        out = []
+        for b in batches:
+            m = predicate(b)
+            if b.keys.device.type == "cpu":
+                idx = m.nonzero(as_tuple=True)[0]
+                out.append(
+                    RecordBatch(
+                        b.keys[idx],
+                        b.ts[idx],
+                        b.vals[idx] if b.vals is not None else None,
+                        max_ts=b.max_ts,
+                        ts_base=b.ts_base,
+                    )
+                )
+                continue
+            k = ext()
+            n = len(b)
+            dev = b.keys.device
+            out_keys = torch.empty(n, dtype=torch.int32, device=dev)
+            out_ts = torch.empty(n, dtype=torch.int64, device=dev)
+            out_vals = torch.empty(
+                n if b.vals is not None else 0,
+                dtype=torch.int64,
+                device=dev,
+            )
+            out_n = torch.zeros(1, dtype=torch.int32, device=dev)
+            k.filter_compact(
+                b.keys, b.ts, b.vals, m.to(torch.uint8), out_keys, out_ts,
+                out_vals, out_n,
+            )
+            kept = int(out_n.item())
+            out.append(
+                RecordBatch(
+                    out_keys[:kept],
+                    out_ts[:kept],
+                    out_vals[:kept] if b.vals is not None else None,
+                    max_ts=b.max_ts,
+                    ts_base=b.ts_base,
+                )
+            )
+        return out
+
+    return op.flat_map_batch("flat_map_batch", up, shim)
 
 
 class _CollectCountsPartition(StatelessSinkPartition[RecordBatch]):

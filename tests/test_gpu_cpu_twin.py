@@ -157,3 +157,49 @@ def test_bench_two_ranks_gloo(tmp_path: Path):
     assert res["n_gpus"] == 2
     assert res["value"] > 0
     assert res["config"]["parallelism"] == "key-hash all-to-allv dp2"
+
+
+def test_filter_batch_cpu():
+    import bytewax_amd.operators as op
+    from bytewax_amd.dataflow import Dataflow
+    from bytewax_amd.gpu.operators import (
+        CollectCountsSink,
+        filter_batch,
+        map_batch,
+    )
+    from bytewax_amd.inputs import DynamicSource, StatelessSourcePartition
+    from bytewax_amd.testing import run_main
+
+    class _OneShot(StatelessSourcePartition):
+        def __init__(self):
+            self.done = False
+
+        def next_batch(self):
+            if self.done:
+                raise StopIteration()
+            self.done = True
+            return [
+                RecordBatch(
+                    torch.arange(10, dtype=torch.int32),
+                    torch.arange(10, dtype=torch.int64),
+                )
+            ]
+
+    class OneShotSource(DynamicSource):
+        def build(self, step_id, worker_index, worker_count):
+            return _OneShot()
+
+    out = []
+    flow = Dataflow("fb")
+    s = op.input("inp", flow, OneShotSource())
+    s = map_batch(
+        "scale",
+        s,
+        lambda b: RecordBatch(b.keys, b.ts * 2, b.vals, b.max_ts, b.ts_base),
+    )
+    s = filter_batch("evens", s, lambda b: b.keys % 2 == 0)
+    op.output("out", s, CollectCountsSink(out))
+    run_main(flow)
+    assert len(out) == 1
+    assert out[0].keys.tolist() == [0, 2, 4, 6, 8]
+    assert out[0].ts.tolist() == [0, 4, 8, 12, 16]

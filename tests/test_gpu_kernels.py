@@ -255,3 +255,33 @@ def test_engine_pipeline_on_gpu():
     from __graft_entry__ import smoke
 
     smoke()
+
+
+def test_filter_compact_gpu():
+    _skip_no_gpu()
+    from bytewax_amd.gpu import RecordBatch
+    from bytewax_amd.gpu.operators import filter_batch  # noqa: F401
+    from bytewax_amd.gpu import ext
+
+    torch.manual_seed(11)
+    n = 1_000_000
+    keys = torch.randint(0, 1000, (n,), dtype=torch.int32).cuda()
+    ts = torch.arange(n, dtype=torch.int64).cuda()
+    mask = (keys % 3 == 0)
+    out_keys = torch.empty(n, dtype=torch.int32, device="cuda")
+    out_ts = torch.empty(n, dtype=torch.int64, device="cuda")
+    out_vals = torch.empty(0, dtype=torch.int64, device="cuda")
+    out_n = torch.zeros(1, dtype=torch.int32, device="cuda")
+    ext().filter_compact(
+        keys, ts, None, mask.to(torch.uint8), out_keys, out_ts, out_vals,
+        out_n,
+    )
+    kept = int(out_n.item())
+    assert kept == int(mask.sum().item())
+    got = sorted(zip(out_keys[:kept].cpu().tolist(), out_ts[:kept].cpu().tolist()))
+    ref = sorted(
+        (k, t)
+        for k, t, m in zip(keys.cpu().tolist(), ts.cpu().tolist(), mask.cpu().tolist())
+        if m
+    )
+    assert got == ref
