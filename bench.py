@@ -49,7 +49,7 @@ def parse_args():
     )
     p.add_argument(
         "--engine",
-        choices=["auto", "python", "native"],
+        choices=["auto", "python", "native", "graph"],
         default="auto",
         help="'native' = C++ step loop (single-GPU); 'python' = the "
         "per-step engine loop (required for multi-GPU exchange); "
@@ -138,11 +138,14 @@ def main():
     engine = args.engine
     if engine == "auto":
         engine = "native" if (world == 1 and on_gpu) else "python"
+    if engine == "graph":
+        # hipGraph latency mode: single-pass COUNT path.
+        args.radix = False
 
     out: List = []
     closed_rows = 0
     lat: List[float] = []
-    if engine == "native":
+    if engine in ("native", "graph"):
         # C++ step loop: the host thread is a pure kernel-launch
         # engine; no Python between steps.
         from bytewax_amd.gpu import WindowAggState, _ms
@@ -167,16 +170,22 @@ def main():
             radix=args.radix,
             max_batch=E,
         )
+        def run_steps(start, count):
+            if engine == "graph":
+                return state.native_run_graph(
+                    part.key_pool, ts_pool, start, count,
+                    args.sim_ms_per_batch,
+                ), None
+            return state.native_run(
+                part.key_pool, ts_pool, start, count, args.sim_ms_per_batch
+            )
+
         # Warmup.
-        r, _ = state.native_run(
-            part.key_pool, ts_pool, 0, W, args.sim_ms_per_batch
-        )
+        r, _ = run_steps(0, W)
         closed_rows += r
         barrier_sync()
         t0 = time.perf_counter()
-        r, step_ns = state.native_run(
-            part.key_pool, ts_pool, W, K, args.sim_ms_per_batch
-        )
+        r, step_ns = run_steps(W, K)
         closed_rows += r
         barrier_sync()
         t1 = time.perf_counter()
@@ -186,9 +195,7 @@ def main():
         # measure under async execution).
         for i in range(args.latency_probes):
             lt0 = time.perf_counter()
-            r, _ = state.native_run(
-                part.key_pool, ts_pool, W + K + i, 1, args.sim_ms_per_batch
-            )
+            r, _ = run_steps(W + K + i, 1)
             closed_rows += r
             torch.cuda.synchronize()
             lat.append((time.perf_counter() - lt0) * 1000.0)

@@ -115,6 +115,10 @@ __device__ __forceinline__ bool hash_add(
   return false;
 }
 
+__global__ void k_bump(int64_t* p, int64_t v) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *p += v;
+}
+
 // Aggregation modes baked at compile time per kernel instantiation.
 enum AggMode { AGG_COUNT = 0, AGG_SUM = 1 };
 
@@ -136,7 +140,11 @@ __global__ void k_window_agg_insert(
                       // reuse one template batch across steps)
     int region_bits,  // table layout (see hash_add)
     unsigned long long* __restrict__ max_ts,  // device scalar (atomicMax)
-    int* __restrict__ error_flag) {
+    int* __restrict__ error_flag,
+    const int64_t* __restrict__ ts_base_dev) {  // extra device-side
+                      // offset, lets a captured hipGraph be replayed
+                      // with advancing timestamps (see k_bump)
+  if (ts_base_dev != nullptr) ts_base += *ts_base_dev;
   int lane = threadIdx.x & (WAVE - 1);
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   int64_t local_max = 0;
@@ -947,7 +955,7 @@ void window_agg_insert(
         (uint64_t)(nslots - 1), align_ms, len_ms, ts_base,
         (int)region_bits,
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
-        error_flag.data_ptr<int32_t>());
+        error_flag.data_ptr<int32_t>(), (const int64_t*)nullptr);
   };
   if (mode == AGG_COUNT && !dedup) launch(k_window_agg_insert<AGG_COUNT, false>);
   else if (mode == AGG_COUNT && dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
@@ -1491,7 +1499,7 @@ int64_t native_run_window_steps(
               (uint64_t)(nslots - 1), align_ms, len_ms, base,
               (int)region_bits,
               (unsigned long long*)max_ts.data_ptr<int64_t>(),
-              error_flag.data_ptr<int32_t>());
+              error_flag.data_ptr<int32_t>(), (const int64_t*)nullptr);
         };
         if (dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
         else launch(k_window_agg_insert<AGG_COUNT, false>);
@@ -1536,6 +1544,156 @@ int64_t native_run_window_steps(
   return total_rows;
 }
 
+// hipGraph variant of the native step loop: one pool cycle of
+// {insert, ts-bump} kernels is captured once and replayed per group —
+// the host cost of a whole pool of steps collapses to one
+// hipGraphLaunch.  Timestamps advance via the device-side offset
+// scalar so the captured graph stays valid across replays.  Window
+// closes happen between replays (their steps are arithmetic).
+// COUNT mode, single-pass insert only (the latency-oriented path).
+int64_t native_run_window_steps_graph(
+    std::vector<torch::Tensor> key_pool,
+    std::vector<torch::Tensor> ts_pool,
+    int64_t start_step,
+    int64_t n_steps,
+    int64_t sim_ms_per_batch,
+    torch::Tensor tkeys,
+    torch::Tensor tvals,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    torch::Tensor out_keys,
+    torch::Tensor out_wins,
+    torch::Tensor out_vals,
+    torch::Tensor out_n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t wait_ms,
+    int64_t closed_horizon_in,
+    torch::Tensor state_out,  // int64 CPU [3]
+    int64_t region_bits,
+    torch::Tensor alt_tkeys,
+    torch::Tensor alt_tvals,
+    torch::Tensor ts_base_dev) {  // int64 device scalar
+  TORCH_CHECK(!key_pool.empty(), "empty key pool");
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  int64_t n = key_pool[0].numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  int pool = (int)key_pool.size();
+  uint64_t mask = (uint64_t)(nslots - 1);
+
+  // The device offset starts at the first step's base.
+  ts_base_dev.fill_(align_ms + start_step * sim_ms_per_batch);
+
+  auto launch_insert = [&](int j) {
+    hipLaunchKernelGGL(
+        (k_window_agg_insert<AGG_COUNT, false>), grid, block, 0, stream,
+        key_pool[j].data_ptr<int32_t>(), ts_pool[j].data_ptr<int64_t>(),
+        (const int64_t*)nullptr, n, (uint64_t*)tkeys.data_ptr<int64_t>(),
+        (unsigned long long*)tvals.data_ptr<int64_t>(), mask, align_ms,
+        len_ms, 0, (int)region_bits,
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(),
+        (const int64_t*)ts_base_dev.data_ptr<int64_t>());
+    hipLaunchKernelGGL(
+        k_bump, dim3(1), dim3(1), 0, stream,
+        ts_base_dev.data_ptr<int64_t>(), sim_ms_per_batch);
+  };
+
+  // NOTE: inserts always target `tkeys`/`tvals` (the tensors captured
+  // into the graph), so closes must migrate live cells back into the
+  // SAME tensors — a double swap: cur -> alt -> cur with a reset in
+  // between keeps probe-chain hygiene while preserving buffer
+  // identity for the graph.
+  int64_t closed_horizon = closed_horizon_in;
+  int64_t total_rows = 0;
+  int* h_n = nullptr;
+  HIP_CHECK(hipHostMalloc((void**)&h_n, sizeof(int), hipHostMallocDefault));
+
+  hipGraph_t graph = nullptr;
+  hipGraphExec_t gexec = nullptr;
+  {
+    pybind11::gil_scoped_release release;
+    HIP_CHECK(hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal));
+    for (int j = 0; j < pool; ++j) launch_insert(j);
+    HIP_CHECK(hipStreamEndCapture(stream, &graph));
+    HIP_CHECK(hipGraphInstantiate(&gexec, graph, nullptr, nullptr, 0));
+
+    auto horizon_after = [&](int64_t step) {
+      int64_t wm = align_ms + (step + 1) * sim_ms_per_batch - 1;
+      return (wm - wait_ms - align_ms) / len_ms;
+    };
+    auto do_close = [&](int64_t horizon) {
+      HIP_CHECK(hipMemsetAsync(out_n.data_ptr<int32_t>(), 0, sizeof(int),
+                               stream));
+      // Migrate live cells out and back so the graph's captured table
+      // pointers stay authoritative.
+      hipLaunchKernelGGL(
+          k_close_migrate, dim3(n_blocks(nslots, 256)), block, 0, stream,
+          (const uint64_t*)tkeys.data_ptr<int64_t>(),
+          (const unsigned long long*)tvals.data_ptr<int64_t>(), nslots,
+          horizon, (uint64_t*)alt_tkeys.data_ptr<int64_t>(),
+          (unsigned long long*)alt_tvals.data_ptr<int64_t>(), mask,
+          (int)region_bits, out_keys.data_ptr<int32_t>(),
+          out_wins.data_ptr<int32_t>(), out_vals.data_ptr<int64_t>(),
+          out_n.data_ptr<int32_t>(), out_keys.numel(),
+          error_flag.data_ptr<int32_t>());
+      HIP_CHECK(hipMemcpyAsync(h_n, out_n.data_ptr<int32_t>(), sizeof(int),
+                               hipMemcpyDeviceToHost, stream));
+      // Reset the primary and migrate back (no emissions: horizon
+      // below everything live now).
+      HIP_CHECK(hipMemsetAsync(tkeys.data_ptr<int64_t>(), 0xFF,
+                               (size_t)nslots * 8, stream));
+      HIP_CHECK(hipMemsetAsync(tvals.data_ptr<int64_t>(), 0,
+                               (size_t)nslots * 8, stream));
+      hipLaunchKernelGGL(
+          k_close_migrate, dim3(n_blocks(nslots, 256)), block, 0, stream,
+          (const uint64_t*)alt_tkeys.data_ptr<int64_t>(),
+          (const unsigned long long*)alt_tvals.data_ptr<int64_t>(), nslots,
+          (int64_t)(-(1LL << 40)), (uint64_t*)tkeys.data_ptr<int64_t>(),
+          (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
+          (int)region_bits, out_keys.data_ptr<int32_t>(),
+          out_wins.data_ptr<int32_t>(), out_vals.data_ptr<int64_t>(),
+          out_n.data_ptr<int32_t>(), out_keys.numel(),
+          error_flag.data_ptr<int32_t>());
+      HIP_CHECK(hipMemsetAsync(alt_tkeys.data_ptr<int64_t>(), 0xFF,
+                               (size_t)nslots * 8, stream));
+      HIP_CHECK(hipMemsetAsync(alt_tvals.data_ptr<int64_t>(), 0,
+                               (size_t)nslots * 8, stream));
+      HIP_CHECK(hipStreamSynchronize(stream));
+      total_rows += *h_n;
+      closed_horizon = horizon;
+    };
+
+    int64_t s = 0;
+    while (s < n_steps) {
+      bool group_ok = (n_steps - s) >= pool &&
+                      horizon_after(start_step + s + pool - 1) ==
+                          closed_horizon;
+      if (group_ok) {
+        HIP_CHECK(hipGraphLaunch(gexec, stream));
+        s += pool;
+      } else {
+        launch_insert((int)((start_step + s) % pool));
+        int64_t h = horizon_after(start_step + s);
+        s += 1;
+        if (h > closed_horizon) do_close(h);
+      }
+    }
+    HIP_CHECK(hipStreamSynchronize(stream));
+  }
+  HIP_CHECK(hipGraphExecDestroy(gexec));
+  HIP_CHECK(hipGraphDestroy(graph));
+  HIP_CHECK(hipHostFree(h_n));
+  auto* st = state_out.data_ptr<int64_t>();
+  st[0] = closed_horizon;
+  st[1] = total_rows;
+  st[2] = 0;  // table identity preserved
+  return total_rows;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("window_agg_insert", &window_agg_insert,
         "Fused window-id + hash-insert + watermark over an event batch");
@@ -1563,6 +1721,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bucket_hist", &bucket_hist, "Per-destination counts for exchange");
   m.def("bucket_scatter", &bucket_scatter,
         "Scatter events into per-destination segments for all-to-allv");
+  m.def("native_run_window_steps_graph", &native_run_window_steps_graph,
+        "hipGraph-captured native step loop (latency mode: one graph "
+        "launch per pool cycle)");
   m.def("native_run_window_steps", &native_run_window_steps,
         "Run N steps of the columnar window pipeline with no Python "
         "between steps (native step loop)");

@@ -566,6 +566,65 @@ class WindowAggState:
             raise RuntimeError(msg)
         return rows, step_ns
 
+    def native_run_graph(
+        self,
+        key_pool,
+        ts_pool,
+        start_step: int,
+        n_steps: int,
+        sim_ms_per_batch: int,
+        wait_ms: int = 0,
+    ) -> int:
+        """Run `n_steps` through the hipGraph-captured native loop: one
+        pool cycle of {insert, timestamp-bump} kernels is captured once
+        and each replay costs a single hipGraphLaunch (latency mode;
+        COUNT, single-pass).  Returns closed rows."""
+        import torch
+
+        if self.cpu or self.radix or self.dedup or self.mode != AGG_COUNT:
+            msg = (
+                "graph mode supports the single-pass COUNT path on GPU"
+            )
+            raise RuntimeError(msg)
+        if not hasattr(self, "_ts_base_dev"):
+            self._ts_base_dev = torch.zeros(
+                1, dtype=torch.int64, device=self.device
+            )
+        state_out = torch.zeros(3, dtype=torch.int64)
+        rows = self.k.native_run_window_steps_graph(
+            list(key_pool),
+            list(ts_pool),
+            start_step,
+            n_steps,
+            sim_ms_per_batch,
+            self.tkeys,
+            self.tvals,
+            self.max_ts_dev,
+            self.error_flag,
+            self.out_keys,
+            self.out_wins,
+            self.out_vals,
+            self.out_n,
+            self.align_ms,
+            self.len_ms,
+            wait_ms,
+            self.closed_horizon,
+            state_out,
+            self.region_bits,
+            self.tkeys_alt,
+            self.tvals_alt,
+            self._ts_base_dev,
+        )
+        self.closed_horizon = int(state_out[0].item())
+        self.max_ts_host = max(
+            self.max_ts_host,
+            self.align_ms + (start_step + n_steps) * sim_ms_per_batch - 1,
+        )
+        if int(self.error_flag.item()) != 0:
+            msg = "keyed window state table overflowed; increase slots_pow"
+            raise RuntimeError(msg)
+        return rows
+
     def restore_from_host(self, snap: Dict[str, Any]) -> None:
         import torch
 
