@@ -1587,9 +1587,9 @@ int64_t native_run_window_steps_graph(
   // The device offset starts at the first step's base.
   ts_base_dev.fill_(align_ms + start_step * sim_ms_per_batch);
 
-  auto launch_insert = [&](int j) {
+  auto launch_insert_on = [&](int j, hipStream_t st) {
     hipLaunchKernelGGL(
-        (k_window_agg_insert<AGG_COUNT, false>), grid, block, 0, stream,
+        (k_window_agg_insert<AGG_COUNT, false>), grid, block, 0, st,
         key_pool[j].data_ptr<int32_t>(), ts_pool[j].data_ptr<int64_t>(),
         (const int64_t*)nullptr, n, (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(), mask, align_ms,
@@ -1598,9 +1598,10 @@ int64_t native_run_window_steps_graph(
         error_flag.data_ptr<int32_t>(),
         (const int64_t*)ts_base_dev.data_ptr<int64_t>());
     hipLaunchKernelGGL(
-        k_bump, dim3(1), dim3(1), 0, stream,
+        k_bump, dim3(1), dim3(1), 0, st,
         ts_base_dev.data_ptr<int64_t>(), sim_ms_per_batch);
   };
+  auto launch_insert = [&](int j) { launch_insert_on(j, stream); };
 
   // NOTE: inserts always target `tkeys`/`tvals` (the tensors captured
   // into the graph), so closes must migrate live cells back into the
@@ -1616,9 +1617,14 @@ int64_t native_run_window_steps_graph(
   hipGraphExec_t gexec = nullptr;
   {
     pybind11::gil_scoped_release release;
-    HIP_CHECK(hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal));
-    for (int j = 0; j < pool; ++j) launch_insert(j);
-    HIP_CHECK(hipStreamEndCapture(stream, &graph));
+    // Capture must happen on a non-default stream; the replays run on
+    // the torch stream.
+    hipStream_t cs = nullptr;
+    HIP_CHECK(hipStreamCreateWithFlags(&cs, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamBeginCapture(cs, hipStreamCaptureModeThreadLocal));
+    for (int j = 0; j < pool; ++j) launch_insert_on(j, cs);
+    HIP_CHECK(hipStreamEndCapture(cs, &graph));
+    HIP_CHECK(hipStreamDestroy(cs));
     HIP_CHECK(hipGraphInstantiate(&gexec, graph, nullptr, nullptr, 0));
 
     auto horizon_after = [&](int64_t step) {
