@@ -154,8 +154,24 @@ def exchange_by_key(
     in_splits = counts.tolist()
     out_splits = recv_counts.tolist()
     m = int(sum(out_splits))
+
+    # Timestamp compression: ship int32 deltas from this rank's base
+    # (a batch spans far less than 2^31 ms) — cuts exchanged bytes by
+    # a third on the biggest column.  Receivers rebuild absolute
+    # timestamps from the per-source-rank bases.
+    if batch.max_ts is not None:
+        # Host-known watermark avoids a device sync; any base within
+        # 2^31 ms of every timestamp works.
+        ts_base = batch.max_ts - (1 << 30)
+    else:
+        ts_base = int(send_ts.min().item()) if n > 0 else 0
+    send_ts32 = (send_ts - ts_base).to(torch.int32)
+    bases = torch.zeros(world, dtype=torch.int64, device=dev)
+    my_base = torch.full((world,), ts_base, dtype=torch.int64, device=dev)
+    dist.all_to_all_single(bases, my_base, group=group)
+
     recv_keys = torch.empty(m, dtype=torch.int32, device=dev)
-    recv_ts = torch.empty(m, dtype=torch.int64, device=dev)
+    recv_ts32 = torch.empty(m, dtype=torch.int32, device=dev)
     works = []
     w = dist.all_to_all_single(
         recv_keys, send_keys, out_splits, in_splits, group=group,
@@ -164,7 +180,7 @@ def exchange_by_key(
     if w is not None:
         works.append(w)
     w = dist.all_to_all_single(
-        recv_ts, send_ts, out_splits, in_splits, group=group,
+        recv_ts32, send_ts32, out_splits, in_splits, group=group,
         async_op=async_op,
     )
     if w is not None:
@@ -178,10 +194,35 @@ def exchange_by_key(
         )
         if w is not None:
             works.append(w)
-    out = RecordBatch(recv_keys, recv_ts, recv_vals, max_ts=batch.max_ts)
+    seg_bases = torch.repeat_interleave(
+        bases, torch.tensor(out_splits, dtype=torch.int64, device=dev)
+    )
+    out = _LazyTsBatch(
+        recv_keys, recv_ts32, seg_bases, recv_vals, batch.max_ts
+    )
     if async_op:
         return (works, out)
-    return out
+    return out.materialize()
+
+
+class _LazyTsBatch:
+    """Deferred timestamp reconstruction so the int64 rebuild happens
+    after the async exchange completes (on the consumer's wait)."""
+
+    def __init__(self, keys, ts32, seg_bases, vals, max_ts):
+        self.keys = keys
+        self.ts32 = ts32
+        self.seg_bases = seg_bases
+        self.vals = vals
+        self.max_ts = max_ts
+
+    def materialize(self) -> RecordBatch:
+        import torch
+
+        ts = self.ts32.to(torch.int64) + self.seg_bases
+        return RecordBatch(
+            self.keys, ts, self.vals, max_ts=self.max_ts
+        )
 
 
 class WindowAggState:

@@ -188,6 +188,8 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         self.pending = None
         for w in works:
             w.wait()
+        if hasattr(batch, "materialize"):
+            batch = batch.materialize()
         self.state.insert(batch)
 
     def on_batch(self, batches: List[RecordBatch]):
