@@ -314,6 +314,128 @@ __global__ void k_radix_scatter(
   }
 }
 
+// One-pass variant: events scatter into fixed-capacity per-region
+// buffers (capacity `cap` each, laid out at region*cap), which removes
+// the separate counting pass and the offsets scan.  A block's chunk
+// that would overflow its region's buffer goes to the overflow spill
+// (aggregated by k_radix_agg's caller via the contended-direct path —
+// statistically empty for uniform keys at cap = 2x mean).
+template <int MODE>
+__global__ void k_radix_scatter_fixed(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ ts,
+    const int64_t* __restrict__ vals,
+    int64_t n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    uint64_t mask,
+    int region_bits,
+    int64_t cap,
+    int* __restrict__ gcursors,       // [n_regions], pre-zeroed
+    uint64_t* __restrict__ ev_packed,  // [n_regions * cap]
+    int64_t* __restrict__ ev_vals,
+    int* __restrict__ ov_cursor,       // [1], pre-zeroed
+    uint64_t* __restrict__ ov_packed,  // overflow spill
+    int64_t* __restrict__ ov_vals,
+    int64_t ov_cap,
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag) {
+  extern __shared__ int lmem[];
+  int nb = (int)(((mask + 1) >> region_bits));
+  int* lhist = lmem;
+  int* lbase = lmem + nb;
+  for (int b = threadIdx.x; b < nb; b += blockDim.x) lhist[b] = 0;
+  __syncthreads();
+  int64_t start = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  int64_t local_max = 0;
+  for (int64_t i = start; i < n; i += stride) {
+    int64_t t = ts[i] + ts_base;
+    if (t > local_max) local_max = t;
+    int64_t win = (t - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    atomicAdd(&lhist[(int)region_of(mix64(packed), mask, region_bits)], 1);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < nb; b += blockDim.x) {
+    int c = lhist[b];
+    int base = -1;
+    if (c > 0) {
+      // Bounded reservation: CAS loop so successful reservations
+      // never exceed `cap` (a plain add+rollback scheme races and
+      // can orphan a later success above the rolled-back cursor).
+      int old = atomicAdd(&gcursors[b], 0);
+      while (old + c <= (int)cap) {
+        int prev = atomicCAS(&gcursors[b], old, old + c);
+        if (prev == old) {
+          base = old;
+          break;
+        }
+        old = prev;
+      }
+    }
+    lbase[b] = base;
+    lhist[b] = 0;
+  }
+  __syncthreads();
+  for (int64_t i = start; i < n; i += stride) {
+    int64_t win = (ts[i] + ts_base - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    int b = (int)region_of(mix64(packed), mask, region_bits);
+    int base = lbase[b];
+    int64_t v = (MODE == AGG_SUM) ? vals[i] : 0;
+    if (base >= 0) {
+      int64_t pos = (int64_t)b * cap + base + atomicAdd(&lhist[b], 1);
+      ev_packed[pos] = packed;
+      if (MODE == AGG_SUM) ev_vals[pos] = v;
+    } else {
+      int opos = atomicAdd(ov_cursor, 1);
+      if (opos < ov_cap) {
+        ov_packed[opos] = packed;
+        if (MODE == AGG_SUM) ov_vals[opos] = v;
+      } else {
+        atomicExch(error_flag, 1);
+      }
+    }
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
+// Aggregate the overflow spill straight into the table (tiny for
+// uniform keys).
+template <int MODE>
+__global__ void k_overflow_agg(
+    const uint64_t* __restrict__ ov_packed,
+    const int64_t* __restrict__ ov_vals,
+    const int* __restrict__ ov_cursor,
+    int64_t ov_cap,
+    uint64_t* __restrict__ tkeys,
+    unsigned long long* __restrict__ tvals,
+    uint64_t mask,
+    int region_bits,
+    int* __restrict__ error_flag) {
+  int64_t n = *ov_cursor;
+  if (n > ov_cap) n = ov_cap;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    unsigned long long inc =
+        (MODE == AGG_SUM) ? (unsigned long long)ov_vals[i] : 1ULL;
+    if (!hash_add(tkeys, tvals, mask, region_bits, ov_packed[i], inc)) {
+      atomicExch(error_flag, 1);
+    }
+  }
+}
+
 template <int MODE>
 __global__ __launch_bounds__(256) void k_radix_agg(
     const uint64_t* __restrict__ ev_packed,
@@ -971,10 +1093,12 @@ void radix_window_insert(
     torch::Tensor tvals,
     torch::Tensor max_ts,
     torch::Tensor error_flag,
-    torch::Tensor counts,     // int32 [n_regions], zeroed here
-    torch::Tensor cursors,    // int32 [n_regions]
-    torch::Tensor ev_packed,  // int64 [>= n]
-    torch::Tensor ev_vals,    // int64 [>= n] (sum mode)
+    torch::Tensor gcursors,   // int32 [n_regions]
+    torch::Tensor ev_packed,  // int64 [n_regions * cap]
+    torch::Tensor ev_vals,    // int64 [n_regions * cap] (sum mode)
+    torch::Tensor ov_cursor,  // int32 [1]
+    torch::Tensor ov_packed,  // int64 overflow spill
+    torch::Tensor ov_vals,
     int64_t align_ms,
     int64_t len_ms,
     int64_t mode,
@@ -990,12 +1114,14 @@ void radix_window_insert(
               "dynamic-LDS-per-workgroup limit");
   int64_t nb = nslots >> region_bits;
   TORCH_CHECK(nb >= 1 && nb <= 8192, "region count out of range");
-  TORCH_CHECK(counts.numel() >= nb, "counts too small");
-  TORCH_CHECK(ev_packed.numel() >= n, "ev_packed too small");
+  TORCH_CHECK(gcursors.numel() >= nb, "gcursors too small");
+  int64_t cap = ev_packed.numel() / nb;
+  TORCH_CHECK(cap * nb >= 2 * n || cap >= n,
+              "scatter buffers too small (need ~2x batch)");
   const int64_t* vptr = nullptr;
   if (mode == AGG_SUM) {
     TORCH_CHECK(vals.has_value(), "sum mode requires vals");
-    TORCH_CHECK(ev_vals.numel() >= n, "ev_vals too small");
+    TORCH_CHECK(ev_vals.numel() >= nb * cap, "ev_vals too small");
     vptr = vals->data_ptr<int64_t>();
   }
   if (n == 0) return;
@@ -1003,50 +1129,58 @@ void radix_window_insert(
   uint64_t mask = (uint64_t)(nslots - 1);
   dim3 block(256);
   dim3 grid(n_blocks(n, 256));
-  counts.zero_();
+  gcursors.narrow(0, 0, nb).zero_();
+  ov_cursor.zero_();
   size_t hist_lds = (size_t)nb * sizeof(int);
-
-  auto hist = [&](auto kern) {
-    hipLaunchKernelGGL(
-        kern, grid, block, hist_lds, stream, keys.data_ptr<int32_t>(),
-        ts.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
-        (int)region_bits, counts.data_ptr<int32_t>(),
-        (unsigned long long*)max_ts.data_ptr<int64_t>());
-  };
-  if (mode == AGG_COUNT) hist(k_radix_hist<AGG_COUNT>);
-  else hist(k_radix_hist<AGG_SUM>);
-
-  // Exclusive prefix sums (device-side, stays on the stream).
-  auto counts64 = counts.narrow(0, 0, nb);
-  auto offsets = at::cumsum(counts64, 0, at::kInt) - counts64;
-  cursors.narrow(0, 0, nb).copy_(offsets);
 
   auto scat = [&](auto kern) {
     hipLaunchKernelGGL(
         kern, grid, block, 2 * hist_lds, stream, keys.data_ptr<int32_t>(),
         ts.data_ptr<int64_t>(), vptr, n, align_ms, len_ms, ts_base, mask,
-        (int)region_bits, cursors.data_ptr<int32_t>(),
+        (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
-        mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr);
+        mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
+        ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ov_vals.data_ptr<int64_t>() : nullptr,
+        ov_packed.numel(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>());
   };
-  if (mode == AGG_COUNT) scat(k_radix_scatter<AGG_COUNT>);
-  else scat(k_radix_scatter<AGG_SUM>);
+  if (mode == AGG_COUNT) scat(k_radix_scatter_fixed<AGG_COUNT>);
+  else scat(k_radix_scatter_fixed<AGG_SUM>);
 
+  // Fixed layout: bucket b's events live at [b*cap, b*cap + count).
+  auto offsets = at::arange(
+      nb, at::TensorOptions().dtype(at::kInt).device(keys.device()));
+  offsets = offsets * (int)cap;
   int region = 1 << region_bits;
   size_t agg_lds = (size_t)region * 16;
-  auto offsets_i32 = offsets.contiguous();
   auto agg = [&](auto kern) {
     hipLaunchKernelGGL(
         kern, dim3((unsigned)nb), block, agg_lds, stream,
         (const uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
-        offsets_i32.data_ptr<int32_t>(), counts.data_ptr<int32_t>(),
+        offsets.data_ptr<int32_t>(), gcursors.data_ptr<int32_t>(),
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
         (int)region_bits, error_flag.data_ptr<int32_t>());
   };
   if (mode == AGG_COUNT) agg(k_radix_agg<AGG_COUNT>);
   else agg(k_radix_agg<AGG_SUM>);
+
+  auto ov = [&](auto kern) {
+    hipLaunchKernelGGL(
+        kern, dim3(64), block, 0, stream,
+        (const uint64_t*)ov_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ov_vals.data_ptr<int64_t>() : nullptr,
+        ov_cursor.data_ptr<int32_t>(), ov_packed.numel(),
+        (uint64_t*)tkeys.data_ptr<int64_t>(),
+        (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
+        (int)region_bits, error_flag.data_ptr<int32_t>());
+  };
+  if (mode == AGG_COUNT) ov(k_overflow_agg<AGG_COUNT>);
+  else ov(k_overflow_agg<AGG_SUM>);
 }
 
 int64_t close_extract(
@@ -1442,10 +1576,12 @@ int64_t native_run_window_steps(
     torch::Tensor state_out,
     int64_t region_bits,
     bool use_radix,
-    c10::optional<torch::Tensor> counts,
-    c10::optional<torch::Tensor> cursors,
+    c10::optional<torch::Tensor> gcursors,
     c10::optional<torch::Tensor> ev_packed,
     c10::optional<torch::Tensor> ev_vals,
+    c10::optional<torch::Tensor> ov_cursor,
+    c10::optional<torch::Tensor> ov_packed,
+    c10::optional<torch::Tensor> ov_vals,
     torch::Tensor alt_tkeys,
     torch::Tensor alt_tvals) {
   TORCH_CHECK(!key_pool.empty(), "empty key pool");
@@ -1487,8 +1623,8 @@ int64_t native_run_window_steps(
       if (use_radix) {
         radix_window_insert(
             keys, ts, c10::nullopt, cur_k, cur_v, max_ts, error_flag,
-            *counts, *cursors, *ev_packed, *ev_vals, align_ms, len_ms,
-            AGG_COUNT, base, region_bits);
+            *gcursors, *ev_packed, *ev_vals, *ov_cursor, *ov_packed,
+            *ov_vals, align_ms, len_ms, AGG_COUNT, base, region_bits);
       } else {
         auto launch = [&](auto kern) {
           hipLaunchKernelGGL(

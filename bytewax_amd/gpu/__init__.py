@@ -263,15 +263,15 @@ class WindowAggState:
         self.region_bits = region_bits if radix else 0
         self.cpu = device.type == "cpu"
         if radix and not self.cpu:
-            n_regions = (1 << slots_pow) >> region_bits
-            if n_regions < 1:
+            self.n_regions = (1 << slots_pow) >> region_bits
+            if self.n_regions < 1:
                 msg = "slots_pow must exceed region_bits"
                 raise ValueError(msg)
-            self.rx_counts = torch.zeros(
-                n_regions, dtype=torch.int32, device=device
+            self.rx_gcursors = torch.zeros(
+                self.n_regions, dtype=torch.int32, device=device
             )
-            self.rx_cursors = torch.zeros(
-                n_regions, dtype=torch.int32, device=device
+            self.rx_ov_cursor = torch.zeros(
+                1, dtype=torch.int32, device=device
             )
             self._alloc_rx(max(max_batch, 1))
         if self.cpu:
@@ -327,14 +327,28 @@ class WindowAggState:
         if mx > self.max_ts_host:
             self.max_ts_host = mx
 
-    def _alloc_rx(self, cap: int) -> None:
+    def _alloc_rx(self, max_batch: int) -> None:
+        """Fixed-capacity per-region scatter buffers (~2.5x the batch)
+        plus the overflow spill."""
         import torch
 
+        per_region = -(-max_batch * 5 // 2) // self.n_regions + 1
+        total = per_region * self.n_regions
+        self.rx_max_batch = max_batch
         self.rx_packed = torch.empty(
-            cap, dtype=torch.int64, device=self.device
+            total, dtype=torch.int64, device=self.device
         )
         self.rx_vals = torch.empty(
-            cap if self.mode == AGG_SUM else 0,
+            total if self.mode == AGG_SUM else 0,
+            dtype=torch.int64,
+            device=self.device,
+        )
+        ov = max(1 << 20, max_batch // 8)
+        self.rx_ov_packed = torch.empty(
+            ov, dtype=torch.int64, device=self.device
+        )
+        self.rx_ov_vals = torch.empty(
+            ov if self.mode == AGG_SUM else 0,
             dtype=torch.int64,
             device=self.device,
         )
@@ -343,7 +357,7 @@ class WindowAggState:
         if self.cpu:
             self._insert_cpu(batch)
             return
-        if self.radix and len(batch) > self.rx_packed.numel():
+        if self.radix and len(batch) > self.rx_max_batch:
             # Scatter buffers grow to fit the largest batch seen
             # (exchange-received batches vary in size).
             self._alloc_rx(int(len(batch) * 5 // 4))
@@ -356,10 +370,12 @@ class WindowAggState:
                 self.tvals,
                 self.max_ts_dev,
                 self.error_flag,
-                self.rx_counts,
-                self.rx_cursors,
+                self.rx_gcursors,
                 self.rx_packed,
                 self.rx_vals,
+                self.rx_ov_cursor,
+                self.rx_ov_packed,
+                self.rx_ov_vals,
                 self.align_ms,
                 self.len_ms,
                 self.mode,
@@ -587,10 +603,12 @@ class WindowAggState:
             state_out,
             self.region_bits,
             self.radix,
-            self.rx_counts if self.radix else None,
-            self.rx_cursors if self.radix else None,
+            self.rx_gcursors if self.radix else None,
             self.rx_packed if self.radix else None,
             self.rx_vals if self.radix else None,
+            self.rx_ov_cursor if self.radix else None,
+            self.rx_ov_packed if self.radix else None,
+            self.rx_ov_vals if self.radix else None,
             self.tkeys_alt,
             self.tvals_alt,
         )
