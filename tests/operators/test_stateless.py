@@ -212,3 +212,30 @@ def test_enrich_cached(entry_point):
     op.output("out", s, TestingSink(out))
     entry_point(flow)
     assert sorted(out) == [("a", 10), ("a", 10), ("b", 20)]
+
+
+def test_user_exception_propagates(entry_point):
+    out = []
+    flow = Dataflow("f")
+    s = op.input("inp", flow, TestingSource([1, 2]))
+
+    def boom(x):
+        raise ValueError("boom")
+
+    s = op.map("boom", s, boom)
+    op.output("out", s, TestingSink(out))
+    with pytest.raises(ValueError, match="boom"):
+        entry_point(flow)
+
+
+def test_mid_flow_merge_of_keyed_streams(entry_point):
+    out = []
+    flow = Dataflow("f")
+    a = op.input("a", flow, TestingSource([("k", 1)]))
+    b = op.input("b", flow, TestingSource([("k", 2)]))
+    m = op.merge("m", a, b)
+    summed = op.fold_final("sum", m, int, lambda acc, v: acc + v)
+    op.output("out", summed, TestingSink(out))
+    entry_point(flow)
+    # fold_final keys on the tuple's key; both items share "k".
+    assert out == [("k", 3)]
