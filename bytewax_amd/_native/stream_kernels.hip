@@ -360,23 +360,12 @@ __global__ void k_radix_scatter_fixed(
   }
   __syncthreads();
   for (int b = threadIdx.x; b < nb; b += blockDim.x) {
-    int c = lhist[b];
-    int base = -1;
-    if (c > 0) {
-      // Bounded reservation: CAS loop so successful reservations
-      // never exceed `cap` (a plain add+rollback scheme races and
-      // can orphan a later success above the rolled-back cursor).
-      int old = atomicAdd(&gcursors[b], 0);
-      while (old + c <= (int)cap) {
-        int prev = atomicCAS(&gcursors[b], old, old + c);
-        if (prev == old) {
-          base = old;
-          break;
-        }
-        old = prev;
-      }
-    }
-    lbase[b] = base;
+    // Unbounded reservation (one atomic per block x bucket; a bounded
+    // CAS loop here storms under 2048-way contention).  Items whose
+    // in-bucket position lands past `cap` spill to overflow; a block
+    // still fills [base, cap) contiguously so the buffer has no holes
+    // and readers clamp the cursor to `cap`.
+    lbase[b] = lhist[b] > 0 ? atomicAdd(&gcursors[b], lhist[b]) : 0;
     lhist[b] = 0;
   }
   __syncthreads();
@@ -385,10 +374,10 @@ __global__ void k_radix_scatter_fixed(
     uint64_t packed =
         ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
     int b = (int)region_of(mix64(packed), mask, region_bits);
-    int base = lbase[b];
+    int64_t in_bucket = lbase[b] + atomicAdd(&lhist[b], 1);
     int64_t v = (MODE == AGG_SUM) ? vals[i] : 0;
-    if (base >= 0) {
-      int64_t pos = (int64_t)b * cap + base + atomicAdd(&lhist[b], 1);
+    if (in_bucket < cap) {
+      int64_t pos = (int64_t)b * cap + in_bucket;
       ev_packed[pos] = packed;
       if (MODE == AGG_SUM) ev_vals[pos] = v;
     } else {
@@ -442,6 +431,7 @@ __global__ __launch_bounds__(256) void k_radix_agg(
     const int64_t* __restrict__ ev_vals,
     const int* __restrict__ offsets,
     const int* __restrict__ counts,
+    int64_t clamp_cap,  // 0 = counts are exact; >0 = clamp (fixed layout)
     uint64_t* __restrict__ tkeys,
     unsigned long long* __restrict__ tvals,
     uint64_t mask,
@@ -459,6 +449,7 @@ __global__ __launch_bounds__(256) void k_radix_agg(
   __syncthreads();
   int b = blockIdx.x;
   int cnt = counts[b];
+  if (clamp_cap > 0 && cnt > (int)clamp_cap) cnt = (int)clamp_cap;
   int start = offsets[b];
   for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
     uint64_t packed = ev_packed[start + j];
@@ -1161,7 +1152,7 @@ void radix_window_insert(
         kern, dim3((unsigned)nb), block, agg_lds, stream,
         (const uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
-        offsets.data_ptr<int32_t>(), gcursors.data_ptr<int32_t>(),
+        offsets.data_ptr<int32_t>(), gcursors.data_ptr<int32_t>(), cap,
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
         (int)region_bits, error_flag.data_ptr<int32_t>());
