@@ -30,6 +30,7 @@
 #include <ATen/hip/HIPContext.h>
 
 #include <chrono>
+#include <cstdlib>
 #include <cstdint>
 #include <vector>
 
@@ -1124,9 +1125,20 @@ void radix_window_insert(
   ov_cursor.zero_();
   size_t hist_lds = (size_t)nb * sizeof(int);
 
+  // Scatter grid: fewer blocks keep fewer segment cursors (and thus
+  // partially-written cache lines) open at once — per-XCD L2 can then
+  // accumulate full lines before eviction.  Tunable while we converge
+  // on the best default (BYTEWAX_SCATTER_BLOCKS).
+  int scatter_blocks = (int)grid.x;
+  if (const char* sb = std::getenv("BYTEWAX_SCATTER_BLOCKS")) {
+    int v = atoi(sb);
+    if (v > 0) scatter_blocks = v;
+  }
+  dim3 sgrid((unsigned)scatter_blocks);
+
   auto scat = [&](auto kern) {
     hipLaunchKernelGGL(
-        kern, grid, block, 2 * hist_lds, stream, keys.data_ptr<int32_t>(),
+        kern, sgrid, block, 2 * hist_lds, stream, keys.data_ptr<int32_t>(),
         ts.data_ptr<int64_t>(), vptr, n, align_ms, len_ms, ts_base, mask,
         (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
