@@ -342,3 +342,41 @@ def test_fold_window_eof_discards_state(recovery_config):
     out.clear()
     run_main(build(), epoch_interval=ZERO_TD, recovery_config=recovery_config)
     assert out == [("ALL", (0, ["c"]))]
+
+
+def test_fold_window_ordered_replays_by_timestamp():
+    """With ordered=True (default) values reach the fold in timestamp
+    order even when they arrive shuffled within the watermark wait."""
+    inp = [(ts(5), "c"), (ts(1), "a"), (ts(3), "b")]
+    out = []
+    flow = Dataflow("f")
+    s = op.input("inp", flow, TestingSource(inp))
+    keyed = op.key_on("k", s, lambda x: "ALL")
+    clock = ec(wait=timedelta(hours=1))
+    tw = TumblingWindower(align_to=ALIGN_TO, length=timedelta(minutes=1))
+    wo = w.fold_window(
+        "fw", keyed, clock, tw,
+        list, lambda acc, x: acc + [x[1]], lambda a, b: a + b,
+        ordered=True,
+    )
+    op.output("out", wo.down, TestingSink(out))
+    run_main(flow)
+    assert out == [("ALL", (0, ["a", "b", "c"]))]
+
+
+def test_fold_window_unordered_keeps_arrival_order():
+    inp = [(ts(5), "c"), (ts(1), "a"), (ts(3), "b")]
+    out = []
+    flow = Dataflow("f")
+    s = op.input("inp", flow, TestingSource(inp))
+    keyed = op.key_on("k", s, lambda x: "ALL")
+    clock = ec(wait=timedelta(hours=1))
+    tw = TumblingWindower(align_to=ALIGN_TO, length=timedelta(minutes=1))
+    wo = w.fold_window(
+        "fw", keyed, clock, tw,
+        list, lambda acc, x: acc + [x[1]], lambda a, b: a + b,
+        ordered=False,
+    )
+    op.output("out", wo.down, TestingSink(out))
+    run_main(flow)
+    assert out == [("ALL", (0, ["c", "a", "b"]))]
