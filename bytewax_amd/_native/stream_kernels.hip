@@ -226,95 +226,6 @@ __global__ void k_window_agg_insert(
 // atomics into ~distinct-keys L2-local ones.
 // ---------------------------------------------------------------------------
 
-template <int MODE>
-__global__ void k_radix_hist(
-    const int32_t* __restrict__ keys,
-    const int64_t* __restrict__ ts,
-    int64_t n,
-    int64_t align_ms,
-    int64_t len_ms,
-    int64_t ts_base,
-    uint64_t mask,
-    int region_bits,
-    int* __restrict__ counts,
-    unsigned long long* __restrict__ max_ts) {
-  extern __shared__ int lhist[];
-  int nb = (int)(((mask + 1) >> region_bits));
-  for (int b = threadIdx.x; b < nb; b += blockDim.x) lhist[b] = 0;
-  __syncthreads();
-  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-  int64_t stride = gridDim.x * (int64_t)blockDim.x;
-  int64_t local_max = 0;
-  for (; i < n; i += stride) {
-    int64_t t = ts[i] + ts_base;
-    if (t > local_max) local_max = t;
-    int64_t win = (t - align_ms) / len_ms;
-    uint64_t packed =
-        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
-    int b = (int)region_of(mix64(packed), mask, region_bits);
-    atomicAdd(&lhist[b], 1);
-  }
-  __syncthreads();
-  for (int b = threadIdx.x; b < nb; b += blockDim.x) {
-    if (lhist[b] > 0) atomicAdd(&counts[b], lhist[b]);
-  }
-  for (int off = WAVE / 2; off > 0; off >>= 1) {
-    int64_t other = __shfl_down((long long)local_max, off);
-    if (other > local_max) local_max = other;
-  }
-  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
-    atomicMax(max_ts, (unsigned long long)local_max);
-  }
-}
-
-template <int MODE>
-__global__ void k_radix_scatter(
-    const int32_t* __restrict__ keys,
-    const int64_t* __restrict__ ts,
-    const int64_t* __restrict__ vals,
-    int64_t n,
-    int64_t align_ms,
-    int64_t len_ms,
-    int64_t ts_base,
-    uint64_t mask,
-    int region_bits,
-    int* __restrict__ cursors,
-    uint64_t* __restrict__ ev_packed,
-    int64_t* __restrict__ ev_vals) {
-  // Two sweeps over this block's (deterministic) grid-stride slice:
-  // count per region into LDS, reserve global ranges, then write.
-  extern __shared__ int lmem[];
-  int nb = (int)(((mask + 1) >> region_bits));
-  int* lhist = lmem;
-  int* lbase = lmem + nb;
-  for (int b = threadIdx.x; b < nb; b += blockDim.x) lhist[b] = 0;
-  __syncthreads();
-  int64_t start = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-  int64_t stride = gridDim.x * (int64_t)blockDim.x;
-  for (int64_t i = start; i < n; i += stride) {
-    int64_t win = (ts[i] + ts_base - align_ms) / len_ms;
-    uint64_t packed =
-        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
-    atomicAdd(&lhist[(int)region_of(mix64(packed), mask, region_bits)], 1);
-  }
-  __syncthreads();
-  for (int b = threadIdx.x; b < nb; b += blockDim.x) {
-    int c = lhist[b];
-    lbase[b] = c > 0 ? atomicAdd(&cursors[b], c) : 0;
-    lhist[b] = 0;  // reused as the block-local write cursor
-  }
-  __syncthreads();
-  for (int64_t i = start; i < n; i += stride) {
-    int64_t win = (ts[i] + ts_base - align_ms) / len_ms;
-    uint64_t packed =
-        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
-    int b = (int)region_of(mix64(packed), mask, region_bits);
-    int pos = lbase[b] + atomicAdd(&lhist[b], 1);
-    ev_packed[pos] = packed;
-    if (MODE == AGG_SUM) ev_vals[pos] = vals[i];
-  }
-}
-
 // One-pass variant: events scatter into fixed-capacity per-region
 // buffers (capacity `cap` each, laid out at region*cap), which removes
 // the separate counting pass and the offsets scan.  A block's chunk
@@ -496,6 +407,37 @@ __global__ __launch_bounds__(256) void k_radix_agg(
   }
 }
 
+// Overflow spill for the stats table (direct contended path).
+__global__ void k_overflow_agg_stats(
+    const uint64_t* __restrict__ ov_packed,
+    const int64_t* __restrict__ ov_vals,
+    const int* __restrict__ ov_cursor,
+    int64_t ov_cap,
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tcnt,
+    long long* __restrict__ tsum,
+    long long* __restrict__ tmin,
+    long long* __restrict__ tmax,
+    uint64_t mask,
+    int* __restrict__ error_flag) {
+  int64_t n = *ov_cursor;
+  if (n > ov_cap) n = ov_cap;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t slot = find_slot(tkeys, mask, ov_packed[i]);
+    if (slot == ~0ULL) {
+      atomicExch(error_flag, 1);
+      continue;
+    }
+    long long v = ov_vals[i];
+    atomicAdd((unsigned long long*)&tcnt[slot], 1ULL);
+    atomicAdd((unsigned long long*)&tsum[slot], (unsigned long long)v);
+    atomicMin(&tmin[slot], v);
+    atomicMax(&tmax[slot], v);
+  }
+}
+
 // Radix aggregation for the 4-accumulator stats table (count / sum /
 // min / max): one workgroup per region, LDS-resident accumulators,
 // one flush per distinct cell.  Events must already be partitioned
@@ -510,6 +452,7 @@ __global__ __launch_bounds__(256) void k_radix_agg_stats(
     long long* __restrict__ tsum,
     long long* __restrict__ tmin,
     long long* __restrict__ tmax,
+    int64_t clamp_cap,
     uint64_t mask,
     int region_bits,
     int* __restrict__ error_flag) {
@@ -531,6 +474,7 @@ __global__ __launch_bounds__(256) void k_radix_agg_stats(
   __syncthreads();
   int b = blockIdx.x;
   int cnt = counts[b];
+  if (clamp_cap > 0 && cnt > (int)clamp_cap) cnt = (int)clamp_cap;
   int start = offsets[b];
   for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
     uint64_t packed = ev_packed[start + j];
@@ -1294,10 +1238,12 @@ void radix_stats_insert(
     torch::Tensor tmax,
     torch::Tensor max_ts,
     torch::Tensor error_flag,
-    torch::Tensor counts,
-    torch::Tensor cursors,
+    torch::Tensor gcursors,
     torch::Tensor ev_packed,
     torch::Tensor ev_vals,
+    torch::Tensor ov_cursor,
+    torch::Tensor ov_packed,
+    torch::Tensor ov_vals,
     int64_t align_ms,
     int64_t len_ms,
     int64_t ts_base,
@@ -1312,42 +1258,55 @@ void radix_stats_insert(
               "stats radix needs 0 < region_bits <= 10 (40 B/slot LDS)");
   int64_t nb = nslots >> region_bits;
   TORCH_CHECK(nb >= 1 && nb <= 8192, "region count out of range");
-  TORCH_CHECK(ev_packed.numel() >= n && ev_vals.numel() >= n,
-              "scatter buffers too small");
+  int64_t cap = ev_packed.numel() / nb;
+  TORCH_CHECK(cap * nb >= 2 * n || cap >= n,
+              "scatter buffers too small (need ~2x batch)");
+  TORCH_CHECK(ev_vals.numel() >= nb * cap, "ev_vals too small");
   if (n == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
   uint64_t mask = (uint64_t)(nslots - 1);
   dim3 block(256);
   dim3 grid(n_blocks(n, 256));
-  counts.zero_();
+  gcursors.narrow(0, 0, nb).zero_();
+  ov_cursor.zero_();
   size_t hist_lds = (size_t)nb * sizeof(int);
   hipLaunchKernelGGL(
-      k_radix_hist<AGG_SUM>, grid, block, hist_lds, stream,
-      keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), n, align_ms,
-      len_ms, ts_base, mask, (int)region_bits, counts.data_ptr<int32_t>(),
-      (unsigned long long*)max_ts.data_ptr<int64_t>());
-  auto counts64 = counts.narrow(0, 0, nb);
-  auto offsets = at::cumsum(counts64, 0, at::kInt) - counts64;
-  cursors.narrow(0, 0, nb).copy_(offsets);
-  hipLaunchKernelGGL(
-      k_radix_scatter<AGG_SUM>, grid, block, 2 * hist_lds, stream,
+      k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
       keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
       vals.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
-      (int)region_bits, cursors.data_ptr<int32_t>(),
-      (uint64_t*)ev_packed.data_ptr<int64_t>(),
-      ev_vals.data_ptr<int64_t>());
+      (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+      (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
+      ov_cursor.data_ptr<int32_t>(),
+      (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
+      ov_packed.numel(),
+      (unsigned long long*)max_ts.data_ptr<int64_t>(),
+      error_flag.data_ptr<int32_t>());
+
+  auto offsets = at::arange(
+      nb, at::TensorOptions().dtype(at::kInt).device(keys.device()));
+  offsets = offsets * (int)cap;
   int region = 1 << region_bits;
   size_t agg_lds = (size_t)region * 40;
-  auto offsets_i32 = offsets.contiguous();
   hipLaunchKernelGGL(
       k_radix_agg_stats, dim3((unsigned)nb), block, agg_lds, stream,
       (const uint64_t*)ev_packed.data_ptr<int64_t>(),
-      ev_vals.data_ptr<int64_t>(), offsets_i32.data_ptr<int32_t>(),
-      counts.data_ptr<int32_t>(), (uint64_t*)tkeys.data_ptr<int64_t>(),
+      ev_vals.data_ptr<int64_t>(), offsets.data_ptr<int32_t>(),
+      gcursors.data_ptr<int32_t>(), (uint64_t*)tkeys.data_ptr<int64_t>(),
       (long long*)tcnt.data_ptr<int64_t>(),
       (long long*)tsum.data_ptr<int64_t>(),
       (long long*)tmin.data_ptr<int64_t>(),
-      (long long*)tmax.data_ptr<int64_t>(), mask, (int)region_bits,
+      (long long*)tmax.data_ptr<int64_t>(), cap, mask, (int)region_bits,
+      error_flag.data_ptr<int32_t>());
+
+  hipLaunchKernelGGL(
+      k_overflow_agg_stats, dim3(64), block, 0, stream,
+      (const uint64_t*)ov_packed.data_ptr<int64_t>(),
+      ov_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
+      ov_packed.numel(), (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (long long*)tcnt.data_ptr<int64_t>(),
+      (long long*)tsum.data_ptr<int64_t>(),
+      (long long*)tmin.data_ptr<int64_t>(),
+      (long long*)tmax.data_ptr<int64_t>(), mask,
       error_flag.data_ptr<int32_t>());
 }
 

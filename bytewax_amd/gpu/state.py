@@ -54,15 +54,20 @@ class StatsAggState:
         self.region_bits = region_bits
         self.nslots = 1 << slots_pow
         if radix:
-            n_regions = self.nslots >> region_bits
-            self.rx_counts = torch.zeros(
-                n_regions, dtype=torch.int32, device=device
+            self.n_regions = self.nslots >> region_bits
+            self.rx_gcursors = torch.zeros(
+                self.n_regions, dtype=torch.int32, device=device
             )
-            self.rx_cursors = torch.zeros(
-                n_regions, dtype=torch.int32, device=device
+            self.rx_ov_cursor = torch.zeros(
+                1, dtype=torch.int32, device=device
             )
+            self.rx_max_batch = 0
             self.rx_packed = torch.empty(1, dtype=torch.int64, device=device)
             self.rx_vals = torch.empty(1, dtype=torch.int64, device=device)
+            self.rx_ov_packed = torch.empty(
+                1, dtype=torch.int64, device=device
+            )
+            self.rx_ov_vals = torch.empty(1, dtype=torch.int64, device=device)
         self.tkeys = torch.full(
             (self.nslots,), -1, dtype=torch.int64, device=device
         )
@@ -94,13 +99,23 @@ class StatsAggState:
         elif self.radix:
             import torch
 
-            if len(batch) > self.rx_packed.numel():
-                cap = int(len(batch) * 5 // 4)
+            if len(batch) > self.rx_max_batch:
+                mb = int(len(batch) * 5 // 4)
+                self.rx_max_batch = mb
+                per_region = -(-mb * 5 // 2) // self.n_regions + 1
+                total = per_region * self.n_regions
                 self.rx_packed = torch.empty(
-                    cap, dtype=torch.int64, device=self.device
+                    total, dtype=torch.int64, device=self.device
                 )
                 self.rx_vals = torch.empty(
-                    cap, dtype=torch.int64, device=self.device
+                    total, dtype=torch.int64, device=self.device
+                )
+                ov = max(1 << 20, mb // 8)
+                self.rx_ov_packed = torch.empty(
+                    ov, dtype=torch.int64, device=self.device
+                )
+                self.rx_ov_vals = torch.empty(
+                    ov, dtype=torch.int64, device=self.device
                 )
             self.k.radix_stats_insert(
                 batch.keys,
@@ -113,10 +128,12 @@ class StatsAggState:
                 self.tmax,
                 self.max_ts_dev,
                 self.error_flag,
-                self.rx_counts,
-                self.rx_cursors,
+                self.rx_gcursors,
                 self.rx_packed,
                 self.rx_vals,
+                self.rx_ov_cursor,
+                self.rx_ov_packed,
+                self.rx_ov_vals,
                 self.align_ms,
                 self.len_ms,
                 batch.ts_base,
