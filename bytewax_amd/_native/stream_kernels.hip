@@ -137,6 +137,9 @@ __global__ void k_window_agg_insert(
     uint64_t mask,
     int64_t align_ms,
     int64_t len_ms,
+    int64_t off_ms,  // window stride; == len_ms for tumbling, < len_ms
+                     // for sliding (an event lands in ceil(len/off)
+                     // windows)
     int64_t ts_base,  // added to every timestamp (columnar sources can
                       // reuse one template batch across steps)
     int region_bits,  // table layout (see hash_add)
@@ -160,9 +163,23 @@ __global__ void k_window_agg_insert(
     uint64_t packed = 0;
     unsigned long long inc = 0;
     if (valid) {
-      int64_t win = (t - align_ms) / len_ms;
+      int64_t win = (t - align_ms) / off_ms;
       packed = ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
       inc = (MODE == AGG_COUNT) ? 1ULL : (unsigned long long)vals[i];
+      if (off_ms < len_ms) {
+        // Sliding: also insert into the earlier windows this event
+        // overlaps ([win_lo, win)).  The newest window is handled by
+        // the shared code below so the DEDUP wave-aggregation path
+        // stays correct for tumbling.
+        int64_t win_lo = (t - align_ms - len_ms) / off_ms + 1;
+        for (int64_t wn = win_lo; wn < win; ++wn) {
+          uint64_t p2 =
+              ((uint64_t)(uint32_t)(int32_t)wn << 32) | (uint32_t)keys[i];
+          if (!hash_add(tkeys, tvals, mask, region_bits, p2, inc)) {
+            atomicExch(error_flag, 1);
+          }
+        }
+      }
     }
 
     bool ok = true;
@@ -981,7 +998,12 @@ void window_agg_insert(
     int64_t mode,
     bool dedup,
     int64_t ts_base,
-    int64_t region_bits) {
+    int64_t region_bits,
+    int64_t off_ms) {
+  if (off_ms <= 0) off_ms = len_ms;
+  TORCH_CHECK(off_ms <= len_ms, "window offset must be <= length");
+  TORCH_CHECK(off_ms == len_ms || !dedup,
+              "sliding windows are incompatible with the dedup path");
   check_dev(keys, torch::kInt32, "keys");
   check_dev(ts, torch::kInt64, "ts");
   check_dev(tkeys, torch::kInt64, "tkeys");
@@ -1010,7 +1032,7 @@ void window_agg_insert(
         keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), vptr, n,
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(),
-        (uint64_t)(nslots - 1), align_ms, len_ms, ts_base,
+        (uint64_t)(nslots - 1), align_ms, len_ms, off_ms, ts_base,
         (int)region_bits,
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>(), (const int64_t*)nullptr);
@@ -1594,7 +1616,7 @@ int64_t native_run_window_steps(
               ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
               (uint64_t*)cur_k.data_ptr<int64_t>(),
               (unsigned long long*)cur_v.data_ptr<int64_t>(),
-              (uint64_t)(nslots - 1), align_ms, len_ms, base,
+              (uint64_t)(nslots - 1), align_ms, len_ms, len_ms, base,
               (int)region_bits,
               (unsigned long long*)max_ts.data_ptr<int64_t>(),
               error_flag.data_ptr<int32_t>(), (const int64_t*)nullptr);
@@ -1691,7 +1713,7 @@ int64_t native_run_window_steps_graph(
         key_pool[j].data_ptr<int32_t>(), ts_pool[j].data_ptr<int64_t>(),
         (const int64_t*)nullptr, n, (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(), mask, align_ms,
-        len_ms, 0, (int)region_bits,
+        len_ms, len_ms, 0, (int)region_bits,
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>(),
         (const int64_t*)ts_base_dev.data_ptr<int64_t>());

@@ -244,17 +244,27 @@ class WindowAggState:
         radix: bool = False,
         region_bits: int = 11,
         max_batch: int = 0,
+        off_ms: int = 0,
     ):
         """:arg radix: Use the radix-partitioned LDS-staged insert path
         (fastest for high-cardinality keys; requires `max_batch`, the
         largest batch size that will be inserted).
         :arg region_bits: log2 of the table-region size when `radix`.
+        :arg off_ms: Window stride; 0 or == len_ms for tumbling,
+            < len_ms for sliding (single-pass insert path only).
         """
         import torch
 
         self.device = device
         self.align_ms = align_ms
         self.len_ms = len_ms
+        self.off_ms = off_ms if off_ms > 0 else len_ms
+        if self.off_ms > len_ms:
+            msg = "window offset must be <= length"
+            raise ValueError(msg)
+        if self.off_ms < len_ms and (radix or dedup):
+            msg = "sliding windows require the plain single-pass path"
+            raise ValueError(msg)
         self.mode = mode
         self.dedup = dedup
         self.max_ts_host = 0  # watermark if batches carry max_ts
@@ -311,7 +321,7 @@ class WindowAggState:
 
         keys = batch.keys.numpy()
         wins = (
-            (batch.ts.numpy() + batch.ts_base - self.align_ms) // self.len_ms
+            (batch.ts.numpy() + batch.ts_base - self.align_ms) // self.off_ms
         ).astype("int64")
         if self.mode == AGG_COUNT:
             vals = np.ones(len(keys), dtype="int64")
@@ -323,6 +333,16 @@ class WindowAggState:
         for p, s in zip(uniq.tolist(), sums.tolist()):
             kw = (int(np.uint32(p & 0xFFFFFFFF)), int(p >> 32))
             self._table[kw] = self._table.get(kw, 0) + s
+        if self.off_ms < self.len_ms:
+            # Sliding: also fold into the earlier overlapped windows.
+            t_arr = batch.ts.numpy() + batch.ts_base
+            lo = (t_arr - self.align_ms - self.len_ms) // self.off_ms + 1
+            for k, w_hi, w_lo, v in zip(
+                keys.tolist(), wins.tolist(), lo.tolist(), vals.tolist()
+            ):
+                for wn in range(int(w_lo), int(w_hi)):
+                    kw = (int(k), int(wn))
+                    self._table[kw] = self._table.get(kw, 0) + int(v)
         mx = int(batch.ts.max().item()) + batch.ts_base if len(batch) else 0
         if mx > self.max_ts_host:
             self.max_ts_host = mx
@@ -397,6 +417,7 @@ class WindowAggState:
                 self.dedup,
                 batch.ts_base,
                 self.region_bits,
+                self.off_ms,
             )
         if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
             self.max_ts_host = batch.max_ts
@@ -425,7 +446,7 @@ class WindowAggState:
                 del self._table[(k, w)]
         keys = torch.tensor([k for k, _w, _v in hit], dtype=torch.int32)
         ts = torch.tensor(
-            [w * self.len_ms + self.align_ms for _k, w, _v in hit],
+            [w * self.off_ms + self.align_ms for _k, w, _v in hit],
             dtype=torch.int64,
         )
         vals = torch.tensor([v for _k, _w, v in hit], dtype=torch.int64)
@@ -448,7 +469,7 @@ class WindowAggState:
             return None
         return RecordBatch(
             self.out_keys[:n].clone(),
-            self.out_wins[:n].to(torch.int64) * self.len_ms + self.align_ms,
+            self.out_wins[:n].to(torch.int64) * self.off_ms + self.align_ms,
             self.out_vals[:n].clone(),
         )
 
@@ -473,7 +494,9 @@ class WindowAggState:
         """Extract all windows fully below the current watermark and
         reclaim their space (migrate live cells to a fresh table)."""
         wm = self.watermark_ms()
-        horizon = (wm - wait_ms - self.align_ms) // self.len_ms
+        horizon = (
+            wm - wait_ms - self.align_ms - self.len_ms
+        ) // self.off_ms + 1
         if horizon <= self.closed_horizon:
             return None
         if self.cpu:
@@ -702,7 +725,9 @@ class WindowAggState:
         if n:
             keys = torch.as_tensor(snap["keys"]).to(self.device)
             wins = torch.as_tensor(snap["wins"]).to(torch.int64)
-            # Window start timestamps reproduce the same window ids.
+            # Timestamps chosen so the kernel's `(t - align) / stride`
+            # reproduces the exact window ids (stride == len disables
+            # sliding expansion during restore).
             ts = (wins * self.len_ms + self.align_ms).to(self.device)
             vals = torch.as_tensor(snap["vals"]).to(self.device)
             self.k.window_agg_insert(
@@ -719,6 +744,7 @@ class WindowAggState:
                 False,
                 0,
                 self.region_bits,
+                self.len_ms,
             )
         self.max_ts_host = snap["max_ts"]
         self.closed_horizon = snap["closed_horizon"]

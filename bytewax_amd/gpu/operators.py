@@ -234,6 +234,7 @@ def keyed_window_agg(
     exchange: Optional[bool] = None,
     radix: bool = False,
     region_bits: int = 11,
+    offset: Optional[timedelta] = None,
 ) -> Stream[RecordBatch]:
     """Keyed tumbling-window aggregation over columnar batches on GPU.
 
@@ -248,6 +249,8 @@ def keyed_window_agg(
     :arg wait: Watermark lateness allowance.
     :arg dedup: Enable wave-level duplicate aggregation (use for
         low-cardinality keys).
+    :arg offset: Window stride for sliding windows (< length puts each
+        event in multiple windows; single-pass insert path).
     :arg exchange: Force the RCCL exchange on/off; default: on iff
         torch.distributed is initialized with world > 1.
     """
@@ -256,7 +259,9 @@ def keyed_window_agg(
     agg_mode = {"count": AGG_COUNT, "sum": AGG_SUM}[mode]
     align_ms = _ms(align_to)
     len_ms = int(length.total_seconds() * 1000)
+    off_ms = int(offset.total_seconds() * 1000) if offset else len_ms
     wait_ms = int(wait.total_seconds() * 1000)
+    sliding = off_ms < len_ms
 
     def make_exchange_flag() -> bool:
         import torch.distributed as dist
@@ -272,10 +277,11 @@ def keyed_window_agg(
             len_ms,
             agg_mode,
             slots_pow=slots_pow,
-            dedup=dedup,
+            dedup=dedup and not sliding,
             out_cap=out_cap,
-            radix=radix and dev.type != "cpu",
+            radix=radix and dev.type != "cpu" and not sliding,
             region_bits=region_bits,
+            off_ms=off_ms,
         )
         return _DeviceWindowLogic(state, wait_ms, ex, resume_state)
 

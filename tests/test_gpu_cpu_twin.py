@@ -203,3 +203,24 @@ def test_filter_batch_cpu():
     assert len(out) == 1
     assert out[0].keys.tolist() == [0, 2, 4, 6, 8]
     assert out[0].ts.tolist() == [0, 4, 8, 12, 16]
+
+
+def test_cpu_sliding_window_count():
+    from bytewax_amd.gpu import AGG_COUNT
+
+    align_ms = _ms(ALIGN)
+    # length 60s, offset 30s: event at t=45s is in windows 0 and 1.
+    state = WindowAggState(
+        torch.device("cpu"), align_ms, 60_000, AGG_COUNT, off_ms=30_000
+    )
+    keys = torch.tensor([7], dtype=torch.int32)
+    ts = torch.tensor([align_ms + 45_000], dtype=torch.int64)
+    state.insert(RecordBatch(keys, ts, max_ts=align_ms + 45_000))
+    out = state.close_all()
+    got = sorted(
+        zip(out.keys.tolist(), out.ts.tolist(), out.vals.tolist())
+    )
+    assert got == [
+        (7, align_ms, 1),
+        (7, align_ms + 30_000, 1),
+    ]

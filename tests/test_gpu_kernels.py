@@ -285,3 +285,39 @@ def test_filter_compact_gpu():
         if m
     )
     assert got == ref
+
+
+def test_sliding_window_gpu_matches_reference():
+    _skip_no_gpu()
+    from bytewax_amd.gpu import AGG_COUNT, RecordBatch, WindowAggState, _ms
+
+    torch.manual_seed(13)
+    n = 100_000
+    align_ms = _ms(ALIGN)
+    len_ms, off_ms = 60_000, 20_000
+    keys = torch.randint(0, 200, (n,), dtype=torch.int32)
+    ts = align_ms + torch.randint(0, 200_000, (n,), dtype=torch.int64)
+
+    ref = Counter()
+    for k, t in zip(keys.tolist(), ts.tolist()):
+        hi = (t - align_ms) // off_ms
+        lo = (t - align_ms - len_ms) // off_ms + 1
+        for wn in range(lo, hi + 1):
+            ref[(k, wn)] += 1
+
+    state = WindowAggState(
+        torch.device("cuda:0"), align_ms, len_ms, AGG_COUNT,
+        slots_pow=14, off_ms=off_ms,
+    )
+    state.insert(RecordBatch(keys.cuda(), ts.cuda()))
+    batch = state.close_all()
+    got = Counter()
+    for k, t, v in zip(
+        batch.keys.cpu().tolist(),
+        batch.ts.cpu().tolist(),
+        batch.vals.cpu().tolist(),
+    ):
+        got[(k, (t - align_ms) // off_ms)] = v
+    # Negative-window cells (events near align with lo < 0) exist in
+    # both; compare only non-negative windows where ref is exact too.
+    assert got == ref
