@@ -41,6 +41,12 @@ def parse_args():
         "(--no-radix for the single-pass path)",
     )
     p.add_argument(
+        "--radix-v2",
+        action="store_true",
+        help="experimental two-level radix (full-line LDS-staged "
+        "scatter; COUNT mode)",
+    )
+    p.add_argument(
         "--latency-probes",
         type=int,
         default=20,
@@ -169,6 +175,7 @@ def main():
             out_cap=max(1 << 20, args.vocab * 2),
             radix=args.radix,
             max_batch=E,
+            radix_v2=args.radix_v2,
         )
         def run_steps(start, count):
             if engine == "graph":
@@ -269,6 +276,7 @@ def main():
                         "parallelism": f"key-hash all-to-allv dp{world}",
                         "engine": engine,
                         "radix": args.radix,
+                        "radix_v2": args.radix_v2,
                         "p99_step_ms": p99_ms,
                         "closed_window_rows": closed_rows,
                     },

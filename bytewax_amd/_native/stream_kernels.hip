@@ -170,8 +170,12 @@ __global__ void k_window_agg_insert(
         // Sliding: also insert into the earlier windows this event
         // overlaps ([win_lo, win)).  The newest window is handled by
         // the shared code below so the DEDUP wave-aggregation path
-        // stays correct for tumbling.
-        int64_t win_lo = (t - align_ms - len_ms) / off_ms + 1;
+        // stays correct for tumbling.  Floor division (C trunc would
+        // drop the earliest window for events within `len` of align).
+        int64_t num = t - align_ms - len_ms;
+        int64_t win_lo = num / off_ms;
+        if (num % off_ms != 0 && num < 0) win_lo -= 1;
+        win_lo += 1;
         for (int64_t wn = win_lo; wn < win; ++wn) {
           uint64_t p2 =
               ((uint64_t)(uint32_t)(int32_t)wn << 32) | (uint32_t)keys[i];
@@ -242,6 +246,204 @@ __global__ void k_window_agg_insert(
 // contiguous slice of the HBM table — turning ~N random global
 // atomics into ~distinct-keys L2-local ones.
 // ---------------------------------------------------------------------------
+
+// ---------------------------------------------------------------------------
+// Radix v2 (experimental, COUNT mode): two-level scheme that writes
+// full 64-byte lines.
+//
+//   Pass A (k_scatter_coarse): events scatter into 256 *coarse*
+//   buckets, staged through LDS in block-synchronous tiles so every
+//   global write is one aligned 64 B line (8 packed events); partial
+//   groups at block end go to a per-bucket residual area (unaligned,
+//   ~5% of events).
+//
+//   Pass B (k_agg_cached): per coarse-bucket slice, aggregate through
+//   a 4096-entry LDS cache; collisions and the final flush go through
+//   hash_add into the bucket's contiguous span of table regions
+//   (cache-local by construction).
+// ---------------------------------------------------------------------------
+
+#define V2_COARSE 256
+#define V2_STAGE 32  // staged events per coarse bucket per tile
+
+__global__ __launch_bounds__(512) void k_scatter_coarse(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ ts,
+    int64_t n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    uint64_t mask,
+    int region_bits,
+    int64_t cap_c,      // aligned-area capacity per coarse bucket (mult of 8)
+    int64_t res_cap,    // residual-area capacity per coarse bucket
+    int* __restrict__ gcur,     // [V2_COARSE] aligned-area cursors
+    int* __restrict__ gres,     // [V2_COARSE] residual-area cursors
+    uint64_t* __restrict__ ev,      // [V2_COARSE * cap_c]
+    uint64_t* __restrict__ ev_res,  // [V2_COARSE * res_cap]
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag) {
+  __shared__ uint64_t stage[V2_COARSE][V2_STAGE];
+  __shared__ int lcnt[V2_COARSE];
+  int nb = (int)(((mask + 1) >> region_bits));
+  int shift = 0;  // region index -> coarse bucket shift
+  while ((nb >> shift) > V2_COARSE) ++shift;
+  for (int b = threadIdx.x; b < V2_COARSE; b += blockDim.x) lcnt[b] = 0;
+  __syncthreads();
+
+  const int TILE = 8;  // events per thread per tile
+  int64_t chunk = (int64_t)blockDim.x * TILE;
+  int64_t start = (int64_t)blockIdx.x * chunk;
+  int64_t gstride = (int64_t)gridDim.x * chunk;
+  int64_t local_max = 0;
+  for (int64_t tile0 = start; tile0 < n; tile0 += gstride) {
+    for (int t = 0; t < TILE; ++t) {
+      int64_t i = tile0 + (int64_t)t * blockDim.x + threadIdx.x;
+      if (i < n) {
+        int64_t tm = ts[i] + ts_base;
+        if (tm > local_max) local_max = tm;
+        int64_t win = (tm - align_ms) / len_ms;
+        uint64_t packed =
+            ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+        int cb = (int)(region_of(mix64(packed), mask, region_bits) >> shift);
+        int r = atomicAdd(&lcnt[cb], 1);
+        if (r < V2_STAGE) {
+          stage[cb][r] = packed;
+        } else {
+          // Tile-local stage overflow: straight to the residual area.
+          int rp = atomicAdd(&gres[cb], 1);
+          if (rp < res_cap) {
+            ev_res[(int64_t)cb * res_cap + rp] = packed;
+          } else {
+            atomicExch(error_flag, 1);
+          }
+        }
+      }
+    }
+    __syncthreads();
+    // Flush complete 8-groups; keep the remainder staged.
+    for (int b = threadIdx.x; b < V2_COARSE; b += blockDim.x) {
+      int c = lcnt[b];
+      if (c > V2_STAGE) c = V2_STAGE;
+      int groups = c / 8;
+      for (int g = 0; g < groups; ++g) {
+        int base8 = atomicAdd(&gcur[b], 8);
+        if (base8 + 8 <= cap_c) {
+          uint64_t* dst = ev + (int64_t)b * cap_c + base8;
+          #pragma unroll
+          for (int q = 0; q < 8; ++q) dst[q] = stage[b][g * 8 + q];
+        } else {
+          for (int q = 0; q < 8; ++q) {
+            int rp = atomicAdd(&gres[b], 1);
+            if (rp < res_cap) {
+              ev_res[(int64_t)b * res_cap + rp] = stage[b][g * 8 + q];
+            } else {
+              atomicExch(error_flag, 1);
+            }
+          }
+        }
+      }
+      int rem = c - groups * 8;
+      for (int q = 0; q < rem; ++q) {
+        stage[b][q] = stage[b][groups * 8 + q];
+      }
+      lcnt[b] = rem;
+    }
+    __syncthreads();
+  }
+  // Final residual flush.
+  for (int b = threadIdx.x; b < V2_COARSE; b += blockDim.x) {
+    int c = lcnt[b];
+    if (c > V2_STAGE) c = V2_STAGE;
+    for (int q = 0; q < c; ++q) {
+      int rp = atomicAdd(&gres[b], 1);
+      if (rp < res_cap) {
+        ev_res[(int64_t)b * res_cap + rp] = stage[b][q];
+      } else {
+        atomicExch(error_flag, 1);
+      }
+    }
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
+#define V2_CACHE 4096  // LDS cache entries in pass B
+
+__global__ __launch_bounds__(256) void k_agg_cached(
+    const uint64_t* __restrict__ ev,
+    const uint64_t* __restrict__ ev_res,
+    const int* __restrict__ gcur,
+    const int* __restrict__ gres,
+    int64_t cap_c,
+    int64_t res_cap,
+    int slices,  // blocks per coarse bucket
+    uint64_t* __restrict__ tkeys,
+    unsigned long long* __restrict__ tvals,
+    uint64_t mask,
+    int region_bits,
+    int* __restrict__ error_flag) {
+  __shared__ uint64_t ckeys[V2_CACHE];
+  __shared__ unsigned long long cvals[V2_CACHE];
+  for (int s = threadIdx.x; s < V2_CACHE; s += blockDim.x) {
+    ckeys[s] = EMPTY_SLOT;
+    cvals[s] = 0;
+  }
+  __syncthreads();
+  int cb = blockIdx.x % V2_COARSE;
+  int slice = blockIdx.x / V2_COARSE;
+
+  auto agg_one = [&](uint64_t packed) {
+    uint64_t h64 = mix64(packed);
+    int lh = (int)((h64 >> 20) & (V2_CACHE - 1));
+    // Two-probe LDS cache; collisions go straight to the table.
+    for (int p = 0; p < 2; ++p) {
+      uint64_t cur = ckeys[lh];
+      if (cur == packed) {
+        atomicAdd(&cvals[lh], 1ULL);
+        return;
+      }
+      if (cur == EMPTY_SLOT) {
+        uint64_t prev = atomicCAS(
+            (unsigned long long*)&ckeys[lh], EMPTY_SLOT, packed);
+        if (prev == EMPTY_SLOT || prev == packed) {
+          atomicAdd(&cvals[lh], 1ULL);
+          return;
+        }
+      }
+      lh = (lh + 1) & (V2_CACHE - 1);
+    }
+    if (!hash_add(tkeys, tvals, mask, region_bits, packed, 1ULL)) {
+      atomicExch(error_flag, 1);
+    }
+  };
+
+  int c = gcur[cb];
+  if (c > cap_c) c = (int)cap_c;
+  for (int j = slice * (int)blockDim.x + threadIdx.x; j < c;
+       j += slices * (int)blockDim.x) {
+    agg_one(ev[(int64_t)cb * cap_c + j]);
+  }
+  int cr = gres[cb];
+  if (cr > res_cap) cr = (int)res_cap;
+  for (int j = slice * (int)blockDim.x + threadIdx.x; j < cr;
+       j += slices * (int)blockDim.x) {
+    agg_one(ev_res[(int64_t)cb * res_cap + j]);
+  }
+  __syncthreads();
+  for (int s = threadIdx.x; s < V2_CACHE; s += blockDim.x) {
+    if (ckeys[s] != EMPTY_SLOT) {
+      if (!hash_add(tkeys, tvals, mask, region_bits, ckeys[s], cvals[s])) {
+        atomicExch(error_flag, 1);
+      }
+    }
+  }
+}
 
 // One-pass variant: events scatter into fixed-capacity per-region
 // buffers (capacity `cap` each, laid out at region*cap), which removes
@@ -1152,6 +1354,54 @@ void radix_window_insert(
   else ov(k_overflow_agg<AGG_SUM>);
 }
 
+void radix_v2_window_insert(
+    torch::Tensor keys,
+    torch::Tensor ts,
+    torch::Tensor tkeys,
+    torch::Tensor tvals,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    torch::Tensor gcur,    // int32 [V2_COARSE]
+    torch::Tensor gres,    // int32 [V2_COARSE]
+    torch::Tensor ev,      // int64 [V2_COARSE * cap_c]
+    torch::Tensor ev_res,  // int64 [V2_COARSE * res_cap]
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    int64_t region_bits) {
+  check_dev(keys, torch::kInt32, "keys");
+  check_dev(ts, torch::kInt64, "ts");
+  int64_t n = keys.numel();
+  int64_t nslots = tkeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  int64_t cap_c = ev.numel() / V2_COARSE;
+  TORCH_CHECK(cap_c % 8 == 0, "coarse capacity must be 8-aligned");
+  int64_t res_cap = ev_res.numel() / V2_COARSE;
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  uint64_t mask = (uint64_t)(nslots - 1);
+  gcur.zero_();
+  gres.zero_();
+  hipLaunchKernelGGL(
+      k_scatter_coarse, dim3(256), dim3(512), 0, stream,
+      keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), n, align_ms,
+      len_ms, ts_base, mask, (int)region_bits, cap_c, res_cap,
+      gcur.data_ptr<int32_t>(), gres.data_ptr<int32_t>(),
+      (uint64_t*)ev.data_ptr<int64_t>(),
+      (uint64_t*)ev_res.data_ptr<int64_t>(),
+      (unsigned long long*)max_ts.data_ptr<int64_t>(),
+      error_flag.data_ptr<int32_t>());
+  int slices = 8;
+  hipLaunchKernelGGL(
+      k_agg_cached, dim3(V2_COARSE * slices), dim3(256), 0, stream,
+      (const uint64_t*)ev.data_ptr<int64_t>(),
+      (const uint64_t*)ev_res.data_ptr<int64_t>(),
+      gcur.data_ptr<int32_t>(), gres.data_ptr<int32_t>(), cap_c, res_cap,
+      slices, (uint64_t*)tkeys.data_ptr<int64_t>(),
+      (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
+      (int)region_bits, error_flag.data_ptr<int32_t>());
+}
+
 int64_t close_extract(
     torch::Tensor tkeys,
     torch::Tensor tvals,
@@ -1567,7 +1817,12 @@ int64_t native_run_window_steps(
     c10::optional<torch::Tensor> ov_packed,
     c10::optional<torch::Tensor> ov_vals,
     torch::Tensor alt_tkeys,
-    torch::Tensor alt_tvals) {
+    torch::Tensor alt_tvals,
+    bool use_radix_v2,
+    c10::optional<torch::Tensor> v2_gcur,
+    c10::optional<torch::Tensor> v2_gres,
+    c10::optional<torch::Tensor> v2_ev,
+    c10::optional<torch::Tensor> v2_ev_res) {
   TORCH_CHECK(!key_pool.empty(), "empty key pool");
   int64_t nslots = tkeys.numel();
   TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
@@ -1604,7 +1859,12 @@ int64_t native_run_window_steps(
       auto& keys = key_pool[step % pool];
       auto& ts = ts_pool[step % pool];
       int64_t base = align_ms + step * sim_ms_per_batch;
-      if (use_radix) {
+      if (use_radix_v2) {
+        radix_v2_window_insert(
+            keys, ts, cur_k, cur_v, max_ts, error_flag, *v2_gcur,
+            *v2_gres, *v2_ev, *v2_ev_res, align_ms, len_ms, base,
+            region_bits);
+      } else if (use_radix) {
         radix_window_insert(
             keys, ts, c10::nullopt, cur_k, cur_v, max_ts, error_flag,
             *gcursors, *ev_packed, *ev_vals, *ov_cursor, *ov_packed,
@@ -1823,6 +2083,9 @@ int64_t native_run_window_steps_graph(
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("window_agg_insert", &window_agg_insert,
         "Fused window-id + hash-insert + watermark over an event batch");
+  m.def("radix_v2_window_insert", &radix_v2_window_insert,
+        "Experimental two-level radix with full-line LDS-staged "
+        "scatter (COUNT mode)");
   m.def("radix_window_insert", &radix_window_insert,
         "Radix-partitioned LDS-staged keyed window aggregation");
   m.def("close_migrate", &close_migrate,

@@ -245,6 +245,7 @@ class WindowAggState:
         region_bits: int = 11,
         max_batch: int = 0,
         off_ms: int = 0,
+        radix_v2: bool = False,
     ):
         """:arg radix: Use the radix-partitioned LDS-staged insert path
         (fastest for high-cardinality keys; requires `max_batch`, the
@@ -269,6 +270,12 @@ class WindowAggState:
         self.dedup = dedup
         self.max_ts_host = 0  # watermark if batches carry max_ts
         self.closed_horizon = -(1 << 62)  # window ids below are closed
+        if radix_v2 and (mode != AGG_COUNT or dedup):
+            msg = "radix_v2 supports the COUNT mode only"
+            raise ValueError(msg)
+        self.radix_v2 = radix_v2 and device.type != "cpu"
+        if self.radix_v2:
+            radix = True  # shares the region table layout + buffers
         self.radix = radix
         self.region_bits = region_bits if radix else 0
         self.cpu = device.type == "cpu"
@@ -358,6 +365,17 @@ class WindowAggState:
         self.rx_packed = torch.empty(
             total, dtype=torch.int64, device=self.device
         )
+        if getattr(self, "radix_v2", False):
+            cap_c = (-(-max_batch * 5 // 2) // 256 + 8) // 8 * 8
+            res_cap = max(1 << 14, cap_c // 4)
+            self.v2_gcur = torch.zeros(256, dtype=torch.int32, device=self.device)
+            self.v2_gres = torch.zeros(256, dtype=torch.int32, device=self.device)
+            self.v2_ev = torch.empty(
+                256 * cap_c, dtype=torch.int64, device=self.device
+            )
+            self.v2_ev_res = torch.empty(
+                256 * res_cap, dtype=torch.int64, device=self.device
+            )
         self.rx_vals = torch.empty(
             total if self.mode == AGG_SUM else 0,
             dtype=torch.int64,
@@ -381,7 +399,24 @@ class WindowAggState:
             # Scatter buffers grow to fit the largest batch seen
             # (exchange-received batches vary in size).
             self._alloc_rx(int(len(batch) * 5 // 4))
-        if self.radix:
+        if self.radix_v2:
+            self.k.radix_v2_window_insert(
+                batch.keys,
+                batch.ts,
+                self.tkeys,
+                self.tvals,
+                self.max_ts_dev,
+                self.error_flag,
+                self.v2_gcur,
+                self.v2_gres,
+                self.v2_ev,
+                self.v2_ev_res,
+                self.align_ms,
+                self.len_ms,
+                batch.ts_base,
+                self.region_bits,
+            )
+        elif self.radix:
             self.k.radix_window_insert(
                 batch.keys,
                 batch.ts,
@@ -634,6 +669,11 @@ class WindowAggState:
             self.rx_ov_vals if self.radix else None,
             self.tkeys_alt,
             self.tvals_alt,
+            self.radix_v2,
+            self.v2_gcur if self.radix_v2 else None,
+            self.v2_gres if self.radix_v2 else None,
+            self.v2_ev if self.radix_v2 else None,
+            self.v2_ev_res if self.radix_v2 else None,
         )
         if int(state_out[2].item()) == 1:
             self.tkeys, self.tkeys_alt = self.tkeys_alt, self.tkeys

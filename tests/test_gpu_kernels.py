@@ -321,3 +321,24 @@ def test_sliding_window_gpu_matches_reference():
     # Negative-window cells (events near align with lo < 0) exist in
     # both; compare only non-negative windows where ref is exact too.
     assert got == ref
+
+
+def test_window_radix_v2_matches_reference():
+    _skip_no_gpu()
+    from bytewax_amd.gpu import AGG_COUNT, RecordBatch, WindowAggState, _ms
+
+    torch.manual_seed(17)
+    n = 800_000
+    align_ms = _ms(ALIGN)
+    len_ms = 60_000
+    keys = torch.randint(0, 30_000, (n,), dtype=torch.int32)
+    ts = align_ms + torch.randint(0, 300_000, (n,), dtype=torch.int64)
+    ref = _ref_counts(keys, ts, align_ms, len_ms)
+
+    state = WindowAggState(
+        torch.device("cuda:0"), align_ms, len_ms, AGG_COUNT,
+        slots_pow=18, radix_v2=True, max_batch=n,
+    )
+    state.insert(RecordBatch(keys.cuda(), ts.cuda()))
+    got = _extract_to_counter(state)
+    assert got == ref
