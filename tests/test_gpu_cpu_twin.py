@@ -224,3 +224,58 @@ def test_cpu_sliding_window_count():
         (7, align_ms, 1),
         (7, align_ms + 30_000, 1),
     ]
+
+
+def test_keyed_stats_agg_pipeline_cpu():
+    import bytewax_amd.operators as op
+    from bytewax_amd.dataflow import Dataflow
+    from bytewax_amd.gpu.operators import (
+        SyntheticEventSource,
+        keyed_stats_agg,
+    )
+    from bytewax_amd.outputs import DynamicSink, StatelessSinkPartition
+    from bytewax_amd.testing import run_main
+
+    class _Collect(StatelessSinkPartition):
+        def __init__(self, ls):
+            self._ls = ls
+
+        def write_batch(self, items):
+            self._ls.extend(items)
+
+    class Collect(DynamicSink):
+        def __init__(self, ls):
+            self._ls = ls
+
+        def build(self, step_id, worker_index, worker_count):
+            return _Collect(self._ls)
+
+    out = []
+    flow = Dataflow("stats_twin")
+    s = op.input(
+        "inp",
+        flow,
+        SyntheticEventSource(
+            events_per_batch=5_000,
+            n_batches=4,
+            vocab=50,
+            align_to=ALIGN,
+            sim_ms_per_batch=1000,
+            device="cpu",
+            with_vals=True,
+        ),
+    )
+    stats = keyed_stats_agg(
+        "stats",
+        s,
+        align_to=ALIGN,
+        length=timedelta(days=365),
+        device="cpu",
+    )
+    op.output("out", stats, Collect(out))
+    run_main(flow)
+    total = sum(int(o["cnt"].sum().item()) for o in out)
+    assert total == 20_000
+    for o in out:
+        assert bool((o["min"] <= o["max"]).all())
+        assert bool((o["sum"] >= o["min"] * o["cnt"]).all())
