@@ -449,7 +449,14 @@ def filter(  # noqa: A001
     up: Stream[X],
     predicate: Callable[[X], bool],
 ) -> Stream[X]:
-    """Keep only some items."""
+    """Keep only the items where `predicate` returns `True`.
+
+    ```python
+    s = op.filter("odds", s, lambda x: x % 2 == 1)
+    ```
+
+    The predicate must return a `bool` (not merely truthy).
+    """
 
     def shim_mapper(x: X) -> Iterable[X]:
         keep = predicate(x)
@@ -556,7 +563,13 @@ def map(  # noqa: A001
     up: Stream[X],
     mapper: Callable[[X], Y],
 ) -> Stream[Y]:
-    """Transform items one-by-one."""
+    """Transform items one-by-one.
+
+    ```python
+    s = op.input("inp", flow, TestingSource([1, 2, 3]))
+    s = op.map("add_one", s, lambda x: x + 1)  # 2, 3, 4
+    ```
+    """
 
     def shim_mapper(xs: List[X]) -> Iterable[Y]:
         return [mapper(x) for x in xs]
@@ -588,7 +601,15 @@ def map_value(
 
 @operator
 def key_on(step_id: str, up: Stream[X], key: Callable[[X], str]) -> KeyedStream[X]:
-    """Add a key for each item, making a {py:obj}`KeyedStream`."""
+    """Add a key for each item, making a {py:obj}`KeyedStream`.
+
+    ```python
+    keyed = op.key_on("k", s, lambda x: x["user"])  # -> (key, item)
+    ```
+
+    The key function must return a `str` (keys route state across
+    workers).
+    """
 
     def shim_mapper(x: X) -> Tuple[str, X]:
         k = key(x)
@@ -732,7 +753,12 @@ def fold_final(
 ) -> KeyedStream[S]:
     """Build an empty accumulator, then combine values into it.
 
-    Only returns results once the upstream is EOF.
+    Only returns results once the upstream is EOF:
+
+    ```python
+    folded = op.fold_final("fold", keyed, lambda: 0, lambda acc, v: acc + v)
+    # ("key1", 1), ("key1", 2), ("key2", 3) -> ("key1", 3), ("key2", 3)
+    ```
     """
 
     def shim_builder(resume_state: Optional[S]) -> _FoldFinalLogic[V, S]:
@@ -866,6 +892,15 @@ def stateful_map(
     mapper: Callable[[Optional[S], V], Tuple[Optional[S], W]],
 ) -> KeyedStream[W]:
     """Transform values one-to-one, referencing a persistent state.
+
+    ```python
+    def running_sum(state, v):
+        state = (state or 0) + v
+        return (state, state)
+
+    sums = op.stateful_map("sum", keyed, running_sum)
+    # ("a", 1), ("a", 2) -> ("a", 1), ("a", 3)
+    ```
 
     :arg mapper: Called whenever a value is encountered from upstream
         with the last state or `None`, and then the value.  Should
