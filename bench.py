@@ -41,6 +41,12 @@ def parse_args():
         "(--no-radix for the single-pass path)",
     )
     p.add_argument(
+        "--dedup",
+        action="store_true",
+        help="wave-level duplicate aggregation (for low-cardinality "
+        "keys; implies --no-radix)",
+    )
+    p.add_argument(
         "--radix-v2",
         action="store_true",
         help="experimental two-level radix (full-line LDS-staged "
@@ -147,6 +153,8 @@ def main():
     if engine == "graph":
         # hipGraph latency mode: single-pass COUNT path.
         args.radix = False
+    if args.dedup:
+        args.radix = False
 
     out: List = []
     closed_rows = 0
@@ -176,6 +184,7 @@ def main():
             radix=args.radix,
             max_batch=E,
             radix_v2=args.radix_v2,
+            dedup=args.dedup,
         )
         def run_steps(start, count):
             if engine == "graph":
