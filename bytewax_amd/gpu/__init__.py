@@ -381,7 +381,11 @@ class WindowAggState:
             dtype=torch.int64,
             device=self.device,
         )
-        ov = max(1 << 20, max_batch // 8)
+        # Sized for the degenerate case where most events hash into a
+        # few regions (tiny key cardinality): the whole batch may
+        # spill and aggregate through the direct path — slow but
+        # correct.  Use `dedup=True` for such workloads instead.
+        ov = max(1 << 20, max_batch)
         self.rx_ov_packed = torch.empty(
             ov, dtype=torch.int64, device=self.device
         )

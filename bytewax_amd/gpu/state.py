@@ -110,7 +110,7 @@ class StatsAggState:
                 self.rx_vals = torch.empty(
                     total, dtype=torch.int64, device=self.device
                 )
-                ov = max(1 << 20, mb // 8)
+                ov = max(1 << 20, mb)
                 self.rx_ov_packed = torch.empty(
                     ov, dtype=torch.int64, device=self.device
                 )
