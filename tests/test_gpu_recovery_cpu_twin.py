@@ -96,7 +96,7 @@ def _counts(out, align_ms):
     return got
 
 
-def _run(recovery_config, source):
+def _run(recovery_config, source, device="cpu"):
     from bytewax_amd.gpu import _ms
 
     align_ms = _ms(ALIGN)
@@ -108,7 +108,7 @@ def _run(recovery_config, source):
         s,
         align_to=ALIGN,
         length=timedelta(minutes=1),
-        device="cpu",
+        device=device,
         exchange=False,
     )
     op.output("out", agg, CollectCountsSink(out))
@@ -152,3 +152,25 @@ def test_window_uninterrupted_baseline(tmp_path):
     align_ms = _ms(ALIGN)
     got = _run(RecoveryConfig(db), ScriptedSource(_mk_batches(align_ms)))
     assert got == EXPECTED
+
+
+@pytest.mark.gpu
+def test_window_state_exactly_once_across_abort_gpu(recovery_config):
+    """Same exactly-once contract with the real device table: the
+    snapshot is the pinned-host spill of HBM state, written into the
+    SQLite store and restored on resume."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from bytewax_amd.gpu import _ms
+
+    align_ms = _ms(ALIGN)
+    batches = [
+        RecordBatch(b.keys.cuda(), b.ts.cuda(), max_ts=b.max_ts)
+        for b in _mk_batches(align_ms)
+    ]
+    got1 = _run(
+        recovery_config, ScriptedSource(batches, abort_at=2), device="cuda"
+    )
+    assert got1 == {}
+    got2 = _run(recovery_config, ScriptedSource(batches), device="cuda")
+    assert got2 == EXPECTED
