@@ -40,6 +40,7 @@ def parse_args():
         help="use the radix-partitioned LDS-staged insert path "
         "(--no-radix for the single-pass path)",
     )
+    p.add_argument("--region-bits", type=int, default=11)
     p.add_argument(
         "--dedup",
         action="store_true",
@@ -185,6 +186,7 @@ def main():
             max_batch=E,
             radix_v2=args.radix_v2,
             dedup=args.dedup,
+            region_bits=args.region_bits,
         )
         def run_steps(start, count):
             if engine == "graph":

@@ -1270,8 +1270,8 @@ void radix_window_insert(
   int64_t nslots = tkeys.numel();
   TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
   TORCH_CHECK(region_bits > 0, "radix path requires region_bits > 0");
-  TORCH_CHECK(region_bits <= 11, "region_bits > 11 exceeds the 64 KiB "
-              "dynamic-LDS-per-workgroup limit");
+  TORCH_CHECK(region_bits <= 12, "region_bits > 12 exceeds the LDS "
+              "budget of the aggregation kernel (16 B/slot)");
   int64_t nb = nslots >> region_bits;
   TORCH_CHECK(nb >= 1 && nb <= 8192, "region count out of range");
   TORCH_CHECK(gcursors.numel() >= nb, "gcursors too small");
