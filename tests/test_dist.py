@@ -75,3 +75,85 @@ def test_two_process_cluster_keyed_state(tmp_path: Path):
     # 4 keys x 5 items each, running sums 1..5 per key.
     expected = sorted(f"{k}={v}" for k in "0123" for v in range(1, 6))
     assert lines == expected
+
+
+@pytest.mark.timeout(240)
+def test_two_process_cluster_recovery_resume(tmp_path: Path):
+    """EOF + resume across a 2-process gloo cluster sharing one
+    recovery directory: keyed state and source positions restore with
+    exactly-once results.  (ABORT is single-process-only by its
+    documented contract — its one-shot latch is in-process state.)"""
+    from bytewax_amd.recovery import init_db_dir
+
+    db = tmp_path / "db"
+    db.mkdir()
+    init_db_dir(db, 4)
+    out_file = tmp_path / "out.txt"
+    flow_file = tmp_path / "flowdef.py"
+    flow_file.write_text(
+        textwrap.dedent(
+            f"""
+            import bytewax_amd.operators as op
+            from bytewax_amd.connectors.files import FileSink
+            from bytewax_amd.dataflow import Dataflow
+            from bytewax_amd.testing import TestingSource
+
+            inp = [(str(i % 2), 1) for i in range(6)]
+            inp.insert(4, TestingSource.EOF())
+
+            flow = Dataflow("dist_rec")
+            s = op.input("inp", flow, TestingSource(inp))
+
+            def running_sum(state, v):
+                state = (state or 0) + v
+                return (state, state)
+
+            s = op.stateful_map("sum", s, running_sum)
+            s = op.map("fmt", s, lambda kv: (kv[0], f"{{kv[0]}}={{kv[1]}}"))
+            op.output("out", s, FileSink({str(out_file)!r}))
+            """
+        )
+    )
+    port = 29100 + os.getpid() % 300
+
+    def launch():
+        addresses = f"127.0.0.1:{port};127.0.0.1:{port + 1}"
+        env = dict(os.environ)
+        env["PYTHONPATH"] = str(Path(__file__).resolve().parent.parent)
+        procs = [
+            subprocess.Popen(
+                [
+                    sys.executable,
+                    "-m",
+                    "bytewax_amd.run",
+                    f"{flow_file}:flow",
+                    "-i",
+                    str(i),
+                    "-a",
+                    addresses,
+                    "-r",
+                    str(db),
+                    "-s",
+                    "0",
+                    "-b",
+                    "0",
+                ],
+                env=env,
+                stdout=subprocess.PIPE,
+                stderr=subprocess.PIPE,
+            )
+            for i in range(2)
+        ]
+        for p in procs:
+            _so, se = p.communicate(timeout=200)
+            assert p.returncode == 0, se.decode()[-2000:]
+
+    launch()  # runs until the EOF sentinel
+    first = sorted(out_file.read_text().splitlines())
+    assert first == ["0=1", "0=2", "1=1", "1=2"]
+
+    launch()  # continuation: state 2 per key, remaining items
+    lines = sorted(out_file.read_text().splitlines())
+    # FileSink truncates to its snapshot offset on resume, so the file
+    # holds the pre-EOF lines plus the continued ones.
+    assert lines == ["0=1", "0=2", "0=3", "1=1", "1=2", "1=3"]
