@@ -316,11 +316,23 @@ class StatsAggState:
 
 
 class HashJoinState:
-    """Two-sided stream join state on device ("last"/"complete")."""
+    """Two-sided stream join state on device ("last"/"complete").
+
+    With ``radix=True`` (default) each side's batch is partitioned
+    into table-region segments first, so the exchange of flags/values
+    stays L2-local per workgroup instead of random across the table.
+    """
 
     N_SIDES = 2
 
-    def __init__(self, device, slots_pow: int = 20, out_cap: int = 1 << 20):
+    def __init__(
+        self,
+        device,
+        slots_pow: int = 20,
+        out_cap: int = 1 << 20,
+        radix: bool = True,
+        region_bits: int = 11,
+    ):
         import torch
 
         self.device = device
@@ -329,7 +341,24 @@ class HashJoinState:
             self._table: Dict[int, list] = {}
             return
         self.k = ext()
+        if radix:
+            slots_pow = max(slots_pow, region_bits + 10)
+        self.radix = radix
+        self.region_bits = region_bits
         self.nslots = 1 << slots_pow
+        if radix:
+            self.n_regions = self.nslots >> region_bits
+            self.rx_gcursors = torch.zeros(
+                self.n_regions, dtype=torch.int32, device=device
+            )
+            self.rx_ov_cursor = torch.zeros(1, dtype=torch.int32, device=device)
+            self.rx_max_batch = 0
+            self.rx_packed = torch.empty(1, dtype=torch.int64, device=device)
+            self.rx_vals = torch.empty(1, dtype=torch.int64, device=device)
+            self.rx_ov_packed = torch.empty(1, dtype=torch.int64, device=device)
+            self.rx_ov_vals = torch.empty(1, dtype=torch.int64, device=device)
+            self.rx_zeros = torch.zeros(1, dtype=torch.int64, device=device)
+            self.rx_maxts = torch.zeros(1, dtype=torch.int64, device=device)
         self.tkeys = torch.full(
             (self.nslots,), -1, dtype=torch.int64, device=device
         )
@@ -352,6 +381,56 @@ class HashJoinState:
                 ent = self._table.setdefault(int(k), [None, None, 0])
                 ent[side] = int(v)
                 ent[2] |= 1 << side
+            return
+        if self.radix:
+            import torch
+
+            n = int(keys.numel())
+            if n > self.rx_max_batch:
+                mb = int(n * 5 // 4)
+                self.rx_max_batch = mb
+                per_region = -(-mb * 5 // 2) // self.n_regions + 1
+                total = per_region * self.n_regions
+                self.rx_packed = torch.empty(
+                    total, dtype=torch.int64, device=self.device
+                )
+                self.rx_vals = torch.empty(
+                    total, dtype=torch.int64, device=self.device
+                )
+                ov = max(1 << 20, mb)
+                self.rx_ov_packed = torch.empty(
+                    ov, dtype=torch.int64, device=self.device
+                )
+                self.rx_ov_vals = torch.empty(
+                    ov, dtype=torch.int64, device=self.device
+                )
+                self.rx_zeros = torch.zeros(
+                    mb, dtype=torch.int64, device=self.device
+                )
+            self.k.radix_join_insert(
+                keys,
+                self.rx_zeros,
+                vals,
+                side,
+                self.N_SIDES,
+                self.tkeys,
+                self.tval0,
+                self.tval1,
+                self.tflags,
+                self.rx_gcursors,
+                self.rx_packed,
+                self.rx_vals,
+                self.rx_ov_cursor,
+                self.rx_ov_packed,
+                self.rx_ov_vals,
+                self.out_keys,
+                self.out_v0,
+                self.out_v1,
+                self.out_n,
+                self.rx_maxts,
+                self.error_flag,
+                self.region_bits,
+            )
             return
         self.k.join_insert(
             keys,

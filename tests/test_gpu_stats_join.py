@@ -132,7 +132,8 @@ def test_stats_gpu_snapshot_roundtrip():
 
 
 @pytest.mark.gpu
-def test_join_gpu_matches_reference():
+@pytest.mark.parametrize("radix", [False, True])
+def test_join_gpu_matches_reference(radix):
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
     torch.manual_seed(7)
@@ -142,7 +143,9 @@ def test_join_gpu_matches_reference():
     rk = torch.randint(0, 50_000, (n,), dtype=torch.int32)
     rv = torch.randint(0, 1000, (n,), dtype=torch.int64)
 
-    st = HashJoinState(torch.device("cuda:0"), slots_pow=17, out_cap=n * 2)
+    st = HashJoinState(
+        torch.device("cuda:0"), slots_pow=17, out_cap=n * 2, radix=radix
+    )
     st.insert(0, lk.cuda(), lv.cuda())
     st.insert(1, rk.cuda(), rv.cuda())
     out = st.take_joined()
