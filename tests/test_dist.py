@@ -157,3 +157,44 @@ def test_two_process_cluster_recovery_resume(tmp_path: Path):
     # FileSink truncates to its snapshot offset on resume, so the file
     # holds the pre-EOF lines plus the continued ones.
     assert lines == ["0=1", "0=2", "0=3", "1=1", "1=2", "1=3"]
+
+
+@pytest.mark.timeout(180)
+def test_testing_module_cluster_launcher(tmp_path: Path):
+    """`python -m bytewax_amd.testing -p2` spawns a working local
+    cluster (parity: reference testing.py:287-383)."""
+    flow_file = tmp_path / "lflow.py"
+    out_file = tmp_path / "out.txt"
+    flow_file.write_text(
+        textwrap.dedent(
+            f"""
+            import bytewax_amd.operators as op
+            from bytewax_amd.connectors.files import FileSink
+            from bytewax_amd.dataflow import Dataflow
+            from bytewax_amd.testing import TestingSource
+
+            flow = Dataflow("launcher")
+            s = op.input("inp", flow, TestingSource([("a", "x"), ("b", "y")]))
+            op.output("out", s, FileSink({str(out_file)!r}))
+            """
+        )
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(Path(__file__).resolve().parent.parent)
+    res = subprocess.run(
+        [
+            sys.executable,
+            "-m",
+            "bytewax_amd.testing",
+            f"{flow_file}:flow",
+            "-p",
+            "2",
+            "-w",
+            "1",
+        ],
+        env=env,
+        capture_output=True,
+        timeout=150,
+    )
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert sorted(out_file.read_text().splitlines()) == ["x", "y"]
