@@ -380,3 +380,71 @@ def test_fold_window_unordered_keeps_arrival_order():
     op.output("out", wo.down, TestingSink(out))
     run_main(flow)
     assert out == [("ALL", (0, ["c", "a", "b"]))]
+
+
+def test_collect_window_set_and_dict(entry_point):
+    inp = [(ts(0), ("k1", 1)), (ts(1), ("k2", 2)), (ts(2), ("k1", 3))]
+    outs, outd = [], []
+    flow = Dataflow("f")
+    s = op.input("inp", flow, TestingSource(inp))
+    keyed = op.key_on("k", s, lambda x: "ALL")
+    vals = op.map_value("strip_ts", keyed, lambda tv: tv[1])
+    # The clock must see timestamps, so collect on the (ts, pair)
+    # stream for `set` and on pairs for `dict` via a second window.
+    tw = TumblingWindower(align_to=ALIGN_TO, length=timedelta(minutes=1))
+    ws = w.collect_window("cs", keyed, ec(), tw, into=set)
+    wd = w.collect_window(
+        "cd",
+        vals,
+        EventClock(
+            ts_getter=lambda _v: ts(0),
+            wait_for_system_duration=timedelta(hours=1),
+        ),
+        tw,
+        into=dict,
+    )
+    op.output("os", ws.down, TestingSink(outs))
+    op.output("od", wd.down, TestingSink(outd))
+    entry_point(flow)
+    assert outs[0][1][1] == set(inp)
+    assert outd[0][1][1] == {"k1": 3, "k2": 2}
+
+
+def test_join_window_running_mode(entry_point):
+    a_inp = [(ts(0), 1), (ts(2), 2)]
+    b_inp = [(ts(1), "x")]
+    out = []
+    flow = Dataflow("f")
+    a = op.input("a", flow, TestingSource(a_inp))
+    b = op.input("b", flow, TestingSource(b_inp))
+    ka = op.key_on("ka", a, lambda x: "K")
+    kb = op.key_on("kb", b, lambda x: "K")
+    clock = EventClock(
+        ts_getter=lambda x: x[0], wait_for_system_duration=ZERO_TD
+    )
+    tw = TumblingWindower(align_to=ALIGN_TO, length=timedelta(minutes=1))
+    wo = w.join_window("jw", clock, tw, ka, kb, emit_mode="running")
+    op.output("out", wo.down, TestingSink(out))
+    entry_point(flow)
+    # Running mode emits on every value (ordered by timestamp).
+    vals = [v for _k, (_w, v) in out]
+    assert vals == [
+        ((ts(0), 1), None),
+        ((ts(0), 1), (ts(1), "x")),
+        ((ts(2), 2), (ts(1), "x")),
+    ]
+
+
+def test_count_window_multiple_keys(entry_point):
+    inp = [(ts(0), "a"), (ts(1), "b"), (ts(2), "a")]
+    out = []
+    flow = Dataflow("f")
+    s = op.input("inp", flow, TestingSource(inp))
+    clock = EventClock(
+        ts_getter=lambda x: x[0], wait_for_system_duration=ZERO_TD
+    )
+    tw = TumblingWindower(align_to=ALIGN_TO, length=timedelta(minutes=1))
+    wo = w.count_window("cw", s, clock, tw, lambda x: x[1])
+    op.output("out", wo.down, TestingSink(out))
+    entry_point(flow)
+    assert sorted(out) == [("a", (0, 2)), ("b", (0, 1))]
