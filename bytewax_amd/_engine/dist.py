@@ -80,19 +80,16 @@ class _DistCtx:
     ) -> Tuple[bool, bool, bool]:
         import torch
 
+        # One MIN all_reduce carries both votes: [eof, 1 - abort]
+        # (abort-any == not min(1 - abort)).
         flags = torch.tensor(
-            [1 if local_eof else 0, 1 if aborting else 0], dtype=torch.int64
+            [1 if local_eof else 0, 0 if aborting else 1], dtype=torch.int64
         )
         self.dist.all_reduce(
             flags, op=self.dist.ReduceOp.MIN, group=self.group
         )
-        eof_flag = int(flags[0].item())
-        abort_any = torch.tensor([1 if aborting else 0], dtype=torch.int64)
-        self.dist.all_reduce(
-            abort_any, op=self.dist.ReduceOp.MAX, group=self.group
-        )
-        abort = bool(int(abort_any.item()))
-        all_eof = bool(eof_flag)
+        all_eof = bool(int(flags[0].item()))
+        abort = not bool(int(flags[1].item()))
         # Single clock: rank 0 decides closing on wall time.
         decision = torch.tensor(
             [

@@ -239,3 +239,16 @@ def test_mid_flow_merge_of_keyed_streams(entry_point):
     entry_point(flow)
     # fold_final keys on the tuple's key; both items share "k".
     assert out == [("k", 3)]
+
+
+def test_merge_many_streams(entry_point):
+    out = []
+    flow = Dataflow("f")
+    ins = [
+        op.input(f"i{i}", flow, TestingSource([i * 10, i * 10 + 1]))
+        for i in range(4)
+    ]
+    m = op.merge("m", *ins)
+    op.output("out", m, TestingSink(out))
+    entry_point(flow)
+    assert sorted(out) == [0, 1, 10, 11, 20, 21, 30, 31]
