@@ -27,6 +27,7 @@
 
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
+#include <pybind11/stl.h>
 #include <ATen/hip/HIPContext.h>
 
 #include <chrono>
@@ -454,10 +455,13 @@ __global__ __launch_bounds__(256) void k_agg_cached(
 // that would overflow its region's buffer goes to the overflow spill
 // (aggregated by k_radix_agg's caller via the contended-direct path —
 // statistically empty for uniform keys at cap = 2x mean).
-template <int MODE>
+// TS is int64_t (absolute ms) or int32_t (deltas from `ts_base`, the
+// wire format of the RCCL exchange — inserting the received segments
+// directly skips the int64 timestamp rebuild entirely).
+template <int MODE, typename TS = int64_t>
 __global__ void k_radix_scatter_fixed(
     const int32_t* __restrict__ keys,
-    const int64_t* __restrict__ ts,
+    const TS* __restrict__ ts,
     const int64_t* __restrict__ vals,
     int64_t n,
     int64_t align_ms,
@@ -485,7 +489,7 @@ __global__ void k_radix_scatter_fixed(
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   int64_t local_max = 0;
   for (int64_t i = start; i < n; i += stride) {
-    int64_t t = ts[i] + ts_base;
+    int64_t t = (int64_t)ts[i] + ts_base;
     if (t > local_max) local_max = t;
     int64_t win = (t - align_ms) / len_ms;
     uint64_t packed =
@@ -504,7 +508,7 @@ __global__ void k_radix_scatter_fixed(
   }
   __syncthreads();
   for (int64_t i = start; i < n; i += stride) {
-    int64_t win = (ts[i] + ts_base - align_ms) / len_ms;
+    int64_t win = ((int64_t)ts[i] + ts_base - align_ms) / len_ms;
     uint64_t packed =
         ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
     int b = (int)region_of(mix64(packed), mask, region_bits);
@@ -1393,9 +1397,23 @@ void radix_window_insert(
     int64_t len_ms,
     int64_t mode,
     int64_t ts_base,
-    int64_t region_bits) {
+    int64_t region_bits,
+    std::vector<int64_t> seg_counts = {},
+    std::vector<int64_t> seg_bases = {}) {
   check_dev(keys, torch::kInt32, "keys");
-  check_dev(ts, torch::kInt64, "ts");
+  // Segmented form: `ts` holds int32 deltas laid out as contiguous
+  // per-source-rank segments (the RCCL all-to-allv wire format);
+  // seg_counts[i] events use seg_bases[i] as their absolute base.
+  // This inserts exchange output without ever materializing int64
+  // timestamps (saves ~3 HBM passes over the biggest column).
+  bool seg32 = !seg_counts.empty();
+  if (seg32) {
+    TORCH_CHECK(seg_counts.size() == seg_bases.size(),
+                "seg_counts/seg_bases length mismatch");
+    check_dev(ts, torch::kInt32, "ts32");
+  } else {
+    check_dev(ts, torch::kInt64, "ts");
+  }
   int64_t n = keys.numel();
   int64_t nslots = tkeys.numel();
   TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
@@ -1427,18 +1445,33 @@ void radix_window_insert(
   // partially-written cache lines) open at once — per-XCD L2 can then
   // accumulate full lines before eviction.  Tunable while we converge
   // on the best default (BYTEWAX_SCATTER_BLOCKS).
-  int scatter_blocks = (int)grid.x;
+  int env_blocks = 0;
   if (const char* sb = std::getenv("BYTEWAX_SCATTER_BLOCKS")) {
     int v = atoi(sb);
-    if (v > 0) scatter_blocks = v;
+    if (v > 0) env_blocks = v;
   }
-  dim3 sgrid((unsigned)scatter_blocks);
 
-  auto scat = [&](auto kern) {
+  struct Seg {
+    int64_t off, n, base;
+  };
+  std::vector<Seg> segs;
+  if (seg32) {
+    int64_t off = 0;
+    for (size_t i = 0; i < seg_counts.size(); ++i) {
+      if (seg_counts[i] > 0) segs.push_back({off, seg_counts[i], seg_bases[i]});
+      off += seg_counts[i];
+    }
+    TORCH_CHECK(off == n, "segment counts must sum to batch length");
+  } else {
+    segs.push_back({0, n, ts_base});
+  }
+
+  auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx) {
     hipLaunchKernelGGL(
-        kern, sgrid, block, 2 * hist_lds, stream, keys.data_ptr<int32_t>(),
-        ts.data_ptr<int64_t>(), vptr, n, align_ms, len_ms, ts_base, mask,
-        (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        kern, dim3(gx), block, 2 * hist_lds, stream,
+        keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
+        vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
+        sg.base, mask, (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
         ov_cursor.data_ptr<int32_t>(),
@@ -1448,8 +1481,23 @@ void radix_window_insert(
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>());
   };
-  if (mode == AGG_COUNT) scat(k_radix_scatter_fixed<AGG_COUNT>);
-  else scat(k_radix_scatter_fixed<AGG_SUM>);
+  for (const Seg& sg : segs) {
+    unsigned gx = (unsigned)n_blocks(sg.n, 256);
+    if (env_blocks > 0 && (unsigned)env_blocks < gx) gx = (unsigned)env_blocks;
+    if (seg32) {
+      const int32_t* t32 = ts.data_ptr<int32_t>();
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_fixed<AGG_COUNT, int32_t>, t32, sg, gx);
+      else
+        scat(k_radix_scatter_fixed<AGG_SUM, int32_t>, t32, sg, gx);
+    } else {
+      const int64_t* t64 = ts.data_ptr<int64_t>();
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_fixed<AGG_COUNT, int64_t>, t64, sg, gx);
+      else
+        scat(k_radix_scatter_fixed<AGG_SUM, int64_t>, t64, sg, gx);
+    }
+  }
 
   // Fixed layout: bucket b's events live at [b*cap, b*cap + count).
   auto offsets = at::arange(

@@ -194,11 +194,13 @@ def exchange_by_key(
         )
         if w is not None:
             works.append(w)
-    seg_bases = torch.repeat_interleave(
-        bases, torch.tensor(out_splits, dtype=torch.int64, device=dev)
-    )
+    # Bases were exchanged synchronously above; reading them to host is
+    # a tiny D2H on the control plane (we already synced on the split
+    # sizes).  Keeping them as scalars lets the radix insert consume
+    # the int32 segments directly — no per-event base tensor at all.
     out = _LazyTsBatch(
-        recv_keys, recv_ts32, seg_bases, recv_vals, batch.max_ts
+        recv_keys, recv_ts32, out_splits, bases.tolist(), recv_vals,
+        batch.max_ts,
     )
     if async_op:
         return (works, out)
@@ -206,20 +208,35 @@ def exchange_by_key(
 
 
 class _LazyTsBatch:
-    """Deferred timestamp reconstruction so the int64 rebuild happens
-    after the async exchange completes (on the consumer's wait)."""
+    """Exchange output in wire format: int32 timestamp deltas laid out
+    as contiguous per-source-rank segments with scalar int64 bases.
 
-    def __init__(self, keys, ts32, seg_bases, vals, max_ts):
+    The radix insert path consumes this directly
+    (:meth:`WindowAggState.insert_lazy` — one scatter launch per
+    segment, no int64 rebuild); other consumers ``materialize()`` the
+    absolute-timestamp :class:`RecordBatch` after the async exchange
+    completes."""
+
+    def __init__(self, keys, ts32, seg_counts, seg_bases, vals, max_ts):
         self.keys = keys
         self.ts32 = ts32
-        self.seg_bases = seg_bases
+        self.seg_counts = list(seg_counts)
+        self.seg_bases = [int(b) for b in seg_bases]
         self.vals = vals
         self.max_ts = max_ts
+
+    def __len__(self) -> int:
+        return self.keys.numel()
 
     def materialize(self) -> RecordBatch:
         import torch
 
-        ts = self.ts32.to(torch.int64) + self.seg_bases
+        ts = self.ts32.to(torch.int64)
+        off = 0
+        for cnt, base in zip(self.seg_counts, self.seg_bases):
+            if cnt and base:
+                ts[off : off + cnt] += base
+            off += cnt
         return RecordBatch(
             self.keys, ts, self.vals, max_ts=self.max_ts
         )
@@ -440,6 +457,8 @@ class WindowAggState:
                 self.mode,
                 batch.ts_base,
                 self.region_bits,
+                [],
+                [],
             )
         else:
             self.k.window_agg_insert(
@@ -460,6 +479,46 @@ class WindowAggState:
             )
         if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
             self.max_ts_host = batch.max_ts
+
+    def insert_lazy(self, lz: "_LazyTsBatch") -> None:
+        """Insert exchange output in wire format (int32 timestamp
+        deltas in per-source-rank segments).
+
+        On the radix path the scatter kernel consumes the segments
+        directly with their scalar bases, skipping the int64 timestamp
+        rebuild (~3 HBM passes over the biggest column per step at
+        world > 1).  Other paths materialize first.
+        """
+        if self.cpu or not self.radix or self.radix_v2:
+            self.insert(lz.materialize())
+            return
+        n = len(lz)
+        if n > self.rx_max_batch:
+            self._alloc_rx(int(n * 5 // 4))
+        self.k.radix_window_insert(
+            lz.keys,
+            lz.ts32,
+            lz.vals,
+            self.tkeys,
+            self.tvals,
+            self.max_ts_dev,
+            self.error_flag,
+            self.rx_gcursors,
+            self.rx_packed,
+            self.rx_vals,
+            self.rx_ov_cursor,
+            self.rx_ov_packed,
+            self.rx_ov_vals,
+            self.align_ms,
+            self.len_ms,
+            self.mode,
+            0,
+            self.region_bits,
+            lz.seg_counts,
+            lz.seg_bases,
+        )
+        if lz.max_ts is not None and lz.max_ts > self.max_ts_host:
+            self.max_ts_host = lz.max_ts
 
     def watermark_ms(self, sync: bool = False) -> int:
         if sync and not self.cpu:

@@ -189,6 +189,10 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         for w in works:
             w.wait()
         if hasattr(batch, "materialize"):
+            if hasattr(self.state, "insert_lazy"):
+                # Wire-format insert: no int64 timestamp rebuild.
+                self.state.insert_lazy(batch)
+                return
             batch = batch.materialize()
         self.state.insert(batch)
 

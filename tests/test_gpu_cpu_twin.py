@@ -279,3 +279,34 @@ def test_keyed_stats_agg_pipeline_cpu():
     for o in out:
         assert bool((o["min"] <= o["max"]).all())
         assert bool((o["sum"] >= o["min"] * o["cnt"]).all())
+
+
+def test_lazy_ts_batch_materialize_segments():
+    """Wire-format batch rebuilds absolute timestamps per segment."""
+    from bytewax_amd.gpu import _LazyTsBatch
+
+    keys = torch.arange(6, dtype=torch.int32)
+    ts32 = torch.tensor([0, 5, -3, 7, 2, 4], dtype=torch.int32)
+    lz = _LazyTsBatch(
+        keys, ts32, [2, 0, 3, 1], [1000, 0, 2000, 50],
+        torch.arange(6, dtype=torch.int64), max_ts=2007,
+    )
+    assert len(lz) == 6
+    b = lz.materialize()
+    assert b.ts.tolist() == [1000, 1005, 1997, 2007, 2002, 54]
+    assert b.keys.tolist() == list(range(6))
+    assert b.max_ts == 2007
+
+
+def test_insert_lazy_cpu_falls_back_to_materialize():
+    from bytewax_amd.gpu import AGG_COUNT, WindowAggState, _LazyTsBatch
+
+    st = WindowAggState(torch.device("cpu"), 0, 100, AGG_COUNT)
+    keys = torch.zeros(4, dtype=torch.int32)
+    ts32 = torch.tensor([10, 20, 110, 120], dtype=torch.int32)
+    st.insert_lazy(_LazyTsBatch(keys, ts32, [2, 2], [0, 100], None, 220))
+    out = st.close_all()
+    # windows 0 [0,100) and 2 [200,300): two events each.
+    assert sorted(
+        zip(out.keys.tolist(), out.ts.tolist(), out.vals.tolist())
+    ) == [(0, 0, 2), (0, 200, 2)]
