@@ -116,12 +116,6 @@ def exchange_by_key(
     world = dist.get_world_size(group)
     if world == 1:
         return ([], batch) if async_op else batch
-    if batch.ts_base != 0:
-        msg = (
-            "exchange_by_key requires ts_base == 0 (bases are "
-            "per-producer; bake them into `ts` before exchanging)"
-        )
-        raise ValueError(msg)
     dev = batch.keys.device
     n = len(batch)
     has_vals = batch.vals is not None
@@ -159,13 +153,18 @@ def exchange_by_key(
     # (a batch spans far less than 2^31 ms) — cuts exchanged bytes by
     # a third on the biggest column.  Receivers rebuild absolute
     # timestamps from the per-source-rank bases.
+    # `send_ts` may itself be relative to batch.ts_base (producers
+    # ship zero-based templates); fold that producer base into the
+    # wire base so absolute timestamps are never materialized here.
     if batch.max_ts is not None:
         # Host-known watermark avoids a device sync; any base within
         # 2^31 ms of every timestamp works.
         ts_base = batch.max_ts - (1 << 30)
     else:
-        ts_base = int(send_ts.min().item()) if n > 0 else 0
-    send_ts32 = (send_ts - ts_base).to(torch.int32)
+        ts_base = (
+            int(send_ts.min().item()) + batch.ts_base if n > 0 else 0
+        )
+    send_ts32 = (send_ts - (ts_base - batch.ts_base)).to(torch.int32)
     bases = torch.zeros(world, dtype=torch.int64, device=dev)
     my_base = torch.full((world,), ts_base, dtype=torch.int64, device=dev)
     dist.all_to_all_single(bases, my_base, group=group)
@@ -287,6 +286,11 @@ class WindowAggState:
         self.dedup = dedup
         self.max_ts_host = 0  # watermark if batches carry max_ts
         self.closed_horizon = -(1 << 62)  # window ids below are closed
+        # Deferred close readback (close_launch/close_resolve).
+        self._close_pending = False
+        self._close_ev = None
+        self._outn_pin = None
+        self._carry_cpu = None
         if radix_v2 and (mode != AGG_COUNT or dedup):
             msg = "radix_v2 supports the COUNT mode only"
             raise ValueError(msg)
@@ -624,8 +628,104 @@ class WindowAggState:
         self.closed_horizon = horizon
         return out
 
+    def close_launch(self, wait_ms: int = 0) -> None:
+        """Launch the due-window close without reading it back.
+
+        The migrate kernel and the row-count D2H land on the stream
+        asynchronously; :meth:`close_resolve` (typically called at the
+        START of the next scheduling step) picks up the emitted rows.
+        This keeps the close latency hidden behind one step of
+        pipeline instead of stalling the stream at every watermark
+        advance — under the per-step Python engine the synchronous
+        readback was the p99 outlier.
+        """
+        import torch
+
+        if self._close_pending:
+            # One in flight at a time; resolve first.
+            return
+        wm = self.watermark_ms()
+        horizon = (
+            wm - wait_ms - self.align_ms - self.len_ms
+        ) // self.off_ms + 1
+        if horizon <= self.closed_horizon:
+            return
+        if self.cpu:
+            self._carry_cpu = self._extract_cpu(
+                -(1 << 40), horizon, delete=True
+            )
+            self.closed_horizon = horizon
+            self._close_pending = self._carry_cpu is not None
+            return
+        self.out_n.zero_()
+        self.k.close_migrate(
+            self.tkeys,
+            self.tvals,
+            self.tkeys_alt,
+            self.tvals_alt,
+            horizon,
+            self.region_bits,
+            self.out_keys,
+            self.out_wins,
+            self.out_vals,
+            self.out_n,
+            self.error_flag,
+        )
+        self.tkeys, self.tkeys_alt = self.tkeys_alt, self.tkeys
+        self.tvals, self.tvals_alt = self.tvals_alt, self.tvals
+        self.tkeys_alt.fill_(-1)
+        self.tvals_alt.zero_()
+        self.closed_horizon = horizon
+        if self._outn_pin is None:
+            self._outn_pin = torch.zeros(
+                2,
+                dtype=torch.int32,
+                pin_memory=torch.cuda.is_available(),
+            )
+        self._outn_pin[0:1].copy_(self.out_n, non_blocking=True)
+        self._outn_pin[1:2].copy_(self.error_flag, non_blocking=True)
+        self._close_ev = torch.cuda.Event()
+        self._close_ev.record()
+        self._close_pending = True
+
+    def close_resolve(self) -> Optional[RecordBatch]:
+        """Read back the rows of the last :meth:`close_launch`.
+
+        Must run before the NEXT close_launch / close_all /
+        snapshot_to_host (those reuse the out staging buffers)."""
+        import torch
+
+        if not self._close_pending:
+            return None
+        self._close_pending = False
+        if self.cpu:
+            out = self._carry_cpu
+            self._carry_cpu = None
+            return out
+        self._close_ev.synchronize()
+        n = int(self._outn_pin[0].item())
+        if int(self._outn_pin[1].item()) != 0:
+            msg = "keyed window state table overflowed; increase slots_pow"
+            raise RuntimeError(msg)
+        if n > self.out_cap:
+            msg = (
+                f"window close produced {n} rows > out_cap "
+                f"{self.out_cap}; increase out_cap"
+            )
+            raise RuntimeError(msg)
+        if n == 0:
+            return None
+        return RecordBatch(
+            self.out_keys[:n].clone(),
+            self.out_wins[:n].to(torch.int64) * self.off_ms + self.align_ms,
+            self.out_vals[:n].clone(),
+        )
+
     def close_all(self) -> Optional[RecordBatch]:
         """EOF: close every window at or above the closed horizon."""
+        if self._close_pending:
+            msg = "close_resolve() the pending close before close_all()"
+            raise RuntimeError(msg)
         if self.cpu:
             return self._extract_cpu(self.closed_horizon, 1 << 40, delete=True)
         out = self._extract_range(self.closed_horizon, 1 << 40)

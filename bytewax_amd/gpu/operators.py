@@ -106,16 +106,22 @@ class _SyntheticPartition(StatelessSourcePartition[RecordBatch]):
         i = self.emitted
         self.emitted += 1
         start = self.align_ms + i * self.sim_ms_per_batch
-        ts = self.ts_template + start
         keys = self.key_pool[i % len(self.key_pool)]
         vals = (
             self.val_pool[i % len(self.val_pool)]
             if self.val_pool is not None
             else None
         )
+        # Zero-based template + scalar base: consumers (insert kernels
+        # and the exchange wire format) apply the base without ever
+        # materializing an absolute int64 timestamp column.
         return [
             RecordBatch(
-                keys, ts, vals, max_ts=start + self.sim_ms_per_batch - 1
+                keys,
+                self.ts_template,
+                vals,
+                max_ts=start + self.sim_ms_per_batch - 1,
+                ts_base=start,
             )
         ]
 
@@ -178,8 +184,34 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         self.wait_ms = wait_ms
         self.exchange = exchange
         self.pending: Optional[tuple] = None  # (works, RecordBatch)
+        # Closed rows resolved during snapshot() (which cannot emit):
+        # re-emitted on the next activation; persisted in the snapshot
+        # so a resume replays them exactly once.
+        self._carry: Optional[RecordBatch] = None
+        # Bounded in-flight window: per-step events let the host stay
+        # at most PIPELINE steps ahead of the device, so the close
+        # readback (and the step latency) never waits on a deep queue.
+        self._evq: List[Any] = []
         if resume is not None:
+            import torch
+
+            resume = dict(resume)
+            carry = resume.pop("__carry__", None)
+            if carry is not None:
+                dev = self.state.device
+                self._carry = RecordBatch(
+                    torch.as_tensor(carry["keys"]).to(dev),
+                    torch.as_tensor(carry["ts"]).to(dev),
+                    torch.as_tensor(carry["vals"]).to(dev),
+                )
             self.state.restore_from_host(resume)
+
+    PIPELINE = 2
+
+    def _take_carry(self, out: List[RecordBatch]) -> None:
+        if self._carry is not None:
+            out.append(self._carry)
+            self._carry = None
 
     def _flush_pending(self) -> None:
         if self.pending is None:
@@ -197,7 +229,13 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         self.state.insert(batch)
 
     def on_batch(self, batches: List[RecordBatch]):
-        out = []
+        import torch
+
+        out: List[RecordBatch] = []
+        self._take_carry(out)
+        closed = self.state.close_resolve()  # previous step's close
+        if closed is not None:
+            out.append(closed)
         for batch in batches:
             if self.exchange:
                 works, exchanged = exchange_by_key(batch, async_op=True)
@@ -205,22 +243,50 @@ class _DeviceWindowLogic(StatefulBatchLogic):
                 self.pending = (works, exchanged)
             else:
                 self.state.insert(batch)
-        closed = self.state.close_due(self.wait_ms)
-        if closed is not None:
-            out.append(closed)
+        self.state.close_launch(self.wait_ms)
+        if not self.state.cpu:
+            ev = torch.cuda.Event()
+            ev.record()
+            self._evq.append(ev)
+            if len(self._evq) > self.PIPELINE:
+                self._evq.pop(0).synchronize()
         return (out, StatefulBatchLogic.RETAIN)
 
     def on_eof(self):
+        out: List[RecordBatch] = []
+        self._take_carry(out)
         self._flush_pending()
-        closed = self.state.close_all()
-        return (
-            [closed] if closed is not None else [],
-            StatefulBatchLogic.RETAIN,
-        )
+        closed = self.state.close_resolve()
+        if closed is not None:
+            out.append(closed)
+        final = self.state.close_all()
+        if final is not None:
+            out.append(final)
+        return (out, StatefulBatchLogic.RETAIN)
 
     def snapshot(self) -> Dict[str, Any]:
+        import torch
+
         self._flush_pending()
-        return self.state.snapshot_to_host()
+        resolved = self.state.close_resolve()
+        if resolved is not None:
+            if self._carry is not None:
+                self._carry = RecordBatch(
+                    torch.cat([self._carry.keys, resolved.keys]),
+                    torch.cat([self._carry.ts, resolved.ts]),
+                    torch.cat([self._carry.vals, resolved.vals]),
+                )
+            else:
+                self._carry = resolved
+        snap = self.state.snapshot_to_host()
+        if self._carry is not None:
+            snap = dict(snap)
+            snap["__carry__"] = {
+                "keys": self._carry.keys.cpu().numpy().copy(),
+                "ts": self._carry.ts.cpu().numpy().copy(),
+                "vals": self._carry.vals.cpu().numpy().copy(),
+            }
+        return snap
 
 
 @operator

@@ -159,7 +159,9 @@ class StatsAggState:
 
     def _insert_cpu(self, batch: RecordBatch) -> None:
         keys = batch.keys.tolist()
-        wins = ((batch.ts - self.align_ms) // self.len_ms).tolist()
+        wins = (
+            (batch.ts + batch.ts_base - self.align_ms) // self.len_ms
+        ).tolist()
         vals = batch.vals.tolist()
         for k, w, v in zip(keys, wins, vals):
             cnt, s, mn, mx = self._table.get(
@@ -169,7 +171,7 @@ class StatsAggState:
                 cnt + 1, s + v, min(mn, v), max(mx, v)
             )
         if len(batch):
-            mx_ts = int(batch.ts.max().item())
+            mx_ts = int(batch.ts.max().item()) + batch.ts_base
             if mx_ts > self.max_ts_host:
                 self.max_ts_host = mx_ts
 
