@@ -310,3 +310,57 @@ def test_insert_lazy_cpu_falls_back_to_materialize():
     assert sorted(
         zip(out.keys.tolist(), out.ts.tolist(), out.vals.tolist())
     ) == [(0, 0, 2), (0, 200, 2)]
+
+
+def test_deferred_close_carry_roundtrips_through_snapshot():
+    """A close resolved during snapshot() may not be emitted in that
+    epoch; its rows ride the snapshot and are re-emitted exactly once
+    after resume."""
+    from bytewax_amd.gpu import AGG_COUNT, RecordBatch, WindowAggState
+    from bytewax_amd.gpu.operators import _DeviceWindowLogic
+
+    def mk_state():
+        return WindowAggState(torch.device("cpu"), 0, 100, AGG_COUNT)
+
+    logic = _DeviceWindowLogic(mk_state(), 0, False, None)
+    # Batch whose max_ts closes window 0 immediately.
+    b = RecordBatch(
+        torch.zeros(3, dtype=torch.int32),
+        torch.tensor([10, 20, 150], dtype=torch.int64),
+        None,
+        max_ts=150,
+    )
+    out, _ = logic.on_batch([b])
+    assert out == []  # close launched, not yet resolved
+    snap = logic.snapshot()
+    assert "__carry__" in snap
+    # Resume in a fresh logic: the carried close is emitted first.
+    logic2 = _DeviceWindowLogic(mk_state(), 0, False, snap)
+    out2, _ = logic2.on_batch(
+        [
+            RecordBatch(
+                torch.zeros(0, dtype=torch.int32),
+                torch.zeros(0, dtype=torch.int64),
+                None,
+                max_ts=151,
+            )
+        ]
+    )
+    rows = [
+        (k, t, v)
+        for batch in out2
+        for k, t, v in zip(
+            batch.keys.tolist(), batch.ts.tolist(), batch.vals.tolist()
+        )
+    ]
+    assert rows == [(0, 0, 2)]
+    # And the live window-1 cell survived in the table.
+    final = logic2.on_eof()[0]
+    rows_eof = [
+        (k, t, v)
+        for batch in final
+        for k, t, v in zip(
+            batch.keys.tolist(), batch.ts.tolist(), batch.vals.tolist()
+        )
+    ]
+    assert rows_eof == [(0, 100, 1)]
