@@ -537,6 +537,69 @@ __global__ void k_radix_scatter_fixed(
   }
 }
 
+// Direct one-pass scatter variant: one global atomicAdd on the
+// region cursor per event, single read of the input (the block-hist
+// version reads the input twice and does two LDS atomics per event).
+// The region cursors are ~n_regions hot words in L2; with ≥2^11
+// regions the per-address serialization is far below the memory
+// time.  Selected via BYTEWAX_SCATTER_DIRECT (measured A/B; see
+// profiles/).
+template <int MODE, typename TS = int64_t>
+__global__ void k_radix_scatter_direct(
+    const int32_t* __restrict__ keys,
+    const TS* __restrict__ ts,
+    const int64_t* __restrict__ vals,
+    int64_t n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    uint64_t mask,
+    int region_bits,
+    int64_t cap,
+    int* __restrict__ gcursors,
+    uint64_t* __restrict__ ev_packed,
+    int64_t* __restrict__ ev_vals,
+    int* __restrict__ ov_cursor,
+    uint64_t* __restrict__ ov_packed,
+    int64_t* __restrict__ ov_vals,
+    int64_t ov_cap,
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag) {
+  int64_t start = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  int64_t local_max = 0;
+  for (int64_t i = start; i < n; i += stride) {
+    int64_t t = (int64_t)ts[i] + ts_base;
+    if (t > local_max) local_max = t;
+    int64_t win = (t - align_ms) / len_ms;
+    uint64_t packed =
+        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    int b = (int)region_of(mix64(packed), mask, region_bits);
+    int64_t in_bucket = atomicAdd(&gcursors[b], 1);
+    int64_t v = (MODE == AGG_SUM) ? vals[i] : 0;
+    if (in_bucket < cap) {
+      int64_t pos = (int64_t)b * cap + in_bucket;
+      ev_packed[pos] = packed;
+      if (MODE == AGG_SUM) ev_vals[pos] = v;
+    } else {
+      int opos = atomicAdd(ov_cursor, 1);
+      if (opos < ov_cap) {
+        ov_packed[opos] = packed;
+        if (MODE == AGG_SUM) ov_vals[opos] = v;
+      } else {
+        atomicExch(error_flag, 1);
+      }
+    }
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
 // Aggregate the overflow spill straight into the table (tiny for
 // uniform keys).
 template <int MODE>
@@ -1450,6 +1513,10 @@ void radix_window_insert(
     int v = atoi(sb);
     if (v > 0) env_blocks = v;
   }
+  bool direct = false;
+  if (const char* sd = std::getenv("BYTEWAX_SCATTER_DIRECT")) {
+    direct = atoi(sd) != 0;
+  }
 
   struct Seg {
     int64_t off, n, base;
@@ -1467,8 +1534,10 @@ void radix_window_insert(
   }
 
   auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx) {
+    // The direct variant uses no LDS; don't charge it occupancy.
+    size_t lds = direct ? 0 : 2 * hist_lds;
     hipLaunchKernelGGL(
-        kern, dim3(gx), block, 2 * hist_lds, stream,
+        kern, dim3(gx), block, lds, stream,
         keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
         vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
         sg.base, mask, (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
@@ -1486,16 +1555,28 @@ void radix_window_insert(
     if (env_blocks > 0 && (unsigned)env_blocks < gx) gx = (unsigned)env_blocks;
     if (seg32) {
       const int32_t* t32 = ts.data_ptr<int32_t>();
-      if (mode == AGG_COUNT)
+      if (direct) {
+        if (mode == AGG_COUNT)
+          scat(k_radix_scatter_direct<AGG_COUNT, int32_t>, t32, sg, gx);
+        else
+          scat(k_radix_scatter_direct<AGG_SUM, int32_t>, t32, sg, gx);
+      } else if (mode == AGG_COUNT) {
         scat(k_radix_scatter_fixed<AGG_COUNT, int32_t>, t32, sg, gx);
-      else
+      } else {
         scat(k_radix_scatter_fixed<AGG_SUM, int32_t>, t32, sg, gx);
+      }
     } else {
       const int64_t* t64 = ts.data_ptr<int64_t>();
-      if (mode == AGG_COUNT)
+      if (direct) {
+        if (mode == AGG_COUNT)
+          scat(k_radix_scatter_direct<AGG_COUNT, int64_t>, t64, sg, gx);
+        else
+          scat(k_radix_scatter_direct<AGG_SUM, int64_t>, t64, sg, gx);
+      } else if (mode == AGG_COUNT) {
         scat(k_radix_scatter_fixed<AGG_COUNT, int64_t>, t64, sg, gx);
-      else
+      } else {
         scat(k_radix_scatter_fixed<AGG_SUM, int64_t>, t64, sg, gx);
+      }
     }
   }
 
