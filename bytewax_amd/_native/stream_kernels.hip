@@ -455,6 +455,44 @@ __global__ __launch_bounds__(256) void k_agg_cached(
 // that would overflow its region's buffer goes to the overflow spill
 // (aggregated by k_radix_agg's caller via the contended-direct path —
 // statistically empty for uniform keys at cap = 2x mean).
+// GCN has no integer divide; `x / len_ms` compiles to a ~60-cycle
+// sequence and the scatter does it twice per event.  The host
+// precomputes a Granlund-Montgomery magic multiplier instead:
+// win = mulhi64(x, m2) is exact for 0 <= x < maxfast (m2 == 0 or
+// x outside the window falls back to the hardware-free division).
+__device__ __forceinline__ int64_t win_of(
+    int64_t t, int64_t align_ms, int64_t len_ms, uint64_t m2,
+    uint64_t maxfast) {
+  int64_t x = t - align_ms;
+  if (m2 != 0 && (uint64_t)x < maxfast) {
+    return (int64_t)__umul64hi((uint64_t)x, m2);
+  }
+  return x / len_ms;  // same truncation semantics as before
+}
+
+// Host side of the magic divider: p = 60 covers numerators < 2^60/e
+// (ms-scale timestamps are < 2^43).  Power-of-two lengths reduce to a
+// plain mulhi shift; len <= 1 disables the fast path.
+static inline void magic_div_u64(
+    int64_t d, uint64_t* m2, uint64_t* maxfast) {
+  *m2 = 0;
+  *maxfast = 0;
+  if (d <= 1) return;
+  if ((d & (d - 1)) == 0) {
+    int s = 0;
+    while ((int64_t(1) << s) < d) ++s;
+    *m2 = (uint64_t)1 << (64 - s);
+    *maxfast = ~(uint64_t)0;
+    return;
+  }
+  const unsigned p = 60;
+  unsigned __int128 one = 1;
+  uint64_t M = (uint64_t)(((one << p) + (uint64_t)d - 1) / (uint64_t)d);
+  uint64_t e = (uint64_t)((unsigned __int128)M * (uint64_t)d - (one << p));
+  *m2 = M << 4;  // M < 2^60 for d >= 2, so the shift is exact
+  *maxfast = e ? (uint64_t)((one << p) / e) : ~(uint64_t)0;
+}
+
 // TS is int64_t (absolute ms) or int32_t (deltas from `ts_base`, the
 // wire format of the RCCL exchange — inserting the received segments
 // directly skips the int64 timestamp rebuild entirely).
@@ -478,7 +516,9 @@ __global__ void k_radix_scatter_fixed(
     int64_t* __restrict__ ov_vals,
     int64_t ov_cap,
     unsigned long long* __restrict__ max_ts,
-    int* __restrict__ error_flag) {
+    int* __restrict__ error_flag,
+    uint64_t win_m2,
+    uint64_t win_maxfast) {
   extern __shared__ int lmem[];
   int nb = (int)(((mask + 1) >> region_bits));
   int* lhist = lmem;
@@ -491,7 +531,7 @@ __global__ void k_radix_scatter_fixed(
   for (int64_t i = start; i < n; i += stride) {
     int64_t t = (int64_t)ts[i] + ts_base;
     if (t > local_max) local_max = t;
-    int64_t win = (t - align_ms) / len_ms;
+    int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
     uint64_t packed =
         ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
     atomicAdd(&lhist[(int)region_of(mix64(packed), mask, region_bits)], 1);
@@ -508,7 +548,8 @@ __global__ void k_radix_scatter_fixed(
   }
   __syncthreads();
   for (int64_t i = start; i < n; i += stride) {
-    int64_t win = ((int64_t)ts[i] + ts_base - align_ms) / len_ms;
+    int64_t win = win_of(
+        (int64_t)ts[i] + ts_base, align_ms, len_ms, win_m2, win_maxfast);
     uint64_t packed =
         ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
     int b = (int)region_of(mix64(packed), mask, region_bits);
@@ -564,14 +605,16 @@ __global__ void k_radix_scatter_direct(
     int64_t* __restrict__ ov_vals,
     int64_t ov_cap,
     unsigned long long* __restrict__ max_ts,
-    int* __restrict__ error_flag) {
+    int* __restrict__ error_flag,
+    uint64_t win_m2,
+    uint64_t win_maxfast) {
   int64_t start = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   int64_t local_max = 0;
   for (int64_t i = start; i < n; i += stride) {
     int64_t t = (int64_t)ts[i] + ts_base;
     if (t > local_max) local_max = t;
-    int64_t win = (t - align_ms) / len_ms;
+    int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
     uint64_t packed =
         ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
     int b = (int)region_of(mix64(packed), mask, region_bits);
@@ -1533,6 +1576,8 @@ void radix_window_insert(
     segs.push_back({0, n, ts_base});
   }
 
+  uint64_t win_m2, win_maxfast;
+  magic_div_u64(len_ms, &win_m2, &win_maxfast);
   auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx) {
     // The direct variant uses no LDS; don't charge it occupancy.
     size_t lds = direct ? 0 : 2 * hist_lds;
@@ -1548,7 +1593,7 @@ void radix_window_insert(
         mode == AGG_SUM ? ov_vals.data_ptr<int64_t>() : nullptr,
         ov_packed.numel(),
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
-        error_flag.data_ptr<int32_t>());
+        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
   };
   for (const Seg& sg : segs) {
     unsigned gx = (unsigned)n_blocks(sg.n, 256);
@@ -1801,6 +1846,8 @@ void radix_stats_insert(
   gcursors.narrow(0, 0, nb).zero_();
   ov_cursor.zero_();
   size_t hist_lds = (size_t)nb * sizeof(int);
+  uint64_t win_m2, win_maxfast;
+  magic_div_u64(len_ms, &win_m2, &win_maxfast);
   hipLaunchKernelGGL(
       k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
       keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
@@ -1811,7 +1858,7 @@ void radix_stats_insert(
       (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
       ov_packed.numel(),
       (unsigned long long*)max_ts.data_ptr<int64_t>(),
-      error_flag.data_ptr<int32_t>());
+      error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
 
   auto offsets = at::arange(
       nb, at::TensorOptions().dtype(at::kInt).device(keys.device()));
@@ -1989,7 +2036,7 @@ void radix_join_insert(
       (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
       ov_packed.numel(),
       (unsigned long long*)max_ts_scratch.data_ptr<int64_t>(),
-      error_flag.data_ptr<int32_t>());
+      error_flag.data_ptr<int32_t>(), (uint64_t)1 << 24, ~(uint64_t)0);
   hipLaunchKernelGGL(
       k_join_region, dim3((unsigned)nb), block, 0, stream,
       (const uint64_t*)ev_packed.data_ptr<int64_t>(),
