@@ -951,3 +951,40 @@ class WindowAggState:
             )
         self.max_ts_host = snap["max_ts"]
         self.closed_horizon = snap["closed_horizon"]
+
+    def reinsert_labeled(self, batch: RecordBatch) -> None:
+        """Re-add spilled accumulator rows (rescale re-exchange).
+
+        ``batch.ts`` carries window LABELS (win * len + align, the
+        snapshot/emission format), and ``vals`` are saved accumulators
+        — added with SUM semantics regardless of mode, exactly like
+        :meth:`restore_from_host`.  Does not advance the watermark or
+        the closed horizon (the caller merges those)."""
+        if len(batch) == 0:
+            return
+        if self.cpu:
+            wins = (
+                (batch.ts + batch.ts_base - self.align_ms) // self.len_ms
+            ).tolist()
+            for k, w, v in zip(
+                batch.keys.tolist(), wins, batch.vals.tolist()
+            ):
+                kw = (int(k), int(w))
+                self._table[kw] = self._table.get(kw, 0) + int(v)
+            return
+        self.k.window_agg_insert(
+            batch.keys,
+            batch.ts,
+            batch.vals,
+            self.tkeys,
+            self.tvals,
+            self.max_ts_dev,
+            self.error_flag,
+            self.align_ms,
+            self.len_ms,
+            AGG_SUM,  # re-add saved accumulators regardless of mode
+            False,
+            batch.ts_base,
+            self.region_bits,
+            self.len_ms,
+        )

@@ -179,11 +179,21 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         wait_ms: int,
         exchange: bool,
         resume: Optional[Dict[str, Any]],
+        shard: Optional[str] = None,
+        world: int = 1,
+        registry: Optional[Any] = None,
     ):
         self.state = state
         self.wait_ms = wait_ms
         self.exchange = exchange
         self.pending: Optional[tuple] = None  # (works, RecordBatch)
+        # Rescale support: `registry` collects spilled rows from
+        # donor logics (shards restored at a different world size);
+        # applied via one aligned exchange on the first activation.
+        self._shard = shard
+        self._world = world
+        self._registry = registry
+        self._applied_rescale = registry is None
         # Closed rows resolved during snapshot() (which cannot emit):
         # re-emitted on the next activation; persisted in the snapshot
         # so a resume replays them exactly once.
@@ -196,6 +206,8 @@ class _DeviceWindowLogic(StatefulBatchLogic):
             import torch
 
             resume = dict(resume)
+            resume.pop("__world__", None)
+            resume.pop("__shard__", None)
             carry = resume.pop("__carry__", None)
             if carry is not None:
                 dev = self.state.device
@@ -212,6 +224,88 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         if self._carry is not None:
             out.append(self._carry)
             self._carry = None
+
+    def _add_carry(self, batch: Optional[RecordBatch]) -> None:
+        import torch
+
+        if batch is None or len(batch) == 0:
+            return
+        if self._carry is None:
+            self._carry = batch
+        else:
+            self._carry = RecordBatch(
+                torch.cat([self._carry.keys, batch.keys]),
+                torch.cat([self._carry.ts, batch.ts]),
+                torch.cat([self._carry.vals, batch.vals]),
+            )
+
+    def _apply_rescale(self) -> None:
+        """Consume donor-shard snapshots (worker count changed since
+        the snapshot was taken): every restored accumulator row is
+        re-exchanged by key hash so it lands on its new owning worker,
+        then re-added with SUM semantics.  Runs exactly once, on the
+        first activation; with the exchange enabled EVERY rank calls
+        the collective exactly once (empty batches are fine), so the
+        all-to-all stays aligned whether or not this rank held
+        donors."""
+        if self._applied_rescale:
+            return
+        self._applied_rescale = True
+        import numpy as np
+        import torch
+
+        rt = self._registry
+        dev = self.state.device
+        keys_l, ts_l, vals_l = [], [], []
+        max_ts = None
+        for entry in rt.rescale_rows:
+            if entry["consumed"]:
+                continue
+            entry["consumed"] = True
+            snap = entry["snap"]
+            if len(snap.get("keys", ())):
+                keys_l.append(np.asarray(snap["keys"], dtype="int32"))
+                wins = np.asarray(snap["wins"], dtype="int64")
+                ts_l.append(
+                    wins * self.state.len_ms + self.state.align_ms
+                )
+                vals_l.append(np.asarray(snap["vals"], dtype="int64"))
+            c = snap.get("__carry__")
+            if c is not None and len(c["keys"]):
+                self._add_carry(
+                    RecordBatch(
+                        torch.as_tensor(c["keys"]).to(dev),
+                        torch.as_tensor(c["ts"]).to(dev),
+                        torch.as_tensor(c["vals"]).to(dev),
+                    )
+                )
+            mts = snap.get("max_ts")
+            if mts is not None:
+                max_ts = mts if max_ts is None else max(max_ts, mts)
+        if not self.exchange and not keys_l:
+            return
+        if keys_l:
+            keys_np = np.concatenate(keys_l)
+            ts_np = np.concatenate(ts_l)
+            vals_np = np.concatenate(vals_l)
+            batch = RecordBatch(
+                torch.as_tensor(keys_np).to(dev),
+                torch.as_tensor(ts_np).to(dev),
+                torch.as_tensor(vals_np).to(dev),
+                max_ts=int(ts_np.max()),
+            )
+        else:
+            batch = RecordBatch(
+                torch.zeros(0, dtype=torch.int32, device=dev),
+                torch.zeros(0, dtype=torch.int64, device=dev),
+                torch.zeros(0, dtype=torch.int64, device=dev),
+                max_ts=0,
+            )
+        if self.exchange:
+            batch = exchange_by_key(batch)
+        self.state.reinsert_labeled(batch)
+        if max_ts is not None and max_ts > self.state.max_ts_host:
+            self.state.max_ts_host = max_ts
 
     def _flush_pending(self) -> None:
         if self.pending is None:
@@ -231,6 +325,7 @@ class _DeviceWindowLogic(StatefulBatchLogic):
     def on_batch(self, batches: List[RecordBatch]):
         import torch
 
+        self._apply_rescale()
         out: List[RecordBatch] = []
         self._take_carry(out)
         closed = self.state.close_resolve()  # previous step's close
@@ -253,6 +348,7 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         return (out, StatefulBatchLogic.RETAIN)
 
     def on_eof(self):
+        self._apply_rescale()
         out: List[RecordBatch] = []
         self._take_carry(out)
         self._flush_pending()
@@ -278,15 +374,74 @@ class _DeviceWindowLogic(StatefulBatchLogic):
                 )
             else:
                 self._carry = resolved
-        snap = self.state.snapshot_to_host()
+        snap = dict(self.state.snapshot_to_host())
         if self._carry is not None:
-            snap = dict(snap)
             snap["__carry__"] = {
                 "keys": self._carry.keys.cpu().numpy().copy(),
                 "ts": self._carry.ts.cpu().numpy().copy(),
                 "vals": self._carry.vals.cpu().numpy().copy(),
             }
+        if self._shard is not None:
+            snap["__shard__"] = self._shard
+            snap["__world__"] = self._world
         return snap
+
+
+class _DonorLogic(StatefulBatchLogic):
+    """A shard restored under a different cluster shape.
+
+    Registers its spilled rows with the worker's rescale registry (the
+    active shard re-exchanges and re-adds them on its first
+    activation) and then lingers, re-snapshotting the rows each epoch,
+    until consumption is durable: it only DISCARDs after the active
+    shard has consumed the rows, so the donor rows and the merged
+    state always change hands inside one atomic epoch commit
+    (exactly-once across crashes at any point)."""
+
+    def __init__(self, snap: Dict[str, Any], rt: Any):
+        self._entry = {"snap": snap, "consumed": False}
+        rt.rescale_rows.append(self._entry)
+
+    def on_batch(self, batches):
+        # Donors hold foreign shard names; live batches always carry
+        # the worker's own shard key.  Receiving one means the shard
+        # keying is broken — dropping data silently is worse.
+        msg = "donor shard received live batches; shard keying broken"
+        raise RuntimeError(msg)
+
+    def on_notify(self):
+        if self._entry["consumed"]:
+            return ([], StatefulBatchLogic.DISCARD)
+        return ([], StatefulBatchLogic.RETAIN)
+
+    def notify_at(self):
+        from datetime import datetime, timezone
+
+        return datetime.now(timezone.utc)
+
+    def snapshot(self) -> Dict[str, Any]:
+        if self._entry["consumed"]:
+            return {"__consumed__": True}
+        return self._entry["snap"]
+
+
+class _ConsumedDonor(StatefulBatchLogic):
+    """Tombstone for an already-consumed donor snapshot: discards on
+    the first timer pass (writing the discard into recovery)."""
+
+    def on_batch(self, batches):  # pragma: no cover — never routed
+        return ([], StatefulBatchLogic.RETAIN)
+
+    def on_notify(self):
+        return ([], StatefulBatchLogic.DISCARD)
+
+    def notify_at(self):
+        from datetime import datetime, timezone
+
+        return datetime.now(timezone.utc)
+
+    def snapshot(self) -> Dict[str, Any]:
+        return {"__consumed__": True}
 
 
 @operator
@@ -331,6 +486,8 @@ def keyed_window_agg(
     :arg exchange: Force the RCCL exchange on/off; default: on iff
         torch.distributed is initialized with world > 1.
     """
+    import threading
+
     import torch
 
     agg_mode = {"count": AGG_COUNT, "sum": AGG_SUM}[mode]
@@ -340,13 +497,49 @@ def keyed_window_agg(
     wait_ms = int(wait.total_seconds() * 1000)
     sliding = off_ms < len_ms
 
-    def make_exchange_flag() -> bool:
-        import torch.distributed as dist
+    # Per-worker-thread runtime: the shard key is resolved LAZILY (the
+    # process group may not exist yet while the user builds the flow),
+    # and the rescale registry collects donor-shard rows (see
+    # `_DeviceWindowLogic._apply_rescale`).
+    _rt = threading.local()
 
-        return dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1
+    def _runtime():
+        if not hasattr(_rt, "shard"):
+            import torch.distributed as dist
+
+            if dist.is_available() and dist.is_initialized():
+                _rt.world = dist.get_world_size()
+                _rt.shard = f"shard-{dist.get_rank()}"
+            else:
+                _rt.world = 1
+                _rt.shard = "shard-0"
+            _rt.rescale_rows = []
+        return _rt
 
     def shim_builder(resume_state):
-        ex = exchange if exchange is not None else make_exchange_flag()
+        rt = _runtime()
+        ex = exchange if exchange is not None else rt.world > 1
+        if resume_state is not None:
+            if resume_state.get("__consumed__"):
+                return _ConsumedDonor()
+            snap_world = resume_state.get("__world__", rt.world)
+            snap_shard = resume_state.get("__shard__", rt.shard)
+            if snap_world != rt.world:
+                # State written under a different cluster shape: its
+                # rows must re-exchange (key ownership changed).
+                if snap_shard != rt.shard:
+                    # Foreign shard name: a pure donor.
+                    return _DonorLogic(resume_state, rt)
+                # Same shard name as this worker's active key: the
+                # engine will route live batches to THIS logic, so it
+                # must be a real one — start it empty and donate the
+                # old rows to the registry like any other donor.
+                rt.rescale_rows.append(
+                    {"snap": resume_state, "consumed": False}
+                )
+                resume_state = None
+            elif snap_shard != rt.shard:
+                return _DonorLogic(resume_state, rt)
         dev = torch.device(device)
         state = WindowAggState(
             dev,
@@ -360,16 +553,17 @@ def keyed_window_agg(
             region_bits=region_bits,
             off_ms=off_ms,
         )
-        return _DeviceWindowLogic(state, wait_ms, ex, resume_state)
+        return _DeviceWindowLogic(
+            state,
+            wait_ms,
+            ex,
+            resume_state,
+            shard=rt.shard,
+            world=rt.world,
+            registry=rt,
+        )
 
-    import torch.distributed as dist
-
-    shard = (
-        f"shard-{dist.get_rank()}"
-        if dist.is_available() and dist.is_initialized()
-        else "shard-0"
-    )
-    keyed = op.map("wrap", up, lambda b: (shard, b))
+    keyed = op.map("wrap", up, lambda b: (_runtime().shard, b))
     agg = op.stateful_batch("agg", keyed, shim_builder)
     return op.map("unwrap", agg, lambda kv: kv[1])
 

@@ -1,0 +1,245 @@
+"""Device-shard rescale: window state snapshotted under one cluster
+shape resumes correctly under another (donor shards re-exchange their
+rows to the new owners; reference pytests/test_recovery.py rescale
+semantics applied to the columnar path)."""
+
+import os
+import subprocess
+import sys
+import textwrap
+from pathlib import Path
+from types import SimpleNamespace
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+from bytewax_amd.gpu import AGG_COUNT, RecordBatch, WindowAggState  # noqa: E402
+from bytewax_amd.gpu.operators import (  # noqa: E402
+    _ConsumedDonor,
+    _DeviceWindowLogic,
+    _DonorLogic,
+)
+from bytewax_amd.operators import StatefulBatchLogic  # noqa: E402
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def _mk_state():
+    return WindowAggState(torch.device("cpu"), 0, 100, AGG_COUNT)
+
+
+def _batch(keys, ts, max_ts=None):
+    return RecordBatch(
+        torch.tensor(keys, dtype=torch.int32),
+        torch.tensor(ts, dtype=torch.int64),
+        None,
+        max_ts=max_ts,
+    )
+
+
+def test_donor_rows_merge_into_active_shard():
+    # "Old world" shard-1 accumulated counts for keys 1 and 2.
+    old = _mk_state()
+    old.insert(_batch([1, 1, 2], [10, 20, 30], max_ts=30))
+    snap = dict(old.snapshot_to_host())
+    snap["__world__"] = 2
+    snap["__shard__"] = "shard-1"
+
+    rt = SimpleNamespace(rescale_rows=[])
+    donor = _DonorLogic(snap, rt)
+    active = _DeviceWindowLogic(
+        _mk_state(), 0, False, None, shard="shard-0", world=1, registry=rt
+    )
+    # New-world traffic for the same window.
+    out, _ = active.on_batch([_batch([1, 3], [40, 50], max_ts=50)])
+    assert out == []
+    final, _ = active.on_eof()
+    rows = sorted(
+        (k, t, v)
+        for b in final
+        for k, t, v in zip(
+            b.keys.tolist(), b.ts.tolist(), b.vals.tolist()
+        )
+    )
+    # Counts merged: key1 = 2(donated)+1, key2 = 1 donated, key3 = 1.
+    assert rows == [(1, 0, 3), (2, 0, 1), (3, 0, 1)]
+    # Donor discards only after consumption (atomic handoff).
+    assert donor.on_notify() == ([], StatefulBatchLogic.DISCARD)
+    assert donor.snapshot() == {"__consumed__": True}
+
+
+def test_unconsumed_donor_retains_and_resnapshots():
+    old = _mk_state()
+    old.insert(_batch([5], [10], max_ts=10))
+    snap = dict(old.snapshot_to_host())
+    snap["__world__"] = 4
+    snap["__shard__"] = "shard-3"
+    rt = SimpleNamespace(rescale_rows=[])
+    donor = _DonorLogic(snap, rt)
+    assert donor.on_notify() == ([], StatefulBatchLogic.RETAIN)
+    assert donor.snapshot() is snap  # rows persist until handed off
+    assert donor.notify_at() is not None
+
+
+def test_consumed_tombstone_discards():
+    t = _ConsumedDonor()
+    assert t.on_notify() == ([], StatefulBatchLogic.DISCARD)
+    assert t.snapshot() == {"__consumed__": True}
+
+
+_FLOW_TEMPLATE = """
+import os
+
+import torch
+
+import bytewax_amd.operators as op
+from bytewax_amd.dataflow import Dataflow
+from bytewax_amd.gpu import RecordBatch
+from bytewax_amd.gpu.operators import keyed_window_agg
+from bytewax_amd.inputs import (
+    AbortExecution,
+    FixedPartitionedSource,
+    StatefulSourcePartition,
+)
+from bytewax_amd.outputs import DynamicSink, StatelessSinkPartition
+from datetime import datetime, timedelta, timezone
+
+PHASE = int(os.environ["RESCALE_PHASE"])
+OUT = os.environ["RESCALE_OUT"]
+ALIGN = datetime(2024, 1, 1, tzinfo=timezone.utc)
+ALIGN_MS = int(ALIGN.timestamp() * 1000)
+N_BATCHES = 4
+ABORT_AT = 3
+
+
+class _Part(StatefulSourcePartition):
+    def __init__(self, resume):
+        self.i = resume if resume is not None else 0
+
+    def next_batch(self):
+        if PHASE == 1 and self.i == ABORT_AT:
+            raise AbortExecution()
+        if self.i >= N_BATCHES:
+            raise StopIteration()
+        self.i += 1
+        keys = torch.arange(10, dtype=torch.int32)
+        ts = torch.full((10,), ALIGN_MS + 1000, dtype=torch.int64)
+        return [RecordBatch(keys, ts, None, max_ts=ALIGN_MS + 1000)]
+
+    def snapshot(self):
+        return self.i
+
+
+class _Src(FixedPartitionedSource):
+    def list_parts(self):
+        return ["p0", "p1"]
+
+    def build_part(self, step_id, part, resume):
+        return _Part(resume)
+
+
+class _Collect(StatelessSinkPartition):
+    def write_batch(self, items):
+        with open(OUT, "a") as f:
+            for b in items:
+                for k, t, v in zip(
+                    b.keys.tolist(), b.ts.tolist(), b.vals.tolist()
+                ):
+                    f.write(f"{k},{t},{v}\\n")
+
+
+class _Sink(DynamicSink):
+    def build(self, step_id, worker_index, worker_count):
+        return _Collect()
+
+
+flow = Dataflow("rescale")
+s = op.input("inp", flow, _Src())
+agg = keyed_window_agg(
+    "win",
+    s,
+    align_to=ALIGN,
+    length=timedelta(seconds=60),
+    mode="count",
+    device="cpu",
+)
+op.output("out", agg, _Sink())
+"""
+
+
+@pytest.mark.timeout(240)
+def test_rescale_two_procs_to_one(tmp_path: Path):
+    flow_file = tmp_path / "rescale_flow.py"
+    flow_file.write_text(textwrap.dedent(_FLOW_TEMPLATE))
+    out_file = tmp_path / "rows.txt"
+    rec_dir = tmp_path / "rec"
+    rec_dir.mkdir()
+    subprocess.run(
+        [sys.executable, "-m", "bytewax_amd.recovery", str(rec_dir), "2"],
+        check=True,
+        env={**os.environ, "PYTHONPATH": str(REPO)},
+        capture_output=True,
+    )
+    port = 29450 + os.getpid() % 400
+    addresses = f"127.0.0.1:{port};127.0.0.1:{port + 1}"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env["RESCALE_PHASE"] = "1"
+    env["RESCALE_OUT"] = str(out_file)
+    procs = [
+        subprocess.Popen(
+            [
+                sys.executable,
+                "-m",
+                "bytewax_amd.run",
+                f"{flow_file}:flow",
+                "-i",
+                str(i),
+                "-a",
+                addresses,
+                "-r",
+                str(rec_dir),
+                "-s",
+                "0",
+                "-b",
+                "0",
+            ],
+            env=env,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE,
+        )
+        for i in range(2)
+    ]
+    for p in procs:
+        stdout, stderr = p.communicate(timeout=200)
+        assert p.returncode == 0, stderr.decode()[-2000:]
+
+    # Phase 2: resume the same recovery store with ONE process.
+    env2 = dict(env)
+    env2["RESCALE_PHASE"] = "2"
+    res = subprocess.run(
+        [
+            sys.executable,
+            "-m",
+            "bytewax_amd.run",
+            f"{flow_file}:flow",
+            "-r",
+            str(rec_dir),
+            "-s",
+            "0",
+            "-b",
+            "0",
+        ],
+        env=env2,
+        capture_output=True,
+        timeout=200,
+    )
+    assert res.returncode == 0, res.stderr.decode()[-2000:]
+
+    totals = {}
+    for line in out_file.read_text().splitlines():
+        k, t, v = line.split(",")
+        totals[int(k)] = totals.get(int(k), 0) + int(v)
+    # 2 parts x 4 batches, one event per key per batch, exactly once.
+    assert totals == {k: 8 for k in range(10)}
