@@ -243,3 +243,55 @@ def test_rescale_two_procs_to_one(tmp_path: Path):
         totals[int(k)] = totals.get(int(k), 0) + int(v)
     # 2 parts x 4 batches, one event per key per batch, exactly once.
     assert totals == {k: 8 for k in range(10)}
+
+
+@pytest.mark.gpu
+def test_donor_rows_merge_on_device():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    dev = torch.device("cuda:0")
+
+    def mk():
+        return WindowAggState(
+            dev, 0, 100, AGG_COUNT, slots_pow=12, radix=True,
+            region_bits=8, max_batch=1024,
+        )
+
+    old = mk()
+    old.insert(
+        RecordBatch(
+            torch.tensor([1, 1, 2], dtype=torch.int32, device=dev),
+            torch.tensor([10, 20, 30], dtype=torch.int64, device=dev),
+            None,
+            max_ts=30,
+        )
+    )
+    snap = dict(old.snapshot_to_host())
+    snap["__world__"] = 2
+    snap["__shard__"] = "shard-1"
+    rt = SimpleNamespace(rescale_rows=[])
+    _DonorLogic(snap, rt)
+    active = _DeviceWindowLogic(
+        mk(), 0, False, None, shard="shard-0", world=1, registry=rt
+    )
+    out, _ = active.on_batch(
+        [
+            RecordBatch(
+                torch.tensor([1, 3], dtype=torch.int32, device=dev),
+                torch.tensor([40, 50], dtype=torch.int64, device=dev),
+                None,
+                max_ts=50,
+            )
+        ]
+    )
+    final, _ = active.on_eof()
+    rows = sorted(
+        (k, t, v)
+        for b in out + final
+        for k, t, v in zip(
+            b.keys.cpu().tolist(),
+            b.ts.cpu().tolist(),
+            b.vals.cpu().tolist(),
+        )
+    )
+    assert rows == [(1, 0, 3), (2, 0, 1), (3, 0, 1)]
