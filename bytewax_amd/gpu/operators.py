@@ -568,17 +568,111 @@ def keyed_window_agg(
     return op.map("unwrap", agg, lambda kv: kv[1])
 
 
+_STATS_COLS = ("keys", "wins", "cnt", "sum", "min", "max")
+
+
 class _DeviceStatsLogic(StatefulBatchLogic):
     """Holds the HBM stats table (1BRC-style count/sum/min/max)."""
 
-    def __init__(self, state, wait_ms: int, exchange: bool, resume):
+    def __init__(
+        self,
+        state,
+        wait_ms: int,
+        exchange: bool,
+        resume,
+        shard: Optional[str] = None,
+        world: int = 1,
+        registry: Optional[Any] = None,
+    ):
         self.state = state
         self.wait_ms = wait_ms
         self.exchange = exchange
+        self._shard = shard
+        self._world = world
+        self._registry = registry
+        self._applied_rescale = registry is None
         if resume is not None:
+            resume = dict(resume)
+            resume.pop("__world__", None)
+            resume.pop("__shard__", None)
             self.state.restore_from_host(resume)
 
+    def _apply_rescale(self) -> None:
+        """Merge donor-shard stats rows after a worker-count change.
+
+        Stats cells carry four accumulators, so instead of the
+        columnar wire exchange this startup-only path all-gathers the
+        (small) spills as host objects; each rank keeps the rows whose
+        key-hash it now owns and merges them (count/sum add, min/max
+        observe — duplicate-safe).  Collective: with the exchange on,
+        EVERY rank calls the all_gather exactly once."""
+        if self._applied_rescale:
+            return
+        self._applied_rescale = True
+        import numpy as np
+        import torch
+
+        from . import _mix64_torch
+
+        rows = []
+        max_ts = None
+        for entry in self._registry.rescale_rows:
+            if entry["consumed"]:
+                continue
+            entry["consumed"] = True
+            snap = entry["snap"]
+            if len(snap.get("keys", ())):
+                rows.append(
+                    {c: np.asarray(snap[c]) for c in _STATS_COLS}
+                )
+            mts = snap.get("max_ts")
+            if mts is not None:
+                max_ts = mts if max_ts is None else max(max_ts, mts)
+        local = (
+            {
+                c: np.concatenate([r[c] for r in rows])
+                for c in _STATS_COLS
+            }
+            if rows
+            else None
+        )
+        if self.exchange:
+            import torch.distributed as dist
+
+            world = dist.get_world_size()
+            rank = dist.get_rank()
+            gathered = [None] * world
+            dist.all_gather_object(gathered, local)
+            parts = []
+            for g in gathered:
+                if g is None:
+                    continue
+                keys_t = torch.as_tensor(g["keys"]).to(torch.int64)
+                mine = (
+                    torch.remainder(_mix64_torch(keys_t), world) == rank
+                ).numpy()
+                if mine.any():
+                    parts.append({c: g[c][mine] for c in _STATS_COLS})
+            merged = (
+                {
+                    c: np.concatenate([p[c] for p in parts])
+                    for c in _STATS_COLS
+                }
+                if parts
+                else None
+            )
+        else:
+            merged = local
+        if merged is not None:
+            self.state.merge_rows(merged)
+        if (
+            max_ts is not None
+            and max_ts > self.state.max_ts_host
+        ):
+            self.state.max_ts_host = max_ts
+
     def on_batch(self, batches):
+        self._apply_rescale()
         out = []
         for batch in batches:
             if self.exchange:
@@ -594,6 +688,7 @@ class _DeviceStatsLogic(StatefulBatchLogic):
         return (out, StatefulBatchLogic.RETAIN)
 
     def on_eof(self):
+        self._apply_rescale()
         closed = self.state.extract(None, clear=True)
         return (
             [closed] if closed is not None else [],
@@ -601,7 +696,11 @@ class _DeviceStatsLogic(StatefulBatchLogic):
         )
 
     def snapshot(self):
-        return self.state.snapshot_to_host()
+        snap = dict(self.state.snapshot_to_host())
+        if self._shard is not None:
+            snap["__shard__"] = self._shard
+            snap["__world__"] = self._world
+        return snap
 
 
 @operator
@@ -623,6 +722,8 @@ def keyed_stats_agg(
     aggregation.  Emits dicts of columnar device tensors
     {keys, wins, cnt, sum, min, max} at window close / EOF.
     """
+    import threading
+
     import torch
 
     from .state import StatsAggState
@@ -632,30 +733,53 @@ def keyed_stats_agg(
     len_ms = int(length.total_seconds() * 1000)
     wait_ms = int(wait.total_seconds() * 1000)
 
-    def shim_builder(resume_state):
-        import torch.distributed as dist
+    _rt = threading.local()
 
-        ex = exchange
-        if ex is None:
-            ex = (
-                dist.is_available()
-                and dist.is_initialized()
-                and dist.get_world_size() > 1
-            )
+    def _runtime():
+        if not hasattr(_rt, "shard"):
+            import torch.distributed as dist
+
+            if dist.is_available() and dist.is_initialized():
+                _rt.world = dist.get_world_size()
+                _rt.shard = f"shard-{dist.get_rank()}"
+            else:
+                _rt.world = 1
+                _rt.shard = "shard-0"
+            _rt.rescale_rows = []
+        return _rt
+
+    def shim_builder(resume_state):
+        rt = _runtime()
+        ex = exchange if exchange is not None else rt.world > 1
+        if resume_state is not None:
+            if resume_state.get("__consumed__"):
+                return _ConsumedDonor()
+            snap_world = resume_state.get("__world__", rt.world)
+            snap_shard = resume_state.get("__shard__", rt.shard)
+            if snap_world != rt.world:
+                if snap_shard != rt.shard:
+                    return _DonorLogic(resume_state, rt)
+                rt.rescale_rows.append(
+                    {"snap": resume_state, "consumed": False}
+                )
+                resume_state = None
+            elif snap_shard != rt.shard:
+                return _DonorLogic(resume_state, rt)
         state = StatsAggState(
             torch.device(device), align_ms, len_ms,
             slots_pow=slots_pow, out_cap=out_cap,
         )
-        return _DeviceStatsLogic(state, wait_ms, ex, resume_state)
+        return _DeviceStatsLogic(
+            state,
+            wait_ms,
+            ex,
+            resume_state,
+            shard=rt.shard,
+            world=rt.world,
+            registry=rt,
+        )
 
-    import torch.distributed as dist
-
-    shard = (
-        f"shard-{dist.get_rank()}"
-        if dist.is_available() and dist.is_initialized()
-        else "shard-0"
-    )
-    keyed = op.map("wrap", up, lambda b: (shard, b))
+    keyed = op.map("wrap", up, lambda b: (_runtime().shard, b))
     agg = op.stateful_batch("agg", keyed, shim_builder)
     return op.map("unwrap", agg, lambda kv: kv[1])
 

@@ -264,11 +264,20 @@ class StatsAggState:
         return out
 
     def restore_from_host(self, snap: Dict[str, Any]) -> None:
-        import torch
-
         self.max_ts_host = snap["max_ts"]
         self.closed_horizon = snap["closed_horizon"]
         if snap["n"] == 0:
+            return
+        self.merge_rows(snap)
+
+    def merge_rows(self, snap: Dict[str, Any]) -> None:
+        """Additively merge saved stats rows into the live table
+        (count/sum add, min/max observe).  Correct for occupied cells
+        and duplicate rows, so rescale can concatenate multiple donor
+        shards' spills before merging."""
+        import torch
+
+        if len(snap.get("keys", ())) == 0:
             return
         if self.cpu:
             for k, w, c, s, mn, mx in zip(
@@ -279,7 +288,17 @@ class StatsAggState:
                 snap["min"].tolist(),
                 snap["max"].tolist(),
             ):
-                self._table[(int(k), int(w))] = (c, s, mn, mx)
+                kw = (int(k), int(w))
+                cur = self._table.get(kw)
+                if cur is None:
+                    self._table[kw] = (c, s, mn, mx)
+                else:
+                    self._table[kw] = (
+                        cur[0] + c,
+                        cur[1] + s,
+                        min(cur[2], mn),
+                        max(cur[3], mx),
+                    )
             return
         wins = torch.as_tensor(snap["wins"]).to(torch.int64)
         ts = (wins * self.len_ms + self.align_ms).to(self.device)

@@ -295,3 +295,205 @@ def test_donor_rows_merge_on_device():
         )
     )
     assert rows == [(1, 0, 3), (2, 0, 1), (3, 0, 1)]
+
+
+def test_stats_donor_rows_merge_with_duplicates():
+    """Stats rescale: donors' spills merge additively (count/sum add,
+    min/max observe), including two donors holding the same cell."""
+    from bytewax_amd.gpu.state import StatsAggState
+    from bytewax_amd.gpu.operators import _DeviceStatsLogic
+
+    def mk():
+        return StatsAggState(torch.device("cpu"), 0, 1 << 40)
+
+    def snap_of(keys, vals, world, shard):
+        st = mk()
+        st.insert(
+            RecordBatch(
+                torch.tensor(keys, dtype=torch.int32),
+                torch.zeros(len(keys), dtype=torch.int64),
+                torch.tensor(vals, dtype=torch.int64),
+                max_ts=1,
+            )
+        )
+        s = dict(st.snapshot_to_host())
+        s["__world__"] = world
+        s["__shard__"] = shard
+        return s
+
+    rt = SimpleNamespace(rescale_rows=[])
+    # Two old shards both saw key 7 (possible after repeated rescales).
+    _DonorLogic(snap_of([7, 7, 8], [10, 30, 5], 2, "shard-1"), rt)
+    rt.rescale_rows.append(
+        {"snap": snap_of([7, 9], [20, 1], 2, "shard-0"), "consumed": False}
+    )
+    active = _DeviceStatsLogic(
+        mk(), 0, False, None, shard="shard-0", world=1, registry=rt
+    )
+    active.on_batch(
+        [
+            RecordBatch(
+                torch.tensor([7], dtype=torch.int32),
+                torch.zeros(1, dtype=torch.int64),
+                torch.tensor([100], dtype=torch.int64),
+                max_ts=2,
+            )
+        ]
+    )
+    out = active.on_eof()[0][0]
+    got = {
+        k: (c, s, mn, mx)
+        for k, c, s, mn, mx in zip(
+            out["keys"].tolist(),
+            out["cnt"].tolist(),
+            out["sum"].tolist(),
+            out["min"].tolist(),
+            out["max"].tolist(),
+        )
+    }
+    assert got == {
+        7: (4, 160, 10, 100),
+        8: (1, 5, 5, 5),
+        9: (1, 1, 1, 1),
+    }
+
+
+_STATS_FLOW_TEMPLATE = """
+import os
+
+import torch
+
+import bytewax_amd.operators as op
+from bytewax_amd.dataflow import Dataflow
+from bytewax_amd.gpu import RecordBatch
+from bytewax_amd.gpu.operators import keyed_stats_agg
+from bytewax_amd.inputs import (
+    AbortExecution,
+    FixedPartitionedSource,
+    StatefulSourcePartition,
+)
+from bytewax_amd.outputs import DynamicSink, StatelessSinkPartition
+from datetime import datetime, timedelta, timezone
+
+PHASE = int(os.environ["RESCALE_PHASE"])
+OUT = os.environ["RESCALE_OUT"]
+ALIGN = datetime(2024, 1, 1, tzinfo=timezone.utc)
+ALIGN_MS = int(ALIGN.timestamp() * 1000)
+N_BATCHES = 4
+ABORT_AT = 3
+
+
+class _Part(StatefulSourcePartition):
+    def __init__(self, resume):
+        self.i = resume if resume is not None else 0
+
+    def next_batch(self):
+        if PHASE == 1 and self.i == ABORT_AT:
+            raise AbortExecution()
+        if self.i >= N_BATCHES:
+            raise StopIteration()
+        self.i += 1
+        keys = torch.arange(10, dtype=torch.int32)
+        ts = torch.full((10,), ALIGN_MS + 1000, dtype=torch.int64)
+        vals = torch.arange(10, dtype=torch.int64) + self.i
+        return [RecordBatch(keys, ts, vals, max_ts=ALIGN_MS + 1000)]
+
+    def snapshot(self):
+        return self.i
+
+
+class _Src(FixedPartitionedSource):
+    def list_parts(self):
+        return ["p0", "p1"]
+
+    def build_part(self, step_id, part, resume):
+        return _Part(resume)
+
+
+class _Collect(StatelessSinkPartition):
+    def write_batch(self, items):
+        with open(OUT, "a") as f:
+            for d in items:
+                for k, c, s in zip(
+                    d["keys"].tolist(), d["cnt"].tolist(), d["sum"].tolist()
+                ):
+                    f.write(f"{k},{c},{s}\\n")
+
+
+class _Sink(DynamicSink):
+    def build(self, step_id, worker_index, worker_count):
+        return _Collect()
+
+
+flow = Dataflow("rescale_stats")
+s = op.input("inp", flow, _Src())
+agg = keyed_stats_agg(
+    "stats",
+    s,
+    align_to=ALIGN,
+    length=timedelta(days=3650),
+    device="cpu",
+)
+op.output("out", agg, _Sink())
+"""
+
+
+@pytest.mark.timeout(240)
+def test_stats_rescale_two_procs_to_one(tmp_path: Path):
+    flow_file = tmp_path / "rescale_stats_flow.py"
+    flow_file.write_text(textwrap.dedent(_STATS_FLOW_TEMPLATE))
+    out_file = tmp_path / "rows.txt"
+    rec_dir = tmp_path / "rec"
+    rec_dir.mkdir()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env["RESCALE_PHASE"] = "1"
+    env["RESCALE_OUT"] = str(out_file)
+    subprocess.run(
+        [sys.executable, "-m", "bytewax_amd.recovery", str(rec_dir), "2"],
+        check=True,
+        env=env,
+        capture_output=True,
+    )
+    port = 29860 + os.getpid() % 100
+    addresses = f"127.0.0.1:{port};127.0.0.1:{port + 1}"
+    procs = [
+        subprocess.Popen(
+            [
+                sys.executable, "-m", "bytewax_amd.run",
+                f"{flow_file}:flow", "-i", str(i), "-a", addresses,
+                "-r", str(rec_dir), "-s", "0", "-b", "0",
+            ],
+            env=env,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE,
+        )
+        for i in range(2)
+    ]
+    for p in procs:
+        stdout, stderr = p.communicate(timeout=200)
+        assert p.returncode == 0, stderr.decode()[-2000:]
+
+    env2 = dict(env)
+    env2["RESCALE_PHASE"] = "2"
+    res = subprocess.run(
+        [
+            sys.executable, "-m", "bytewax_amd.run",
+            f"{flow_file}:flow",
+            "-r", str(rec_dir), "-s", "0", "-b", "0",
+        ],
+        env=env2,
+        capture_output=True,
+        timeout=200,
+    )
+    assert res.returncode == 0, res.stderr.decode()[-2000:]
+
+    cnts, sums = {}, {}
+    for line in out_file.read_text().splitlines():
+        k, c, s = (int(x) for x in line.split(","))
+        cnts[k] = cnts.get(k, 0) + c
+        sums[k] = sums.get(k, 0) + s
+    # 2 parts x 4 batches: count 8 per key; vals k+i for i in 1..4
+    # twice: sum = 2 * (4k + 10).
+    assert cnts == {k: 8 for k in range(10)}
+    assert sums == {k: 2 * (4 * k + 10) for k in range(10)}
