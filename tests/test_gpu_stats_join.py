@@ -180,3 +180,23 @@ def test_join_gpu_matches_reference(radix):
     for k, (a, b) in got.items():
         assert a in lvals_by_key[k]
         assert b in rvals_by_key[k]
+
+
+@pytest.mark.gpu
+def test_stats_merge_rows_into_occupied_cells_gpu():
+    """merge_rows (rescale path) must compose with live cells and
+    duplicate rows on device exactly like the host reference."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    align_ms = _ms(ALIGN)
+    keys, ts, vals = _mk_events(20_000, 50, 120_000, seed=9)
+    ref = _ref_stats(keys, ts, vals, align_ms, 60_000)
+    st = StatsAggState(torch.device("cuda:0"), align_ms, 60_000, slots_pow=11)
+    half = 10_000
+    st.insert(RecordBatch(keys[:half].cuda(), ts[:half].cuda(), vals[:half].cuda()))
+    # Spill the second half through a twin and merge its rows in.
+    twin = StatsAggState(torch.device("cuda:0"), align_ms, 60_000, slots_pow=11)
+    twin.insert(RecordBatch(keys[half:].cuda(), ts[half:].cuda(), vals[half:].cuda()))
+    st.merge_rows(twin.snapshot_to_host())
+    got = _got_stats({k: t.cpu() for k, t in st.extract().items()})
+    assert got == ref
