@@ -787,9 +787,11 @@ def keyed_stats_agg(
 class _DeviceJoinLogic(StatefulBatchLogic):
     """Holds the HBM join table; values are (side, RecordBatch)."""
 
-    def __init__(self, state, exchange: bool):
+    def __init__(self, state, exchange: bool, resume=None):
         self.state = state
         self.exchange = exchange
+        if resume is not None:
+            self.state.restore_from_host(resume)
 
     def on_batch(self, side_batches):
         for side, batch in side_batches:
@@ -806,9 +808,9 @@ class _DeviceJoinLogic(StatefulBatchLogic):
         )
 
     def snapshot(self):
-        # Join state snapshots ride the generic host spill: extract
-        # live per-side values.  (Device path; see HashJoinState.)
-        return None
+        # Live single-side cells spill to host; completed pairs were
+        # already emitted and reset, so nothing re-emits on resume.
+        return self.state.snapshot_to_host()
 
     def on_eof(self):
         return ([], StatefulBatchLogic.RETAIN)
@@ -856,6 +858,7 @@ def stream_join(
                 torch.device(device), slots_pow=slots_pow, out_cap=out_cap
             ),
             ex,
+            resume_state,
         )
 
     import torch.distributed as dist

@@ -469,6 +469,71 @@ class HashJoinState:
             self.error_flag,
         )
 
+    def snapshot_to_host(self) -> Dict[str, Any]:
+        """Spill live (single-side) cells to host for recovery.
+
+        Completed pairs reset their flags on emission
+        (k_join_insert), so the live state is exactly the cells with
+        a nonzero side mask.  Reuses the (drained) output buffers as
+        extraction scratch."""
+        import numpy as np
+        import torch
+
+        if self.cpu:
+            rows = [
+                (k, e[0] or 0, e[1] or 0, e[2])
+                for k, e in self._table.items()
+                if e[2]
+            ]
+            return {
+                "keys": np.array([r[0] for r in rows], dtype="int32"),
+                "v0": np.array([r[1] for r in rows], dtype="int64"),
+                "v1": np.array([r[2] for r in rows], dtype="int64"),
+                "flags": np.array([r[3] for r in rows], dtype="int32"),
+            }
+        if not hasattr(self, "_snap_flags"):
+            self._snap_flags = torch.empty(
+                self.out_cap, dtype=torch.int32, device=self.device
+            )
+            self._snap_n = torch.zeros(
+                1, dtype=torch.int32, device=self.device
+            )
+        self._snap_n.zero_()
+        self.k.join_extract(
+            self.tkeys,
+            self.tval0,
+            self.tval1,
+            self.tflags,
+            self.out_keys,
+            self.out_v0,
+            self.out_v1,
+            self._snap_flags,
+            self._snap_n,
+        )
+        n = int(self._snap_n.item())
+        if n > self.out_cap:
+            msg = f"join snapshot produced {n} rows > out_cap"
+            raise RuntimeError(msg)
+        return {
+            "keys": self.out_keys[:n].cpu().numpy().copy(),
+            "v0": self.out_v0[:n].cpu().numpy().copy(),
+            "v1": self.out_v1[:n].cpu().numpy().copy(),
+            "flags": self._snap_flags[:n].cpu().numpy().copy(),
+        }
+
+    def restore_from_host(self, snap: Dict[str, Any]) -> None:
+        """Re-insert each spilled side presence.  Single-side cells
+        cannot re-complete, so nothing re-emits."""
+        import torch
+
+        flags = snap["flags"]
+        for side, col in ((0, "v0"), (1, "v1")):
+            m = (flags & (1 << side)) != 0
+            if m.any():
+                keys = torch.as_tensor(snap["keys"][m]).to(self.device)
+                vals = torch.as_tensor(snap[col][m]).to(self.device)
+                self.insert(side, keys, vals)
+
     def take_joined(self):
         """Drain completed (key, v0, v1) rows."""
         import torch

@@ -200,3 +200,53 @@ def test_stats_merge_rows_into_occupied_cells_gpu():
     st.merge_rows(twin.snapshot_to_host())
     got = _got_stats({k: t.cpu() for k, t in st.extract().items()})
     assert got == ref
+
+
+def test_join_snapshot_restores_half_pairs_cpu():
+    """Half-pairs survive snapshot/restore and complete exactly once
+    after resume; completed pairs do not re-emit."""
+    st = HashJoinState(torch.device("cpu"))
+    st.insert(0, torch.tensor([1, 2], dtype=torch.int32),
+              torch.tensor([10, 20], dtype=torch.int64))
+    st.insert(1, torch.tensor([2], dtype=torch.int32),
+              torch.tensor([200], dtype=torch.int64))
+    assert st.take_joined()[0].tolist() == [2]  # pair 2 emitted
+    snap = st.snapshot_to_host()
+    # Only key 1 (left side) is live.
+    assert snap["keys"].tolist() == [1]
+    assert snap["flags"].tolist() == [1]
+
+    st2 = HashJoinState(torch.device("cpu"))
+    st2.restore_from_host(snap)
+    assert st2.take_joined() is None  # restore must not emit
+    st2.insert(1, torch.tensor([1], dtype=torch.int32),
+               torch.tensor([100], dtype=torch.int64))
+    keys, v0, v1 = st2.take_joined()
+    assert (keys.tolist(), v0.tolist(), v1.tolist()) == ([1], [10], [100])
+
+
+@pytest.mark.gpu
+def test_join_snapshot_roundtrip_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    torch.manual_seed(5)
+    n = 20_000
+    lk = torch.randint(0, 30_000, (n,), dtype=torch.int32)
+    lv = torch.randint(0, 1000, (n,), dtype=torch.int64)
+    st = HashJoinState(torch.device("cuda:0"), slots_pow=16, out_cap=n * 2)
+    st.insert(0, lk.cuda(), lv.cuda())
+    assert st.take_joined() is None
+    snap = st.snapshot_to_host()
+    assert set(snap["keys"].tolist()) == set(lk.tolist())
+    assert (snap["flags"] == 1).all()
+
+    st2 = HashJoinState(torch.device("cuda:0"), slots_pow=16, out_cap=n * 2)
+    st2.restore_from_host(snap)
+    assert st2.take_joined() is None
+    rk = torch.unique(lk)
+    st2.insert(
+        1, rk.cuda(), torch.full((len(rk),), 7, dtype=torch.int64).cuda()
+    )
+    keys, v0, v1 = st2.take_joined()
+    assert set(keys.cpu().tolist()) == set(rk.tolist())
+    assert (v1.cpu() == 7).all()
