@@ -106,28 +106,30 @@ def test_keyed_window_agg_pipeline_cpu():
 
 
 @pytest.mark.timeout(300)
-def test_bench_two_ranks_gloo(tmp_path: Path):
-    """The exact bench.py path on 2 CPU ranks over gloo: validates the
-    collective alignment (exchange per step, split-size exchange,
-    vote/EOF) that the driver's 8-GPU scale run depends on."""
+@pytest.mark.parametrize("world", [2, 4])
+def test_bench_ranks_gloo(tmp_path: Path, world):
+    """The exact bench.py path on 2 and 4 CPU ranks over gloo:
+    validates the collective alignment (exchange per step, split-size
+    exchange, vote/EOF) that the driver's 8-GPU scale run depends
+    on."""
     repo = Path(__file__).resolve().parent.parent
     env = dict(os.environ)
     env["PYTHONPATH"] = str(repo)
     env["MASTER_ADDR"] = "127.0.0.1"
-    env["MASTER_PORT"] = str(29650 + os.getpid() % 200)
+    env["MASTER_PORT"] = str(29650 + (os.getpid() + world) % 200)
     procs = []
-    for rank in range(2):
+    for rank in range(world):
         e = dict(env)
         e["RANK"] = str(rank)
         e["LOCAL_RANK"] = str(rank)
-        e["WORLD_SIZE"] = "2"
+        e["WORLD_SIZE"] = str(world)
         procs.append(
             subprocess.Popen(
                 [
                     sys.executable,
                     str(repo / "bench.py"),
                     "--gpus",
-                    "2",
+                    str(world),
                     "--steps",
                     "3",
                     "--warmup",
@@ -154,9 +156,9 @@ def test_bench_two_ranks_gloo(tmp_path: Path):
         if ln.startswith("{")
     ][-1]
     res = json.loads(line)
-    assert res["n_gpus"] == 2
+    assert res["n_gpus"] == world
     assert res["value"] > 0
-    assert res["config"]["parallelism"] == "key-hash all-to-allv dp2"
+    assert res["config"]["parallelism"] == f"key-hash all-to-allv dp{world}"
 
 
 def test_filter_batch_cpu():
