@@ -42,6 +42,12 @@ def parse_args():
     )
     p.add_argument("--region-bits", type=int, default=11)
     p.add_argument(
+        "--no-pipeline",
+        action="store_true",
+        help="disable the two-stream scatter/agg pipeline (native "
+        "radix engine)",
+    )
+    p.add_argument(
         "--dedup",
         action="store_true",
         help="wave-level duplicate aggregation (for low-cardinality "
@@ -195,7 +201,8 @@ def main():
                     args.sim_ms_per_batch,
                 ), None
             return state.native_run(
-                part.key_pool, ts_pool, start, count, args.sim_ms_per_batch
+                part.key_pool, ts_pool, start, count,
+                args.sim_ms_per_batch, pipelined=not args.no_pipeline,
             )
 
         # Warmup.

@@ -786,10 +786,16 @@ class WindowAggState:
         n_steps: int,
         sim_ms_per_batch: int,
         wait_ms: int = 0,
+        pipelined: bool = True,
     ):
         """Run `n_steps` of the window pipeline through the native C++
         step loop (no Python between steps).  Single-worker path; the
         multi-GPU exchange path uses the per-step Python loop.
+
+        With ``pipelined`` (radix only), the scatter of step N+1 runs
+        on a second HIP stream into a second region-buffer set while
+        the aggregation of step N drains the first — the agg stage
+        hides behind the scatter.
 
         Returns (closed_rows, step_launch_ns: torch int64 tensor).
         """
@@ -798,6 +804,12 @@ class WindowAggState:
         if self.cpu:
             msg = "native_run requires a device table"
             raise RuntimeError(msg)
+        pipe = pipelined and self.radix and not self.radix_v2
+        if pipe and not hasattr(self, "rx_gcursors2"):
+            self.rx_gcursors2 = torch.zeros_like(self.rx_gcursors)
+            self.rx_packed2 = torch.empty_like(self.rx_packed)
+            self.rx_ov_cursor2 = torch.zeros_like(self.rx_ov_cursor)
+            self.rx_ov_packed2 = torch.empty_like(self.rx_ov_packed)
         step_ns = torch.zeros(n_steps, dtype=torch.int64)
         state_out = torch.zeros(3, dtype=torch.int64)
         rows = self.k.native_run_window_steps(
@@ -837,6 +849,11 @@ class WindowAggState:
             self.v2_gres if self.radix_v2 else None,
             self.v2_ev if self.radix_v2 else None,
             self.v2_ev_res if self.radix_v2 else None,
+            pipe,
+            self.rx_gcursors2 if pipe else None,
+            self.rx_packed2 if pipe else None,
+            self.rx_ov_cursor2 if pipe else None,
+            self.rx_ov_packed2 if pipe else None,
         )
         if int(state_out[2].item()) == 1:
             self.tkeys, self.tkeys_alt = self.tkeys_alt, self.tkeys
