@@ -705,12 +705,21 @@ class _Worker:
                     try:
                         for src in self.input_execs:
                             progressed |= src.poll(now, self._emit)
+                        for sf in self.stateful_execs:
+                            for stream_id, out_items in sf.fire_timers(
+                                now
+                            ):
+                                progressed = True
+                                self._emit(stream_id, out_items)
                     except AbortExecution as ex:
                         abort_exc = ex
-                    for sf in self.stateful_execs:
-                        for stream_id, out_items in sf.fire_timers(now):
-                            progressed = True
-                            self._emit(stream_id, out_items)
+                    except Exception as ex:  # noqa: BLE001
+                        # A user function raised: vote the whole
+                        # cluster into an abort so peers exit cleanly
+                        # (reference run.rs:273-304 panic hook +
+                        # shutdown flag) instead of dying on a broken
+                        # collective; this rank re-raises the error.
+                        abort_exc = ex
                 self._exchange_rounds()
                 local_eof = all(e.eof_all() for e in self.input_execs)
                 close, all_eof, abort = self.ctx.vote_close(

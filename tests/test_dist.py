@@ -198,3 +198,85 @@ def test_testing_module_cluster_launcher(tmp_path: Path):
     )
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     assert sorted(out_file.read_text().splitlines()) == ["x", "y"]
+
+
+@pytest.mark.timeout(180)
+def test_one_rank_failure_aborts_cluster_cleanly(tmp_path: Path):
+    """A user exception on one rank votes the cluster into an abort:
+    the failing rank re-raises, the peer exits cleanly and promptly
+    (no hang on a broken collective). Reference run.rs:273-304 panic
+    hook semantics."""
+    flow_file = tmp_path / "failing_flow.py"
+    flow_file.write_text(
+        textwrap.dedent(
+            """
+            import bytewax_amd.operators as op
+            from bytewax_amd.connectors.stdio import StdOutSink
+            from bytewax_amd.dataflow import Dataflow
+            from bytewax_amd.inputs import (
+                FixedPartitionedSource,
+                StatefulSourcePartition,
+            )
+
+            class _Part(StatefulSourcePartition):
+                def __init__(self, part, resume):
+                    self.part = part
+                    self.i = resume if resume is not None else 0
+
+                def next_batch(self):
+                    if self.i >= 5:
+                        raise StopIteration()
+                    self.i += 1
+                    return [(self.part, self.i)]
+
+                def snapshot(self):
+                    return self.i
+
+            class Src(FixedPartitionedSource):
+                def list_parts(self):
+                    return ["p0", "p1"]
+
+                def build_part(self, step_id, part, resume):
+                    return _Part(part, resume)
+
+            def boom(item):
+                part, i = item
+                if part == "p1" and i == 3:
+                    raise ValueError("poison item")
+                return item
+
+            flow = Dataflow("fail_test")
+            s = op.input("inp", flow, Src())
+            s = op.map("boom", s, boom)
+            op.output("out", s, StdOutSink())
+            """
+        )
+    )
+    port = 29500 + os.getpid() % 300
+    addresses = f"127.0.0.1:{port};127.0.0.1:{port + 1}"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(Path(__file__).resolve().parent.parent)
+    procs = [
+        subprocess.Popen(
+            [
+                sys.executable,
+                "-m",
+                "bytewax_amd.run",
+                f"{flow_file}:flow",
+                "-i",
+                str(i),
+                "-a",
+                addresses,
+            ],
+            env=env,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE,
+        )
+        for i in range(2)
+    ]
+    outs = [p.communicate(timeout=120) for p in procs]
+    rcs = [p.returncode for p in procs]
+    # Parts assign round-robin over sorted names: p1 -> worker 1.
+    assert rcs[1] != 0
+    assert "poison item" in outs[1][1].decode()
+    assert rcs[0] == 0, outs[0][1].decode()[-1500:]
