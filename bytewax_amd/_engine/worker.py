@@ -523,6 +523,9 @@ class _Worker:
         self.fifo: deque = deque()
         self.outbox: Dict[int, List[Tuple[int, int, List[Any]]]] = {}
         self.execs: List[_Exec] = []
+        # A user exception caught mid-drain; resolved to an abort vote
+        # at the next epoch-close round.
+        self._poison: Optional[BaseException] = None
         w, n = ctx.worker_index, ctx.worker_count
 
         for step in graph.steps:
@@ -654,6 +657,25 @@ class _Worker:
         from .._metrics import metrics_enabled, observe_batch
 
         instrumented = metrics_enabled()
+        if self._poison is not None:
+            # Aborting: discard queued work but keep the collective
+            # schedule aligned (exchange rounds still run with empty
+            # outboxes so peers don't block on a missing collective).
+            self.fifo.clear()
+            self.outbox.clear()
+            return
+        try:
+            self._drain_inner(instrumented, observe_batch)
+        except Exception as ex:  # noqa: BLE001
+            # A user function raised mid-epoch: poison this worker so
+            # the next epoch-close vote aborts the whole cluster
+            # (reference run.rs:273-304 panic hook + shutdown flag);
+            # this rank re-raises the error after the vote.
+            self._poison = ex
+            self.fifo.clear()
+            self.outbox.clear()
+
+    def _drain_inner(self, instrumented, observe_batch) -> None:
         while self.fifo:
             step_idx, input_idx, items = self.fifo.popleft()
             ex = self.execs[step_idx]
@@ -722,13 +744,14 @@ class _Worker:
                         abort_exc = ex
                 self._exchange_rounds()
                 local_eof = all(e.eof_all() for e in self.input_execs)
+                failing = abort_exc or self._poison
                 close, all_eof, abort = self.ctx.vote_close(
-                    local_eof, deadline, abort_exc is not None
+                    local_eof, deadline, failing is not None
                 )
                 if abort:
                     # Abort the whole execution: no snapshot for the
                     # open epoch; resume replays it.
-                    self.ctx.fail(abort_exc or AbortExecution())
+                    self.ctx.fail(failing or AbortExecution())
                     raise _Interrupted()
                 if close:
                     break
