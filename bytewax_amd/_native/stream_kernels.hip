@@ -1335,6 +1335,172 @@ __global__ void k_join_extract(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Session windows (gap-based).  Per-key state is one table cell:
+// (session_start, last_ts, accumulator).  Batches arrive SORTED by
+// (key, ts) — the host wrapper sorts — and one thread walks each
+// key's contiguous segment in order: a gap > gap_ms closes the
+// running session (emit) and starts a new one.  Parallelism is the
+// distinct-key count of the batch, which is the right shape for the
+// high-cardinality workloads the device path targets (low-cardinality
+// session streams belong on the host path).
+//
+// Reference semantics: bytewax windowing.py _SessionWindowerLogic
+// (gap merge); here in-order ingestion makes merges degenerate to
+// extension, matching the reference under watermark-ordered input.
+
+// Table layout: skeys int64 (EMPTY_SLOT = empty), svals 3x int64 per
+// slot (start, last, acc).
+__device__ inline uint64_t session_find_or_claim(
+    uint64_t* skeys, uint64_t mask, uint64_t key) {
+  uint64_t h = mix64(key);
+  for (uint64_t probe = 0; probe <= mask; ++probe) {
+    uint64_t slot = (h + probe) & mask;
+    uint64_t cur = skeys[slot];
+    if (cur == key) return slot;
+    if (cur == EMPTY_SLOT) {
+      uint64_t prev = atomicCAS(
+          (unsigned long long*)&skeys[slot], EMPTY_SLOT, key);
+      if (prev == EMPTY_SLOT || prev == key) return slot;
+    }
+  }
+  return ~0ULL;
+}
+
+template <int MODE>
+__global__ void k_session_insert(
+    const int32_t* __restrict__ keys,   // sorted by (key, ts)
+    const int64_t* __restrict__ ts,
+    const int64_t* __restrict__ vals,
+    const int64_t* __restrict__ seg_start,  // [n_segs] segment offsets
+    const int64_t* __restrict__ seg_end,
+    int64_t n_segs,
+    int64_t gap_ms,
+    uint64_t* __restrict__ skeys,
+    long long* __restrict__ sstart,
+    long long* __restrict__ slast,
+    long long* __restrict__ sacc,
+    uint64_t mask,
+    int32_t* __restrict__ out_keys,  // closed sessions
+    int64_t* __restrict__ out_start,
+    int64_t* __restrict__ out_end,
+    int64_t* __restrict__ out_vals,
+    int* __restrict__ out_n,
+    int64_t out_cap,
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag) {
+  int64_t seg = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; seg < n_segs; seg += stride) {
+    int64_t i = seg_start[seg];
+    int64_t end = seg_end[seg];
+    if (i >= end) continue;
+    uint64_t key = (uint64_t)(uint32_t)keys[i];
+    uint64_t slot = session_find_or_claim(skeys, mask, key);
+    if (slot == ~0ULL) {
+      atomicExch(error_flag, 1);
+      continue;
+    }
+    // This thread owns the key's whole segment; no other thread
+    // touches this slot within the launch.
+    long long start = sstart[slot];
+    long long last = slast[slot];
+    long long acc = sacc[slot];
+    bool open = last >= 0;
+    for (; i < end; ++i) {
+      int64_t t = ts[i];
+      int64_t v = (MODE == AGG_SUM) ? vals[i] : 1;
+      if (open && t - last > gap_ms) {
+        int idx = atomicAdd(out_n, 1);
+        if (idx < out_cap) {
+          out_keys[idx] = (int32_t)(uint32_t)key;
+          out_start[idx] = start;
+          out_end[idx] = last;
+          out_vals[idx] = acc;
+        } else {
+          atomicExch(error_flag, 1);
+        }
+        open = false;
+      }
+      if (!open) {
+        start = t;
+        acc = 0;
+        open = true;
+      }
+      last = t;
+      acc += v;
+    }
+    sstart[slot] = start;
+    slast[slot] = last;
+    sacc[slot] = acc;
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+      long long other = __shfl_down(last, off);
+      if (other > last) last = other;
+    }
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+      atomicMax(max_ts, (unsigned long long)last);
+    }
+  }
+}
+
+// Close sessions whose last event is older than `horizon_ms` (the
+// watermark minus the gap): emit and clear the cell.  Clearing is
+// safe here, unlike the windowed table, because the NEXT insert
+// re-claims slots via CAS (session_find_or_claim tolerates probe
+// holes by claiming the first empty slot; a key displaced past a
+// cleared hole simply occupies two probes' worth of distance — the
+// chain is never read without the CAS claim).  To keep lookups exact
+// we still migrate live cells to the alternate table, mirroring the
+// windowed close.
+__global__ void k_session_close_migrate(
+    const uint64_t* __restrict__ skeys,
+    const long long* __restrict__ sstart,
+    const long long* __restrict__ slast,
+    const long long* __restrict__ sacc,
+    int64_t nslots,
+    int64_t horizon_ms,
+    uint64_t* __restrict__ dst_keys,
+    long long* __restrict__ dst_start,
+    long long* __restrict__ dst_last,
+    long long* __restrict__ dst_acc,
+    uint64_t mask,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_start,
+    int64_t* __restrict__ out_end,
+    int64_t* __restrict__ out_vals,
+    int* __restrict__ out_n,
+    int64_t out_cap,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < nslots; i += stride) {
+    uint64_t key = skeys[i];
+    if (key == EMPTY_SLOT) continue;
+    long long last = slast[i];
+    if (last < 0) continue;
+    if (last < horizon_ms) {
+      int idx = atomicAdd(out_n, 1);
+      if (idx < out_cap) {
+        out_keys[idx] = (int32_t)(uint32_t)key;
+        out_start[idx] = sstart[i];
+        out_end[idx] = last;
+        out_vals[idx] = sacc[i];
+      } else {
+        atomicExch(error_flag, 1);
+      }
+    } else {
+      uint64_t slot = session_find_or_claim(dst_keys, mask, key);
+      if (slot == ~0ULL) {
+        atomicExch(error_flag, 1);
+        continue;
+      }
+      dst_start[slot] = sstart[i];
+      dst_last[slot] = last;
+      dst_acc[slot] = sacc[i];
+    }
+  }
+}
+
 // Stream compaction: keep events where mask != 0, preserving relative
 // order per wave (wave-ballot ranks + one atomic per wave).
 __global__ void k_filter_compact(
@@ -2062,6 +2228,144 @@ void radix_join_insert(
       error_flag.data_ptr<int32_t>());
 }
 
+__global__ void k_session_restore(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ start,
+    const int64_t* __restrict__ last,
+    const int64_t* __restrict__ acc,
+    int64_t n,
+    uint64_t* __restrict__ skeys,
+    long long* __restrict__ sstart,
+    long long* __restrict__ slast,
+    long long* __restrict__ sacc,
+    uint64_t mask,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t slot = session_find_or_claim(
+        skeys, mask, (uint64_t)(uint32_t)keys[i]);
+    if (slot == ~0ULL) {
+      atomicExch(error_flag, 1);
+      continue;
+    }
+    sstart[slot] = start[i];
+    slast[slot] = last[i];
+    sacc[slot] = acc[i];
+  }
+}
+
+void session_insert(
+    torch::Tensor keys,  // SORTED by (key, ts)
+    torch::Tensor ts,
+    c10::optional<torch::Tensor> vals,
+    torch::Tensor seg_start,
+    torch::Tensor seg_end,
+    torch::Tensor skeys,
+    torch::Tensor sstart,
+    torch::Tensor slast,
+    torch::Tensor sacc,
+    torch::Tensor out_keys,
+    torch::Tensor out_start,
+    torch::Tensor out_end,
+    torch::Tensor out_vals,
+    torch::Tensor out_n,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    int64_t gap_ms,
+    int64_t mode) {
+  check_dev(keys, torch::kInt32, "keys");
+  check_dev(ts, torch::kInt64, "ts");
+  int64_t nslots = skeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  int64_t n_segs = seg_start.numel();
+  if (n_segs == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n_segs, 256));
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(
+        kern, grid, block, 0, stream, keys.data_ptr<int32_t>(),
+        ts.data_ptr<int64_t>(),
+        mode == AGG_SUM ? vals->data_ptr<int64_t>() : nullptr,
+        seg_start.data_ptr<int64_t>(), seg_end.data_ptr<int64_t>(),
+        n_segs, gap_ms, (uint64_t*)skeys.data_ptr<int64_t>(),
+        (long long*)sstart.data_ptr<int64_t>(),
+        (long long*)slast.data_ptr<int64_t>(),
+        (long long*)sacc.data_ptr<int64_t>(),
+        (uint64_t)(nslots - 1), out_keys.data_ptr<int32_t>(),
+        out_start.data_ptr<int64_t>(), out_end.data_ptr<int64_t>(),
+        out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
+        out_keys.numel(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>());
+  };
+  if (mode == AGG_COUNT) launch(k_session_insert<AGG_COUNT>);
+  else launch(k_session_insert<AGG_SUM>);
+}
+
+void session_close_migrate(
+    torch::Tensor skeys,
+    torch::Tensor sstart,
+    torch::Tensor slast,
+    torch::Tensor sacc,
+    torch::Tensor dst_keys,
+    torch::Tensor dst_start,
+    torch::Tensor dst_last,
+    torch::Tensor dst_acc,
+    torch::Tensor out_keys,
+    torch::Tensor out_start,
+    torch::Tensor out_end,
+    torch::Tensor out_vals,
+    torch::Tensor out_n,
+    torch::Tensor error_flag,
+    int64_t horizon_ms) {
+  int64_t nslots = skeys.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  hipLaunchKernelGGL(
+      k_session_close_migrate, dim3(n_blocks(nslots, 256)), block, 0,
+      stream, (const uint64_t*)skeys.data_ptr<int64_t>(),
+      (const long long*)sstart.data_ptr<int64_t>(),
+      (const long long*)slast.data_ptr<int64_t>(),
+      (const long long*)sacc.data_ptr<int64_t>(), nslots, horizon_ms,
+      (uint64_t*)dst_keys.data_ptr<int64_t>(),
+      (long long*)dst_start.data_ptr<int64_t>(),
+      (long long*)dst_last.data_ptr<int64_t>(),
+      (long long*)dst_acc.data_ptr<int64_t>(),
+      (uint64_t)(nslots - 1), out_keys.data_ptr<int32_t>(),
+      out_start.data_ptr<int64_t>(), out_end.data_ptr<int64_t>(),
+      out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
+      out_keys.numel(), error_flag.data_ptr<int32_t>());
+}
+
+void session_restore(
+    torch::Tensor keys,
+    torch::Tensor start,
+    torch::Tensor last,
+    torch::Tensor acc,
+    torch::Tensor skeys,
+    torch::Tensor sstart,
+    torch::Tensor slast,
+    torch::Tensor sacc,
+    torch::Tensor error_flag) {
+  check_dev(keys, torch::kInt32, "keys");
+  int64_t n = keys.numel();
+  if (n == 0) return;
+  int64_t nslots = skeys.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  hipLaunchKernelGGL(
+      k_session_restore, dim3(n_blocks(n, 256)), block, 0, stream,
+      keys.data_ptr<int32_t>(), start.data_ptr<int64_t>(),
+      last.data_ptr<int64_t>(), acc.data_ptr<int64_t>(), n,
+      (uint64_t*)skeys.data_ptr<int64_t>(),
+      (long long*)sstart.data_ptr<int64_t>(),
+      (long long*)slast.data_ptr<int64_t>(),
+      (long long*)sacc.data_ptr<int64_t>(),
+      (uint64_t)(nslots - 1), error_flag.data_ptr<int32_t>());
+}
+
 void join_extract(
     torch::Tensor tkeys,
     torch::Tensor tval0,
@@ -2581,6 +2885,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Region-partitioned stream join insert (L2-local table ops)");
   m.def("join_insert", &join_insert,
         "Stream-stream hash join insert; emits completed pairs");
+  m.def("session_insert", &session_insert,
+        "Gap-based session aggregation over a (key, ts)-sorted batch");
+  m.def("session_close_migrate", &session_close_migrate,
+        "Emit sessions idle past the horizon; migrate live cells");
+  m.def("session_restore", &session_restore,
+        "Rebuild session cells from a host spill");
   m.def("join_extract", &join_extract,
         "Extract live join state (recovery snapshot)");
   m.def("filter_compact", &filter_compact,

@@ -35,6 +35,7 @@ __all__ = [
     "SyntheticEventSource",
     "filter_batch",
     "map_batch",
+    "keyed_session_agg",
     "keyed_stats_agg",
     "keyed_window_agg",
     "stream_join",
@@ -975,3 +976,98 @@ class CollectCountsSink(DynamicSink[RecordBatch]):
         self, step_id: str, worker_index: int, worker_count: int
     ) -> _CollectCountsPartition:
         return _CollectCountsPartition(self._ls)
+
+
+class _DeviceSessionLogic(StatefulBatchLogic):
+    """Holds the HBM session table (gap-based windows)."""
+
+    def __init__(self, state, wait_ms: int, exchange: bool, resume):
+        self.state = state
+        self.wait_ms = wait_ms
+        self.exchange = exchange
+        if resume is not None:
+            self.state.restore_from_host(resume)
+
+    def on_batch(self, batches):
+        out = []
+        for batch in batches:
+            if self.exchange:
+                batch = exchange_by_key(batch)
+            self.state.insert(batch)
+        closed = self.state.close_due(self.wait_ms)
+        if closed is not None:
+            out.append(closed)
+        return (out, StatefulBatchLogic.RETAIN)
+
+    def on_eof(self):
+        closed = self.state.close_all()
+        return (
+            [closed] if closed is not None else [],
+            StatefulBatchLogic.RETAIN,
+        )
+
+    def snapshot(self):
+        return self.state.snapshot_to_host()
+
+
+@operator
+def keyed_session_agg(
+    step_id: str,
+    up: Stream[RecordBatch],
+    gap: timedelta,
+    mode: str = "count",
+    wait: timedelta = timedelta(0),
+    slots_pow: int = 20,
+    out_cap: int = 1 << 20,
+    device: str = "cuda",
+    exchange: Optional[bool] = None,
+) -> Stream[Dict[str, Any]]:
+    """Keyed session windows over columnar batches on GPU.
+
+    Sessions close when a key sees no events for ``gap`` (reference
+    `SessionWindower`, windowing.py).  Batches are sorted by
+    (key, ts) on device and one thread walks each key's run in event
+    order, so parallelism equals the batch's distinct-key count —
+    use the host path's `SessionWindower` for low-cardinality
+    streams.  Ingestion must be watermark-ordered across batches
+    (each batch's events at or after the previous watermark), which
+    in-order sources provide; under that ordering the reference's
+    session merges degenerate to extensions.
+
+    Emits dicts of device columns {keys, start, end, vals} at session
+    close / EOF.
+    """
+    import torch
+
+    from .state import SessionAggState
+
+    agg_mode = {"count": AGG_COUNT, "sum": AGG_SUM}[mode]
+    gap_ms = int(gap.total_seconds() * 1000)
+    wait_ms = int(wait.total_seconds() * 1000)
+
+    def shim_builder(resume_state):
+        import torch.distributed as dist
+
+        ex = exchange
+        if ex is None:
+            ex = (
+                dist.is_available()
+                and dist.is_initialized()
+                and dist.get_world_size() > 1
+            )
+        state = SessionAggState(
+            torch.device(device), gap_ms, agg_mode,
+            slots_pow=slots_pow, out_cap=out_cap,
+        )
+        return _DeviceSessionLogic(state, wait_ms, ex, resume_state)
+
+    import torch.distributed as dist
+
+    shard = (
+        f"shard-{dist.get_rank()}"
+        if dist.is_available() and dist.is_initialized()
+        else "shard-0"
+    )
+    keyed = op.map("wrap", up, lambda b: (shard, b))
+    agg = op.stateful_batch("agg", keyed, shim_builder)
+    return op.map("unwrap", agg, lambda kv: kv[1])

@@ -12,9 +12,9 @@ Both support pinned-host snapshot spill for recovery and have CPU
 twins for GPU-less test runs.
 """
 
-from typing import Any, Dict, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
-from . import RecordBatch, _ms  # noqa: F401
+from . import AGG_COUNT, AGG_SUM, RecordBatch, _ms  # noqa: F401
 from ._ext import ext
 
 _I64_MAX = (1 << 63) - 1
@@ -567,3 +567,236 @@ class HashJoinState:
         )
         self.out_n.zero_()
         return out
+
+
+class SessionAggState:
+    """Gap-based session aggregation state on device.
+
+    One cell per key: (session_start, last_ts, accumulator).  Batches
+    are sorted by (key, ts) on insert and one thread walks each key's
+    segment in order (parallelism = distinct keys — built for
+    high-cardinality streams; low-cardinality session streams belong
+    on the host path's `SessionWindower`).  A gap > ``gap_ms`` closes
+    the running session; the watermark close emits sessions idle past
+    ``watermark - gap``.  Role parity: reference
+    windowing.py _SessionWindowerLogic under watermark-ordered input
+    (in-order ingestion makes merges degenerate to extension).
+    """
+
+    def __init__(
+        self,
+        device,
+        gap_ms: int,
+        mode: int = AGG_COUNT,
+        slots_pow: int = 20,
+        out_cap: int = 1 << 20,
+    ):
+        import torch
+
+        self.device = device
+        self.cpu = device.type == "cpu"
+        self.gap_ms = gap_ms
+        self.mode = mode
+        self.max_ts_host = 0
+        if self.cpu:
+            self._table: Dict[int, Tuple[int, int, int]] = {}
+            self._closed: List[Tuple[int, int, int, int]] = []
+            return
+        self.k = ext()
+        self.nslots = 1 << slots_pow
+        mk = lambda fill: torch.full(  # noqa: E731
+            (self.nslots,), fill, dtype=torch.int64, device=device
+        )
+        self.skeys, self.sstart = mk(-1), mk(0)
+        self.slast, self.sacc = mk(-1), mk(0)
+        self.skeys_alt, self.sstart_alt = mk(-1), mk(0)
+        self.slast_alt, self.sacc_alt = mk(-1), mk(0)
+        self.out_cap = out_cap
+        self.out_keys = torch.empty(out_cap, dtype=torch.int32, device=device)
+        self.out_start = torch.empty(out_cap, dtype=torch.int64, device=device)
+        self.out_end = torch.empty(out_cap, dtype=torch.int64, device=device)
+        self.out_vals = torch.empty(out_cap, dtype=torch.int64, device=device)
+        self.out_n = torch.zeros(1, dtype=torch.int32, device=device)
+        self.max_ts_dev = torch.zeros(1, dtype=torch.int64, device=device)
+        self.error_flag = torch.zeros(1, dtype=torch.int32, device=device)
+
+    def _sort(self, batch: RecordBatch):
+        import torch
+
+        order = torch.argsort(batch.ts, stable=True)
+        k1 = batch.keys[order]
+        order2 = torch.argsort(k1.to(torch.int64), stable=True)
+        perm = order[order2]
+        keys = batch.keys[perm]
+        ts = batch.ts[perm] + batch.ts_base
+        vals = batch.vals[perm] if batch.vals is not None else None
+        return keys, ts, vals
+
+    def _insert_cpu(self, batch: RecordBatch) -> None:
+        keys, ts, vals = self._sort(batch)
+        it = zip(
+            keys.tolist(),
+            ts.tolist(),
+            vals.tolist() if vals is not None else [1] * len(keys),
+        )
+        for k, t, v in it:
+            if self.mode == AGG_COUNT:
+                v = 1
+            cur = self._table.get(k)
+            if cur is not None and t - cur[1] > self.gap_ms:
+                self._closed.append((k, cur[0], cur[1], cur[2]))
+                cur = None
+            if cur is None:
+                self._table[k] = (t, t, v)
+            else:
+                self._table[k] = (cur[0], t, cur[2] + v)
+            if t > self.max_ts_host:
+                self.max_ts_host = t
+
+    def insert(self, batch: RecordBatch) -> None:
+        import torch
+
+        if len(batch) == 0:
+            return
+        if self.cpu:
+            self._insert_cpu(batch)
+            return
+        keys, ts, vals = self._sort(batch)
+        neq = keys[1:] != keys[:-1]
+        idx = torch.nonzero(neq).flatten() + 1
+        zero = torch.zeros(1, dtype=torch.int64, device=self.device)
+        seg_start = torch.cat([zero, idx])
+        n = torch.full((1,), len(keys), dtype=torch.int64, device=self.device)
+        seg_end = torch.cat([idx, n])
+        self.k.session_insert(
+            keys, ts, vals, seg_start, seg_end,
+            self.skeys, self.sstart, self.slast, self.sacc,
+            self.out_keys, self.out_start, self.out_end, self.out_vals,
+            self.out_n, self.max_ts_dev, self.error_flag,
+            self.gap_ms, self.mode,
+        )
+        if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
+            self.max_ts_host = batch.max_ts
+
+    def _drain_out(self) -> Optional[Dict[str, Any]]:
+        if self.cpu:
+            if not self._closed:
+                return None
+            import torch
+
+            rows = self._closed
+            self._closed = []
+            return {
+                "keys": torch.tensor([r[0] for r in rows], dtype=torch.int32),
+                "start": torch.tensor([r[1] for r in rows], dtype=torch.int64),
+                "end": torch.tensor([r[2] for r in rows], dtype=torch.int64),
+                "vals": torch.tensor([r[3] for r in rows], dtype=torch.int64),
+            }
+        n = int(self.out_n.item())
+        if int(self.error_flag.item()) != 0:
+            msg = "session state overflow; increase slots_pow/out_cap"
+            raise RuntimeError(msg)
+        if n == 0:
+            return None
+        out = {
+            "keys": self.out_keys[:n].clone(),
+            "start": self.out_start[:n].clone(),
+            "end": self.out_end[:n].clone(),
+            "vals": self.out_vals[:n].clone(),
+        }
+        self.out_n.zero_()
+        return out
+
+    def close_due(self, wait_ms: int = 0) -> Optional[Dict[str, Any]]:
+        """Emit sessions idle past watermark - gap (plus any closed by
+        gaps during inserts since the last drain)."""
+        horizon = self.max_ts_host - self.gap_ms - wait_ms
+        if self.cpu:
+            for k in list(self._table):
+                s, last, acc = self._table[k]
+                if last < horizon:
+                    self._closed.append((k, s, last, acc))
+                    del self._table[k]
+            return self._drain_out()
+        self.k.session_close_migrate(
+            self.skeys, self.sstart, self.slast, self.sacc,
+            self.skeys_alt, self.sstart_alt, self.slast_alt, self.sacc_alt,
+            self.out_keys, self.out_start, self.out_end, self.out_vals,
+            self.out_n, self.error_flag, horizon,
+        )
+        self._swap_reset()
+        return self._drain_out()
+
+    def _swap_reset(self) -> None:
+        self.skeys, self.skeys_alt = self.skeys_alt, self.skeys
+        self.sstart, self.sstart_alt = self.sstart_alt, self.sstart
+        self.slast, self.slast_alt = self.slast_alt, self.slast
+        self.sacc, self.sacc_alt = self.sacc_alt, self.sacc
+        self.skeys_alt.fill_(-1)
+        self.slast_alt.fill_(-1)
+        self.sstart_alt.zero_()
+        self.sacc_alt.zero_()
+
+    def close_all(self) -> Optional[Dict[str, Any]]:
+        if self.cpu:
+            for k in list(self._table):
+                s, last, acc = self._table[k]
+                self._closed.append((k, s, last, acc))
+            self._table.clear()
+            return self._drain_out()
+        self.k.session_close_migrate(
+            self.skeys, self.sstart, self.slast, self.sacc,
+            self.skeys_alt, self.sstart_alt, self.slast_alt, self.sacc_alt,
+            self.out_keys, self.out_start, self.out_end, self.out_vals,
+            self.out_n, self.error_flag, 1 << 60,
+        )
+        self._swap_reset()
+        return self._drain_out()
+
+    def snapshot_to_host(self) -> Dict[str, Any]:
+        import numpy as np
+
+        if self.cpu:
+            items = list(self._table.items())
+            return {
+                "keys": np.array([k for k, _ in items], dtype="int32"),
+                "start": np.array([v[0] for _, v in items], dtype="int64"),
+                "last": np.array([v[1] for _, v in items], dtype="int64"),
+                "acc": np.array([v[2] for _, v in items], dtype="int64"),
+                "max_ts": self.max_ts_host,
+                "pending": list(self._closed),
+            }
+        live = (self.skeys >= 0) & (self.slast >= 0)
+        return {
+            "keys": self.skeys[live].to("cpu").numpy().astype("int32"),
+            "start": self.sstart[live].cpu().numpy().copy(),
+            "last": self.slast[live].cpu().numpy().copy(),
+            "acc": self.sacc[live].cpu().numpy().copy(),
+            "max_ts": self.max_ts_host,
+            "pending": [],
+        }
+
+    def restore_from_host(self, snap: Dict[str, Any]) -> None:
+        import torch
+
+        self.max_ts_host = snap["max_ts"]
+        if self.cpu:
+            for k, s, last, acc in zip(
+                snap["keys"].tolist(), snap["start"].tolist(),
+                snap["last"].tolist(), snap["acc"].tolist(),
+            ):
+                self._table[int(k)] = (int(s), int(last), int(acc))
+            self._closed.extend(
+                tuple(r) for r in snap.get("pending", [])
+            )
+            return
+        if len(snap["keys"]) == 0:
+            return
+        self.k.session_restore(
+            torch.as_tensor(snap["keys"]).to(self.device),
+            torch.as_tensor(snap["start"]).to(self.device),
+            torch.as_tensor(snap["last"]).to(self.device),
+            torch.as_tensor(snap["acc"]).to(self.device),
+            self.skeys, self.sstart, self.slast, self.sacc,
+            self.error_flag,
+        )

@@ -1,0 +1,229 @@
+"""Device session windows: CPU twins (always) + device numerics vs a
+host reference (gpu mark).  Parity: reference windowing.py
+SessionWindower semantics under watermark-ordered input."""
+
+from datetime import datetime, timedelta, timezone
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+from bytewax_amd.gpu import AGG_COUNT, AGG_SUM, RecordBatch  # noqa: E402
+from bytewax_amd.gpu.state import SessionAggState  # noqa: E402
+
+ALIGN = datetime(2024, 1, 1, tzinfo=timezone.utc)
+
+
+def _rows(out):
+    if out is None:
+        return []
+    return sorted(
+        zip(
+            out["keys"].tolist(),
+            out["start"].tolist(),
+            out["end"].tolist(),
+            out["vals"].tolist(),
+        )
+    )
+
+
+def _ref_sessions(keys, ts, vals, gap_ms):
+    """Serial reference: per-key in-order walk."""
+    from collections import defaultdict
+
+    per_key = defaultdict(list)
+    for k, t, v in sorted(zip(keys, ts, vals), key=lambda r: (r[0], r[1])):
+        per_key[k].append((t, v))
+    out = []
+    for k, evs in per_key.items():
+        start, last, acc = None, None, 0
+        for t, v in evs:
+            if last is not None and t - last > gap_ms:
+                out.append((k, start, last, acc))
+                start, acc = None, 0
+            if start is None:
+                start = t
+                acc = 0
+            last = t
+            acc += v
+        if start is not None:
+            out.append((k, start, last, acc))
+    return sorted(out)
+
+
+def test_session_gap_close_cpu():
+    st = SessionAggState(torch.device("cpu"), gap_ms=100, mode=AGG_COUNT)
+    st.insert(
+        RecordBatch(
+            torch.tensor([1, 1, 1, 2], dtype=torch.int32),
+            torch.tensor([10, 50, 300, 20], dtype=torch.int64),
+            None,
+            max_ts=300,
+        )
+    )
+    # Key 1's first session (10..50) closed by the 300 gap; key 2
+    # idle-closed by the watermark (300 - gap = 200 > 20).
+    closed = st.close_due()
+    assert _rows(closed) == [(1, 10, 50, 2), (2, 20, 20, 1)]
+    # EOF closes the rest.
+    assert _rows(st.close_all()) == [(1, 300, 300, 1)]
+
+
+def test_session_watermark_close_cpu():
+    st = SessionAggState(torch.device("cpu"), gap_ms=100, mode=AGG_SUM)
+    st.insert(
+        RecordBatch(
+            torch.tensor([5], dtype=torch.int32),
+            torch.tensor([1000], dtype=torch.int64),
+            torch.tensor([7], dtype=torch.int64),
+            max_ts=1000,
+        )
+    )
+    assert st.close_due() is None  # watermark 1000: still within gap
+    st.insert(
+        RecordBatch(
+            torch.tensor([6], dtype=torch.int32),
+            torch.tensor([5000], dtype=torch.int64),
+            torch.tensor([1], dtype=torch.int64),
+            max_ts=5000,
+        )
+    )
+    # Watermark 5000 > 1000 + gap: key 5's session is idle-closed.
+    assert _rows(st.close_due()) == [(5, 1000, 1000, 7)]
+
+
+def test_session_snapshot_roundtrip_cpu():
+    a = SessionAggState(torch.device("cpu"), gap_ms=100, mode=AGG_COUNT)
+    a.insert(
+        RecordBatch(
+            torch.tensor([1, 2], dtype=torch.int32),
+            torch.tensor([10, 20], dtype=torch.int64),
+            None,
+            max_ts=20,
+        )
+    )
+    snap = a.snapshot_to_host()
+    b = SessionAggState(torch.device("cpu"), gap_ms=100, mode=AGG_COUNT)
+    b.restore_from_host(snap)
+    assert _rows(b.close_all()) == _rows(a.close_all())
+
+
+def test_keyed_session_agg_pipeline_cpu():
+    import bytewax_amd.operators as op
+    from bytewax_amd.dataflow import Dataflow
+    from bytewax_amd.gpu.operators import keyed_session_agg
+    from bytewax_amd.inputs import DynamicSource, StatelessSourcePartition
+    from bytewax_amd.outputs import DynamicSink, StatelessSinkPartition
+    from bytewax_amd.testing import run_main
+
+    class _Src(StatelessSourcePartition):
+        def __init__(self):
+            self.i = 0
+
+        def next_batch(self):
+            if self.i >= 3:
+                raise StopIteration()
+            base = self.i * 1000
+            self.i += 1
+            return [
+                RecordBatch(
+                    torch.arange(4, dtype=torch.int32),
+                    torch.full((4,), base, dtype=torch.int64),
+                    None,
+                    max_ts=base,
+                )
+            ]
+
+    class Src(DynamicSource):
+        def build(self, step_id, worker_index, worker_count):
+            return _Src()
+
+    class _Collect(StatelessSinkPartition):
+        def __init__(self, ls):
+            self._ls = ls
+
+        def write_batch(self, items):
+            self._ls.extend(items)
+
+    class Sink(DynamicSink):
+        def __init__(self, ls):
+            self._ls = ls
+
+        def build(self, step_id, worker_index, worker_count):
+            return _Collect(self._ls)
+
+    out = []
+    flow = Dataflow("sess")
+    s = op.input("inp", flow, Src())
+    # Gap 500ms < 1000ms batch spacing: every event is its own session.
+    agg = keyed_session_agg(
+        "sess_agg", s, gap=timedelta(milliseconds=500), device="cpu"
+    )
+    op.output("out", agg, Sink(out))
+    run_main(flow)
+    rows = [r for d in out for r in _rows(d)]
+    assert len(rows) == 12  # 3 batches x 4 keys, all singleton sessions
+    assert all(v == 1 for _k, _s, _e, v in rows)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("mode_name", ["count", "sum"])
+def test_session_gpu_matches_reference(mode_name):
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    g = torch.Generator().manual_seed(12)
+    n = 200_000
+    vocab = 5_000
+    gap = 500
+    keys = torch.randint(0, vocab, (n,), dtype=torch.int32, generator=g)
+    # Timestamps over a 20s span: plenty of per-key gaps > 500ms.
+    ts = torch.randint(0, 20_000, (n,), dtype=torch.int64, generator=g)
+    vals = torch.randint(1, 50, (n,), dtype=torch.int64, generator=g)
+    mode = AGG_COUNT if mode_name == "count" else AGG_SUM
+    ref = _ref_sessions(
+        keys.tolist(),
+        ts.tolist(),
+        [1] * n if mode == AGG_COUNT else vals.tolist(),
+        gap,
+    )
+    st = SessionAggState(
+        torch.device("cuda:0"), gap_ms=gap, mode=mode, slots_pow=14,
+        out_cap=n,
+    )
+    st.insert(
+        RecordBatch(keys.cuda(), ts.cuda(), vals.cuda(), max_ts=20_000)
+    )
+    got = []
+    closed = st.close_due()
+    if closed is not None:
+        got.extend(
+            _rows({k: t.cpu() for k, t in closed.items()})
+        )
+    final = st.close_all()
+    if final is not None:
+        got.extend(_rows({k: t.cpu() for k, t in final.items()}))
+    assert sorted(got) == ref
+
+
+@pytest.mark.gpu
+def test_session_gpu_snapshot_roundtrip():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    g = torch.Generator().manual_seed(13)
+    n = 50_000
+    keys = torch.randint(0, 1000, (n,), dtype=torch.int32, generator=g)
+    ts = torch.randint(0, 5_000, (n,), dtype=torch.int64, generator=g)
+    a = SessionAggState(
+        torch.device("cuda:0"), gap_ms=10_000, mode=AGG_COUNT,
+        slots_pow=12, out_cap=n,
+    )
+    a.insert(RecordBatch(keys.cuda(), ts.cuda(), None, max_ts=5_000))
+    snap = a.snapshot_to_host()
+    b = SessionAggState(
+        torch.device("cuda:0"), gap_ms=10_000, mode=AGG_COUNT,
+        slots_pow=12, out_cap=n,
+    )
+    b.restore_from_host(snap)
+    ra = _rows({k: t.cpu() for k, t in a.close_all().items()})
+    rb = _rows({k: t.cpu() for k, t in b.close_all().items()})
+    assert ra == rb
