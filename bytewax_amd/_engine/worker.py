@@ -425,24 +425,34 @@ class _StatefulBatchExec(_Exec):
 
     def process(self, input_idx, items, epoch):
         grouped: Dict[str, List[Any]] = {}
+        get = grouped.get
         for kv in items:
-            if (
-                not isinstance(kv, tuple)
-                or len(kv) != 2
-            ):
+            # Hot loop: exact-type checks first (`type(x) is` is much
+            # cheaper than isinstance), subclass fallbacks second.
+            if type(kv) is not tuple and not isinstance(kv, tuple):
+                msg = (
+                    f"step {self.step.step_id!r} requires `(key, value)` "
+                    f"2-tuples from upstream; got a {type(kv)!r} instead"
+                )
+                raise TypeError(msg)
+            if len(kv) != 2:
                 msg = (
                     f"step {self.step.step_id!r} requires `(key, value)` "
                     f"2-tuples from upstream; got a {type(kv)!r} instead"
                 )
                 raise TypeError(msg)
             k, v = kv
-            if not isinstance(k, str):
+            if type(k) is not str and not isinstance(k, str):
                 msg = (
                     f"step {self.step.step_id!r} requires keys to be `str`; "
                     f"got a {type(k)!r} instead"
                 )
                 raise TypeError(msg)
-            grouped.setdefault(k, []).append(v)
+            lst = get(k)
+            if lst is None:
+                grouped[k] = [v]
+            else:
+                lst.append(v)
         out: List[Tuple[str, Any]] = []
         for k in sorted(grouped):
             logic = self.logics.get(k)
