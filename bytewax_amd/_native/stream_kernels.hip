@@ -1824,6 +1824,159 @@ void radix_window_insert(
   else ov(k_overflow_agg<AGG_SUM>);
 }
 
+// Split radix insert for the per-step Python engine's pipelined
+// path: `radix_scatter_only` launches the (memset + per-segment
+// scatter) on the CURRENT stream — the caller runs it under a side
+// torch.cuda.Stream — and `radix_agg_only` drains a buffer set into
+// the table on whatever stream is current.  Stream/event
+// orchestration (scatter N+1 overlapping agg N, buffer-parity reuse
+// fences, NCCL completion ordering) lives in Python with torch
+// events, which keeps it per-state and lets the exchange wait be
+// recorded on the side stream only.
+void radix_scatter_only(
+    torch::Tensor keys,
+    torch::Tensor ts,
+    c10::optional<torch::Tensor> vals,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    torch::Tensor gcursors,
+    torch::Tensor ev_packed,
+    torch::Tensor ev_vals,
+    torch::Tensor ov_cursor,
+    torch::Tensor ov_packed,
+    torch::Tensor ov_vals,
+    int64_t nslots,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t mode,
+    int64_t ts_base,
+    int64_t region_bits,
+    std::vector<int64_t> seg_counts,
+    std::vector<int64_t> seg_bases) {
+  check_dev(keys, torch::kInt32, "keys");
+  bool seg32 = !seg_counts.empty();
+  if (seg32) {
+    TORCH_CHECK(seg_counts.size() == seg_bases.size(),
+                "seg_counts/seg_bases length mismatch");
+    check_dev(ts, torch::kInt32, "ts32");
+  } else {
+    check_dev(ts, torch::kInt64, "ts");
+  }
+  int64_t n = keys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  int64_t nb = nslots >> region_bits;
+  int64_t cap = ev_packed.numel() / nb;
+  TORCH_CHECK(cap * nb >= 2 * n || cap >= n, "scatter buffers too small");
+  const int64_t* vptr = nullptr;
+  if (mode == AGG_SUM) {
+    TORCH_CHECK(vals.has_value(), "sum mode requires vals");
+    vptr = vals->data_ptr<int64_t>();
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  uint64_t mask = (uint64_t)(nslots - 1);
+  dim3 block(256);
+  uint64_t win_m2, win_maxfast;
+  magic_div_u64(len_ms, &win_m2, &win_maxfast);
+  size_t hist_lds = (size_t)nb * sizeof(int);
+  HIP_CHECK(hipMemsetAsync(
+      gcursors.data_ptr<int32_t>(), 0, (size_t)nb * sizeof(int), stream));
+  HIP_CHECK(hipMemsetAsync(
+      ov_cursor.data_ptr<int32_t>(), 0, sizeof(int), stream));
+  if (n == 0) return;
+
+  struct Seg {
+    int64_t off, n, base;
+  };
+  std::vector<Seg> segs;
+  if (seg32) {
+    int64_t off = 0;
+    for (size_t i = 0; i < seg_counts.size(); ++i) {
+      if (seg_counts[i] > 0) segs.push_back({off, seg_counts[i], seg_bases[i]});
+      off += seg_counts[i];
+    }
+  } else {
+    segs.push_back({0, n, ts_base});
+  }
+  auto scat = [&](auto kern, auto tsptr, const Seg& sg) {
+    hipLaunchKernelGGL(
+        kern, dim3((unsigned)n_blocks(sg.n, 256)), block, 2 * hist_lds,
+        stream, keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
+        vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
+        sg.base, mask, (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
+        ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ov_vals.data_ptr<int64_t>() : nullptr,
+        ov_packed.numel(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+  };
+  for (const Seg& sg : segs) {
+    if (seg32) {
+      const int32_t* t32 = ts.data_ptr<int32_t>();
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_fixed<AGG_COUNT, int32_t>, t32, sg);
+      else
+        scat(k_radix_scatter_fixed<AGG_SUM, int32_t>, t32, sg);
+    } else {
+      const int64_t* t64 = ts.data_ptr<int64_t>();
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_fixed<AGG_COUNT, int64_t>, t64, sg);
+      else
+        scat(k_radix_scatter_fixed<AGG_SUM, int64_t>, t64, sg);
+    }
+  }
+}
+
+void radix_agg_only(
+    torch::Tensor tkeys,
+    torch::Tensor tvals,
+    torch::Tensor error_flag,
+    torch::Tensor gcursors,
+    torch::Tensor ev_packed,
+    torch::Tensor ev_vals,
+    torch::Tensor ov_cursor,
+    torch::Tensor ov_packed,
+    torch::Tensor ov_vals,
+    int64_t mode,
+    int64_t region_bits) {
+  int64_t nslots = tkeys.numel();
+  int64_t nb = nslots >> region_bits;
+  int64_t cap = ev_packed.numel() / nb;
+  auto stream = at::hip::getCurrentHIPStream();
+  uint64_t mask = (uint64_t)(nslots - 1);
+  dim3 block(256);
+  auto offsets = at::arange(
+      nb, at::TensorOptions().dtype(at::kInt).device(tkeys.device()));
+  offsets = offsets * (int)cap;
+  size_t agg_lds = (size_t)(1 << region_bits) * 16;
+  auto agg = [&](auto kern) {
+    hipLaunchKernelGGL(
+        kern, dim3((unsigned)nb), block, agg_lds, stream,
+        (const uint64_t*)ev_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
+        offsets.data_ptr<int32_t>(), gcursors.data_ptr<int32_t>(), cap,
+        (uint64_t*)tkeys.data_ptr<int64_t>(),
+        (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
+        (int)region_bits, error_flag.data_ptr<int32_t>());
+  };
+  if (mode == AGG_COUNT) agg(k_radix_agg<AGG_COUNT>);
+  else agg(k_radix_agg<AGG_SUM>);
+  auto ov = [&](auto kern) {
+    hipLaunchKernelGGL(
+        kern, dim3(64), block, 0, stream,
+        (const uint64_t*)ov_packed.data_ptr<int64_t>(),
+        mode == AGG_SUM ? ov_vals.data_ptr<int64_t>() : nullptr,
+        ov_cursor.data_ptr<int32_t>(), ov_packed.numel(),
+        (uint64_t*)tkeys.data_ptr<int64_t>(),
+        (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
+        (int)region_bits, error_flag.data_ptr<int32_t>());
+  };
+  if (mode == AGG_COUNT) ov(k_overflow_agg<AGG_COUNT>);
+  else ov(k_overflow_agg<AGG_SUM>);
+}
+
 void radix_v2_window_insert(
     torch::Tensor keys,
     torch::Tensor ts,
@@ -2866,6 +3019,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("radix_v2_window_insert", &radix_v2_window_insert,
         "Experimental two-level radix with full-line LDS-staged "
         "scatter (COUNT mode)");
+  m.def("radix_scatter_only", &radix_scatter_only,
+        "Scatter stage of the radix insert on the current stream");
+  m.def("radix_agg_only", &radix_agg_only,
+        "Aggregation stage of the radix insert on the current stream");
   m.def("radix_window_insert", &radix_window_insert,
         "Radix-partitioned LDS-staged keyed window aggregation");
   m.def("close_migrate", &close_migrate,

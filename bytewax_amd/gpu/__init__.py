@@ -415,6 +415,14 @@ class WindowAggState:
             dtype=torch.int64,
             device=self.device,
         )
+        if hasattr(self, "rx_packed2"):
+            # Keep the pipelined second buffer set in sync on growth.
+            self.rx_gcursors2 = torch.zeros_like(self.rx_gcursors)
+            self.rx_packed2 = torch.empty_like(self.rx_packed)
+            self.rx_vals2 = torch.empty_like(self.rx_vals)
+            self.rx_ov_cursor2 = torch.zeros_like(self.rx_ov_cursor)
+            self.rx_ov_packed2 = torch.empty_like(self.rx_ov_packed)
+            self.rx_ov_vals2 = torch.empty_like(self.rx_ov_vals)
 
     def insert(self, batch: RecordBatch) -> None:
         if self.cpu:
@@ -483,6 +491,115 @@ class WindowAggState:
             )
         if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
             self.max_ts_host = batch.max_ts
+
+    def _ensure_pipe(self) -> None:
+        import torch
+
+        if getattr(self, "_pipe_ready", False):
+            return
+        self._pipe_ready = True
+        self._sc_stream = torch.cuda.Stream(self.device)
+        self._pipe_parity = 0
+        self._ev_sc = [torch.cuda.Event(), torch.cuda.Event()]
+        self._ev_ag = [torch.cuda.Event(), torch.cuda.Event()]
+        self._ag_recorded = [False, False]
+        if not hasattr(self, "rx_packed2"):
+            self.rx_gcursors2 = torch.zeros_like(self.rx_gcursors)
+            self.rx_packed2 = torch.empty_like(self.rx_packed)
+            self.rx_vals2 = torch.empty_like(self.rx_vals)
+            self.rx_ov_cursor2 = torch.zeros_like(self.rx_ov_cursor)
+            self.rx_ov_packed2 = torch.empty_like(self.rx_ov_packed)
+            self.rx_ov_vals2 = torch.empty_like(self.rx_ov_vals)
+
+    def _pipe_bufs(self, par: int):
+        if par == 0:
+            return (
+                self.rx_gcursors,
+                self.rx_packed,
+                self.rx_vals,
+                self.rx_ov_cursor,
+                self.rx_ov_packed,
+                self.rx_ov_vals,
+            )
+        return (
+            self.rx_gcursors2,
+            self.rx_packed2,
+            self.rx_vals2,
+            self.rx_ov_cursor2,
+            self.rx_ov_packed2,
+            self.rx_ov_vals2,
+        )
+
+    def insert_pipelined(
+        self,
+        keys,
+        ts,
+        vals,
+        ts_base: int,
+        seg_counts,
+        seg_bases,
+        max_ts: Optional[int],
+        works=None,
+    ) -> None:
+        """Radix insert with the scatter on a side stream (the per-step
+        engine's twin of the native loop's two-stream pipeline).
+
+        The scatter of this call overlaps the aggregation of the
+        PREVIOUS call (alternating buffer parity; events fence buffer
+        reuse).  ``works`` (async exchange handles) are waited on the
+        side stream, so the NCCL completion gates only the scatter —
+        not the previous step's aggregation."""
+        import torch
+
+        n = int(keys.numel())
+        if n > self.rx_max_batch:
+            self._alloc_rx(int(n * 5 // 4))
+        self._ensure_pipe()
+        par = self._pipe_parity
+        self._pipe_parity ^= 1
+        bufs = self._pipe_bufs(par)
+        cur = torch.cuda.current_stream(self.device)
+        with torch.cuda.stream(self._sc_stream):
+            if works:
+                for w in works:
+                    w.wait()
+            if self._ag_recorded[par]:
+                self._sc_stream.wait_event(self._ev_ag[par])
+            # Consumed on a non-allocating stream: pin lifetimes.
+            keys.record_stream(self._sc_stream)
+            ts.record_stream(self._sc_stream)
+            if vals is not None:
+                vals.record_stream(self._sc_stream)
+            self.k.radix_scatter_only(
+                keys,
+                ts,
+                vals,
+                self.max_ts_dev,
+                self.error_flag,
+                *bufs,
+                self.nslots,
+                self.align_ms,
+                self.len_ms,
+                self.mode,
+                ts_base,
+                self.region_bits,
+                list(seg_counts),
+                [int(b) for b in seg_bases],
+            )
+            self._ev_sc[par].record(self._sc_stream)
+        cur.wait_event(self._ev_sc[par])
+        self.k.radix_agg_only(
+            self.tkeys,
+            self.tvals,
+            self.error_flag,
+            *bufs,
+            self.mode,
+            self.region_bits,
+        )
+        self._ev_ag[par].record(cur)
+        self._ag_recorded[par] = True
+        if max_ts is not None and max_ts > self.max_ts_host:
+            self.max_ts_host = max_ts
 
     def insert_lazy(self, lz: "_LazyTsBatch") -> None:
         """Insert exchange output in wire format (int32 timestamp

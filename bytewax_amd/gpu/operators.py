@@ -308,11 +308,38 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         if max_ts is not None and max_ts > self.state.max_ts_host:
             self.state.max_ts_host = max_ts
 
+    def _use_pipe(self) -> bool:
+        import os
+
+        st = self.state
+        return (
+            not st.cpu
+            and st.radix
+            and not getattr(st, "radix_v2", False)
+            and os.environ.get("BYTEWAX_PY_PIPELINE", "1") != "0"
+        )
+
     def _flush_pending(self) -> None:
         if self.pending is None:
             return
         works, batch = self.pending
         self.pending = None
+        if self._use_pipe():
+            # Two-stream insert: the exchange completion and this
+            # batch's scatter land on the side stream, overlapping the
+            # previous batch's aggregation on the compute stream.
+            if hasattr(batch, "materialize"):
+                self.state.insert_pipelined(
+                    batch.keys, batch.ts32, batch.vals, 0,
+                    batch.seg_counts, batch.seg_bases, batch.max_ts,
+                    works,
+                )
+            else:
+                self.state.insert_pipelined(
+                    batch.keys, batch.ts, batch.vals, batch.ts_base,
+                    [], [], batch.max_ts, works,
+                )
+            return
         for w in works:
             w.wait()
         if hasattr(batch, "materialize"):
@@ -337,6 +364,11 @@ class _DeviceWindowLogic(StatefulBatchLogic):
                 works, exchanged = exchange_by_key(batch, async_op=True)
                 self._flush_pending()
                 self.pending = (works, exchanged)
+            elif self._use_pipe():
+                self.state.insert_pipelined(
+                    batch.keys, batch.ts, batch.vals, batch.ts_base,
+                    [], [], batch.max_ts, None,
+                )
             else:
                 self.state.insert(batch)
         self.state.close_launch(self.wait_ms)

@@ -437,3 +437,62 @@ def test_native_pipelined_matches_serial():
     # 24 steps x 5 windows/step worth of events over 5k keys: every
     # (key, window) cell must appear exactly once in closed+tail.
     assert rows_a > 0
+
+
+def test_python_engine_pipelined_insert_matches_serial(monkeypatch):
+    """The per-step engine's two-stream insert must equal the serial
+    insert exactly (same flow, env-toggled)."""
+    _skip_no_gpu()
+    from datetime import timedelta
+
+    import bytewax_amd.operators as op
+    from bytewax_amd.dataflow import Dataflow
+    from bytewax_amd.gpu.operators import (
+        CollectCountsSink,
+        SyntheticEventSource,
+        keyed_window_agg,
+    )
+    from bytewax_amd.testing import run_main
+
+    def run(pipe: str):
+        monkeypatch.setenv("BYTEWAX_PY_PIPELINE", pipe)
+        out = []
+        flow = Dataflow("pipe_eq")
+        s = op.input(
+            "inp",
+            flow,
+            SyntheticEventSource(
+                events_per_batch=2_000_000,
+                n_batches=12,
+                vocab=50_000,
+                align_to=ALIGN,
+                sim_ms_per_batch=30_000,
+            ),
+        )
+        agg = keyed_window_agg(
+            "agg",
+            s,
+            align_to=ALIGN,
+            length=timedelta(minutes=1),
+            mode="count",
+            slots_pow=18,
+            radix=True,
+            out_cap=1 << 21,
+        )
+        op.output("out", agg, CollectCountsSink(out))
+        run_main(flow)
+        rows = sorted(
+            (k, t, v)
+            for b in out
+            for k, t, v in zip(
+                b.keys.cpu().tolist(),
+                b.ts.cpu().tolist(),
+                b.vals.cpu().tolist(),
+            )
+        )
+        return rows
+
+    serial = run("0")
+    piped = run("1")
+    assert serial == piped
+    assert len(serial) > 0
