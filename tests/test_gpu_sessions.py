@@ -227,3 +227,113 @@ def test_session_gpu_snapshot_roundtrip():
     ra = _rows({k: t.cpu() for k, t in a.close_all().items()})
     rb = _rows({k: t.cpu() for k, t in b.close_all().items()})
     assert ra == rb
+
+
+@pytest.mark.timeout(300)
+def test_sessions_two_ranks_gloo(tmp_path):
+    """keyed_session_agg across 2 gloo ranks: the sync exchange is
+    collective per step; totals must be exact."""
+    import os
+    import subprocess
+    import sys
+    import textwrap
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    prog = tmp_path / "sess2.py"
+    prog.write_text(
+        textwrap.dedent(
+            """
+            import os
+            import torch
+            import torch.distributed as dist
+
+            import bytewax_amd.operators as op
+            from bytewax_amd.dataflow import Dataflow
+            from bytewax_amd.gpu import RecordBatch
+            from bytewax_amd.gpu.operators import keyed_session_agg
+            from bytewax_amd.inputs import (
+                DynamicSource,
+                StatelessSourcePartition,
+            )
+            from bytewax_amd.outputs import (
+                DynamicSink,
+                StatelessSinkPartition,
+            )
+            from bytewax_amd.testing import run_main
+            from datetime import timedelta
+
+            dist.init_process_group("gloo")
+            rank = dist.get_rank()
+
+            class _Src(StatelessSourcePartition):
+                def __init__(self):
+                    self.i = 0
+
+                def next_batch(self):
+                    if self.i >= 3:
+                        raise StopIteration()
+                    base = self.i * 1000
+                    self.i += 1
+                    # Both ranks emit the same 6 keys each step.
+                    return [
+                        RecordBatch(
+                            torch.arange(6, dtype=torch.int32),
+                            torch.full((6,), base, dtype=torch.int64),
+                            None,
+                            max_ts=base,
+                        )
+                    ]
+
+            class Src(DynamicSource):
+                def build(self, step_id, wi, wc):
+                    return _Src()
+
+            total = []
+
+            class _Sink(StatelessSinkPartition):
+                def write_batch(self, items):
+                    for d in items:
+                        total.extend(d["vals"].tolist())
+
+            class Sink(DynamicSink):
+                def build(self, step_id, wi, wc):
+                    return _Sink()
+
+            flow = Dataflow("sess2")
+            s = op.input("inp", flow, Src())
+            agg = keyed_session_agg(
+                "sess", s, gap=timedelta(milliseconds=500), device="cpu"
+            )
+            op.output("out", agg, Sink())
+            run_main(flow)
+            t = torch.tensor([sum(total)], dtype=torch.int64)
+            dist.all_reduce(t)
+            if rank == 0:
+                # 3 steps x 6 keys x 2 ranks, singleton sessions of
+                # count 2 (both ranks' copies merge on the owner).
+                assert int(t.item()) == 36, int(t.item())
+                print("SESS2 OK")
+            dist.destroy_process_group()
+            """
+        )
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(repo)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["MASTER_PORT"] = str(29880 + os.getpid() % 100)
+    procs = []
+    for rank in range(2):
+        e = dict(env, RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK=str(rank))
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, str(prog)],
+                env=e,
+                stdout=subprocess.PIPE,
+                stderr=subprocess.PIPE,
+            )
+        )
+    outs = [p.communicate(timeout=240) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se.decode()[-1500:]
+    assert "SESS2 OK" in outs[0][0].decode()
