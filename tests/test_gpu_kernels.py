@@ -496,3 +496,45 @@ def test_python_engine_pipelined_insert_matches_serial(monkeypatch):
     piped = run("1")
     assert serial == piped
     assert len(serial) > 0
+
+
+def test_insert_pipelined_buffer_growth_matches_serial():
+    """Growing batches force a mid-run scatter-buffer reallocation on
+    the pipelined path; results must still match the serial insert."""
+    _skip_no_gpu()
+    from bytewax_amd.gpu import AGG_COUNT, RecordBatch, WindowAggState, _ms
+
+    align_ms = _ms(ALIGN)
+    dev = torch.device("cuda:0")
+    sizes = [100_000, 150_000, 450_000, 200_000]
+    g = torch.Generator(device="cuda").manual_seed(21)
+    batches = []
+    for i, n in enumerate(sizes):
+        keys = torch.randint(0, 10_000, (n,), dtype=torch.int32,
+                             generator=g, device=dev)
+        ts = torch.randint(0, 5_000, (n,), dtype=torch.int64,
+                           generator=g, device=dev)
+        batches.append(
+            RecordBatch(keys, ts, None, max_ts=(i + 1) * 5_000 - 1,
+                        ts_base=align_ms + i * 5_000)
+        )
+
+    def run(pipelined):
+        st = WindowAggState(
+            dev, align_ms, 1000, AGG_COUNT, slots_pow=16, radix=True,
+            region_bits=9, max_batch=sizes[0],
+        )
+        for b in batches:
+            if pipelined:
+                st.insert_pipelined(
+                    b.keys, b.ts, b.vals, b.ts_base, [], [], b.max_ts
+                )
+            else:
+                st.insert(b)
+        rest = st.close_all()
+        return sorted(
+            zip(rest.keys.cpu().tolist(), rest.ts.cpu().tolist(),
+                rest.vals.cpu().tolist())
+        )
+
+    assert run(False) == run(True)
