@@ -553,6 +553,10 @@ class WindowAggState:
 
         n = int(keys.numel())
         if n > self.rx_max_batch:
+            if getattr(self, "_pipe_ready", False):
+                # Retire in-flight side-stream work before freeing the
+                # old buffers it may still be reading.
+                torch.cuda.synchronize(self.device)
             self._alloc_rx(int(n * 5 // 4))
         self._ensure_pipe()
         par = self._pipe_parity

@@ -521,7 +521,7 @@ def test_insert_pipelined_buffer_growth_matches_serial():
 
     def run(pipelined):
         st = WindowAggState(
-            dev, align_ms, 1000, AGG_COUNT, slots_pow=16, radix=True,
+            dev, align_ms, 1000, AGG_COUNT, slots_pow=19, radix=True,
             region_bits=9, max_batch=sizes[0],
         )
         for b in batches:
