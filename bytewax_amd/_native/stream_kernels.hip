@@ -2381,6 +2381,78 @@ void radix_join_insert(
       error_flag.data_ptr<int32_t>());
 }
 
+// Batch-level session merge: when the gap is at least the batch's
+// time span, no session boundary can fall INSIDE the batch, so each
+// key's per-batch (count, min_ts, max_ts) — computed by the radix
+// stats kernels at full speed — is merged as one unit.  Requires
+// watermark-ordered batches (each batch's events at/after the
+// previous batch's), the same contract as the sorted walk.
+__global__ void k_session_merge_batch(
+    const int32_t* __restrict__ keys,
+    const int64_t* __restrict__ cnt,
+    const int64_t* __restrict__ mn_ts,
+    const int64_t* __restrict__ mx_ts,
+    const int* __restrict__ n_rows,
+    int64_t ts_shift,  // add to mn/mx (zero-based batch timestamps)
+    int64_t gap_ms,
+    uint64_t* __restrict__ skeys,
+    long long* __restrict__ sstart,
+    long long* __restrict__ slast,
+    long long* __restrict__ sacc,
+    uint64_t mask,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_start,
+    int64_t* __restrict__ out_end,
+    int64_t* __restrict__ out_vals,
+    int* __restrict__ out_n,
+    int64_t out_cap,
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag) {
+  int64_t n = *n_rows;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  int64_t local_max = 0;
+  for (; i < n; i += stride) {
+    uint64_t key = (uint64_t)(uint32_t)keys[i];
+    uint64_t slot = session_find_or_claim(skeys, mask, key);
+    if (slot == ~0ULL) {
+      atomicExch(error_flag, 1);
+      continue;
+    }
+    long long lo = mn_ts[i] + ts_shift;
+    long long hi = mx_ts[i] + ts_shift;
+    long long c = cnt[i];
+    if (hi > local_max) local_max = hi;
+    long long last = slast[slot];
+    if (last >= 0 && lo - last > gap_ms) {
+      int idx = atomicAdd(out_n, 1);
+      if (idx < out_cap) {
+        out_keys[idx] = (int32_t)(uint32_t)key;
+        out_start[idx] = sstart[slot];
+        out_end[idx] = last;
+        out_vals[idx] = sacc[slot];
+      } else {
+        atomicExch(error_flag, 1);
+      }
+      last = -1;
+    }
+    if (last < 0) {
+      sstart[slot] = lo;
+      sacc[slot] = c;
+    } else {
+      sacc[slot] += c;
+    }
+    slast[slot] = hi;
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
 __global__ void k_session_restore(
     const int32_t* __restrict__ keys,
     const int64_t* __restrict__ start,
@@ -2490,6 +2562,45 @@ void session_close_migrate(
       out_start.data_ptr<int64_t>(), out_end.data_ptr<int64_t>(),
       out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
       out_keys.numel(), error_flag.data_ptr<int32_t>());
+}
+
+void session_merge_batch(
+    torch::Tensor keys,
+    torch::Tensor cnt,
+    torch::Tensor mn_ts,
+    torch::Tensor mx_ts,
+    torch::Tensor n_rows,
+    torch::Tensor skeys,
+    torch::Tensor sstart,
+    torch::Tensor slast,
+    torch::Tensor sacc,
+    torch::Tensor out_keys,
+    torch::Tensor out_start,
+    torch::Tensor out_end,
+    torch::Tensor out_vals,
+    torch::Tensor out_n,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    int64_t ts_shift,
+    int64_t gap_ms) {
+  int64_t nslots = skeys.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  hipLaunchKernelGGL(
+      k_session_merge_batch, dim3(n_blocks(out_keys.numel(), 256)),
+      block, 0, stream, keys.data_ptr<int32_t>(),
+      cnt.data_ptr<int64_t>(), mn_ts.data_ptr<int64_t>(),
+      mx_ts.data_ptr<int64_t>(), n_rows.data_ptr<int32_t>(), ts_shift,
+      gap_ms, (uint64_t*)skeys.data_ptr<int64_t>(),
+      (long long*)sstart.data_ptr<int64_t>(),
+      (long long*)slast.data_ptr<int64_t>(),
+      (long long*)sacc.data_ptr<int64_t>(),
+      (uint64_t)(nslots - 1), out_keys.data_ptr<int32_t>(),
+      out_start.data_ptr<int64_t>(), out_end.data_ptr<int64_t>(),
+      out_vals.data_ptr<int64_t>(), out_n.data_ptr<int32_t>(),
+      out_keys.numel(),
+      (unsigned long long*)max_ts.data_ptr<int64_t>(),
+      error_flag.data_ptr<int32_t>());
 }
 
 void session_restore(
@@ -3048,6 +3159,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Emit sessions idle past the horizon; migrate live cells");
   m.def("session_restore", &session_restore,
         "Rebuild session cells from a host spill");
+  m.def("session_merge_batch", &session_merge_batch,
+        "Merge per-key per-batch (count, min_ts, max_ts) rows into "
+        "the session table (gap >= batch span fast path)");
   m.def("join_extract", &join_extract,
         "Extract live join state (recovery snapshot)");
   m.def("filter_compact", &filter_compact,

@@ -337,3 +337,59 @@ def test_sessions_two_ranks_gloo(tmp_path):
     for p, (so, se) in zip(procs, outs):
         assert p.returncode == 0, se.decode()[-1500:]
     assert "SESS2 OK" in outs[0][0].decode()
+
+
+@pytest.mark.gpu
+def test_session_batch_fast_path_matches_sort_path():
+    """gap >= batch span triggers the stats-kernel fast path; results
+    must equal the sorted walk on identical data (absolute-ts batches
+    force the sort path)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    dev = torch.device("cuda:0")
+    gap = 5_000
+    span = 1_000
+    bases = [0, 1_000, 10_000, 11_000, 30_000]  # jumps split sessions
+    g = torch.Generator().manual_seed(17)
+    raw = []
+    for base in bases:
+        n = 40_000
+        keys = torch.randint(0, 2_000, (n,), dtype=torch.int32, generator=g)
+        ts0 = torch.randint(0, span, (n,), dtype=torch.int64, generator=g)
+        raw.append((keys, ts0, base))
+
+    def run(fast):
+        st = SessionAggState(
+            dev, gap_ms=gap, mode=AGG_COUNT, slots_pow=13, out_cap=1 << 20
+        )
+        got = []
+        for keys, ts0, base in raw:
+            if fast:
+                b = RecordBatch(
+                    keys.cuda(), ts0.cuda(), None,
+                    max_ts=base + span, ts_base=base,
+                )
+            else:
+                b = RecordBatch(
+                    keys.cuda(), (ts0 + base).cuda(), None,
+                    max_ts=base + span,
+                )
+            st.insert(b)
+            out = st.close_due()
+            if out is not None:
+                got.extend(_rows({k: t.cpu() for k, t in out.items()}))
+        fin = st.close_all()
+        if fin is not None:
+            got.extend(_rows({k: t.cpu() for k, t in fin.items()}))
+        return sorted(got)
+
+    slow = run(False)
+    fast = run(True)
+    assert fast == slow
+    ref = _ref_sessions(
+        [k for keys, _t, _b in raw for k in keys.tolist()],
+        [t + b for _k, ts0, b in raw for t in ts0.tolist()],
+        [1] * sum(len(keys) for keys, _t, _b in raw),
+        gap,
+    )
+    assert fast == ref
