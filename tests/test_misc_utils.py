@@ -34,3 +34,23 @@ def test_route_part_stable():
     assert _route_part("step", "key", 7) == _route_part("step", "key", 7)
     spread = {_route_part("s", f"k{i}", 8) for i in range(100)}
     assert len(spread) == 8
+
+
+def test_ttl_cache_expires_entries():
+    from datetime import datetime, timedelta, timezone
+
+    from bytewax_amd.operators import TTLCache
+
+    t = [datetime(2024, 1, 1, tzinfo=timezone.utc)]
+    calls = []
+
+    def getter(k):
+        calls.append(k)
+        return f"v-{k}-{len(calls)}"
+
+    cache = TTLCache(getter, lambda: t[0], timedelta(seconds=10))
+    assert cache.get("a") == "v-a-1"
+    assert cache.get("a") == "v-a-1"  # cached
+    t[0] += timedelta(seconds=11)
+    assert cache.get("a") == "v-a-2"  # expired, re-fetched
+    assert calls == ["a", "a"]
