@@ -497,3 +497,66 @@ def test_stats_rescale_two_procs_to_one(tmp_path: Path):
     # twice: sum = 2 * (4k + 10).
     assert cnts == {k: 8 for k in range(10)}
     assert sums == {k: 2 * (4 * k + 10) for k in range(10)}
+
+
+@pytest.mark.timeout(240)
+def test_rescale_one_proc_to_two(tmp_path: Path):
+    """Upscale: state written at world=1 resumes on a 2-process
+    cluster; the donor re-exchange splits the restored rows to their
+    new owners with the collective aligned on both ranks (the rank
+    with no donors still participates)."""
+    flow_file = tmp_path / "rescale_flow.py"
+    flow_file.write_text(textwrap.dedent(_FLOW_TEMPLATE))
+    out_file = tmp_path / "rows.txt"
+    rec_dir = tmp_path / "rec"
+    rec_dir.mkdir()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env["RESCALE_PHASE"] = "1"
+    env["RESCALE_OUT"] = str(out_file)
+    subprocess.run(
+        [sys.executable, "-m", "bytewax_amd.recovery", str(rec_dir), "2"],
+        check=True,
+        env=env,
+        capture_output=True,
+    )
+    # Phase 1: ONE process, aborts mid-window.
+    res = subprocess.run(
+        [
+            sys.executable, "-m", "bytewax_amd.run",
+            f"{flow_file}:flow",
+            "-r", str(rec_dir), "-s", "0", "-b", "0",
+        ],
+        env=env,
+        capture_output=True,
+        timeout=200,
+    )
+    assert res.returncode == 0, res.stderr.decode()[-2000:]
+
+    # Phase 2: TWO processes resume the same store.
+    env2 = dict(env)
+    env2["RESCALE_PHASE"] = "2"
+    port = 29960 + os.getpid() % 30
+    addresses = f"127.0.0.1:{port};127.0.0.1:{port + 1}"
+    procs = [
+        subprocess.Popen(
+            [
+                sys.executable, "-m", "bytewax_amd.run",
+                f"{flow_file}:flow", "-i", str(i), "-a", addresses,
+                "-r", str(rec_dir), "-s", "0", "-b", "0",
+            ],
+            env=env2,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE,
+        )
+        for i in range(2)
+    ]
+    for p in procs:
+        stdout, stderr = p.communicate(timeout=200)
+        assert p.returncode == 0, stderr.decode()[-2000:]
+
+    totals = {}
+    for line in out_file.read_text().splitlines():
+        k, t, v = line.split(",")
+        totals[int(k)] = totals.get(int(k), 0) + int(v)
+    assert totals == {k: 8 for k in range(10)}
