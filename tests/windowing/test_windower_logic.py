@@ -112,3 +112,35 @@ def test_snapshot_roundtrip_sliding():
     snap = a.snapshot()
     b = windower.build(snap)
     assert list(b.close_for(ts(10))) == [(0, WindowMetadata(ts(0), ts(10)))]
+
+
+def test_event_clock_watermark_monotonic_and_waits():
+    """EventClock watermark = max event ts - wait + elapsed system
+    time, and never regresses (reference windowing.py
+    _EventClockLogic)."""
+    from datetime import datetime, timedelta, timezone
+
+    from bytewax_amd.operators.windowing import EventClock
+    from bytewax_amd.testing import TimeTestingGetter
+
+    t0 = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    getter = TimeTestingGetter(t0)
+    clock = EventClock(
+        ts_getter=lambda x: x,
+        wait_for_system_duration=timedelta(seconds=10),
+        now_getter=getter.get,
+    )
+    logic = clock.build(None)
+    ev = t0 + timedelta(seconds=100)
+    logic.before_batch()
+    ts, _wm = logic.on_item(ev)
+    assert ts == ev
+    wm0 = logic.on_notify()
+    assert wm0 == ev - timedelta(seconds=10)
+    # System time advances without events: watermark advances too.
+    getter.advance(timedelta(seconds=4))
+    assert logic.on_notify() == wm0 + timedelta(seconds=4)
+    # An older event cannot regress the watermark.
+    logic.before_batch()
+    logic.on_item(t0)
+    assert logic.on_notify() >= wm0 + timedelta(seconds=4)
