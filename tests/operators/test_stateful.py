@@ -214,3 +214,31 @@ def test_stateful_requires_str_key(entry_point):
     op.output("out", s, TestingSink(out))
     with pytest.raises(TypeError):
         entry_point(flow)
+
+
+def test_stateful_flat_map_expands_and_discards(entry_point):
+    """stateful_flat_map: one-to-many with per-key state; returning
+    None state discards the key (reference operators/__init__.py
+    stateful_flat_map)."""
+    import bytewax_amd.operators as op
+    from bytewax_amd.dataflow import Dataflow
+    from bytewax_amd.testing import TestingSink, TestingSource
+
+    out = []
+    flow = Dataflow("sfm")
+    s = op.input(
+        "inp", flow, TestingSource([("k", 1), ("k", 2), ("k", 3)])
+    )
+
+    def dup_until_two(state, v):
+        seen = (state or 0) + 1
+        # Emit v repeated `seen` times; discard state after 2 values.
+        new_state = None if seen >= 2 else seen
+        return (new_state, [v] * seen)
+
+    s = op.stateful_flat_map("fm", s, dup_until_two)
+    op.output("out", s, TestingSink(out))
+    entry_point(flow)
+    # k=1 (seen 1 -> [1]), k=2 (seen 2 -> [2,2], discard),
+    # k=3 (fresh, seen 1 -> [3]).
+    assert sorted(out) == [("k", 1), ("k", 2), ("k", 2), ("k", 3)]
