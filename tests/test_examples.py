@@ -44,3 +44,18 @@ def test_gpu_wordcount_example_cpu_twin():
     res = _run([sys.executable, "examples/gpu_wordcount.py"])
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     assert "counted 20000000 events" in res.stdout.decode()
+
+
+def test_onebrc_example_cpu_twin():
+    res = _run(
+        [
+            sys.executable,
+            "examples/onebrc_gpu.py",
+            "--rows",
+            "200000",
+            "--rows-per-batch",
+            "50000",
+        ]
+    )
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "aggregated 200000 rows over 10000 stations" in res.stdout.decode()
