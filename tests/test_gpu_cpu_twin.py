@@ -366,3 +366,29 @@ def test_deferred_close_carry_roundtrips_through_snapshot():
         )
     ]
     assert rows_eof == [(0, 100, 1)]
+
+
+def test_close_due_respects_wait_allowance():
+    """`wait` (lateness allowance) delays window close by that much
+    watermark time (reference windowing wait_for_system_duration
+    analog on the columnar path)."""
+    align_ms = _ms(ALIGN)
+    st = WindowAggState(torch.device("cpu"), align_ms, 60_000, AGG_COUNT)
+    st.insert(
+        RecordBatch(
+            torch.tensor([1], dtype=torch.int32),
+            torch.tensor([align_ms + 10], dtype=torch.int64),
+            max_ts=align_ms + 70_000,
+        )
+    )
+    # Watermark 70s: window 0 closes without allowance...
+    assert st.close_due(wait_ms=20_000) is None  # ...but not with 20s
+    st.insert(
+        RecordBatch(
+            torch.tensor([1], dtype=torch.int32),
+            torch.tensor([align_ms + 15], dtype=torch.int64),
+            max_ts=align_ms + 90_000,
+        )
+    )
+    out = st.close_due(wait_ms=20_000)  # watermark 90s - 20s >= 60s
+    assert out is not None and out.vals.tolist() == [2]
