@@ -427,6 +427,13 @@ class WindowAggState:
     def insert(self, batch: RecordBatch) -> None:
         if self.cpu:
             self._insert_cpu(batch)
+            # Same contract as the device path: a host-known max_ts
+            # advances the watermark even past the event timestamps.
+            if (
+                batch.max_ts is not None
+                and batch.max_ts > self.max_ts_host
+            ):
+                self.max_ts_host = batch.max_ts
             return
         if self.radix and len(batch) > self.rx_max_batch:
             # Scatter buffers grow to fit the largest batch seen

@@ -704,6 +704,11 @@ class SessionAggState:
             return
         if self.cpu:
             self._insert_cpu(batch)
+            if (
+                batch.max_ts is not None
+                and batch.max_ts > self.max_ts_host
+            ):
+                self.max_ts_host = batch.max_ts
             return
         if (
             self.mode == AGG_COUNT
