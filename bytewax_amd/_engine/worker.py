@@ -55,6 +55,15 @@ def _now() -> datetime:
 
 
 def _route_key(key: str, worker_count: int) -> int:
+    # "shard-<i>" is a reserved namespace used by the GPU columnar
+    # operators to pin rank i's device batches to worker i: routing it
+    # by rank (not hash) guarantees the batch never crosses the data
+    # plane regardless of worker count (adler32 only lines up for
+    # power-of-two worlds by arithmetic accident).
+    if key.startswith("shard-"):
+        suffix = key[6:]
+        if suffix.isdigit():
+            return int(suffix) % worker_count
     return zlib.adler32(key.encode()) % worker_count
 
 

@@ -1023,13 +1023,14 @@ __global__ void k_stats_insert(
     uint64_t mask,
     int64_t align_ms,
     int64_t len_ms,
+    int64_t ts_base,
     unsigned long long* __restrict__ max_ts,
     int* __restrict__ error_flag) {
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   int64_t local_max = 0;
   for (; i < n; i += stride) {
-    int64_t t = ts[i];
+    int64_t t = ts[i] + ts_base;
     if (t > local_max) local_max = t;
     int64_t win = (t - align_ms) / len_ms;
     uint64_t packed =
@@ -2097,7 +2098,8 @@ void stats_insert(
     torch::Tensor max_ts,
     torch::Tensor error_flag,
     int64_t align_ms,
-    int64_t len_ms) {
+    int64_t len_ms,
+    int64_t ts_base) {
   check_dev(keys, torch::kInt32, "keys");
   check_dev(ts, torch::kInt64, "ts");
   check_dev(vals, torch::kInt64, "vals");
@@ -2117,7 +2119,7 @@ void stats_insert(
       (long long*)tsum.data_ptr<int64_t>(),
       (long long*)tmin.data_ptr<int64_t>(),
       (long long*)tmax.data_ptr<int64_t>(), (uint64_t)(nslots - 1),
-      align_ms, len_ms,
+      align_ms, len_ms, ts_base,
       (unsigned long long*)max_ts.data_ptr<int64_t>(),
       error_flag.data_ptr<int32_t>());
 }

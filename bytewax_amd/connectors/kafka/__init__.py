@@ -208,7 +208,12 @@ class _KafkaSourcePartition(
             )
             item = ksm if error is None else KafkaError(error, ksm)
             batch.append(item)
-            last_offset = msg.offset()
+            # Error events can report offset -1/None; advancing (or
+            # rewinding) from those would re-deliver already-emitted
+            # messages after a resume.
+            off = msg.offset()
+            if off is not None and off >= 0:
+                last_offset = off
         if last_offset is not None:
             self._offset = last_offset + 1
         return batch

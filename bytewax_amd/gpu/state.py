@@ -153,6 +153,7 @@ class StatsAggState:
                 self.error_flag,
                 self.align_ms,
                 self.len_ms,
+                batch.ts_base,
             )
         if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
             self.max_ts_host = batch.max_ts
@@ -315,12 +316,12 @@ class StatsAggState:
         self.k.stats_insert(
             keys, ts, mn, self.tkeys, self.tcnt, self.tsum, self.tmin,
             self.tmax, self.max_ts_dev, self.error_flag,
-            self.align_ms, self.len_ms,
+            self.align_ms, self.len_ms, 0,
         )
         self.k.stats_insert(
             keys, ts, mx, self.tkeys, self.tcnt, self.tsum, self.tmin,
             self.tmax, self.max_ts_dev, self.error_flag,
-            self.align_ms, self.len_ms,
+            self.align_ms, self.len_ms, 0,
         )
         # Correct cnt/sum: current slots have cnt=2, sum=min+max; the
         # delta columns below restore the snapshotted values exactly.
@@ -830,13 +831,27 @@ class SessionAggState:
                 "pending": list(self._closed),
             }
         live = (self.skeys >= 0) & (self.slast >= 0)
+        # Sessions closed by gaps during insert() but not yet drained
+        # sit in the out buffer; persist them (without draining — a
+        # snapshot must not mutate) so a restore re-emits them exactly
+        # once, mirroring the CPU twin's `_closed` list.
+        pend_n = int(self.out_n.item())
+        pending = [
+            (int(k), int(s), int(e), int(v))
+            for k, s, e, v in zip(
+                self.out_keys[:pend_n].cpu().tolist(),
+                self.out_start[:pend_n].cpu().tolist(),
+                self.out_end[:pend_n].cpu().tolist(),
+                self.out_vals[:pend_n].cpu().tolist(),
+            )
+        ]
         return {
             "keys": self.skeys[live].to("cpu").numpy().astype("int32"),
             "start": self.sstart[live].cpu().numpy().copy(),
             "last": self.slast[live].cpu().numpy().copy(),
             "acc": self.sacc[live].cpu().numpy().copy(),
             "max_ts": self.max_ts_host,
-            "pending": [],
+            "pending": pending,
         }
 
     def restore_from_host(self, snap: Dict[str, Any]) -> None:
@@ -853,6 +868,28 @@ class SessionAggState:
                 tuple(r) for r in snap.get("pending", [])
             )
             return
+        pending = snap.get("pending", [])
+        if pending:
+            # Re-stage snapshotted closed-but-undrained sessions into
+            # the out buffer; the next drain re-emits them.
+            n0 = int(self.out_n.item())
+            n1 = n0 + len(pending)
+            if n1 > self.out_cap:
+                msg = "session restore overflows out_cap"
+                raise RuntimeError(msg)
+            self.out_keys[n0:n1] = torch.tensor(
+                [r[0] for r in pending], dtype=torch.int32
+            ).to(self.device)
+            self.out_start[n0:n1] = torch.tensor(
+                [r[1] for r in pending], dtype=torch.int64
+            ).to(self.device)
+            self.out_end[n0:n1] = torch.tensor(
+                [r[2] for r in pending], dtype=torch.int64
+            ).to(self.device)
+            self.out_vals[n0:n1] = torch.tensor(
+                [r[3] for r in pending], dtype=torch.int64
+            ).to(self.device)
+            self.out_n.fill_(n1)
         if len(snap["keys"]) == 0:
             return
         self.k.session_restore(
