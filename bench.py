@@ -29,6 +29,15 @@ def parse_args():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--events-per-batch", type=int, default=64_000_000)
+    p.add_argument(
+        "--batches-per-poll",
+        type=int,
+        default=150,
+        help="RecordBatches the source hands the engine per "
+        "scheduling step; one timed step = this many batches "
+        "(amortizes per-step engine overhead and keeps the timed "
+        "region >= ~5 s at the default steps)",
+    )
     p.add_argument("--vocab", type=int, default=1_000_000)
     p.add_argument("--window-sec", type=int, default=60)
     p.add_argument("--sim-ms-per-batch", type=int, default=5000)
@@ -68,11 +77,12 @@ def parse_args():
     )
     p.add_argument(
         "--engine",
-        choices=["auto", "python", "native", "graph"],
+        choices=["auto", "dataflow", "python", "native", "graph"],
         default="auto",
-        help="'native' = C++ step loop (single-GPU); 'python' = the "
-        "per-step engine loop (required for multi-GPU exchange); "
-        "'auto' picks native for world==1 on GPU",
+        help="'dataflow' (= 'python', the default): the full dataflow "
+        "engine — source -> keyed_window_agg -> sink under run_main "
+        "(single- and multi-GPU); 'native' = the bare C++ step loop "
+        "(kernel-pipeline comparison, single-GPU only)",
     )
     return p.parse_args()
 
@@ -120,15 +130,19 @@ def main():
             if on_gpu:
                 torch.cuda.synchronize()
 
+    B = args.batches_per_poll
     timings = {"t0": None, "t1": None, "step_starts": []}
 
     class _BenchPartition(_SyntheticPartition):
         """Source partition with timing hooks: polls are the step
-        boundaries (processing of batch i completes before poll
-        i+1)."""
+        boundaries (processing of the batches handed out at poll i
+        completes before poll i+1)."""
+
+        polls = 0
 
         def next_batch(self):
-            i = self.emitted
+            i = self.polls
+            self.polls += 1
             if i == W:
                 barrier_sync()
                 timings["t0"] = time.perf_counter()
@@ -152,11 +166,12 @@ def main():
                 args.sim_ms_per_batch,
                 _ms(align),
                 seed=42 + rank * 7919,
+                per_poll=B,
             )
 
     engine = args.engine
-    if engine == "auto":
-        engine = "native" if (world == 1 and on_gpu) else "python"
+    if engine in ("auto", "dataflow"):
+        engine = "python"
     if engine == "graph":
         # hipGraph latency mode: single-pass COUNT path.
         args.radix = False
@@ -205,22 +220,23 @@ def main():
                 args.sim_ms_per_batch, pipelined=not args.no_pipeline,
             )
 
-        # Warmup.
-        r, _ = run_steps(0, W)
+        # Warmup.  One timed step = B batches (matching the dataflow
+        # engine's batches-per-poll) so `ms_per_step` is comparable.
+        r, _ = run_steps(0, W * B)
         closed_rows += r
         barrier_sync()
         t0 = time.perf_counter()
-        r, step_ns = run_steps(W, K)
+        r, step_ns = run_steps(W * B, K * B)
         closed_rows += r
         barrier_sync()
         t1 = time.perf_counter()
         timings["t0"], timings["t1"] = t0, t1
-        # Latency probes: single steps with a sync each, after the
+        # Latency probes: single batches with a sync each, after the
         # throughput region (launch-to-launch gaps are not a latency
         # measure under async execution).
         for i in range(args.latency_probes):
             lt0 = time.perf_counter()
-            r, _ = run_steps(W + K + i, 1)
+            r, _ = run_steps((W + K) * B + i, 1)
             closed_rows += r
             torch.cuda.synchronize()
             lat.append((time.perf_counter() - lt0) * 1000.0)
@@ -263,7 +279,7 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    total_events = K * E * world
+    total_events = K * B * E * world
     events_per_sec = total_events / elapsed
     ms_per_step = elapsed / K * 1000.0
 
@@ -288,11 +304,17 @@ def main():
                     "config": {
                         "model": "tumbling-window wordcount (keyed count, 60s windows)",
                         "events_per_batch_per_gpu": E,
+                        "batches_per_step": B,
                         "vocab": args.vocab,
                         "window_sec": args.window_sec,
                         "sim_ms_per_batch": args.sim_ms_per_batch,
                         "parallelism": f"key-hash all-to-allv dp{world}",
-                        "engine": engine,
+                        "engine": (
+                            "dataflow (run_main: source -> "
+                            "keyed_window_agg -> sink)"
+                            if engine == "python"
+                            else engine
+                        ),
                         "radix": args.radix,
                         "radix_v2": args.radix_v2,
                         "p99_step_ms": p99_ms,

@@ -61,12 +61,14 @@ class _SyntheticPartition(StatelessSourcePartition[RecordBatch]):
         seed: int,
         pool: int = 8,
         vals: bool = False,
+        per_poll: int = 1,
     ):
         import torch
 
         self.device = device
         self.n_batches = n_batches
         self.emitted = 0
+        self.per_poll = per_poll
         self.sim_ms_per_batch = sim_ms_per_batch
         self.align_ms = align_ms
         # Generate pools directly on the target device (CPU-side
@@ -101,11 +103,7 @@ class _SyntheticPartition(StatelessSourcePartition[RecordBatch]):
                 for _ in range(pool)
             ]
 
-    def next_batch(self) -> List[RecordBatch]:
-        if self.n_batches is not None and self.emitted >= self.n_batches:
-            raise StopIteration()
-        i = self.emitted
-        self.emitted += 1
+    def _mk_batch(self, i: int) -> RecordBatch:
         start = self.align_ms + i * self.sim_ms_per_batch
         keys = self.key_pool[i % len(self.key_pool)]
         vals = (
@@ -116,15 +114,27 @@ class _SyntheticPartition(StatelessSourcePartition[RecordBatch]):
         # Zero-based template + scalar base: consumers (insert kernels
         # and the exchange wire format) apply the base without ever
         # materializing an absolute int64 timestamp column.
-        return [
-            RecordBatch(
-                keys,
-                self.ts_template,
-                vals,
-                max_ts=start + self.sim_ms_per_batch - 1,
-                ts_base=start,
-            )
-        ]
+        return RecordBatch(
+            keys,
+            self.ts_template,
+            vals,
+            max_ts=start + self.sim_ms_per_batch - 1,
+            ts_base=start,
+        )
+
+    def next_batch(self) -> List[RecordBatch]:
+        if self.n_batches is not None and self.emitted >= self.n_batches:
+            raise StopIteration()
+        out = []
+        for _ in range(self.per_poll):
+            if (
+                self.n_batches is not None
+                and self.emitted >= self.n_batches
+            ):
+                break
+            out.append(self._mk_batch(self.emitted))
+            self.emitted += 1
+        return out
 
 
 @dataclass
@@ -146,6 +156,7 @@ class SyntheticEventSource(DynamicSource):
     device: str = "cuda"
     seed: int = 42
     with_vals: bool = False
+    per_poll: int = 1
 
     def build(
         self, step_id: str, worker_index: int, worker_count: int
@@ -162,6 +173,7 @@ class SyntheticEventSource(DynamicSource):
             _ms(self.align_to),
             self.seed + worker_index * 7919,
             vals=self.with_vals,
+            per_poll=self.per_poll,
         )
 
 
@@ -356,10 +368,15 @@ class _DeviceWindowLogic(StatefulBatchLogic):
         self._apply_rescale()
         out: List[RecordBatch] = []
         self._take_carry(out)
-        closed = self.state.close_resolve()  # previous step's close
-        if closed is not None:
-            out.append(closed)
+        # Close cadence is per BATCH (not per poll): sources may hand
+        # the engine many batches per scheduling step, and the window
+        # table is sized for the close-every-window working set.  The
+        # close of batch i resolves while batch i+1 inserts (one-step
+        # deferred readback).
         for batch in batches:
+            closed = self.state.close_resolve()
+            if closed is not None:
+                out.append(closed)
             if self.exchange:
                 works, exchanged = exchange_by_key(batch, async_op=True)
                 self._flush_pending()
@@ -371,13 +388,13 @@ class _DeviceWindowLogic(StatefulBatchLogic):
                 )
             else:
                 self.state.insert(batch)
-        self.state.close_launch(self.wait_ms)
-        if not self.state.cpu:
-            ev = torch.cuda.Event()
-            ev.record()
-            self._evq.append(ev)
-            if len(self._evq) > self.PIPELINE:
-                self._evq.pop(0).synchronize()
+            self.state.close_launch(self.wait_ms)
+            if not self.state.cpu:
+                ev = torch.cuda.Event()
+                ev.record()
+                self._evq.append(ev)
+                if len(self._evq) > self.PIPELINE:
+                    self._evq.pop(0).synchronize()
         return (out, StatefulBatchLogic.RETAIN)
 
     def on_eof(self):
