@@ -31,6 +31,7 @@
 #include <ATen/hip/HIPContext.h>
 
 #include <chrono>
+#include <type_traits>
 #include <cstdlib>
 #include <cstdint>
 #include <vector>
@@ -49,6 +50,33 @@ static inline int n_blocks(int64_t n, int block) {
   if (b > 2048) b = 2048;
   if (b < 1) b = 1;
   return (int)b;
+}
+
+// Scatter-variant selection for the radix window path (A/B-able via
+// env until the measured default settles; see profiles/).
+enum ScatterKind { SCAT_FIXED = 0, SCAT_DIRECT = 1, SCAT_STAGED = 2 };
+
+static ScatterKind scatter_kind_env() {
+  if (const char* s = std::getenv("BYTEWAX_SCATTER")) {
+    if (s[0] == 'd') return SCAT_DIRECT;
+    if (s[0] == 'f') return SCAT_FIXED;
+    if (s[0] == 's') return SCAT_STAGED;
+  }
+  if (const char* sd = std::getenv("BYTEWAX_SCATTER_DIRECT")) {
+    if (atoi(sd) != 0) return SCAT_DIRECT;
+  }
+  return SCAT_STAGED;
+}
+
+// Extra segment coarsening: scatter segments cover 2^coarse table
+// regions each (longer runs per segment; the agg kernel stages
+// 2^(region_bits+coarse) LDS slots per block).
+static int scatter_coarse_bits(ScatterKind kind) {
+  if (const char* cb = std::getenv("BYTEWAX_SCATTER_COARSE_BITS")) {
+    int v = atoi(cb);
+    if (v >= 0 && v <= 4) return v;
+  }
+  return kind == SCAT_STAGED ? 2 : 0;
 }
 
 __device__ __forceinline__ uint64_t mix64(uint64_t x) {
@@ -643,6 +671,175 @@ __global__ void k_radix_scatter_direct(
   }
 }
 
+// Line-staged scatter: kills the partial-line write amplification of
+// the scattered 8 B segment writes (round-1 PMC: 7.7x — ~2 GB of
+// actual HBM writes for a 256 MB payload).  Two invariants make every
+// `ev_packed` line a single full-line HBM write:
+//   1. the shared segment cursors advance only in SC_GRAN-event
+//      (64 B) units, so each line is exclusively owned by the one
+//      block that reserved it (no cross-XCD line sharing);
+//   2. a block writes all 8 entries of its granule within one tile
+//      phase, so the line completes in the local XCD's L2 (and is
+//      evicted once, whole) instead of staying open across the whole
+//      input sweep.
+// Events stage per segment in LDS; residual (< SC_GRAN) events per
+// segment ride along across tiles and flush as sentinel-padded
+// granules at the end (the aggregation kernels skip EMPTY_SLOT).
+// Segments may be COARSER than table regions (seg_bits >
+// region_bits): fewer, longer runs; the agg kernel then covers
+// 2^(seg_bits-region_bits) regions per block.
+#define SC_GRAN 8  // events per cursor reservation = one 64 B line
+
+template <int MODE, typename TS = int64_t>
+__global__ __launch_bounds__(256) void k_radix_scatter_staged(
+    const int32_t* __restrict__ keys,
+    const TS* __restrict__ ts,
+    const int64_t* __restrict__ vals,  // nullptr for COUNT
+    int64_t n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t ts_base,
+    uint64_t mask,
+    int seg_bits,
+    int64_t cap,  // per-segment capacity; multiple of SC_GRAN
+    int* __restrict__ gcursors,        // [nseg], zeroed before 1st call
+    uint64_t* __restrict__ ev_packed,  // [nseg * cap]
+    int64_t* __restrict__ ev_vals,     // [nseg * cap] (SUM)
+    int* __restrict__ ov_cursor,
+    uint64_t* __restrict__ ov_packed,
+    int64_t* __restrict__ ov_vals,
+    int64_t ov_cap,
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag,
+    uint64_t win_m2,
+    uint64_t win_maxfast) {
+  extern __shared__ char smem[];
+  const int nseg = (int)(((mask + 1) >> seg_bits));
+  uint64_t* res = (uint64_t*)smem;  // [nseg][SC_GRAN] residual staging
+  int64_t* res_v =
+      (int64_t*)(smem + (size_t)nseg * SC_GRAN * 8);  // SUM only
+  int* res_cnt =
+      (int*)(smem + (size_t)nseg * SC_GRAN * 8 *
+                        (MODE == AGG_SUM ? 2 : 1));  // [nseg]
+  int* lhist = res_cnt + nseg;  // [nseg] tile hist, then grant
+  int* lbase = lhist + nseg;    // [nseg] granted global base
+  int* lofs = lbase + nseg;     // [nseg] virtual-position cursor
+  for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+    res_cnt[b] = 0;
+    lhist[b] = 0;
+  }
+  __syncthreads();
+
+  // One granule write, with capacity split to the overflow spill.
+  auto emit = [&](int b, int64_t gpos, uint64_t packed, int64_t v) {
+    if (gpos < cap) {
+      int64_t pos = (int64_t)b * cap + gpos;
+      ev_packed[pos] = packed;
+      if (MODE == AGG_SUM) ev_vals[pos] = v;
+    } else if (packed != EMPTY_SLOT) {
+      int opos = atomicAdd(ov_cursor, 1);
+      if (opos < ov_cap) {
+        ov_packed[opos] = packed;
+        if (MODE == AGG_SUM) ov_vals[opos] = v;
+      } else {
+        atomicExch(error_flag, 1);
+      }
+    }
+  };
+
+  const int U = 16;
+  const int64_t tile = (int64_t)blockDim.x * U;
+  int64_t local_max = 0;
+  for (int64_t t0 = (int64_t)blockIdx.x * tile; t0 < n;
+       t0 += (int64_t)gridDim.x * tile) {
+    uint64_t pk[U];
+    int64_t pv[U];
+    int sg[U];
+    // P1: load, classify, histogram.
+    for (int u = 0; u < U; ++u) {
+      int64_t i = t0 + (int64_t)u * blockDim.x + threadIdx.x;
+      sg[u] = -1;
+      if (i >= n) continue;
+      int64_t t = (int64_t)ts[i] + ts_base;
+      if (t > local_max) local_max = t;
+      int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+      uint64_t packed =
+          ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+      pk[u] = packed;
+      if (MODE == AGG_SUM) pv[u] = vals[i];
+      int b = (int)region_of(mix64(packed), mask, seg_bits);
+      sg[u] = b;
+      atomicAdd(&lhist[b], 1);
+    }
+    __syncthreads();
+    // P2: per-segment granule reservation.
+    for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+      int tot = res_cnt[b] + lhist[b];
+      int grant = tot & ~(SC_GRAN - 1);
+      lbase[b] = grant > 0 ? atomicAdd(&gcursors[b], grant) : 0;
+      lofs[b] = res_cnt[b];
+      lhist[b] = grant;
+    }
+    __syncthreads();
+    // P3a: drain old residual into the granted granule (granule >=
+    // SC_GRAN > residual when granted, so all-or-nothing).
+    for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+      int grant = lhist[b];
+      if (grant > 0) {
+        int rc = res_cnt[b];
+        int gb = lbase[b];
+        for (int j = 0; j < rc; ++j) {
+          emit(b, (int64_t)gb + j, res[(size_t)b * SC_GRAN + j],
+               MODE == AGG_SUM ? res_v[(size_t)b * SC_GRAN + j] : 0);
+        }
+        res_cnt[b] = 0;
+      }
+    }
+    __syncthreads();
+    // P3b: place this tile's events at their virtual positions.
+    for (int u = 0; u < U; ++u) {
+      int b = sg[u];
+      if (b < 0) continue;
+      int vpos = atomicAdd(&lofs[b], 1);
+      int grant = lhist[b];
+      if (vpos < grant) {
+        emit(b, (int64_t)lbase[b] + vpos, pk[u],
+             MODE == AGG_SUM ? pv[u] : 0);
+      } else {
+        res[(size_t)b * SC_GRAN + (vpos - grant)] = pk[u];
+        if (MODE == AGG_SUM) res_v[(size_t)b * SC_GRAN + (vpos - grant)] = pv[u];
+      }
+    }
+    __syncthreads();
+    // P3c: carry the leftover count; reset the histogram.
+    for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+      res_cnt[b] = lofs[b] - lhist[b];
+      lhist[b] = 0;
+    }
+    __syncthreads();
+  }
+  // Final flush: pad each non-empty residual to a full granule with
+  // EMPTY_SLOT sentinels (skipped by the agg kernels).
+  for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+    int rc = res_cnt[b];
+    if (rc == 0) continue;
+    int gb = atomicAdd(&gcursors[b], SC_GRAN);
+    for (int j = 0; j < SC_GRAN; ++j) {
+      uint64_t p = j < rc ? res[(size_t)b * SC_GRAN + j] : EMPTY_SLOT;
+      emit(b, (int64_t)gb + j, p,
+           (MODE == AGG_SUM && j < rc) ? res_v[(size_t)b * SC_GRAN + j]
+                                       : 0);
+    }
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
 // Aggregate the overflow spill straight into the table (tiny for
 // uniform keys).
 template <int MODE>
@@ -670,7 +867,7 @@ __global__ void k_overflow_agg(
 }
 
 template <int MODE>
-__global__ __launch_bounds__(256) void k_radix_agg(
+__global__ __launch_bounds__(1024) void k_radix_agg(
     const uint64_t* __restrict__ ev_packed,
     const int64_t* __restrict__ ev_vals,
     const int* __restrict__ offsets,
@@ -680,9 +877,11 @@ __global__ __launch_bounds__(256) void k_radix_agg(
     unsigned long long* __restrict__ tvals,
     uint64_t mask,
     int region_bits,
+    int lds_bits,  // LDS staging-table slots (>= region_bits when one
+                   // scatter segment spans several table regions)
     int* __restrict__ error_flag) {
   extern __shared__ char smem[];
-  int region = 1 << region_bits;
+  int region = 1 << lds_bits;
   uint64_t* lkeys = (uint64_t*)smem;
   unsigned long long* lvals =
       (unsigned long long*)(smem + (size_t)region * sizeof(uint64_t));
@@ -697,6 +896,7 @@ __global__ __launch_bounds__(256) void k_radix_agg(
   int start = offsets[b];
   for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
     uint64_t packed = ev_packed[start + j];
+    if (packed == EMPTY_SLOT) continue;  // staged-scatter pad
     unsigned long long inc =
         (MODE == AGG_SUM) ? (unsigned long long)ev_vals[start + j] : 1ULL;
     uint64_t h64 = mix64(packed);
@@ -721,15 +921,16 @@ __global__ __launch_bounds__(256) void k_radix_agg(
       lh = (lh + 1) & (region - 1);
     }
     if (!done) {
-      // LDS region full (only possible when the global region is at
-      // least as full): flush straight to the global region.
+      // LDS staging full (only possible when the global regions are
+      // at least as full): flush straight to the global region.
       if (!hash_add(tkeys, tvals, mask, region_bits, packed, inc)) {
         atomicExch(error_flag, 1);
       }
     }
   }
   __syncthreads();
-  // Flush distinct keys once into the region's contiguous HBM slice.
+  // Flush distinct keys once into each key's own table region (the
+  // segment's regions are contiguous in HBM, so flushes stay local).
   for (int s = threadIdx.x; s < region; s += blockDim.x) {
     if (lkeys[s] != EMPTY_SLOT) {
       if (!hash_add(tkeys, tvals, mask, region_bits, lkeys[s], lvals[s])) {
@@ -1696,13 +1897,9 @@ void radix_window_insert(
   int64_t nb = nslots >> region_bits;
   TORCH_CHECK(nb >= 1 && nb <= 8192, "region count out of range");
   TORCH_CHECK(gcursors.numel() >= nb, "gcursors too small");
-  int64_t cap = ev_packed.numel() / nb;
-  TORCH_CHECK(cap * nb >= 2 * n || cap >= n,
-              "scatter buffers too small (need ~2x batch)");
   const int64_t* vptr = nullptr;
   if (mode == AGG_SUM) {
     TORCH_CHECK(vals.has_value(), "sum mode requires vals");
-    TORCH_CHECK(ev_vals.numel() >= nb * cap, "ev_vals too small");
     vptr = vals->data_ptr<int64_t>();
   }
   if (n == 0) return;
@@ -1710,9 +1907,44 @@ void radix_window_insert(
   uint64_t mask = (uint64_t)(nslots - 1);
   dim3 block(256);
   dim3 grid(n_blocks(n, 256));
-  gcursors.narrow(0, 0, nb).zero_();
+
+  ScatterKind kind = scatter_kind_env();
+  int coarse = scatter_coarse_bits(kind);
+  int seg_bits = (int)region_bits + coarse;
+  int64_t nseg = nslots >> seg_bits;
+  // LDS budgets: staged scatter stages SC_GRAN packed (+vals) per
+  // segment; the agg kernel stages 16 B per LDS slot.
+  size_t staged_lds = (size_t)nseg * SC_GRAN * 8 *
+                          (mode == AGG_SUM ? 2 : 1) +
+                      4 * (size_t)nseg * sizeof(int);
+  while (kind == SCAT_STAGED &&
+         (nseg < 1 || staged_lds > 160 * 1024 || seg_bits > 13)) {
+    if (coarse > 0 && seg_bits > 13) {
+      coarse -= 1;
+    } else {
+      kind = SCAT_FIXED;
+      coarse = 0;
+    }
+    seg_bits = (int)region_bits + coarse;
+    nseg = nslots >> seg_bits;
+    staged_lds = (size_t)nseg * SC_GRAN * 8 * (mode == AGG_SUM ? 2 : 1) +
+                 4 * (size_t)nseg * sizeof(int);
+  }
+  if (seg_bits > 13) {
+    coarse = 0;
+    seg_bits = (int)region_bits;
+    nseg = nb;
+  }
+  int64_t cap = ev_packed.numel() / nseg;
+  if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
+  TORCH_CHECK(cap * nseg >= 2 * n || cap >= n,
+              "scatter buffers too small (need ~2x batch)");
+  if (mode == AGG_SUM) {
+    TORCH_CHECK(ev_vals.numel() >= nseg * cap, "ev_vals too small");
+  }
+  gcursors.narrow(0, 0, nseg).zero_();
   ov_cursor.zero_();
-  size_t hist_lds = (size_t)nb * sizeof(int);
+  size_t hist_lds = (size_t)nseg * sizeof(int);
 
   // Scatter grid: fewer blocks keep fewer segment cursors (and thus
   // partially-written cache lines) open at once — per-XCD L2 can then
@@ -1722,10 +1954,6 @@ void radix_window_insert(
   if (const char* sb = std::getenv("BYTEWAX_SCATTER_BLOCKS")) {
     int v = atoi(sb);
     if (v > 0) env_blocks = v;
-  }
-  bool direct = false;
-  if (const char* sd = std::getenv("BYTEWAX_SCATTER_DIRECT")) {
-    direct = atoi(sd) != 0;
   }
 
   struct Seg {
@@ -1745,14 +1973,13 @@ void radix_window_insert(
 
   uint64_t win_m2, win_maxfast;
   magic_div_u64(len_ms, &win_m2, &win_maxfast);
-  auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx) {
-    // The direct variant uses no LDS; don't charge it occupancy.
-    size_t lds = direct ? 0 : 2 * hist_lds;
+  auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx,
+                  size_t lds) {
     hipLaunchKernelGGL(
         kern, dim3(gx), block, lds, stream,
         keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
         vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
-        sg.base, mask, (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
         ov_cursor.data_ptr<int32_t>(),
@@ -1762,58 +1989,64 @@ void radix_window_insert(
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
   };
-  for (const Seg& sg : segs) {
+  auto scat_any = [&](auto tsptr, const Seg& sg) {
+    using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
     unsigned gx = (unsigned)n_blocks(sg.n, 256);
     if (env_blocks > 0 && (unsigned)env_blocks < gx) gx = (unsigned)env_blocks;
-    if (seg32) {
-      const int32_t* t32 = ts.data_ptr<int32_t>();
-      if (direct) {
-        if (mode == AGG_COUNT)
-          scat(k_radix_scatter_direct<AGG_COUNT, int32_t>, t32, sg, gx);
-        else
-          scat(k_radix_scatter_direct<AGG_SUM, int32_t>, t32, sg, gx);
-      } else if (mode == AGG_COUNT) {
-        scat(k_radix_scatter_fixed<AGG_COUNT, int32_t>, t32, sg, gx);
-      } else {
-        scat(k_radix_scatter_fixed<AGG_SUM, int32_t>, t32, sg, gx);
-      }
+    if (kind == SCAT_STAGED) {
+      // One tile = blockDim * 16 events; enough blocks to fill the
+      // chip at the LDS-bounded occupancy, few enough to keep the
+      // end-of-kernel residual padding small.
+      unsigned gs = (unsigned)((sg.n + 4095) / 4096);
+      if (gs > 512) gs = 512;
+      if (gs < 1) gs = 1;
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_staged<AGG_COUNT, TSV>, tsptr, sg, gs,
+             staged_lds);
+      else
+        scat(k_radix_scatter_staged<AGG_SUM, TSV>, tsptr, sg, gs,
+             staged_lds);
+    } else if (kind == SCAT_DIRECT) {
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_direct<AGG_COUNT, TSV>, tsptr, sg, gx, 0);
+      else
+        scat(k_radix_scatter_direct<AGG_SUM, TSV>, tsptr, sg, gx, 0);
     } else {
-      const int64_t* t64 = ts.data_ptr<int64_t>();
-      if (direct) {
-        if (mode == AGG_COUNT)
-          scat(k_radix_scatter_direct<AGG_COUNT, int64_t>, t64, sg, gx);
-        else
-          scat(k_radix_scatter_direct<AGG_SUM, int64_t>, t64, sg, gx);
-      } else if (mode == AGG_COUNT) {
-        scat(k_radix_scatter_fixed<AGG_COUNT, int64_t>, t64, sg, gx);
-      } else {
-        scat(k_radix_scatter_fixed<AGG_SUM, int64_t>, t64, sg, gx);
-      }
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_fixed<AGG_COUNT, TSV>, tsptr, sg, gx,
+             2 * hist_lds);
+      else
+        scat(k_radix_scatter_fixed<AGG_SUM, TSV>, tsptr, sg, gx,
+             2 * hist_lds);
     }
+  };
+  for (const Seg& sg : segs) {
+    if (seg32) scat_any(ts.data_ptr<int32_t>(), sg);
+    else scat_any(ts.data_ptr<int64_t>(), sg);
   }
 
-  // Fixed layout: bucket b's events live at [b*cap, b*cap + count).
+  // Fixed layout: segment b's events live at [b*cap, b*cap + count).
   auto offsets = at::arange(
-      nb, at::TensorOptions().dtype(at::kInt).device(keys.device()));
+      nseg, at::TensorOptions().dtype(at::kInt).device(keys.device()));
   offsets = offsets * (int)cap;
-  int region = 1 << region_bits;
-  size_t agg_lds = (size_t)region * 16;
+  size_t agg_lds = (size_t)16 << seg_bits;
+  dim3 agg_block(seg_bits >= 12 ? 1024 : 256);
   auto agg = [&](auto kern) {
     hipLaunchKernelGGL(
-        kern, dim3((unsigned)nb), block, agg_lds, stream,
+        kern, dim3((unsigned)nseg), agg_block, agg_lds, stream,
         (const uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
         offsets.data_ptr<int32_t>(), gcursors.data_ptr<int32_t>(), cap,
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
-        (int)region_bits, error_flag.data_ptr<int32_t>());
+        (int)region_bits, seg_bits, error_flag.data_ptr<int32_t>());
   };
   if (mode == AGG_COUNT) agg(k_radix_agg<AGG_COUNT>);
   else agg(k_radix_agg<AGG_SUM>);
 
   auto ov = [&](auto kern) {
     hipLaunchKernelGGL(
-        kern, dim3(64), block, 0, stream,
+        kern, dim3(64), dim3(256), 0, stream,
         (const uint64_t*)ov_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ov_vals.data_ptr<int64_t>() : nullptr,
         ov_cursor.data_ptr<int32_t>(), ov_packed.numel(),
@@ -1866,8 +2099,6 @@ void radix_scatter_only(
   int64_t n = keys.numel();
   TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
   int64_t nb = nslots >> region_bits;
-  int64_t cap = ev_packed.numel() / nb;
-  TORCH_CHECK(cap * nb >= 2 * n || cap >= n, "scatter buffers too small");
   const int64_t* vptr = nullptr;
   if (mode == AGG_SUM) {
     TORCH_CHECK(vals.has_value(), "sum mode requires vals");
@@ -1878,9 +2109,40 @@ void radix_scatter_only(
   dim3 block(256);
   uint64_t win_m2, win_maxfast;
   magic_div_u64(len_ms, &win_m2, &win_maxfast);
-  size_t hist_lds = (size_t)nb * sizeof(int);
+
+  // Same variant/segmentation decision as radix_agg_only (both read
+  // the env once per call; one process = one configuration).
+  ScatterKind kind = scatter_kind_env();
+  int coarse = scatter_coarse_bits(kind);
+  int seg_bits = (int)region_bits + coarse;
+  int64_t nseg = nslots >> seg_bits;
+  size_t staged_lds = (size_t)nseg * SC_GRAN * 8 *
+                          (mode == AGG_SUM ? 2 : 1) +
+                      4 * (size_t)nseg * sizeof(int);
+  while (kind == SCAT_STAGED &&
+         (nseg < 1 || staged_lds > 160 * 1024 || seg_bits > 13)) {
+    if (coarse > 0 && seg_bits > 13) {
+      coarse -= 1;
+    } else {
+      kind = SCAT_FIXED;
+      coarse = 0;
+    }
+    seg_bits = (int)region_bits + coarse;
+    nseg = nslots >> seg_bits;
+    staged_lds = (size_t)nseg * SC_GRAN * 8 * (mode == AGG_SUM ? 2 : 1) +
+                 4 * (size_t)nseg * sizeof(int);
+  }
+  if (seg_bits > 13) {
+    coarse = 0;
+    seg_bits = (int)region_bits;
+    nseg = nb;
+  }
+  int64_t cap = ev_packed.numel() / nseg;
+  if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
+  TORCH_CHECK(cap * nseg >= 2 * n || cap >= n, "scatter buffers too small");
+  size_t hist_lds = (size_t)nseg * sizeof(int);
   HIP_CHECK(hipMemsetAsync(
-      gcursors.data_ptr<int32_t>(), 0, (size_t)nb * sizeof(int), stream));
+      gcursors.data_ptr<int32_t>(), 0, (size_t)nseg * sizeof(int), stream));
   HIP_CHECK(hipMemsetAsync(
       ov_cursor.data_ptr<int32_t>(), 0, sizeof(int), stream));
   if (n == 0) return;
@@ -1898,12 +2160,13 @@ void radix_scatter_only(
   } else {
     segs.push_back({0, n, ts_base});
   }
-  auto scat = [&](auto kern, auto tsptr, const Seg& sg) {
+  auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx,
+                  size_t lds) {
     hipLaunchKernelGGL(
-        kern, dim3((unsigned)n_blocks(sg.n, 256)), block, 2 * hist_lds,
+        kern, dim3(gx), block, lds,
         stream, keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
         vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
-        sg.base, mask, (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
         ov_cursor.data_ptr<int32_t>(),
@@ -1913,20 +2176,37 @@ void radix_scatter_only(
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
   };
-  for (const Seg& sg : segs) {
-    if (seg32) {
-      const int32_t* t32 = ts.data_ptr<int32_t>();
+  auto scat_any = [&](auto tsptr, const Seg& sg) {
+    using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
+    if (kind == SCAT_STAGED) {
+      unsigned gs = (unsigned)((sg.n + 4095) / 4096);
+      if (gs > 512) gs = 512;
+      if (gs < 1) gs = 1;
       if (mode == AGG_COUNT)
-        scat(k_radix_scatter_fixed<AGG_COUNT, int32_t>, t32, sg);
+        scat(k_radix_scatter_staged<AGG_COUNT, TSV>, tsptr, sg, gs,
+             staged_lds);
       else
-        scat(k_radix_scatter_fixed<AGG_SUM, int32_t>, t32, sg);
+        scat(k_radix_scatter_staged<AGG_SUM, TSV>, tsptr, sg, gs,
+             staged_lds);
+    } else if (kind == SCAT_DIRECT) {
+      unsigned gx = (unsigned)n_blocks(sg.n, 256);
+      if (mode == AGG_COUNT)
+        scat(k_radix_scatter_direct<AGG_COUNT, TSV>, tsptr, sg, gx, 0);
+      else
+        scat(k_radix_scatter_direct<AGG_SUM, TSV>, tsptr, sg, gx, 0);
     } else {
-      const int64_t* t64 = ts.data_ptr<int64_t>();
+      unsigned gx = (unsigned)n_blocks(sg.n, 256);
       if (mode == AGG_COUNT)
-        scat(k_radix_scatter_fixed<AGG_COUNT, int64_t>, t64, sg);
+        scat(k_radix_scatter_fixed<AGG_COUNT, TSV>, tsptr, sg, gx,
+             2 * hist_lds);
       else
-        scat(k_radix_scatter_fixed<AGG_SUM, int64_t>, t64, sg);
+        scat(k_radix_scatter_fixed<AGG_SUM, TSV>, tsptr, sg, gx,
+             2 * hist_lds);
     }
+  };
+  for (const Seg& sg : segs) {
+    if (seg32) scat_any(ts.data_ptr<int32_t>(), sg);
+    else scat_any(ts.data_ptr<int64_t>(), sg);
   }
 }
 
@@ -1944,29 +2224,55 @@ void radix_agg_only(
     int64_t region_bits) {
   int64_t nslots = tkeys.numel();
   int64_t nb = nslots >> region_bits;
-  int64_t cap = ev_packed.numel() / nb;
   auto stream = at::hip::getCurrentHIPStream();
   uint64_t mask = (uint64_t)(nslots - 1);
-  dim3 block(256);
+  // Mirror radix_scatter_only's segmentation decision.
+  ScatterKind kind = scatter_kind_env();
+  int coarse = scatter_coarse_bits(kind);
+  int seg_bits = (int)region_bits + coarse;
+  int64_t nseg = nslots >> seg_bits;
+  size_t staged_lds = (size_t)nseg * SC_GRAN * 8 *
+                          (mode == AGG_SUM ? 2 : 1) +
+                      4 * (size_t)nseg * sizeof(int);
+  while (kind == SCAT_STAGED &&
+         (nseg < 1 || staged_lds > 160 * 1024 || seg_bits > 13)) {
+    if (coarse > 0 && seg_bits > 13) {
+      coarse -= 1;
+    } else {
+      kind = SCAT_FIXED;
+      coarse = 0;
+    }
+    seg_bits = (int)region_bits + coarse;
+    nseg = nslots >> seg_bits;
+    staged_lds = (size_t)nseg * SC_GRAN * 8 * (mode == AGG_SUM ? 2 : 1) +
+                 4 * (size_t)nseg * sizeof(int);
+  }
+  if (seg_bits > 13) {
+    seg_bits = (int)region_bits;
+    nseg = nb;
+  }
+  int64_t cap = ev_packed.numel() / nseg;
+  if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
   auto offsets = at::arange(
-      nb, at::TensorOptions().dtype(at::kInt).device(tkeys.device()));
+      nseg, at::TensorOptions().dtype(at::kInt).device(tkeys.device()));
   offsets = offsets * (int)cap;
-  size_t agg_lds = (size_t)(1 << region_bits) * 16;
+  size_t agg_lds = (size_t)16 << seg_bits;
+  dim3 agg_block(seg_bits >= 12 ? 1024 : 256);
   auto agg = [&](auto kern) {
     hipLaunchKernelGGL(
-        kern, dim3((unsigned)nb), block, agg_lds, stream,
+        kern, dim3((unsigned)nseg), agg_block, agg_lds, stream,
         (const uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
         offsets.data_ptr<int32_t>(), gcursors.data_ptr<int32_t>(), cap,
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(), mask,
-        (int)region_bits, error_flag.data_ptr<int32_t>());
+        (int)region_bits, seg_bits, error_flag.data_ptr<int32_t>());
   };
   if (mode == AGG_COUNT) agg(k_radix_agg<AGG_COUNT>);
   else agg(k_radix_agg<AGG_SUM>);
   auto ov = [&](auto kern) {
     hipLaunchKernelGGL(
-        kern, dim3(64), block, 0, stream,
+        kern, dim3(64), dim3(256), 0, stream,
         (const uint64_t*)ov_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ov_vals.data_ptr<int64_t>() : nullptr,
         ov_cursor.data_ptr<int32_t>(), ov_packed.numel(),
@@ -2817,17 +3123,46 @@ int64_t native_run_window_steps(
   hipEvent_t ev_sc[2] = {nullptr, nullptr};
   hipEvent_t ev_ag[2] = {nullptr, nullptr};
   torch::Tensor agg_offsets;
-  int64_t nb = 0, cap = 0;
+  int64_t nb = 0, cap = 0, nseg = 0;
+  int seg_bits = (int)region_bits;
+  ScatterKind kind = SCAT_FIXED;
+  size_t staged_lds = 0;
   uint64_t mask = (uint64_t)(nslots - 1);
   uint64_t win_m2 = 0, win_maxfast = 0;
   if (pipe) {
     nb = nslots >> region_bits;
-    cap = ev_packed->numel() / nb;
     TORCH_CHECK(ev_packed2->numel() == ev_packed->numel(),
                 "pipelined scatter buffers must match");
     magic_div_u64(len_ms, &win_m2, &win_maxfast);
+    // Same scatter-variant/segmentation decision as radix_window_insert
+    // (COUNT mode here).
+    kind = scatter_kind_env();
+    int coarse = scatter_coarse_bits(kind);
+    seg_bits = (int)region_bits + coarse;
+    nseg = nslots >> seg_bits;
+    staged_lds =
+        (size_t)nseg * SC_GRAN * 8 + 4 * (size_t)nseg * sizeof(int);
+    while (kind == SCAT_STAGED &&
+           (nseg < 1 || staged_lds > 160 * 1024 || seg_bits > 13)) {
+      if (coarse > 0 && seg_bits > 13) {
+        coarse -= 1;
+      } else {
+        kind = SCAT_FIXED;
+        coarse = 0;
+      }
+      seg_bits = (int)region_bits + coarse;
+      nseg = nslots >> seg_bits;
+      staged_lds =
+          (size_t)nseg * SC_GRAN * 8 + 4 * (size_t)nseg * sizeof(int);
+    }
+    if (seg_bits > 13) {
+      seg_bits = (int)region_bits;
+      nseg = nb;
+    }
+    cap = ev_packed->numel() / nseg;
+    if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
     agg_offsets = at::arange(
-        nb,
+        nseg,
         at::TensorOptions().dtype(at::kInt).device(tkeys.device()));
     agg_offsets = agg_offsets * (int)cap;
     HIP_CHECK(hipStreamCreateWithFlags(&sc_stream, hipStreamNonBlocking));
@@ -2866,32 +3201,61 @@ int64_t native_run_window_steps(
           HIP_CHECK(hipStreamWaitEvent(sc_stream, ev_ag[par], 0));
         }
         HIP_CHECK(hipMemsetAsync(
-            gcur.data_ptr<int32_t>(), 0, (size_t)nb * sizeof(int),
+            gcur.data_ptr<int32_t>(), 0, (size_t)nseg * sizeof(int),
             sc_stream));
         HIP_CHECK(hipMemsetAsync(
             ovc.data_ptr<int32_t>(), 0, sizeof(int), sc_stream));
-        size_t hist_lds = (size_t)nb * sizeof(int);
-        hipLaunchKernelGGL(
-            k_radix_scatter_fixed<AGG_COUNT>, grid, block, 2 * hist_lds,
-            sc_stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-            (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
-            (int)region_bits, cap, gcur.data_ptr<int32_t>(),
-            (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
-            ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
-            (int64_t*)nullptr, ovp.numel(),
-            (unsigned long long*)max_ts.data_ptr<int64_t>(),
-            error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+        if (kind == SCAT_STAGED) {
+          unsigned gs = (unsigned)((n + 4095) / 4096);
+          if (gs > 512) gs = 512;
+          if (gs < 1) gs = 1;
+          hipLaunchKernelGGL(
+              k_radix_scatter_staged<AGG_COUNT>, dim3(gs), block,
+              staged_lds, sc_stream, keys.data_ptr<int32_t>(),
+              ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
+              align_ms, len_ms, base, mask, seg_bits, cap,
+              gcur.data_ptr<int32_t>(), (uint64_t*)evp.data_ptr<int64_t>(),
+              (int64_t*)nullptr, ovc.data_ptr<int32_t>(),
+              (uint64_t*)ovp.data_ptr<int64_t>(), (int64_t*)nullptr,
+              ovp.numel(),
+              (unsigned long long*)max_ts.data_ptr<int64_t>(),
+              error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+        } else if (kind == SCAT_DIRECT) {
+          hipLaunchKernelGGL(
+              k_radix_scatter_direct<AGG_COUNT>, grid, block, 0,
+              sc_stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
+              (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
+              seg_bits, cap, gcur.data_ptr<int32_t>(),
+              (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
+              ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
+              (int64_t*)nullptr, ovp.numel(),
+              (unsigned long long*)max_ts.data_ptr<int64_t>(),
+              error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+        } else {
+          size_t hist_lds = (size_t)nseg * sizeof(int);
+          hipLaunchKernelGGL(
+              k_radix_scatter_fixed<AGG_COUNT>, grid, block, 2 * hist_lds,
+              sc_stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
+              (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
+              seg_bits, cap, gcur.data_ptr<int32_t>(),
+              (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
+              ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
+              (int64_t*)nullptr, ovp.numel(),
+              (unsigned long long*)max_ts.data_ptr<int64_t>(),
+              error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+        }
         HIP_CHECK(hipEventRecord(ev_sc[par], sc_stream));
         HIP_CHECK(hipStreamWaitEvent(stream, ev_sc[par], 0));
-        size_t agg_lds = (size_t)(1 << region_bits) * 16;
+        size_t agg_lds = (size_t)16 << seg_bits;
+        dim3 agg_block(seg_bits >= 12 ? 1024 : 256);
         hipLaunchKernelGGL(
-            k_radix_agg<AGG_COUNT>, dim3((unsigned)nb), block, agg_lds,
-            stream, (const uint64_t*)evp.data_ptr<int64_t>(),
+            k_radix_agg<AGG_COUNT>, dim3((unsigned)nseg), agg_block,
+            agg_lds, stream, (const uint64_t*)evp.data_ptr<int64_t>(),
             (const int64_t*)nullptr, agg_offsets.data_ptr<int32_t>(),
             gcur.data_ptr<int32_t>(), cap,
             (uint64_t*)cur_k.data_ptr<int64_t>(),
             (unsigned long long*)cur_v.data_ptr<int64_t>(), mask,
-            (int)region_bits, error_flag.data_ptr<int32_t>());
+            (int)region_bits, seg_bits, error_flag.data_ptr<int32_t>());
         hipLaunchKernelGGL(
             k_overflow_agg<AGG_COUNT>, dim3(64), block, 0, stream,
             (const uint64_t*)ovp.data_ptr<int64_t>(),

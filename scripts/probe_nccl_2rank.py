@@ -1,0 +1,74 @@
+"""Probe: can RCCL run 2 ranks on one physical MI355X?
+
+Functional validation of the exchange code path (bucketing kernel +
+all_to_all_single over the real NCCL/RCCL backend) without an 8-GPU
+node: both ranks map to cuda:0.  Prints per-rank checksums of the
+exchanged batch so correctness can be compared against the
+single-process expectation.
+"""
+
+import os
+import sys
+
+import torch
+import torch.multiprocessing as mp
+
+
+def expected(world: int):
+    """What every rank together should hold after the exchange."""
+    from bytewax_amd.gpu import RecordBatch, _mix64_torch
+
+    per = {}
+    for rank in range(world):
+        g = torch.Generator().manual_seed(1234 + rank)
+        keys = torch.randint(0, 1000, (4096,), dtype=torch.int32, generator=g)
+        ts = torch.arange(4096, dtype=torch.int64) + rank
+        dst = torch.remainder(_mix64_torch(keys.to(torch.int64)), world)
+        for d in range(world):
+            m = dst == d
+            k, t = keys[m], ts[m]
+            s = per.setdefault(d, [0, 0, 0])
+            s[0] += int(k.numel())
+            s[1] += int(k.to(torch.int64).sum())
+            s[2] += int(t.sum())
+    return per
+
+
+def worker(rank: int, world: int):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = "29571"
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    torch.cuda.set_device(0)
+    import torch.distributed as dist
+
+    dist.init_process_group("nccl", rank=rank, world_size=world)
+    from bytewax_amd.gpu import RecordBatch, exchange_by_key
+
+    g = torch.Generator().manual_seed(1234 + rank)
+    keys = torch.randint(0, 1000, (4096,), dtype=torch.int32, generator=g)
+    ts = torch.arange(4096, dtype=torch.int64) + rank
+    batch = RecordBatch(
+        keys.to("cuda:0"), ts.to("cuda:0"), max_ts=int(ts.max())
+    )
+    out = exchange_by_key(batch)
+    n = len(out)
+    ksum = int(out.keys.to(torch.int64).sum().item())
+    tsum = int(out.ts.sum().item())
+    exp = expected(world)[rank]
+    ok = [n, ksum, tsum] == exp
+    print(
+        f"RANK {rank}: n={n} ksum={ksum} tsum={tsum} "
+        f"expected={exp} OK={ok}",
+        flush=True,
+    )
+    dist.barrier()
+    dist.destroy_process_group()
+    if not ok:
+        sys.exit(1)
+
+
+if __name__ == "__main__":
+    world = int(sys.argv[1]) if len(sys.argv) > 1 else 2
+    mp.spawn(worker, args=(world,), nprocs=world, join=True)
+    print("PROBE_2RANK_RESULT: PASS", flush=True)
