@@ -32,7 +32,7 @@ def parse_args():
     p.add_argument(
         "--batches-per-poll",
         type=int,
-        default=150,
+        default=200,
         help="RecordBatches the source hands the engine per "
         "scheduling step; one timed step = this many batches "
         "(amortizes per-step engine overhead and keeps the timed "

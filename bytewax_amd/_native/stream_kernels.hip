@@ -1785,6 +1785,159 @@ __global__ void k_bucket_scatter(
 }
 
 // ---------------------------------------------------------------------------
+// String dictionary encode (str keys -> dense int32 ids, on device).
+//
+// The reference's key contract is `str` (reference src/operators.rs:
+// 363-439 extract_key); the columnar fast path keys RecordBatches by
+// int32 id.  These kernels bridge the two: string bytes (+offsets)
+// land on device once, each string is 128-bit hashed and mapped to a
+// dense id through an open-address device dictionary.
+//
+// Exactness: the device table stores the full 128-bit hash (lo is
+// the CAS claim word, hi the published check word).  Newly created
+// ids are read back as (id, batch_index) pairs so the host keeps the
+// authoritative id->string list; the Python layer verifies each new
+// id against its own exact dict and fails loudly on a 128-bit
+// collision (probability ~1e-26 at 1e6 keys) instead of aggregating
+// wrong.
+//
+// Race-free without spins: insert and lookup are SEPARATE kernel
+// launches.  k_dict_insert only guarantees every distinct string has
+// a claimed slot with published (hi, id) by kernel end (threads that
+// lose the CAS to the same hash exit; the winner's hi/id stores are
+// globally visible at the kernel boundary).  k_dict_lookup then
+// resolves every event's id with plain loads.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void str_hash128(
+    const uint8_t* __restrict__ s, int len, uint64_t* h_lo,
+    uint64_t* h_hi) {
+  // FNV-1a over the bytes, finalized twice with independent seeds.
+  uint64_t h = 0xcbf29ce484222325ULL;
+  for (int i = 0; i < len; ++i) {
+    h ^= (uint64_t)s[i];
+    h *= 0x100000001b3ULL;
+  }
+  uint64_t lo = mix64(h ^ 0x9e3779b97f4a7c15ULL);
+  if (lo == EMPTY_SLOT) lo = 0;  // reserve the empty sentinel
+  *h_lo = lo;
+  *h_hi = mix64(h + 0x2545f4914f6cdd1dULL);
+}
+
+__global__ void k_dict_insert(
+    const uint8_t* __restrict__ bytes,
+    const int64_t* __restrict__ offs,  // [n+1]
+    int64_t n,
+    uint64_t* __restrict__ dlo,   // [nslots] CAS claim words
+    uint64_t* __restrict__ dhi,   // [nslots] published check words
+    int32_t* __restrict__ dids,   // [nslots] published ids
+    uint64_t mask,
+    int* __restrict__ counter,        // next id
+    int32_t* __restrict__ new_ids,    // [new_cap] readback: id
+    int32_t* __restrict__ new_idx,    // [new_cap] readback: batch index
+    int* __restrict__ new_n,
+    int64_t new_cap,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    int len = (int)(offs[i + 1] - offs[i]);
+    uint64_t lo, hi;
+    str_hash128(bytes + offs[i], len, &lo, &hi);
+    uint64_t h = lo & mask;
+    for (uint64_t probes = 0; probes <= mask; ++probes) {
+      uint64_t cur = dlo[h];
+      if (cur == lo) break;  // claimed (here or elsewhere): winner publishes
+      if (cur == EMPTY_SLOT) {
+        uint64_t prev = atomicCAS((unsigned long long*)&dlo[h],
+                                  EMPTY_SLOT, lo);
+        if (prev == EMPTY_SLOT) {
+          int id = atomicAdd(counter, 1);
+          dhi[h] = hi;
+          dids[h] = id;
+          int r = atomicAdd(new_n, 1);
+          if (r < new_cap) {
+            new_ids[r] = id;
+            new_idx[r] = (int32_t)i;
+          } else {
+            atomicExch(error_flag, 3);  // readback buffer overflow
+          }
+          break;
+        }
+        if (prev == lo) break;
+      }
+      h = (h + 1) & mask;
+      if (probes == mask) atomicExch(error_flag, 1);  // table full
+    }
+  }
+}
+
+__global__ void k_dict_lookup(
+    const uint8_t* __restrict__ bytes,
+    const int64_t* __restrict__ offs,
+    int64_t n,
+    const uint64_t* __restrict__ dlo,
+    const uint64_t* __restrict__ dhi,
+    const int32_t* __restrict__ dids,
+    uint64_t mask,
+    int32_t* __restrict__ out_ids,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    int len = (int)(offs[i + 1] - offs[i]);
+    uint64_t lo, hi;
+    str_hash128(bytes + offs[i], len, &lo, &hi);
+    uint64_t h = lo & mask;
+    int32_t id = -1;
+    for (uint64_t probes = 0; probes <= mask; ++probes) {
+      uint64_t cur = dlo[h];
+      if (cur == lo && dhi[h] == hi) {
+        id = dids[h];
+        break;
+      }
+      if (cur == EMPTY_SLOT) break;
+      h = (h + 1) & mask;
+    }
+    out_ids[i] = id;
+    if (id < 0) atomicExch(error_flag, 2);  // unpublished / collision
+  }
+}
+
+// Restore path: re-claim slots with PINNED ids (snapshot order must
+// be reproduced exactly across restarts).
+__global__ void k_dict_insert_pinned(
+    const uint8_t* __restrict__ bytes,
+    const int64_t* __restrict__ offs,
+    const int32_t* __restrict__ ids,
+    int64_t n,
+    uint64_t* __restrict__ dlo,
+    uint64_t* __restrict__ dhi,
+    int32_t* __restrict__ dids,
+    uint64_t mask,
+    int* __restrict__ error_flag) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    int len = (int)(offs[i + 1] - offs[i]);
+    uint64_t lo, hi;
+    str_hash128(bytes + offs[i], len, &lo, &hi);
+    uint64_t h = lo & mask;
+    for (uint64_t probes = 0; probes <= mask; ++probes) {
+      uint64_t prev = atomicCAS((unsigned long long*)&dlo[h],
+                                EMPTY_SLOT, lo);
+      if (prev == EMPTY_SLOT) {
+        dhi[h] = hi;
+        dids[h] = ids[i];
+        break;
+      }
+      h = (h + 1) & mask;
+      if (probes == mask) atomicExch(error_flag, 1);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Host-side wrappers (torch extension API)
 // ---------------------------------------------------------------------------
 
@@ -3490,7 +3643,79 @@ int64_t native_run_window_steps_graph(
   return total_rows;
 }
 
+// ---- String dictionary wrappers ----
+
+int64_t dict_encode(
+    torch::Tensor bytes,    // uint8 device [total_bytes]
+    torch::Tensor offs,     // int64 device [n+1]
+    torch::Tensor dlo,      // int64 device [nslots] (init EMPTY)
+    torch::Tensor dhi,      // int64 device [nslots]
+    torch::Tensor dids,     // int32 device [nslots]
+    torch::Tensor counter,  // int32 device [1]
+    torch::Tensor new_ids,  // int32 device [new_cap]
+    torch::Tensor new_idx,  // int32 device [new_cap]
+    torch::Tensor new_n,    // int32 device [1]
+    torch::Tensor out_ids,  // int32 device [n]
+    torch::Tensor error_flag) {
+  check_dev(bytes, torch::kUInt8, "bytes");
+  check_dev(offs, torch::kInt64, "offs");
+  check_dev(dlo, torch::kInt64, "dlo");
+  check_dev(out_ids, torch::kInt32, "out_ids");
+  int64_t n = offs.numel() - 1;
+  int64_t nslots = dlo.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "dict size must be 2^k");
+  if (n == 0) return 0;
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  uint64_t mask = (uint64_t)(nslots - 1);
+  new_n.zero_();
+  hipLaunchKernelGGL(
+      k_dict_insert, grid, block, 0, stream,
+      bytes.data_ptr<uint8_t>(), offs.data_ptr<int64_t>(), n,
+      (uint64_t*)dlo.data_ptr<int64_t>(),
+      (uint64_t*)dhi.data_ptr<int64_t>(), dids.data_ptr<int32_t>(), mask,
+      counter.data_ptr<int32_t>(), new_ids.data_ptr<int32_t>(),
+      new_idx.data_ptr<int32_t>(), new_n.data_ptr<int32_t>(),
+      new_ids.numel(), error_flag.data_ptr<int32_t>());
+  hipLaunchKernelGGL(
+      k_dict_lookup, grid, block, 0, stream,
+      bytes.data_ptr<uint8_t>(), offs.data_ptr<int64_t>(), n,
+      (const uint64_t*)dlo.data_ptr<int64_t>(),
+      (const uint64_t*)dhi.data_ptr<int64_t>(), dids.data_ptr<int32_t>(),
+      mask, out_ids.data_ptr<int32_t>(), error_flag.data_ptr<int32_t>());
+  return n;
+}
+
+void dict_restore(
+    torch::Tensor bytes,
+    torch::Tensor offs,
+    torch::Tensor ids,  // int32 device [n] pinned id per string
+    torch::Tensor dlo,
+    torch::Tensor dhi,
+    torch::Tensor dids,
+    torch::Tensor error_flag) {
+  check_dev(bytes, torch::kUInt8, "bytes");
+  check_dev(offs, torch::kInt64, "offs");
+  check_dev(ids, torch::kInt32, "ids");
+  int64_t n = offs.numel() - 1;
+  int64_t nslots = dlo.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "dict size must be 2^k");
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(
+      k_dict_insert_pinned, dim3(n_blocks(n, 256)), dim3(256), 0, stream,
+      bytes.data_ptr<uint8_t>(), offs.data_ptr<int64_t>(),
+      ids.data_ptr<int32_t>(), n, (uint64_t*)dlo.data_ptr<int64_t>(),
+      (uint64_t*)dhi.data_ptr<int64_t>(), dids.data_ptr<int32_t>(),
+      (uint64_t)(nslots - 1), error_flag.data_ptr<int32_t>());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("dict_encode", &dict_encode,
+        "device string-dictionary encode: 2-phase insert+lookup");
+  m.def("dict_restore", &dict_restore,
+        "device string-dictionary restore with pinned ids");
   m.def("window_agg_insert", &window_agg_insert,
         "Fused window-id + hash-insert + watermark over an event batch");
   m.def("radix_v2_window_insert", &radix_v2_window_insert,

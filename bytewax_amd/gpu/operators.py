@@ -38,6 +38,7 @@ __all__ = [
     "keyed_session_agg",
     "keyed_stats_agg",
     "keyed_window_agg",
+    "keyed_window_agg_str",
     "stream_join",
 ]
 
@@ -614,6 +615,119 @@ def keyed_window_agg(
         )
 
     keyed = op.map("wrap", up, lambda b: (_runtime().shard, b))
+    agg = op.stateful_batch("agg", keyed, shim_builder)
+    return op.map("unwrap", agg, lambda kv: kv[1])
+
+
+class _StrWindowLogic(StatefulBatchLogic):
+    """Dictionary-encode str-keyed host batches into the HBM window
+    table; decode ids back to strings at window close.
+
+    Items are ``(strings, ts[, vals])`` host batches (strings: list of
+    str or a packed ``(uint8 bytes, int64 offsets)`` pair; ts:
+    epoch-ms ints).  Emissions are ``(key_str, win_start_ms, value)``
+    tuples — the same rows the host windowing path produces for a
+    keyed count/sum fold (reference windowing.py count_window /
+    fold_window semantics under watermark close).
+    """
+
+    def __init__(self, sdict, state, wait_ms: int, resume):
+        self.sdict = sdict
+        self.state = state
+        self.wait_ms = wait_ms
+        if resume is not None:
+            self.sdict.restore(resume["dict"])
+            self.state.restore_from_host(resume["win"])
+
+    def _emit(self, closed) -> List[tuple]:
+        if closed is None:
+            return []
+        keys = self.sdict.decode(closed.keys)
+        ts = closed.ts.cpu().tolist()
+        vals = closed.vals.cpu().tolist()
+        return list(zip(keys, ts, vals))
+
+    def on_batch(self, batches):
+        import torch
+
+        out: List[tuple] = []
+        for item in batches:
+            strings, ts = item[0], item[1]
+            vals = item[2] if len(item) > 2 else None
+            ids = self.sdict.encode(strings)
+            dev = self.state.device
+            ts_t = torch.as_tensor(ts, dtype=torch.int64).to(dev)
+            vals_t = (
+                torch.as_tensor(vals, dtype=torch.int64).to(dev)
+                if vals is not None
+                else None
+            )
+            self.state.insert(
+                RecordBatch(ids.to(dev), ts_t, vals_t)
+            )
+        closed = self.state.close_due(self.wait_ms)
+        out.extend(self._emit(closed))
+        return (out, StatefulBatchLogic.RETAIN)
+
+    def on_eof(self):
+        return (
+            self._emit(self.state.close_all()),
+            StatefulBatchLogic.RETAIN,
+        )
+
+    def snapshot(self):
+        return {
+            "dict": self.sdict.snapshot(),
+            "win": self.state.snapshot_to_host(),
+        }
+
+
+@operator
+def keyed_window_agg_str(
+    step_id: str,
+    up: Stream,
+    align_to: datetime,
+    length: timedelta,
+    mode: str = "count",
+    wait: timedelta = timedelta(0),
+    slots_pow: int = 20,
+    dict_slots_pow: int = 21,
+    out_cap: int = 1 << 20,
+    device: str = "cuda",
+) -> Stream:
+    """Str-keyed tumbling-window aggregation on GPU.
+
+    The string-keyed twin of :func:`keyed_window_agg`: the reference's
+    key contract is `str` (reference src/operators.rs:363-439), so
+    this lowers str-keyed streams onto the columnar kernels through a
+    device string dictionary (:class:`bytewax_amd.gpu.strings.
+    StringDict`) — hash/dedupe on device, id->str decode on close.
+
+    Upstream items: ``(strings, ts[, vals])`` host batches.  Output:
+    ``(key_str, win_start_ms, value)`` tuples at watermark close,
+    matching the host windowing path's rows exactly (gpu test
+    `tests/test_gpu_strings.py`).
+    """
+    import torch
+
+    from .strings import StringDict
+
+    agg_mode = {"count": AGG_COUNT, "sum": AGG_SUM}[mode]
+    align_ms = _ms(align_to)
+    len_ms = int(length.total_seconds() * 1000)
+    wait_ms = int(wait.total_seconds() * 1000)
+
+    def shim_builder(resume_state):
+        dev = torch.device(device)
+        sdict = StringDict(dev, slots_pow=dict_slots_pow)
+        state = WindowAggState(
+            dev, align_ms, len_ms, agg_mode,
+            slots_pow=slots_pow, out_cap=out_cap,
+            radix=dev.type != "cpu",
+        )
+        return _StrWindowLogic(sdict, state, wait_ms, resume_state)
+
+    keyed = op.map("wrap", up, lambda b: ("shard-0", b))
     agg = op.stateful_batch("agg", keyed, shim_builder)
     return op.map("unwrap", agg, lambda kv: kv[1])
 

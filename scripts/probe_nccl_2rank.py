@@ -9,9 +9,17 @@ single-process expectation.
 
 import os
 import sys
+from pathlib import Path
 
 import torch
 import torch.multiprocessing as mp
+
+REPO = str(Path(__file__).resolve().parent.parent)
+# Spawned workers re-import this module; make the package importable
+# in them regardless of how the parent was launched.
+os.environ["PYTHONPATH"] = REPO + os.pathsep + os.environ.get("PYTHONPATH", "")
+if REPO not in sys.path:
+    sys.path.insert(0, REPO)
 
 
 def expected(world: int):
