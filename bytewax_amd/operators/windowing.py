@@ -61,6 +61,8 @@ ZERO_TD = timedelta(seconds=0)
 UTC_MIN = datetime.min.replace(tzinfo=timezone.utc)
 """Minimum possible UTC date time."""
 UTC_MAX = datetime.max.replace(tzinfo=timezone.utc)
+
+_EPOCH_UTC = datetime(1970, 1, 1, tzinfo=timezone.utc)
 """Maximum possible UTC date time."""
 
 LATE_SESSION_ID: int = -1
@@ -84,8 +86,14 @@ __all__ = [
     "WindowOut",
     "Windower",
     "WindowerLogic",
+    "COLUMNAR_WINDOW_ID",
+    "COUNT_FOLD",
+    "DeviceFoldable",
+    "SUM_FOLD",
     "collect_window",
     "count_window",
+    "device_count",
+    "device_sum",
     "fold_window",
     "join_window",
     "max_window",
@@ -795,6 +803,218 @@ class WindowOut(Generic[V, W_co]):
     """Metadata about closed windows, sub-keyed by window ID."""
 
 
+#: Window-id placeholder on emissions from the columnar (GPU) lowering
+#: of `fold_window`/`count_window`: one emission carries a whole
+#: RecordBatch of closed `(key_id, window_start_ms, value)` rows, so a
+#: single scalar window id does not apply.
+COLUMNAR_WINDOW_ID = -2
+
+
+class DeviceFoldable:
+    """A fold recognized by the columnar (GPU) lowering.
+
+    Behaves as a plain host folder callable, so one flow definition
+    runs on both engines: over Python-object streams the host
+    `_FoldWindowLogic` calls it per item; over
+    :class:`bytewax_amd.gpu.RecordBatch` streams `fold_window`
+    dispatches to the fused HIP window kernels instead
+    (`WindowAggState`), chosen by stream type at runtime.
+    """
+
+    def __init__(self, mode: str, host_fn: Callable):
+        self.mode = mode
+        self._host_fn = host_fn
+
+    def __call__(self, acc, value):
+        return self._host_fn(acc, value)
+
+    def __repr__(self) -> str:  # pragma: no cover - cosmetic
+        return f"DeviceFoldable({self.mode!r})"
+
+
+#: Sum fold: host = `acc + value`; columnar = fused kernel sum of the
+#: batch `vals` column.
+SUM_FOLD = DeviceFoldable("sum", lambda acc, value: acc + value)
+#: Count fold: host = `acc + 1`; columnar = fused kernel count.
+COUNT_FOLD = DeviceFoldable("count", lambda acc, _value: acc + 1)
+
+
+def device_sum(value_getter: Callable = _identity) -> DeviceFoldable:
+    """A `fold_window` sum fold that lowers to the HIP kernels over
+    RecordBatch streams; `value_getter` extracts the number from each
+    host-path item (columnar batches use their `vals` column)."""
+    return DeviceFoldable(
+        "sum", lambda acc, value: acc + value_getter(value)
+    )
+
+
+def device_count() -> DeviceFoldable:
+    """A `fold_window` count fold that lowers to the HIP kernels over
+    RecordBatch streams."""
+    return COUNT_FOLD
+
+
+@dataclass
+class _ColumnarSpec:
+    """Device-lowering parameters resolved from clock + windower."""
+
+    mode: str
+    align_ms: int
+    len_ms: int
+    off_ms: int
+    wait_ms: int
+
+    @staticmethod
+    def resolve(mode, clock, windower) -> Optional["_ColumnarSpec"]:
+        """Lowerable iff the fold is a device fold, time comes from an
+        EventClock (RecordBatch.ts IS the event timestamp), and the
+        windower is tumbling/sliding (sessions keep the dedicated
+        `gpu.operators.keyed_session_agg`)."""
+        if mode is None or not isinstance(clock, EventClock):
+            return None
+        if isinstance(windower, TumblingWindower):
+            len_ms = int(windower.length.total_seconds() * 1000)
+            off_ms = len_ms
+        elif isinstance(windower, SlidingWindower):
+            len_ms = int(windower.length.total_seconds() * 1000)
+            off_ms = int(windower.offset.total_seconds() * 1000)
+        else:
+            return None
+        align_ms = int(
+            (windower.align_to - _EPOCH_UTC).total_seconds() * 1000
+        )
+        wait_ms = int(
+            clock.wait_for_system_duration.total_seconds() * 1000
+        )
+        return _ColumnarSpec(mode, align_ms, len_ms, off_ms, wait_ms)
+
+    def build(self, resume) -> "_ColumnarWindowLogic":
+        return _ColumnarWindowLogic(self, resume)
+
+
+class _ColumnarWindowLogic(StatefulBatchLogic):
+    """The columnar engine behind a lowered `fold_window`.
+
+    Values are RecordBatches; emissions are
+    `(COLUMNAR_WINDOW_ID, "E", closed_rows_RecordBatch)` — the same
+    tagged-tuple protocol `_WindowLogic` uses, so the `window`
+    operator's unwrappers split the streams identically.  Ingestion
+    must be watermark-ordered (in-order columnar sources); the late
+    stream is not populated on this path.
+    """
+
+    def __init__(self, spec: _ColumnarSpec, resume):
+        self.spec = spec
+        self.state = None  # built on first batch (device known then)
+        self._resume = resume
+
+    def _ensure(self, batch) -> None:
+        if self.state is not None:
+            return
+        from ..gpu import AGG_COUNT, AGG_SUM, WindowAggState
+
+        dev = batch.keys.device
+        on_gpu = dev.type != "cpu"
+        sliding = self.spec.off_ms < self.spec.len_ms
+        self.state = WindowAggState(
+            dev,
+            self.spec.align_ms,
+            self.spec.len_ms,
+            AGG_COUNT if self.spec.mode == "count" else AGG_SUM,
+            slots_pow=22 if on_gpu else 16,
+            out_cap=1 << (22 if on_gpu else 16),
+            radix=on_gpu and not sliding,
+            off_ms=self.spec.off_ms,
+        )
+        if self._resume is not None:
+            self.state.restore_from_host(self._resume)
+            self._resume = None
+
+    def on_batch(self, values) -> Tuple[Iterable, bool]:
+        events: List = []
+        for batch in values:
+            self._ensure(batch)
+            self.state.insert(batch)
+        closed = self.state.close_due(self.spec.wait_ms)
+        if closed is not None:
+            events.append((COLUMNAR_WINDOW_ID, "E", closed))
+        return (events, False)
+
+    def on_notify(self) -> Tuple[Iterable, bool]:
+        return ([], False)
+
+    def on_eof(self) -> Tuple[Iterable, bool]:
+        if self.state is None:
+            return ([], True)
+        final = self.state.close_all()
+        if final is None:
+            return ([], True)
+        return ([(COLUMNAR_WINDOW_ID, "E", final)], True)
+
+    def notify_at(self) -> Optional[datetime]:
+        return None
+
+    def snapshot(self):
+        if self.state is None:
+            return {"__columnar__": self._resume}
+        return {"__columnar__": self.state.snapshot_to_host()}
+
+
+class _AutoWindowLogic(StatefulBatchLogic):
+    """Engine dispatch for `window`: host `_WindowLogic` for
+    Python-object values, the fused HIP kernels for RecordBatch values
+    — decided on the first batch (streams are dynamically typed), and
+    persisted through snapshots."""
+
+    def __init__(self, host_builder, spec: Optional[_ColumnarSpec], resume):
+        self._host_builder = host_builder
+        self._spec = spec
+        self._impl: Optional[StatefulBatchLogic] = None
+        if resume is not None:
+            if isinstance(resume, dict) and "__columnar__" in resume:
+                self._impl = spec.build(resume["__columnar__"])
+            else:
+                self._impl = host_builder(resume)
+
+    def _pick(self, values) -> None:
+        if self._impl is not None:
+            return
+        v = values[0] if values else None
+        if (
+            self._spec is not None
+            and hasattr(v, "keys")
+            and hasattr(v, "ts")
+            and hasattr(v, "ts_base")
+        ):
+            self._impl = self._spec.build(None)
+        else:
+            self._impl = self._host_builder(None)
+
+    def on_batch(self, values):
+        self._pick(values)
+        return self._impl.on_batch(values)
+
+    def on_notify(self):
+        if self._impl is None:
+            return ([], False)
+        return self._impl.on_notify()
+
+    def on_eof(self):
+        if self._impl is None:
+            return ([], True)
+        return self._impl.on_eof()
+
+    def notify_at(self):
+        if self._impl is None:
+            return None
+        return self._impl.notify_at()
+
+    def snapshot(self):
+        if self._impl is None:
+            self._impl = self._host_builder(None)
+        return self._impl.snapshot()
+
+
 def _unwrap_emit(id_typ_obj):
     window_id, typ, obj = id_typ_obj
     return (window_id, obj) if typ == "E" else None
@@ -818,6 +1038,7 @@ def window(
     windower: Windower[Any],
     builder: Callable[[Optional[S]], WindowLogic[V, W, S]],
     ordered: bool = True,
+    device_mode: Optional[str] = None,
 ) -> WindowOut[V, W]:
     """Advanced generic windowing operator.
 
@@ -830,7 +1051,9 @@ def window(
     :returns: Window result streams.
     """
 
-    def shim_builder(resume_state: Optional[_WindowSnapshot]) -> _WindowLogic:
+    spec = _ColumnarSpec.resolve(device_mode, clock, windower)
+
+    def host_builder(resume_state: Optional[_WindowSnapshot]) -> _WindowLogic:
         if resume_state is not None:
             clock_logic = clock.build(resume_state.clock_state)
             windower_logic = windower.build(resume_state.windower_state)
@@ -850,6 +1073,11 @@ def window(
         return _WindowLogic(
             clock.build(None), windower.build(None), builder, ordered
         )
+
+    def shim_builder(resume_state) -> StatefulBatchLogic:
+        if spec is None:
+            return host_builder(resume_state)
+        return _AutoWindowLogic(host_builder, spec, resume_state)
 
     events = op.stateful_batch("stateful_batch", up, shim_builder)
     downs = op.filter_map_value("unwrap_down", events, _unwrap_emit)
@@ -915,8 +1143,10 @@ def fold_window(
         state = resume_state if resume_state is not None else builder()
         return _FoldWindowLogic(folder, merger, state)
 
+    mode = folder.mode if isinstance(folder, DeviceFoldable) else None
     return window(
-        "window", up, clock, windower, shim_builder, ordered=ordered
+        "window", up, clock, windower, shim_builder, ordered=ordered,
+        device_mode=mode,
     )
 
 
@@ -999,7 +1229,7 @@ def count_window(
         clock,
         windower,
         int,
-        lambda s, _x: s + 1,
+        COUNT_FOLD,
         lambda a, b: a + b,
         ordered=False,
     )

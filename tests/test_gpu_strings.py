@@ -43,15 +43,15 @@ def _host_counts(events, window_sec=60):
     out = []
     flow = Dataflow("host")
     s = op.input("inp", flow, TestingSource(inp))
-    keyed = op.key_on("k", s, lambda x: x[1])
     clock = EventClock(
         ts_getter=lambda x: x[0],
         wait_for_system_duration=timedelta(0),
     )
     wo = w.count_window(
-        "cw", keyed,
+        "cw", s,
         clock,
         TumblingWindower(align_to=ALIGN, length=timedelta(seconds=window_sec)),
+        key=lambda x: x[1],
     )
     op.output("out", wo.down, TestingSink(out))
     run_main(flow)
