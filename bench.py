@@ -194,6 +194,8 @@ def main():
             args.sim_ms_per_batch,
             _ms(align),
             seed=42 + rank * 7919,
+            # hipGraph capture reads absolute int64 template columns.
+            ts32=(engine != "graph"),
         )
         # Zero-based ts templates; the kernel applies the per-step base.
         ts_pool = [part.ts_template] * len(part.key_pool)

@@ -158,10 +158,10 @@ __device__ __forceinline__ uint64_t find_slot_r(
     uint64_t* __restrict__ tkeys, uint64_t mask, int region_bits,
     uint64_t packed);
 
-template <int MODE, bool DEDUP>
+template <int MODE, bool DEDUP, typename TS = int64_t>
 __global__ void k_window_agg_insert(
     const int32_t* __restrict__ keys,
-    const int64_t* __restrict__ ts,
+    const TS* __restrict__ ts,
     const int64_t* __restrict__ vals,  // nullptr for COUNT
     int64_t n,
     uint64_t* __restrict__ tkeys,
@@ -190,7 +190,7 @@ __global__ void k_window_agg_insert(
   int64_t first = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   for (int64_t i = first; i - lane < n; i += stride) {
     bool valid = i < n;
-    int64_t t = valid ? (ts[i] + ts_base) : 0;
+    int64_t t = valid ? ((int64_t)ts[i] + ts_base) : 0;
     if (t > local_max) local_max = t;
     uint64_t packed = 0;
     unsigned long long inc = 0;
@@ -1762,15 +1762,16 @@ __global__ void k_bucket_hist(
 // `cursors` must be pre-loaded with the exclusive prefix sums of the
 // destination counts.  Order within a destination is not preserved
 // (items within an epoch are unordered).
+template <typename TS = int64_t>
 __global__ void k_bucket_scatter(
     const int32_t* __restrict__ keys,
-    const int64_t* __restrict__ ts,
+    const TS* __restrict__ ts,
     const int64_t* __restrict__ vals,  // may be nullptr
     int64_t n,
     int world,
     int* __restrict__ cursors,
     int32_t* __restrict__ out_keys,
-    int64_t* __restrict__ out_ts,
+    TS* __restrict__ out_ts,
     int64_t* __restrict__ out_vals) {
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
@@ -1968,7 +1969,8 @@ void window_agg_insert(
   TORCH_CHECK(off_ms == len_ms || !dedup,
               "sliding windows are incompatible with the dedup path");
   check_dev(keys, torch::kInt32, "keys");
-  check_dev(ts, torch::kInt64, "ts");
+  bool ts32 = ts.scalar_type() == torch::kInt32;
+  if (!ts32) check_dev(ts, torch::kInt64, "ts");
   check_dev(tkeys, torch::kInt64, "tkeys");
   check_dev(tvals, torch::kInt64, "tvals");
   check_dev(max_ts, torch::kInt64, "max_ts");
@@ -1989,10 +1991,10 @@ void window_agg_insert(
   auto stream = at::hip::getCurrentHIPStream();
   dim3 block(256);
   dim3 grid(n_blocks(n, 256));
-  auto launch = [&](auto kern) {
+  auto launch = [&](auto kern, auto tsptr) {
     hipLaunchKernelGGL(
         kern, grid, block, 0, stream,
-        keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), vptr, n,
+        keys.data_ptr<int32_t>(), tsptr, vptr, n,
         (uint64_t*)tkeys.data_ptr<int64_t>(),
         (unsigned long long*)tvals.data_ptr<int64_t>(),
         (uint64_t)(nslots - 1), align_ms, len_ms, off_ms, ts_base,
@@ -2000,10 +2002,19 @@ void window_agg_insert(
         (unsigned long long*)max_ts.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>(), (const int64_t*)nullptr);
   };
-  if (mode == AGG_COUNT && !dedup) launch(k_window_agg_insert<AGG_COUNT, false>);
-  else if (mode == AGG_COUNT && dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
-  else if (mode == AGG_SUM && !dedup) launch(k_window_agg_insert<AGG_SUM, false>);
-  else launch(k_window_agg_insert<AGG_SUM, true>);
+  auto dispatch = [&](auto tsptr) {
+    using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
+    if (mode == AGG_COUNT && !dedup)
+      launch(k_window_agg_insert<AGG_COUNT, false, TSV>, tsptr);
+    else if (mode == AGG_COUNT && dedup)
+      launch(k_window_agg_insert<AGG_COUNT, true, TSV>, tsptr);
+    else if (mode == AGG_SUM && !dedup)
+      launch(k_window_agg_insert<AGG_SUM, false, TSV>, tsptr);
+    else
+      launch(k_window_agg_insert<AGG_SUM, true, TSV>, tsptr);
+  };
+  if (ts32) dispatch(ts.data_ptr<int32_t>());
+  else dispatch(ts.data_ptr<int64_t>());
 }
 
 void radix_window_insert(
@@ -2034,11 +2045,12 @@ void radix_window_insert(
   // This inserts exchange output without ever materializing int64
   // timestamps (saves ~3 HBM passes over the biggest column).
   bool seg32 = !seg_counts.empty();
+  bool ts32 = ts.scalar_type() == torch::kInt32;
   if (seg32) {
     TORCH_CHECK(seg_counts.size() == seg_bases.size(),
                 "seg_counts/seg_bases length mismatch");
     check_dev(ts, torch::kInt32, "ts32");
-  } else {
+  } else if (!ts32) {
     check_dev(ts, torch::kInt64, "ts");
   }
   int64_t n = keys.numel();
@@ -2174,7 +2186,7 @@ void radix_window_insert(
     }
   };
   for (const Seg& sg : segs) {
-    if (seg32) scat_any(ts.data_ptr<int32_t>(), sg);
+    if (seg32 || ts32) scat_any(ts.data_ptr<int32_t>(), sg);
     else scat_any(ts.data_ptr<int64_t>(), sg);
   }
 
@@ -2242,11 +2254,12 @@ void radix_scatter_only(
     std::vector<int64_t> seg_bases) {
   check_dev(keys, torch::kInt32, "keys");
   bool seg32 = !seg_counts.empty();
+  bool ts32 = ts.scalar_type() == torch::kInt32;
   if (seg32) {
     TORCH_CHECK(seg_counts.size() == seg_bases.size(),
                 "seg_counts/seg_bases length mismatch");
     check_dev(ts, torch::kInt32, "ts32");
-  } else {
+  } else if (!ts32) {
     check_dev(ts, torch::kInt64, "ts");
   }
   int64_t n = keys.numel();
@@ -2358,7 +2371,7 @@ void radix_scatter_only(
     }
   };
   for (const Seg& sg : segs) {
-    if (seg32) scat_any(ts.data_ptr<int32_t>(), sg);
+    if (seg32 || ts32) scat_any(ts.data_ptr<int32_t>(), sg);
     else scat_any(ts.data_ptr<int64_t>(), sg);
   }
 }
@@ -3167,7 +3180,10 @@ void bucket_scatter(
     torch::Tensor out_ts,
     torch::Tensor out_vals) {
   check_dev(keys, torch::kInt32, "keys");
-  check_dev(ts, torch::kInt64, "ts");
+  bool ts32 = ts.scalar_type() == torch::kInt32;
+  if (!ts32) check_dev(ts, torch::kInt64, "ts");
+  TORCH_CHECK(out_ts.scalar_type() == ts.scalar_type(),
+              "out_ts dtype must match ts");
   check_dev(cursors, torch::kInt32, "cursors");
   int64_t n = keys.numel();
   if (n == 0) return;
@@ -3179,11 +3195,21 @@ void bucket_scatter(
   auto stream = at::hip::getCurrentHIPStream();
   dim3 block(256);
   dim3 grid(n_blocks(n, 256));
-  hipLaunchKernelGGL(
-      k_bucket_scatter, grid, block, 0, stream,
-      keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), vptr, n, (int)world,
-      cursors.data_ptr<int32_t>(), out_keys.data_ptr<int32_t>(),
-      out_ts.data_ptr<int64_t>(), out_vals.data_ptr<int64_t>());
+  if (ts32) {
+    hipLaunchKernelGGL(
+        k_bucket_scatter<int32_t>, grid, block, 0, stream,
+        keys.data_ptr<int32_t>(), ts.data_ptr<int32_t>(), vptr, n,
+        (int)world, cursors.data_ptr<int32_t>(),
+        out_keys.data_ptr<int32_t>(), out_ts.data_ptr<int32_t>(),
+        out_vals.data_ptr<int64_t>());
+  } else {
+    hipLaunchKernelGGL(
+        k_bucket_scatter<int64_t>, grid, block, 0, stream,
+        keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), vptr, n,
+        (int)world, cursors.data_ptr<int32_t>(),
+        out_keys.data_ptr<int32_t>(), out_ts.data_ptr<int64_t>(),
+        out_vals.data_ptr<int64_t>());
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -3358,45 +3384,53 @@ int64_t native_run_window_steps(
             sc_stream));
         HIP_CHECK(hipMemsetAsync(
             ovc.data_ptr<int32_t>(), 0, sizeof(int), sc_stream));
-        if (kind == SCAT_STAGED) {
-          unsigned gs = (unsigned)((n + 4095) / 4096);
-          if (gs > 512) gs = 512;
-          if (gs < 1) gs = 1;
-          hipLaunchKernelGGL(
-              k_radix_scatter_staged<AGG_COUNT>, dim3(gs), block,
-              staged_lds, sc_stream, keys.data_ptr<int32_t>(),
-              ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
-              align_ms, len_ms, base, mask, seg_bits, cap,
-              gcur.data_ptr<int32_t>(), (uint64_t*)evp.data_ptr<int64_t>(),
-              (int64_t*)nullptr, ovc.data_ptr<int32_t>(),
-              (uint64_t*)ovp.data_ptr<int64_t>(), (int64_t*)nullptr,
-              ovp.numel(),
-              (unsigned long long*)max_ts.data_ptr<int64_t>(),
-              error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
-        } else if (kind == SCAT_DIRECT) {
-          hipLaunchKernelGGL(
-              k_radix_scatter_direct<AGG_COUNT>, grid, block, 0,
-              sc_stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-              (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
-              seg_bits, cap, gcur.data_ptr<int32_t>(),
-              (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
-              ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
-              (int64_t*)nullptr, ovp.numel(),
-              (unsigned long long*)max_ts.data_ptr<int64_t>(),
-              error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
-        } else {
-          size_t hist_lds = (size_t)nseg * sizeof(int);
-          hipLaunchKernelGGL(
-              k_radix_scatter_fixed<AGG_COUNT>, grid, block, 2 * hist_lds,
-              sc_stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-              (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
-              seg_bits, cap, gcur.data_ptr<int32_t>(),
-              (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
-              ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
-              (int64_t*)nullptr, ovp.numel(),
-              (unsigned long long*)max_ts.data_ptr<int64_t>(),
-              error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
-        }
+        auto scat_step = [&](auto tsptr) {
+          using TSV =
+              std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
+          if (kind == SCAT_STAGED) {
+            unsigned gs = (unsigned)((n + 4095) / 4096);
+            if (gs > 512) gs = 512;
+            if (gs < 1) gs = 1;
+            hipLaunchKernelGGL(
+                (k_radix_scatter_staged<AGG_COUNT, TSV>), dim3(gs), block,
+                staged_lds, sc_stream, keys.data_ptr<int32_t>(),
+                tsptr, (const int64_t*)nullptr, n,
+                align_ms, len_ms, base, mask, seg_bits, cap,
+                gcur.data_ptr<int32_t>(), (uint64_t*)evp.data_ptr<int64_t>(),
+                (int64_t*)nullptr, ovc.data_ptr<int32_t>(),
+                (uint64_t*)ovp.data_ptr<int64_t>(), (int64_t*)nullptr,
+                ovp.numel(),
+                (unsigned long long*)max_ts.data_ptr<int64_t>(),
+                error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+          } else if (kind == SCAT_DIRECT) {
+            hipLaunchKernelGGL(
+                (k_radix_scatter_direct<AGG_COUNT, TSV>), grid, block, 0,
+                sc_stream, keys.data_ptr<int32_t>(), tsptr,
+                (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
+                seg_bits, cap, gcur.data_ptr<int32_t>(),
+                (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
+                ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
+                (int64_t*)nullptr, ovp.numel(),
+                (unsigned long long*)max_ts.data_ptr<int64_t>(),
+                error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+          } else {
+            size_t hist_lds = (size_t)nseg * sizeof(int);
+            hipLaunchKernelGGL(
+                (k_radix_scatter_fixed<AGG_COUNT, TSV>), grid, block,
+                2 * hist_lds,
+                sc_stream, keys.data_ptr<int32_t>(), tsptr,
+                (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
+                seg_bits, cap, gcur.data_ptr<int32_t>(),
+                (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
+                ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
+                (int64_t*)nullptr, ovp.numel(),
+                (unsigned long long*)max_ts.data_ptr<int64_t>(),
+                error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+          }
+        };
+        if (ts.scalar_type() == torch::kInt32)
+          scat_step(ts.data_ptr<int32_t>());
+        else scat_step(ts.data_ptr<int64_t>());
         HIP_CHECK(hipEventRecord(ev_sc[par], sc_stream));
         HIP_CHECK(hipStreamWaitEvent(stream, ev_sc[par], 0));
         size_t agg_lds = (size_t)16 << seg_bits;
@@ -3423,10 +3457,10 @@ int64_t native_run_window_steps(
             *gcursors, *ev_packed, *ev_vals, *ov_cursor, *ov_packed,
             *ov_vals, align_ms, len_ms, AGG_COUNT, base, region_bits);
       } else {
-        auto launch = [&](auto kern) {
+        auto launch = [&](auto kern, auto tsptr) {
           hipLaunchKernelGGL(
               kern, grid, block, 0, stream, keys.data_ptr<int32_t>(),
-              ts.data_ptr<int64_t>(), (const int64_t*)nullptr, n,
+              tsptr, (const int64_t*)nullptr, n,
               (uint64_t*)cur_k.data_ptr<int64_t>(),
               (unsigned long long*)cur_v.data_ptr<int64_t>(),
               (uint64_t)(nslots - 1), align_ms, len_ms, len_ms, base,
@@ -3434,8 +3468,16 @@ int64_t native_run_window_steps(
               (unsigned long long*)max_ts.data_ptr<int64_t>(),
               error_flag.data_ptr<int32_t>(), (const int64_t*)nullptr);
         };
-        if (dedup) launch(k_window_agg_insert<AGG_COUNT, true>);
-        else launch(k_window_agg_insert<AGG_COUNT, false>);
+        auto disp = [&](auto tsptr) {
+          using TSV =
+              std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
+          if (dedup)
+            launch(k_window_agg_insert<AGG_COUNT, true, TSV>, tsptr);
+          else launch(k_window_agg_insert<AGG_COUNT, false, TSV>, tsptr);
+        };
+        if (ts.scalar_type() == torch::kInt32)
+          disp(ts.data_ptr<int32_t>());
+        else disp(ts.data_ptr<int64_t>());
       }
 
       int64_t wm = base + sim_ms_per_batch - 1;

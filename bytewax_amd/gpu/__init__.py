@@ -52,7 +52,10 @@ class RecordBatch:
     """
 
     keys: Any  # torch.Tensor int32
-    ts: Any  # torch.Tensor int64
+    #: int64 absolute ms, or int32 deltas relative to `ts_base` (the
+    #: compact wire/template form — halves the timestamp read traffic
+    #: of the insert kernels).
+    ts: Any  # torch.Tensor int64 | int32
     vals: Optional[Any] = None  # torch.Tensor int64
     max_ts: Optional[int] = None
     #: Scalar added to every `ts` on the fly by consuming kernels —
@@ -128,7 +131,7 @@ def exchange_by_key(
         offsets = torch.cumsum(counts, 0, dtype=torch.int32) - counts
         cursors = offsets.clone()
         send_keys = torch.empty(n, dtype=torch.int32, device=dev)
-        send_ts = torch.empty(n, dtype=torch.int64, device=dev)
+        send_ts = torch.empty(n, dtype=batch.ts.dtype, device=dev)
         send_vals = torch.empty(
             n if has_vals else 0, dtype=torch.int64, device=dev
         )
@@ -156,15 +159,20 @@ def exchange_by_key(
     # `send_ts` may itself be relative to batch.ts_base (producers
     # ship zero-based templates); fold that producer base into the
     # wire base so absolute timestamps are never materialized here.
-    if batch.max_ts is not None:
+    if batch.ts.dtype == torch.int32:
+        # Already wire format: int32 deltas relative to ts_base.
+        ts_base = batch.ts_base
+        send_ts32 = send_ts
+    elif batch.max_ts is not None:
         # Host-known watermark avoids a device sync; any base within
         # 2^31 ms of every timestamp works.
         ts_base = batch.max_ts - (1 << 30)
+        send_ts32 = (send_ts - (ts_base - batch.ts_base)).to(torch.int32)
     else:
         ts_base = (
             int(send_ts.min().item()) + batch.ts_base if n > 0 else 0
         )
-    send_ts32 = (send_ts - (ts_base - batch.ts_base)).to(torch.int32)
+        send_ts32 = (send_ts - (ts_base - batch.ts_base)).to(torch.int32)
     bases = torch.zeros(world, dtype=torch.int64, device=dev)
     my_base = torch.full((world,), ts_base, dtype=torch.int64, device=dev)
     dist.all_to_all_single(bases, my_base, group=group)
@@ -348,8 +356,9 @@ class WindowAggState:
         import numpy as np
 
         keys = batch.keys.numpy()
+        ts64 = batch.ts.numpy().astype("int64", copy=False)
         wins = (
-            (batch.ts.numpy() + batch.ts_base - self.align_ms) // self.off_ms
+            (ts64 + batch.ts_base - self.align_ms) // self.off_ms
         ).astype("int64")
         if self.mode == AGG_COUNT:
             vals = np.ones(len(keys), dtype="int64")
@@ -363,7 +372,7 @@ class WindowAggState:
             self._table[kw] = self._table.get(kw, 0) + s
         if self.off_ms < self.len_ms:
             # Sliding: also fold into the earlier overlapped windows.
-            t_arr = batch.ts.numpy() + batch.ts_base
+            t_arr = batch.ts.numpy().astype("int64", copy=False) + batch.ts_base
             lo = (t_arr - self.align_ms - self.len_ms) // self.off_ms + 1
             for k, w_hi, w_lo, v in zip(
                 keys.tolist(), wins.tolist(), lo.tolist(), vals.tolist()
@@ -439,10 +448,12 @@ class WindowAggState:
             # Scatter buffers grow to fit the largest batch seen
             # (exchange-received batches vary in size).
             self._alloc_rx(int(len(batch) * 5 // 4))
+        import torch
+
         if self.radix_v2:
             self.k.radix_v2_window_insert(
                 batch.keys,
-                batch.ts,
+                batch.ts.to(torch.int64),
                 self.tkeys,
                 self.tvals,
                 self.max_ts_dev,

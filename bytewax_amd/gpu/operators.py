@@ -63,6 +63,7 @@ class _SyntheticPartition(StatelessSourcePartition[RecordBatch]):
         pool: int = 8,
         vals: bool = False,
         per_poll: int = 1,
+        ts32: bool = True,
     ):
         import torch
 
@@ -86,10 +87,16 @@ class _SyntheticPartition(StatelessSourcePartition[RecordBatch]):
             )
             for _ in range(pool)
         ]
+        # Zero-based timestamp template; int32 by default (spans <
+        # sim_ms_per_batch ms << 2^31) — halves the hot path's
+        # timestamp read traffic.  int64 for consumers that need
+        # absolute columns (hipGraph capture path).
         base = torch.arange(events_per_batch, dtype=torch.int64, device=device)
         self.ts_template = (
             base * sim_ms_per_batch
         ) // max(events_per_batch, 1)
+        if ts32:
+            self.ts_template = self.ts_template.to(torch.int32)
         self.val_pool = None
         if vals:
             self.val_pool = [
@@ -158,6 +165,7 @@ class SyntheticEventSource(DynamicSource):
     seed: int = 42
     with_vals: bool = False
     per_poll: int = 1
+    ts32: bool = True
 
     def build(
         self, step_id: str, worker_index: int, worker_count: int
@@ -175,6 +183,7 @@ class SyntheticEventSource(DynamicSource):
             self.seed + worker_index * 7919,
             vals=self.with_vals,
             per_poll=self.per_poll,
+            ts32=self.ts32,
         )
 
 
@@ -1087,6 +1096,11 @@ def filter_batch(
                 )
                 continue
             k = ext()
+            if b.ts.dtype != torch.int64:
+                b = RecordBatch(
+                    b.keys, b.ts.to(torch.int64), b.vals,
+                    max_ts=b.max_ts, ts_base=b.ts_base,
+                )
             n = len(b)
             dev = b.keys.device
             out_keys = torch.empty(n, dtype=torch.int32, device=dev)

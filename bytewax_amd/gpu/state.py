@@ -91,9 +91,20 @@ class StatsAggState:
         self.out_n = torch.zeros(1, dtype=torch.int32, device=device)
 
     def insert(self, batch: RecordBatch) -> None:
+        import torch
+
         if batch.vals is None:
             msg = "stats aggregation requires a `vals` column"
             raise ValueError(msg)
+        if not self.cpu and batch.ts.dtype != torch.int64:
+            # Stats kernels take absolute int64 timestamps.
+            batch = RecordBatch(
+                batch.keys,
+                batch.ts.to(torch.int64),
+                batch.vals,
+                max_ts=batch.max_ts,
+                ts_base=batch.ts_base,
+            )
         if self.cpu:
             self._insert_cpu(batch)
         elif self.radix:
@@ -160,8 +171,11 @@ class StatsAggState:
 
     def _insert_cpu(self, batch: RecordBatch) -> None:
         keys = batch.keys.tolist()
+        import torch
+
         wins = (
-            (batch.ts + batch.ts_base - self.align_ms) // self.len_ms
+            (batch.ts.to(torch.int64) + batch.ts_base - self.align_ms)
+            // self.len_ms
         ).tolist()
         vals = batch.vals.tolist()
         for k, w, v in zip(keys, wins, vals):
@@ -629,7 +643,7 @@ class SessionAggState:
         order2 = torch.argsort(k1.to(torch.int64), stable=True)
         perm = order[order2]
         keys = batch.keys[perm]
-        ts = batch.ts[perm] + batch.ts_base
+        ts = batch.ts[perm].to(torch.int64) + batch.ts_base
         vals = batch.vals[perm] if batch.vals is not None else None
         return keys, ts, vals
 
@@ -676,9 +690,8 @@ class SessionAggState:
         bs.tmax.fill_(_I64_MIN)
         # vals := zero-based timestamps, so min/max are the per-key
         # batch time range (shifted by ts_base at merge).
-        bs.insert(
-            RecordBatch(batch.keys, batch.ts, batch.ts, ts_base=0)
-        )
+        ts64 = batch.ts.to(torch.int64)
+        bs.insert(RecordBatch(batch.keys, ts64, ts64, ts_base=0))
         bs.out_n.zero_()
         self.k.stats_extract(
             bs.tkeys, bs.tcnt, bs.tsum, bs.tmin, bs.tmax,
