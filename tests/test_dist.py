@@ -280,3 +280,69 @@ def test_one_rank_failure_aborts_cluster_cleanly(tmp_path: Path):
     assert rcs[1] != 0
     assert "poison item" in outs[1][1].decode()
     assert rcs[0] == 0, outs[0][1].decode()[-1500:]
+
+
+@pytest.mark.timeout(180)
+def test_threads_times_processes_cluster(tmp_path: Path):
+    """`-w 2 -a <2 procs>` = 4 global workers: 2 worker THREADS per
+    process over the gloo mesh (reference Cluster{threads, process,
+    addresses}, src/run.rs:259-271).  Keyed state partitions across
+    all 4; results are exact."""
+    out_file = tmp_path / "out.txt"
+    flow_file = tmp_path / "flowdef.py"
+    flow_file.write_text(
+        textwrap.dedent(
+            f"""
+            import bytewax_amd.operators as op
+            from bytewax_amd.connectors.files import FileSink
+            from bytewax_amd.dataflow import Dataflow
+            from bytewax_amd.testing import TestingSource
+
+            inp = [(str(i % 8), 1) for i in range(40)]
+
+            flow = Dataflow("dist_hybrid")
+            s = op.input("inp", flow, TestingSource(inp))
+
+            def running_sum(state, v):
+                state = (state or 0) + v
+                return (state, state)
+
+            s = op.stateful_map("sum", s, running_sum)
+            s = op.map("fmt", s, lambda kv: (kv[0], f"{{kv[0]}}={{kv[1]}}"))
+            op.output("out", s, FileSink({str(out_file)!r}))
+            """
+        )
+    )
+    port = 29700 + os.getpid() % 200
+    addresses = f"127.0.0.1:{port};127.0.0.1:{port + 1}"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(Path(__file__).resolve().parent.parent)
+    procs = [
+        subprocess.Popen(
+            [
+                sys.executable,
+                "-m",
+                "bytewax_amd.run",
+                f"{flow_file}:flow",
+                "-i",
+                str(i),
+                "-a",
+                addresses,
+                "-w",
+                "2",
+            ],
+            env=env,
+            stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE,
+        )
+        for i in range(2)
+    ]
+    for p in procs:
+        _so, se = p.communicate(timeout=150)
+        assert p.returncode == 0, f"proc failed: {se.decode()[-2000:]}"
+
+    lines = sorted(out_file.read_text().splitlines())
+    expected = sorted(
+        f"{k}={v}" for k in "01234567" for v in range(1, 6)
+    )
+    assert lines == expected
