@@ -99,7 +99,8 @@ def _bucket_cpu(batch: RecordBatch, world: int):
 
 
 def exchange_by_key(
-    batch: RecordBatch, group=None, async_op: bool = False
+    batch: RecordBatch, group=None, async_op: bool = False,
+    force: bool = False,
 ):
     """Exchange a batch across all workers so each key lands on its
     owning worker: bucketing kernel → RCCL all-to-allv over xGMI (gloo
@@ -117,7 +118,10 @@ def exchange_by_key(
     import torch.distributed as dist
 
     world = dist.get_world_size(group)
-    if world == 1:
+    if world == 1 and not force:
+        # `force` runs the full bucket+collective path even at world 1
+        # (self-copy) — used to validate the RCCL wiring on a 1-GPU
+        # lease where 2 ranks per device are refused.
         return ([], batch) if async_op else batch
     dev = batch.keys.device
     n = len(batch)
