@@ -150,7 +150,10 @@ __global__ void k_bump(int64_t* p, int64_t v) {
 }
 
 // Aggregation modes baked at compile time per kernel instantiation.
-enum AggMode { AGG_COUNT = 0, AGG_SUM = 1 };
+enum AggMode { AGG_COUNT = 0, AGG_SUM = 1,
+               // Scatter-only mode for the fused session path:
+               // packed = key (no window), value = absolute ts.
+               AGG_TS = 2 };
 
 __device__ __forceinline__ uint64_t find_slot(
     uint64_t* __restrict__ tkeys, uint64_t mask, uint64_t packed);
@@ -559,9 +562,13 @@ __global__ void k_radix_scatter_fixed(
   for (int64_t i = start; i < n; i += stride) {
     int64_t t = (int64_t)ts[i] + ts_base;
     if (t > local_max) local_max = t;
-    int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
-    uint64_t packed =
-        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    uint64_t packed;
+    if (MODE == AGG_TS) {
+      packed = (uint64_t)(uint32_t)keys[i];
+    } else {
+      int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+      packed = ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    }
     atomicAdd(&lhist[(int)region_of(mix64(packed), mask, region_bits)], 1);
   }
   __syncthreads();
@@ -576,22 +583,26 @@ __global__ void k_radix_scatter_fixed(
   }
   __syncthreads();
   for (int64_t i = start; i < n; i += stride) {
-    int64_t win = win_of(
-        (int64_t)ts[i] + ts_base, align_ms, len_ms, win_m2, win_maxfast);
-    uint64_t packed =
-        ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    int64_t t = (int64_t)ts[i] + ts_base;
+    uint64_t packed;
+    if (MODE == AGG_TS) {
+      packed = (uint64_t)(uint32_t)keys[i];
+    } else {
+      int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+      packed = ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+    }
     int b = (int)region_of(mix64(packed), mask, region_bits);
     int64_t in_bucket = lbase[b] + atomicAdd(&lhist[b], 1);
-    int64_t v = (MODE == AGG_SUM) ? vals[i] : 0;
+    int64_t v = (MODE == AGG_SUM) ? vals[i] : (MODE == AGG_TS ? t : 0);
     if (in_bucket < cap) {
       int64_t pos = (int64_t)b * cap + in_bucket;
       ev_packed[pos] = packed;
-      if (MODE == AGG_SUM) ev_vals[pos] = v;
+      if (MODE != AGG_COUNT) ev_vals[pos] = v;
     } else {
       int opos = atomicAdd(ov_cursor, 1);
       if (opos < ov_cap) {
         ov_packed[opos] = packed;
-        if (MODE == AGG_SUM) ov_vals[opos] = v;
+        if (MODE != AGG_COUNT) ov_vals[opos] = v;
       } else {
         atomicExch(error_flag, 1);
       }
@@ -717,10 +728,10 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
   const int nseg = (int)(((mask + 1) >> seg_bits));
   uint64_t* res = (uint64_t*)smem;  // [nseg][SC_GRAN] residual staging
   int64_t* res_v =
-      (int64_t*)(smem + (size_t)nseg * SC_GRAN * 8);  // SUM only
+      (int64_t*)(smem + (size_t)nseg * SC_GRAN * 8);  // SUM/TS only
   int* res_cnt =
       (int*)(smem + (size_t)nseg * SC_GRAN * 8 *
-                        (MODE == AGG_SUM ? 2 : 1));  // [nseg]
+                        (MODE != AGG_COUNT ? 2 : 1));  // [nseg]
   int* lhist = res_cnt + nseg;  // [nseg] tile hist, then grant
   int* lbase = lhist + nseg;    // [nseg] granted global base
   int* lofs = lbase + nseg;     // [nseg] virtual-position cursor
@@ -735,12 +746,12 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
     if (gpos < cap) {
       int64_t pos = (int64_t)b * cap + gpos;
       ev_packed[pos] = packed;
-      if (MODE == AGG_SUM) ev_vals[pos] = v;
+      if (MODE != AGG_COUNT) ev_vals[pos] = v;
     } else if (packed != EMPTY_SLOT) {
       int opos = atomicAdd(ov_cursor, 1);
       if (opos < ov_cap) {
         ov_packed[opos] = packed;
-        if (MODE == AGG_SUM) ov_vals[opos] = v;
+        if (MODE != AGG_COUNT) ov_vals[opos] = v;
       } else {
         atomicExch(error_flag, 1);
       }
@@ -762,11 +773,17 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
       if (i >= n) continue;
       int64_t t = (int64_t)ts[i] + ts_base;
       if (t > local_max) local_max = t;
-      int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
-      uint64_t packed =
-          ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+      uint64_t packed;
+      if (MODE == AGG_TS) {
+        packed = (uint64_t)(uint32_t)keys[i];
+      } else {
+        int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+        packed =
+            ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+      }
       pk[u] = packed;
       if (MODE == AGG_SUM) pv[u] = vals[i];
+      else if (MODE == AGG_TS) pv[u] = t;
       int b = (int)region_of(mix64(packed), mask, seg_bits);
       sg[u] = b;
       atomicAdd(&lhist[b], 1);
@@ -790,7 +807,7 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
         int gb = lbase[b];
         for (int j = 0; j < rc; ++j) {
           emit(b, (int64_t)gb + j, res[(size_t)b * SC_GRAN + j],
-               MODE == AGG_SUM ? res_v[(size_t)b * SC_GRAN + j] : 0);
+               MODE != AGG_COUNT ? res_v[(size_t)b * SC_GRAN + j] : 0);
         }
         res_cnt[b] = 0;
       }
@@ -804,10 +821,10 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
       int grant = lhist[b];
       if (vpos < grant) {
         emit(b, (int64_t)lbase[b] + vpos, pk[u],
-             MODE == AGG_SUM ? pv[u] : 0);
+             MODE != AGG_COUNT ? pv[u] : 0);
       } else {
         res[(size_t)b * SC_GRAN + (vpos - grant)] = pk[u];
-        if (MODE == AGG_SUM) res_v[(size_t)b * SC_GRAN + (vpos - grant)] = pv[u];
+        if (MODE != AGG_COUNT) res_v[(size_t)b * SC_GRAN + (vpos - grant)] = pv[u];
       }
     }
     __syncthreads();
@@ -827,8 +844,8 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
     for (int j = 0; j < SC_GRAN; ++j) {
       uint64_t p = j < rc ? res[(size_t)b * SC_GRAN + j] : EMPTY_SLOT;
       emit(b, (int64_t)gb + j, p,
-           (MODE == AGG_SUM && j < rc) ? res_v[(size_t)b * SC_GRAN + j]
-                                       : 0);
+           (MODE != AGG_COUNT && j < rc) ? res_v[(size_t)b * SC_GRAN + j]
+                                         : 0);
     }
   }
   for (int off = WAVE / 2; off > 0; off >>= 1) {
@@ -2101,7 +2118,18 @@ void radix_window_insert(
     nseg = nb;
   }
   int64_t cap = ev_packed.numel() / nseg;
-  if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
+  if (kind == SCAT_STAGED) {
+    cap &= ~(int64_t)(SC_GRAN - 1);
+    if (cap < SC_GRAN) {
+      // Tiny buffers (small batches): granule rounding would zero the
+      // per-segment capacity; the fixed variant has no granularity.
+      kind = SCAT_FIXED;
+      coarse = 0;
+      seg_bits = (int)region_bits;
+      nseg = nb;
+      cap = ev_packed.numel() / nseg;
+    }
+  }
   TORCH_CHECK(cap * nseg >= 2 * n || cap >= n,
               "scatter buffers too small (need ~2x batch)");
   if (mode == AGG_SUM) {
@@ -2162,8 +2190,9 @@ void radix_window_insert(
       // One tile = blockDim * 16 events; enough blocks to fill the
       // chip at the LDS-bounded occupancy, few enough to keep the
       // end-of-kernel residual padding small.
+      unsigned cap_gs = env_blocks > 0 ? (unsigned)env_blocks : 512u;
       unsigned gs = (unsigned)((sg.n + 4095) / 4096);
-      if (gs > 512) gs = 512;
+      if (gs > cap_gs) gs = cap_gs;
       if (gs < 1) gs = 1;
       if (mode == AGG_COUNT)
         scat(k_radix_scatter_staged<AGG_COUNT, TSV>, tsptr, sg, gs,
@@ -2304,7 +2333,16 @@ void radix_scatter_only(
     nseg = nb;
   }
   int64_t cap = ev_packed.numel() / nseg;
-  if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
+  if (kind == SCAT_STAGED) {
+    cap &= ~(int64_t)(SC_GRAN - 1);
+    if (cap < SC_GRAN) {
+      kind = SCAT_FIXED;
+      coarse = 0;
+      seg_bits = (int)region_bits;
+      nseg = nb;
+      cap = ev_packed.numel() / nseg;
+    }
+  }
   TORCH_CHECK(cap * nseg >= 2 * n || cap >= n, "scatter buffers too small");
   size_t hist_lds = (size_t)nseg * sizeof(int);
   HIP_CHECK(hipMemsetAsync(
@@ -2418,7 +2456,15 @@ void radix_agg_only(
     nseg = nb;
   }
   int64_t cap = ev_packed.numel() / nseg;
-  if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
+  if (kind == SCAT_STAGED) {
+    cap &= ~(int64_t)(SC_GRAN - 1);
+    if (cap < SC_GRAN) {
+      kind = SCAT_FIXED;
+      seg_bits = (int)region_bits;
+      nseg = nb;
+      cap = ev_packed.numel() / nseg;
+    }
+  }
   auto offsets = at::arange(
       nseg, at::TensorOptions().dtype(at::kInt).device(tkeys.device()));
   offsets = offsets * (int)cap;
@@ -3339,7 +3385,15 @@ int64_t native_run_window_steps(
       nseg = nb;
     }
     cap = ev_packed->numel() / nseg;
-    if (kind == SCAT_STAGED) cap &= ~(int64_t)(SC_GRAN - 1);
+    if (kind == SCAT_STAGED) {
+      cap &= ~(int64_t)(SC_GRAN - 1);
+      if (cap < SC_GRAN) {
+        kind = SCAT_FIXED;
+        seg_bits = (int)region_bits;
+        nseg = nb;
+        cap = ev_packed->numel() / nseg;
+      }
+    }
     agg_offsets = at::arange(
         nseg,
         at::TensorOptions().dtype(at::kInt).device(tkeys.device()));

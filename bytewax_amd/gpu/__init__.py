@@ -393,7 +393,12 @@ class WindowAggState:
         plus the overflow spill."""
         import torch
 
-        per_region = -(-max_batch * 5 // 2) // self.n_regions + 1
+        # Floor of 4 granule lines per region so the staged scatter's
+        # 8-event reservation granularity always has working capacity
+        # even for tiny states/batches.
+        per_region = max(
+            32, -(-max_batch * 5 // 2) // self.n_regions + 1
+        )
         total = per_region * self.n_regions
         self.rx_max_batch = max_batch
         self.rx_packed = torch.empty(
