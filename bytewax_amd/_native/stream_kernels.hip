@@ -3000,6 +3000,169 @@ __global__ void k_session_restore(
   }
 }
 
+// Fused radix session aggregation (the fast path's round-2 rework):
+// an AGG_TS scatter partitions (key, t) pairs into per-segment runs;
+// one block then LDS-aggregates its segment's per-key
+// (count, min_ts, max_ts) and merges each distinct key ONCE into the
+// session table with the k_session_merge_batch semantics.  Replaces
+// the stats-table round trip (insert + extract + merge + 5 table
+// clears per batch).  Contract: gap >= batch time span and
+// watermark-ordered batches (same as the batch merge).  min/max
+// guards keep multi-merge (overflow spill) within one batch exact.
+__device__ inline void session_merge_cell(
+    uint64_t* skeys, long long* sstart, long long* slast,
+    long long* sacc, uint64_t mask, uint64_t key, long long lo,
+    long long hi, long long c, int64_t gap_ms,
+    int32_t* out_keys, int64_t* out_start, int64_t* out_end,
+    int64_t* out_vals, int* out_n, int64_t out_cap, int* error_flag) {
+  uint64_t slot = session_find_or_claim(skeys, mask, key);
+  if (slot == ~0ULL) {
+    atomicExch(error_flag, 1);
+    return;
+  }
+  long long last = slast[slot];
+  if (last >= 0 && lo - last > gap_ms) {
+    int idx = atomicAdd(out_n, 1);
+    if (idx < out_cap) {
+      out_keys[idx] = (int32_t)(uint32_t)key;
+      out_start[idx] = sstart[slot];
+      out_end[idx] = last;
+      out_vals[idx] = sacc[slot];
+    } else {
+      atomicExch(error_flag, 1);
+    }
+    last = -1;
+  }
+  if (last < 0) {
+    sstart[slot] = lo;
+    sacc[slot] = c;
+    slast[slot] = hi;
+  } else {
+    sacc[slot] += c;
+    if (lo < sstart[slot]) sstart[slot] = lo;
+    if (hi > slast[slot]) slast[slot] = hi;
+  }
+}
+
+__global__ __launch_bounds__(1024) void k_radix_session_agg(
+    const uint64_t* __restrict__ ev_packed,  // keys (AGG_TS packing)
+    const int64_t* __restrict__ ev_vals,     // absolute timestamps
+    const int* __restrict__ offsets,
+    const int* __restrict__ counts,
+    int64_t clamp_cap,
+    int lds_bits,
+    uint64_t* __restrict__ skeys,
+    long long* __restrict__ sstart,
+    long long* __restrict__ slast,
+    long long* __restrict__ sacc,
+    uint64_t mask,
+    int64_t gap_ms,
+    int* __restrict__ ov_cursor,
+    uint64_t* __restrict__ ov_packed,
+    int64_t* __restrict__ ov_vals,
+    int64_t ov_cap,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_start,
+    int64_t* __restrict__ out_end,
+    int64_t* __restrict__ out_vals,
+    int* __restrict__ out_n,
+    int64_t out_cap,
+    int* __restrict__ error_flag) {
+  extern __shared__ char smem[];
+  int slots = 1 << lds_bits;
+  uint64_t* lkeys = (uint64_t*)smem;
+  long long* lcnt = (long long*)(smem + (size_t)slots * 8);
+  long long* lmn = lcnt + slots;
+  long long* lmx = lmn + slots;
+  for (int s = threadIdx.x; s < slots; s += blockDim.x) {
+    lkeys[s] = EMPTY_SLOT;
+    lcnt[s] = 0;
+    lmn[s] = 0x7FFFFFFFFFFFFFFFLL;
+    lmx[s] = -0x7FFFFFFFFFFFFFFFLL;
+  }
+  __syncthreads();
+  int b = blockIdx.x;
+  int cnt = counts[b];
+  if (clamp_cap > 0 && cnt > (int)clamp_cap) cnt = (int)clamp_cap;
+  int start = offsets[b];
+  for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
+    uint64_t key = ev_packed[start + j];
+    if (key == EMPTY_SLOT) continue;  // staged-scatter pad
+    long long t = ev_vals[start + j];
+    uint64_t h64 = mix64(key);
+    int lh = (int)((h64 >> 32) & (slots - 1));
+    bool done = false;
+    for (int p = 0; p < slots; ++p) {
+      uint64_t cur = lkeys[lh];
+      if (cur == EMPTY_SLOT) {
+        uint64_t prev = atomicCAS(
+            (unsigned long long*)&lkeys[lh], EMPTY_SLOT, key);
+        cur = (prev == EMPTY_SLOT) ? key : prev;
+      }
+      if (cur == key) {
+        atomicAdd((unsigned long long*)&lcnt[lh], 1ULL);
+        atomicMin(&lmn[lh], t);
+        atomicMax(&lmx[lh], t);
+        done = true;
+        break;
+      }
+      lh = (lh + 1) & (slots - 1);
+    }
+    if (!done) {
+      // LDS staging full: spill to the (sequential) overflow walk.
+      int opos = atomicAdd(ov_cursor, 1);
+      if (opos < ov_cap) {
+        ov_packed[opos] = key;
+        ov_vals[opos] = t;
+      } else {
+        atomicExch(error_flag, 1);
+      }
+    }
+  }
+  __syncthreads();
+  for (int s = threadIdx.x; s < slots; s += blockDim.x) {
+    if (lkeys[s] == EMPTY_SLOT) continue;
+    session_merge_cell(
+        skeys, sstart, slast, sacc, mask, lkeys[s], lmn[s], lmx[s],
+        lcnt[s], gap_ms, out_keys, out_start, out_end, out_vals, out_n,
+        out_cap, error_flag);
+  }
+}
+
+// Sequential overflow walk: one thread merges spilled (key, t) pairs
+// per event — order within the batch does not matter under the
+// gap >= span contract (no close can trigger between same-batch
+// merges; min/max guards keep start/last exact).  Slow by design;
+// reached only under extreme key skew or LDS overflow.
+__global__ void k_session_ov_walk(
+    const uint64_t* __restrict__ ov_packed,
+    const int64_t* __restrict__ ov_vals,
+    const int* __restrict__ ov_cursor,
+    int64_t ov_cap,
+    uint64_t* __restrict__ skeys,
+    long long* __restrict__ sstart,
+    long long* __restrict__ slast,
+    long long* __restrict__ sacc,
+    uint64_t mask,
+    int64_t gap_ms,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_start,
+    int64_t* __restrict__ out_end,
+    int64_t* __restrict__ out_vals,
+    int* __restrict__ out_n,
+    int64_t out_cap,
+    int* __restrict__ error_flag) {
+  if (blockIdx.x != 0 || threadIdx.x != 0) return;
+  int64_t n = *ov_cursor;
+  if (n > ov_cap) n = ov_cap;
+  for (int64_t i = 0; i < n; ++i) {
+    session_merge_cell(
+        skeys, sstart, slast, sacc, mask, ov_packed[i], ov_vals[i],
+        ov_vals[i], 1, gap_ms, out_keys, out_start, out_end, out_vals,
+        out_n, out_cap, error_flag);
+  }
+}
+
 void session_insert(
     torch::Tensor keys,  // SORTED by (key, ts)
     torch::Tensor ts,
@@ -3739,6 +3902,128 @@ int64_t native_run_window_steps_graph(
   return total_rows;
 }
 
+// Fused radix session insert: AGG_TS scatter -> per-segment LDS
+// session aggregation -> sequential overflow walk.  One call per
+// batch; emissions accumulate in the out buffers until drained.
+void session_radix_insert(
+    torch::Tensor keys,
+    torch::Tensor ts,  // int32 template or int64 absolute
+    int64_t ts_base,
+    int64_t gap_ms,
+    torch::Tensor skeys,
+    torch::Tensor sstart,
+    torch::Tensor slast,
+    torch::Tensor sacc,
+    torch::Tensor gcursors,
+    torch::Tensor ev_packed,
+    torch::Tensor ev_vals,
+    torch::Tensor ov_cursor,
+    torch::Tensor ov_packed,
+    torch::Tensor ov_vals,
+    torch::Tensor out_keys,
+    torch::Tensor out_start,
+    torch::Tensor out_end,
+    torch::Tensor out_vals,
+    torch::Tensor out_n,
+    torch::Tensor max_ts,
+    torch::Tensor error_flag,
+    int64_t seg_bits) {
+  check_dev(keys, torch::kInt32, "keys");
+  bool ts32 = ts.scalar_type() == torch::kInt32;
+  if (!ts32) check_dev(ts, torch::kInt64, "ts");
+  int64_t n = keys.numel();
+  int64_t nslots = skeys.numel();
+  TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
+  TORCH_CHECK(seg_bits > 0 && seg_bits < 40, "bad seg_bits");
+  int64_t nseg = nslots >> seg_bits;
+  TORCH_CHECK(nseg >= 1 && nseg <= 8192, "segment count out of range");
+  TORCH_CHECK(gcursors.numel() >= nseg, "gcursors too small");
+  int64_t cap = ev_packed.numel() / nseg;
+  bool staged = true;
+  cap &= ~(int64_t)(SC_GRAN - 1);
+  if (cap < SC_GRAN) {
+    staged = false;
+    cap = ev_packed.numel() / nseg;
+  }
+  TORCH_CHECK(cap * nseg >= 2 * n || cap >= n,
+              "session scatter buffers too small");
+  TORCH_CHECK(ev_vals.numel() >= nseg * cap, "ev_vals too small");
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  uint64_t mask = (uint64_t)(nslots - 1);
+  gcursors.narrow(0, 0, nseg).zero_();
+  ov_cursor.zero_();
+  uint64_t win_m2, win_maxfast;
+  magic_div_u64((int64_t)1 << 40, &win_m2, &win_maxfast);
+  size_t staged_lds = (size_t)nseg * SC_GRAN * 8 * 2 +
+                      4 * (size_t)nseg * sizeof(int);
+  auto scat = [&](auto kern, auto tsptr, unsigned gx, size_t lds) {
+    hipLaunchKernelGGL(
+        kern, dim3(gx), dim3(256), lds, stream,
+        keys.data_ptr<int32_t>(), tsptr, (const int64_t*)nullptr, n, 0,
+        (int64_t)1 << 40, ts_base, mask, (int)seg_bits, cap,
+        gcursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(),
+        ev_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(),
+        ov_vals.data_ptr<int64_t>(), ov_packed.numel(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+  };
+  auto scat_any = [&](auto tsptr) {
+    using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
+    if (staged && staged_lds <= 160 * 1024) {
+      unsigned gs = (unsigned)((n + 4095) / 4096);
+      if (gs > 512) gs = 512;
+      if (gs < 1) gs = 1;
+      scat(k_radix_scatter_staged<AGG_TS, TSV>, tsptr, gs, staged_lds);
+    } else {
+      size_t hist_lds = (size_t)nseg * sizeof(int);
+      scat(k_radix_scatter_fixed<AGG_TS, TSV>, tsptr,
+           (unsigned)n_blocks(n, 256), 2 * hist_lds);
+    }
+  };
+  if (ts32) scat_any(ts.data_ptr<int32_t>());
+  else scat_any(ts.data_ptr<int64_t>());
+
+  auto offsets = at::arange(
+      nseg, at::TensorOptions().dtype(at::kInt).device(keys.device()));
+  offsets = offsets * (int)cap;
+  // LDS staging sized ~2x the expected distinct keys per segment.
+  int lds_bits = 12;
+  while ((int64_t)1 << lds_bits > nslots >> 1 && lds_bits > 6) lds_bits--;
+  size_t agg_lds = (size_t)32 << lds_bits;
+  hipLaunchKernelGGL(
+      k_radix_session_agg, dim3((unsigned)nseg),
+      dim3(lds_bits >= 12 ? 1024 : 256), agg_lds, stream,
+      (const uint64_t*)ev_packed.data_ptr<int64_t>(),
+      ev_vals.data_ptr<int64_t>(), offsets.data_ptr<int32_t>(),
+      gcursors.data_ptr<int32_t>(), cap, lds_bits,
+      (uint64_t*)skeys.data_ptr<int64_t>(),
+      (long long*)sstart.data_ptr<int64_t>(),
+      (long long*)slast.data_ptr<int64_t>(),
+      (long long*)sacc.data_ptr<int64_t>(), mask, gap_ms,
+      ov_cursor.data_ptr<int32_t>(),
+      (uint64_t*)ov_packed.data_ptr<int64_t>(),
+      ov_vals.data_ptr<int64_t>(), ov_packed.numel(),
+      out_keys.data_ptr<int32_t>(), out_start.data_ptr<int64_t>(),
+      out_end.data_ptr<int64_t>(), out_vals.data_ptr<int64_t>(),
+      out_n.data_ptr<int32_t>(), out_keys.numel(),
+      error_flag.data_ptr<int32_t>());
+  hipLaunchKernelGGL(
+      k_session_ov_walk, dim3(1), dim3(64), 0, stream,
+      (const uint64_t*)ov_packed.data_ptr<int64_t>(),
+      ov_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
+      ov_packed.numel(), (uint64_t*)skeys.data_ptr<int64_t>(),
+      (long long*)sstart.data_ptr<int64_t>(),
+      (long long*)slast.data_ptr<int64_t>(),
+      (long long*)sacc.data_ptr<int64_t>(), mask, gap_ms,
+      out_keys.data_ptr<int32_t>(), out_start.data_ptr<int64_t>(),
+      out_end.data_ptr<int64_t>(), out_vals.data_ptr<int64_t>(),
+      out_n.data_ptr<int32_t>(), out_keys.numel(),
+      error_flag.data_ptr<int32_t>());
+}
+
 // ---- String dictionary wrappers ----
 
 int64_t dict_encode(
@@ -3808,6 +4093,9 @@ void dict_restore(
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("session_radix_insert", &session_radix_insert,
+        "fused radix session insert: AGG_TS scatter + per-segment LDS "
+        "session aggregation + sequential overflow walk");
   m.def("dict_encode", &dict_encode,
         "device string-dictionary encode: 2-phase insert+lookup");
   m.def("dict_restore", &dict_restore,

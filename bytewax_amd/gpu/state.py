@@ -670,43 +670,66 @@ class SessionAggState:
 
     def _insert_batch_fast(self, batch: RecordBatch) -> None:
         """Batch-level fast path (COUNT, gap >= batch span): no
-        session boundary can fall inside the batch, so the per-key
-        (count, min_ts, max_ts) — computed by the radix stats kernels
-        at full speed, no sort — merges as one unit per key."""
+        session boundary can fall inside the batch, so each key's
+        per-batch (count, min_ts, max_ts) merges as one unit — fused
+        into ONE radix pass: AGG_TS scatter partitions (key, t) pairs
+        into segments, one block LDS-aggregates each segment and
+        merges its distinct keys straight into the session table
+        (round-2 rework: replaces the stats-table round trip and its
+        five per-batch table clears)."""
         import torch
 
-        if not hasattr(self, "_bs"):
-            # Internal per-batch stats aggregator; windowing disabled
-            # by a window longer than any ms-epoch timestamp.
-            self._bs = StatsAggState(
-                self.device, 0, 1 << 41,
-                slots_pow=(self.nslots.bit_length() - 1),
-                out_cap=self.out_cap,
+        n = len(batch)
+        if not hasattr(self, "rx_gcursors"):
+            # Segment count ~nslots/2^seg_bits; LDS staging in the agg
+            # covers up to 2^12 distinct keys per segment.
+            self.seg_bits = max(self.nslots.bit_length() - 1 - 10, 1)
+            nseg = self.nslots >> self.seg_bits
+            self.rx_gcursors = torch.zeros(
+                nseg, dtype=torch.int32, device=self.device
             )
-        bs = self._bs
-        bs.tkeys.fill_(-1)
-        bs.tcnt.zero_()
-        bs.tmin.fill_(_I64_MAX)
-        bs.tmax.fill_(_I64_MIN)
-        # vals := zero-based timestamps, so min/max are the per-key
-        # batch time range (shifted by ts_base at merge).
-        ts64 = batch.ts.to(torch.int64)
-        bs.insert(RecordBatch(batch.keys, ts64, ts64, ts_base=0))
-        bs.out_n.zero_()
-        self.k.stats_extract(
-            bs.tkeys, bs.tcnt, bs.tsum, bs.tmin, bs.tmax,
-            -(1 << 39), 1 << 39,
-            bs.out_keys, bs.out_wins,
-            bs.out["cnt"], bs.out["sum"], bs.out["min"], bs.out["max"],
-            bs.out_n,
-        )
-        self.k.session_merge_batch(
-            bs.out_keys, bs.out["cnt"], bs.out["min"], bs.out["max"],
-            bs.out_n,
+            self.rx_ov_cursor = torch.zeros(
+                1, dtype=torch.int32, device=self.device
+            )
+            self.rx_max_batch = 0
+            self.rx_packed = torch.empty(1, dtype=torch.int64, device=self.device)
+            self.rx_vals = torch.empty(1, dtype=torch.int64, device=self.device)
+            self.rx_ov_packed = torch.empty(
+                1, dtype=torch.int64, device=self.device
+            )
+            self.rx_ov_vals = torch.empty(
+                1, dtype=torch.int64, device=self.device
+            )
+        if n > self.rx_max_batch:
+            mb = int(n * 5 // 4)
+            self.rx_max_batch = mb
+            nseg = self.nslots >> self.seg_bits
+            per_seg = max(64, -(-mb * 5 // 2) // nseg + 1)
+            total = per_seg * nseg
+            self.rx_packed = torch.empty(
+                total, dtype=torch.int64, device=self.device
+            )
+            self.rx_vals = torch.empty(
+                total, dtype=torch.int64, device=self.device
+            )
+            ov = max(1 << 20, mb)
+            self.rx_ov_packed = torch.empty(
+                ov, dtype=torch.int64, device=self.device
+            )
+            self.rx_ov_vals = torch.empty(
+                ov, dtype=torch.int64, device=self.device
+            )
+        self.k.session_radix_insert(
+            batch.keys,
+            batch.ts,
+            batch.ts_base,
+            self.gap_ms,
             self.skeys, self.sstart, self.slast, self.sacc,
+            self.rx_gcursors, self.rx_packed, self.rx_vals,
+            self.rx_ov_cursor, self.rx_ov_packed, self.rx_ov_vals,
             self.out_keys, self.out_start, self.out_end, self.out_vals,
             self.out_n, self.max_ts_dev, self.error_flag,
-            batch.ts_base, self.gap_ms,
+            self.seg_bits,
         )
         if batch.max_ts is not None and batch.max_ts > self.max_ts_host:
             self.max_ts_host = batch.max_ts
