@@ -138,6 +138,18 @@ def branch(
     :arg predicate: Called on each item; which output stream the item
         is sent to depends on the truthiness of the return value.
     :returns: A {py:obj}`BranchOut` with `trues` and `falses` streams.
+
+    Example:
+
+    >>> flow = Dataflow("branch_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, 2, 3, 4]))
+    >>> b = op.branch("is_even", s, lambda x: x % 2 == 0)
+    >>> evens, odds = [], []
+    >>> op.output("e", b.trues, TestingSink(evens))
+    >>> op.output("o", b.falses, TestingSink(odds))
+    >>> run_main(flow)
+    >>> (evens, odds)
+    ([2, 4], [1, 3])
     """
     if not callable(predicate):
         msg = "`predicate` must be callable"
@@ -156,6 +168,18 @@ def flat_map_batch(
     The mapper sees batches of items in the order the engine happens to
     batch them — use this to amortize per-call overhead of expensive
     vectorized transformations.
+
+    Example:
+
+    >>> flow = Dataflow("flat_map_batch_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, 2]))
+    >>> s = op.flat_map_batch(
+    ...     "expand", s, lambda xs: [x * 10 for x in xs]
+    ... )
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    10
+    20
     """
     if not callable(mapper):
         msg = "`mapper` must be callable"
@@ -201,7 +225,19 @@ def inspect_debug(
 
 @operator(_core=True)
 def merge(step_id: str, *ups: Stream[Any]) -> Stream[Any]:
-    """Combine multiple streams together into one."""
+    """Combine multiple streams together into one.
+
+    Example:
+
+    >>> flow = Dataflow("merge_eg")
+    >>> a = op.input("a", flow, TestingSource([1, 2]))
+    >>> b = op.input("b", flow, TestingSource([10, 20]))
+    >>> out = []
+    >>> op.output("out", op.merge("merge", a, b), TestingSink(out))
+    >>> run_main(flow)
+    >>> sorted(out)
+    [1, 2, 10, 20]
+    """
     if len(ups) < 1:
         msg = "`merge` needs at least one upstream"
         raise TypeError(msg)
@@ -229,6 +265,17 @@ def redistribute(step_id: str, up: Stream[X]) -> Stream[X]:
     Use to rebalance skewed load after a filter or a skewed-key
     section.  On the GPU engine this is an RCCL all-to-all with random
     bucket assignment (reference operators.rs:345-361).
+
+    Example:
+
+    >>> flow = Dataflow("redistribute_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, 2, 3]))
+    >>> s = op.redistribute("spread", s)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    1
+    2
+    3
     """
     return "down"  # type: ignore[return-value]
 
@@ -391,7 +438,19 @@ def flat_map(
     up: Stream[X],
     mapper: Callable[[X], Iterable[Y]],
 ) -> Stream[Y]:
-    """Transform items one-to-many."""
+    """Transform items one-to-many.
+
+    Example:
+
+    >>> flow = Dataflow("flat_map_eg")
+    >>> s = op.input("inp", flow, TestingSource(["hello world", "hi"]))
+    >>> s = op.flat_map("split", s, str.split)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    hello
+    world
+    hi
+    """
 
     def shim_mapper(xs: List[X]) -> Iterable[Y]:
         for x in xs:
@@ -407,7 +466,18 @@ def flat_map_value(
     up: KeyedStream[V],
     mapper: Callable[[V], Iterable[W]],
 ) -> KeyedStream[W]:
-    """Transform values one-to-many."""
+    """Transform values one-to-many.
+
+    Example:
+
+    >>> flow = Dataflow("flat_map_value_eg")
+    >>> s = op.input("inp", flow, TestingSource([("k", "a b")]))
+    >>> s = op.flat_map_value("split", s, str.split)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('k', 'a')
+    ('k', 'b')
+    """
 
     def shim_mapper(k_v: Tuple[str, V]) -> Iterable[Tuple[str, W]]:
         try:
@@ -429,7 +499,19 @@ def flatten(
     step_id: str,
     up: Stream[Iterable[X]],
 ) -> Stream[X]:
-    """Move all sub-items up a level."""
+    """Move all sub-items up a level.
+
+    Example:
+
+    >>> flow = Dataflow("flatten_eg")
+    >>> s = op.input("inp", flow, TestingSource([[1, 2], [3]]))
+    >>> s = op.flatten("flatten", s)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    1
+    2
+    3
+    """
 
     def shim_mapper(x: Iterable[X]) -> Iterable[X]:
         if not isinstance(x, Iterable):
@@ -456,6 +538,16 @@ def filter(  # noqa: A001
     ```
 
     The predicate must return a `bool` (not merely truthy).
+
+    Example:
+
+    >>> flow = Dataflow("filter_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, 2, 3, 4]))
+    >>> s = op.filter("keep_even", s, lambda x: x % 2 == 0)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    2
+    4
     """
 
     def shim_mapper(x: X) -> Iterable[X]:
@@ -480,7 +572,17 @@ def filter_value(
     up: KeyedStream[V],
     predicate: Callable[[V], bool],
 ) -> KeyedStream[V]:
-    """Keep only some values."""
+    """Keep only some values.
+
+    Example:
+
+    >>> flow = Dataflow("filter_value_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", 1), ("a", 2)]))
+    >>> s = op.filter_value("keep_big", s, lambda v: v > 1)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 2)
+    """
 
     def shim_mapper(v: V) -> Iterable[V]:
         keep = predicate(v)
@@ -504,7 +606,23 @@ def filter_map(
     up: Stream[X],
     mapper: Callable[[X], Optional[Y]],
 ) -> Stream[Y]:
-    """A one-to-maybe-one transformation; `None` is discarded."""
+    """A one-to-maybe-one transformation; `None` is discarded.
+
+    Example:
+
+    >>> def parse(x):
+    ...     try:
+    ...         return int(x)
+    ...     except ValueError:
+    ...         return None
+    >>> flow = Dataflow("filter_map_eg")
+    >>> s = op.input("inp", flow, TestingSource(["1", "x", "3"]))
+    >>> s = op.filter_map("parse", s, parse)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    1
+    3
+    """
 
     def shim_mapper(x: X) -> Iterable[Y]:
         y = mapper(x)
@@ -521,7 +639,19 @@ def filter_map_value(
     up: KeyedStream[V],
     mapper: Callable[[V], Optional[W]],
 ) -> KeyedStream[W]:
-    """Transform values one-to-maybe-one; `None` is discarded."""
+    """Transform values one-to-maybe-one; `None` is discarded.
+
+    Example:
+
+    >>> flow = Dataflow("filter_map_value_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", "1"), ("a", "x")]))
+    >>> s = op.filter_map_value(
+    ...     "parse", s, lambda v: int(v) if v.isdigit() else None
+    ... )
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 1)
+    """
 
     def shim_mapper(v: V) -> Iterable[W]:
         w = mapper(v)
@@ -541,6 +671,16 @@ def inspect(
     """Observe items for debugging.
 
     The default inspector prints ``{step_id}: {item!r}``.
+
+    Example:
+
+    >>> flow = Dataflow("inspect_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, 2]))
+    >>> s = op.inspect("check", s)
+    >>> op.output("out", s, TestingSink([]))
+    >>> run_main(flow)
+    inspect_eg.check: 1
+    inspect_eg.check: 2
     """
     if inspector is None:
 
@@ -569,6 +709,17 @@ def map(  # noqa: A001
     s = op.input("inp", flow, TestingSource([1, 2, 3]))
     s = op.map("add_one", s, lambda x: x + 1)  # 2, 3, 4
     ```
+
+    Example:
+
+    >>> flow = Dataflow("map_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, 2, 3]))
+    >>> s = op.map("add_one", s, lambda x: x + 1)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    2
+    3
+    4
     """
 
     def shim_mapper(xs: List[X]) -> Iterable[Y]:
@@ -583,7 +734,18 @@ def map_value(
     up: KeyedStream[V],
     mapper: Callable[[V], W],
 ) -> KeyedStream[W]:
-    """Transform values one-by-one."""
+    """Transform values one-by-one.
+
+    Example:
+
+    >>> flow = Dataflow("map_value_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", 1), ("b", 2)]))
+    >>> s = op.map_value("add_one", s, lambda v: v + 1)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 2)
+    ('b', 3)
+    """
 
     def shim_mapper(k_v: Tuple[str, V]) -> Tuple[str, W]:
         try:
@@ -609,6 +771,16 @@ def key_on(step_id: str, up: Stream[X], key: Callable[[X], str]) -> KeyedStream[
 
     The key function must return a `str` (keys route state across
     workers).
+
+    Example:
+
+    >>> flow = Dataflow("key_on_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, 2]))
+    >>> s = op.key_on("key", s, str)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('1', 1)
+    ('2', 2)
     """
 
     def shim_mapper(x: X) -> Tuple[str, X]:
@@ -627,7 +799,18 @@ def key_on(step_id: str, up: Stream[X], key: Callable[[X], str]) -> KeyedStream[
 
 @operator
 def key_rm(step_id: str, up: KeyedStream[X]) -> Stream[X]:
-    """Discard keys from a keyed stream."""
+    """Discard keys from a keyed stream.
+
+    Example:
+
+    >>> flow = Dataflow("key_rm_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", 1), ("b", 2)]))
+    >>> s = op.key_rm("unkey", s)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    1
+    2
+    """
 
     def shim_mapper(k_v: Tuple[str, X]) -> X:
         _k, v = k_v
@@ -716,6 +899,20 @@ def collect(
         was not reached.
     :arg max_size: Emit the list once it reaches this size, even if
         `timeout` was not reached.
+
+    Example:
+
+    >>> flow = Dataflow("collect_eg")
+    >>> s = op.input(
+    ...     "inp", flow, TestingSource([("a", 1), ("a", 2), ("a", 3)])
+    ... )
+    >>> s = op.collect(
+    ...     "collect", s, timeout=timedelta(seconds=10), max_size=2
+    ... )
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', [1, 2])
+    ('a', [3])
     """
 
     def shim_builder(
@@ -759,6 +956,15 @@ def fold_final(
     folded = op.fold_final("fold", keyed, lambda: 0, lambda acc, v: acc + v)
     # ("key1", 1), ("key1", 2), ("key2", 3) -> ("key1", 3), ("key2", 3)
     ```
+
+    Example:
+
+    >>> flow = Dataflow("fold_final_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", 1), ("a", 2)]))
+    >>> s = op.fold_final("fold", s, list, lambda acc, v: acc + [v])
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', [1, 2])
     """
 
     def shim_builder(resume_state: Optional[S]) -> _FoldFinalLogic[V, S]:
@@ -778,6 +984,18 @@ def reduce_final(
 
     Like {py:obj}`fold_final` but the first value is the initial
     accumulator.
+
+    Example:
+
+    >>> flow = Dataflow("reduce_final_eg")
+    >>> s = op.input(
+    ...     "inp", flow, TestingSource([("a", 1), ("a", 2), ("b", 5)])
+    ... )
+    >>> s = op.reduce_final("sum", s, lambda a, b: a + b)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 3)
+    ('b', 5)
     """
 
     def pre_folder(acc: List[V], v: V) -> List[V]:
@@ -802,6 +1020,16 @@ def count_final(
 
     Only works on finite streams; results are only emitted once the
     upstream is EOF.
+
+    Example:
+
+    >>> flow = Dataflow("count_final_eg")
+    >>> s = op.input("inp", flow, TestingSource(["a", "a", "b"]))
+    >>> s = op.count_final("count", s, lambda x: x)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 2)
+    ('b', 1)
     """
     keyed = map("extract_key", up, lambda x: (key(x), 1))
     return fold_final("count", keyed, int, lambda s, x: s + x)
@@ -813,7 +1041,17 @@ def max_final(
     up: KeyedStream[V],
     by: Callable[[V], Any] = _identity,
 ) -> KeyedStream[V]:
-    """Find the maximum value for each key; emitted at EOF."""
+    """Find the maximum value for each key; emitted at EOF.
+
+    Example:
+
+    >>> flow = Dataflow("max_final_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", 1), ("a", 4)]))
+    >>> s = op.max_final("max", s)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 4)
+    """
     return reduce_final(
         "reduce_final", up, lambda a, b: a if by(a) >= by(b) else b
     )
@@ -825,7 +1063,17 @@ def min_final(
     up: KeyedStream[V],
     by: Callable[[V], Any] = _identity,
 ) -> KeyedStream[V]:
-    """Find the minimum value for each key; emitted at EOF."""
+    """Find the minimum value for each key; emitted at EOF.
+
+    Example:
+
+    >>> flow = Dataflow("min_final_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", 1), ("a", 4)]))
+    >>> s = op.min_final("min", s)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 1)
+    """
     return reduce_final(
         "reduce_final", up, lambda a, b: a if by(a) <= by(b) else b
     )
@@ -877,6 +1125,23 @@ def stateful_flat_map(
         with the last state or `None`, and then the value.  Should
         return a 2-tuple of `(updated_state, emit_values)`.  If the
         updated state is `None`, discard it.
+
+    Example:
+
+    >>> def dedupe(state, v):
+    ...     seen = state or set()
+    ...     out = [] if v in seen else [v]
+    ...     seen.add(v)
+    ...     return (seen, out)
+    >>> flow = Dataflow("stateful_flat_map_eg")
+    >>> s = op.input(
+    ...     "inp", flow, TestingSource([("a", 1), ("a", 1), ("a", 2)])
+    ... )
+    >>> s = op.stateful_flat_map("dedupe", s, dedupe)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 1)
+    ('a', 2)
     """
 
     def shim_builder(resume_state: Optional[S]) -> _StatefulMapLogic[V, W, S]:
@@ -906,6 +1171,22 @@ def stateful_map(
         with the last state or `None`, and then the value.  Should
         return a 2-tuple of `(updated_state, emit_value)`.  If the
         updated state is `None`, discard it.
+
+    Example:
+
+    >>> def running_sum(state, v):
+    ...     state = (state or 0) + v
+    ...     return (state, state)
+    >>> flow = Dataflow("stateful_map_eg")
+    >>> s = op.input(
+    ...     "inp", flow, TestingSource([("a", 1), ("a", 2), ("b", 5)])
+    ... )
+    >>> s = op.stateful_map("sum", s, running_sum)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 1)
+    ('a', 3)
+    ('b', 5)
     """
 
     def shim_builder(resume_state: Optional[S]) -> _StatefulMapLogic[V, W, S]:
@@ -953,7 +1234,23 @@ def enrich_cached(
     ttl: timedelta = timedelta.max,
     _now_getter: Callable[[], datetime] = _get_system_utc,
 ) -> Stream[Y]:
-    """Enrich / join items using a cached lookup to an external service."""
+    """Enrich / join items using a cached lookup to an external service.
+
+    Example:
+
+    >>> def lookup(k):
+    ...     return {"1": "alice", "2": "bob"}[k]
+    >>> flow = Dataflow("enrich_eg")
+    >>> s = op.input("inp", flow, TestingSource(["1", "2", "1"]))
+    >>> s = op.enrich_cached(
+    ...     "names", s, lookup, lambda cache, k: cache.get(k)
+    ... )
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    alice
+    bob
+    alice
+    """
     cache = TTLCache(getter, _now_getter, ttl)
 
     def shim_mapper(xs: Iterable[X]) -> Iterable[Y]:
@@ -1068,6 +1365,16 @@ def join(
         the most recent, "product" all of them.
     :arg emit_mode: "complete" emits once all sides have a value and
         resets; "final" emits at EOF; "running" emits on every item.
+
+    Example:
+
+    >>> flow = Dataflow("join_eg")
+    >>> names = op.input("n", flow, TestingSource([("1", "alice")]))
+    >>> emails = op.input("e", flow, TestingSource([("1", "a@x.io")]))
+    >>> s = op.join("join", names, emails)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('1', ('alice', 'a@x.io'))
     """
     if insert_mode not in typing.get_args(JoinInsertMode):
         msg = f"unknown join insert mode {insert_mode!r}"

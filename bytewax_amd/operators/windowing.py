@@ -1137,6 +1137,35 @@ def fold_window(
     :arg folder: Combines a new value into the accumulator.
     :arg merger: Combines two accumulators when windows merge
         (session windows).
+
+    Example:
+
+    >>> align = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    >>> inp = [
+    ...     (align + timedelta(seconds=1), "a"),
+    ...     (align + timedelta(seconds=2), "b"),
+    ...     (align + timedelta(seconds=61), "c"),
+    ... ]
+    >>> flow = Dataflow("fold_window_eg")
+    >>> s = op.input("inp", flow, TestingSource(inp))
+    >>> keyed = op.key_on("key", s, lambda x: "ALL")
+    >>> clock = win.EventClock(
+    ...     ts_getter=lambda x: x[0],
+    ...     wait_for_system_duration=timedelta(0),
+    ... )
+    >>> wo = win.fold_window(
+    ...     "fold",
+    ...     keyed,
+    ...     clock,
+    ...     win.TumblingWindower(align_to=align, length=timedelta(minutes=1)),
+    ...     list,
+    ...     lambda acc, x: acc + [x[1]],
+    ...     lambda a, b: a + b,
+    ... )
+    >>> op.output("out", wo.down, StdOutSink())
+    >>> run_main(flow)
+    ('ALL', (0, ['a', 'b']))
+    ('ALL', (1, ['c']))
     """
 
     def shim_builder(resume_state: Optional[S]) -> _FoldWindowLogic[V, S]:
@@ -1192,7 +1221,34 @@ def collect_window(
     ordered: bool = True,
 ) -> WindowOut[V, Any]:
     """Collect items in a window into a container (`list`, `set` or
-    `dict`; for `dict` the values must be `(key, value)` 2-tuples)."""
+    `dict`; for `dict` the values must be `(key, value)` 2-tuples).
+
+    Example:
+
+    >>> align = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    >>> inp = [
+    ...     (align + timedelta(seconds=s), v)
+    ...     for s, v in [(1, 1), (2, 2), (61, 3)]
+    ... ]
+    >>> flow = Dataflow("collect_window_eg")
+    >>> s = op.input("inp", flow, TestingSource(inp))
+    >>> keyed = op.key_on("key", s, lambda x: "ALL")
+    >>> clock = win.EventClock(
+    ...     ts_getter=lambda x: x[0],
+    ...     wait_for_system_duration=timedelta(0),
+    ... )
+    >>> wo = win.collect_window(
+    ...     "collect",
+    ...     keyed,
+    ...     clock,
+    ...     win.TumblingWindower(align_to=align, length=timedelta(minutes=1)),
+    ... )
+    >>> out = []
+    >>> op.output("out", wo.down, TestingSink(out))
+    >>> run_main(flow)
+    >>> [(k, (wid, [v for _ts, v in items])) for k, (wid, items) in out]
+    [('ALL', (0, [1, 2])), ('ALL', (1, [3]))]
+    """
     if into is list:
         folder, merger = _collect_list_folder, (lambda a, b: a + b)
     elif into is set:
@@ -1221,6 +1277,27 @@ def count_window(
     :arg key: Called on each item to route the counts.
     :returns: Window result streams; downstream contains `(key,
         (window_id, count))` once the window closes.
+
+    Example:
+
+    >>> align = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    >>> inp = [align, align + timedelta(seconds=30), align + timedelta(seconds=65)]
+    >>> flow = Dataflow("count_window_eg")
+    >>> s = op.input("inp", flow, TestingSource(inp))
+    >>> clock = win.EventClock(
+    ...     ts_getter=lambda x: x, wait_for_system_duration=timedelta(0)
+    ... )
+    >>> wo = win.count_window(
+    ...     "count",
+    ...     s,
+    ...     clock,
+    ...     win.TumblingWindower(align_to=align, length=timedelta(minutes=1)),
+    ...     key=lambda x: "ALL",
+    ... )
+    >>> op.output("out", wo.down, StdOutSink())
+    >>> run_main(flow)
+    ('ALL', (0, 2))
+    ('ALL', (1, 1))
     """
     keyed = op.key_on("key", up, key)
     return fold_window(
@@ -1243,7 +1320,35 @@ def max_window(
     windower: Windower[Any],
     by=_identity,
 ) -> WindowOut[V, V]:
-    """Find the maximum value for each key per window."""
+    """Find the maximum value for each key per window.
+
+    Example:
+
+    >>> align = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    >>> inp = [
+    ...     (align + timedelta(seconds=s), v)
+    ...     for s, v in [(1, 5), (2, 9), (61, 3)]
+    ... ]
+    >>> flow = Dataflow("max_window_eg")
+    >>> s = op.input("inp", flow, TestingSource(inp))
+    >>> keyed = op.key_on("key", s, lambda x: "ALL")
+    >>> clock = win.EventClock(
+    ...     ts_getter=lambda x: x[0],
+    ...     wait_for_system_duration=timedelta(0),
+    ... )
+    >>> wo = win.max_window(
+    ...     "max",
+    ...     keyed,
+    ...     clock,
+    ...     win.TumblingWindower(align_to=align, length=timedelta(minutes=1)),
+    ...     by=lambda x: x[1],
+    ... )
+    >>> out = []
+    >>> op.output("out", wo.down, TestingSink(out))
+    >>> run_main(flow)
+    >>> [(k, (wid, v[1])) for k, (wid, v) in out]
+    [('ALL', (0, 9)), ('ALL', (1, 3))]
+    """
     return reduce_window(
         "reduce_window", up, clock, windower, partial(max, key=by)
     )
@@ -1273,7 +1378,35 @@ def reduce_window(
 ) -> WindowOut[V, V]:
     """Distill all values for a key in a window down into a single
     value; like {py:obj}`fold_window` but the first value is the
-    initial accumulator."""
+    initial accumulator.
+
+    Example:
+
+    >>> align = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    >>> inp = [
+    ...     (align + timedelta(seconds=s), v)
+    ...     for s, v in [(1, 5), (2, 9), (61, 3)]
+    ... ]
+    >>> flow = Dataflow("reduce_window_eg")
+    >>> s = op.input("inp", flow, TestingSource(inp))
+    >>> keyed = op.key_on("key", s, lambda x: "ALL")
+    >>> clock = win.EventClock(
+    ...     ts_getter=lambda x: x[0],
+    ...     wait_for_system_duration=timedelta(0),
+    ... )
+    >>> wo = win.reduce_window(
+    ...     "sum",
+    ...     keyed,
+    ...     clock,
+    ...     win.TumblingWindower(align_to=align, length=timedelta(minutes=1)),
+    ...     lambda a, b: (a[0], a[1] + b[1]),
+    ... )
+    >>> out = []
+    >>> op.output("out", wo.down, TestingSink(out))
+    >>> run_main(flow)
+    >>> [(k, (wid, v[1])) for k, (wid, v) in out]
+    [('ALL', (0, 14)), ('ALL', (1, 3))]
+    """
 
     def shim_folder(s, v):
         return v if s is None else reducer(s, v)
