@@ -59,3 +59,46 @@ def test_onebrc_example_cpu_twin():
     )
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     assert "aggregated 200000 rows over 10000 stations" in res.stdout.decode()
+
+
+def test_csv_input_example():
+    res = _run([sys.executable, "examples/csv_input.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "'instance':" in res.stdout.decode()
+
+
+def test_join_example():
+    res = _run([sys.executable, "examples/join.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert (
+        "('123', ('Bumble', 'bee@example.com', 'yellow', 'buzz'))"
+        in res.stdout.decode()
+    )
+
+
+def test_anomaly_detector_example():
+    res = _run([sys.executable, "examples/anomaly_detector.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    out = res.stdout.decode()
+    assert "v_metric:" in out and "hz_metric:" in out
+
+
+def test_periodic_input_example():
+    res = _run([sys.executable, "examples/periodic_input.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert res.stdout.decode().count("delay (ms):") == 8  # 2 flows x 4
+
+
+def test_orderbook_example():
+    res = _run([sys.executable, "examples/orderbook.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    out = res.stdout.decode()
+    assert "BTC-USD" in out or "ETH-USD" in out
+    assert "Summary(" in out
+
+
+def test_search_session_example():
+    res = _run([sys.executable, "examples/search_session.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    out = res.stdout.decode()
+    assert "('1', 1.0)" in out and "('2', 0.0)" in out
