@@ -674,8 +674,11 @@ class _Worker:
 
     def _drain(self) -> None:
         from .._metrics import metrics_enabled, observe_batch
+        from ..tracing import tracing_active
 
-        instrumented = metrics_enabled()
+        self._metrics_on = metrics_enabled()
+        self._tracing_on = tracing_active()
+        instrumented = self._metrics_on or self._tracing_on
         if self._poison is not None:
             # Aborting: discard queued work but keep the collective
             # schedule aligned (exchange rounds still run with empty
@@ -706,13 +709,26 @@ class _Worker:
                     n_out += len(out_items)
                     if out_items:
                         self._emit(stream_id, out_items)
-                observe_batch(
-                    ex.step.step_id,
-                    self.ctx.worker_index,
-                    len(items),
-                    n_out,
-                    time.perf_counter() - t0,
-                )
+                t1 = time.perf_counter()
+                if self._metrics_on:
+                    observe_batch(
+                        ex.step.step_id,
+                        self.ctx.worker_index,
+                        len(items),
+                        n_out,
+                        t1 - t0,
+                    )
+                if self._tracing_on:
+                    from ..tracing import record_operator_span
+
+                    record_operator_span(
+                        ex.step.step_id,
+                        self.ctx.worker_index,
+                        len(items),
+                        n_out,
+                        t0,
+                        t1,
+                    )
             else:
                 for stream_id, out_items in ex.process(
                     input_idx, items, self.epoch
