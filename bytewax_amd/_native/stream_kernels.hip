@@ -992,7 +992,7 @@ __global__ void k_overflow_agg_stats(
 // min / max): one workgroup per region, LDS-resident accumulators,
 // one flush per distinct cell.  Events must already be partitioned
 // (k_radix_scatter with values).
-__global__ __launch_bounds__(256) void k_radix_agg_stats(
+__global__ __launch_bounds__(1024) void k_radix_agg_stats(
     const uint64_t* __restrict__ ev_packed,
     const int64_t* __restrict__ ev_vals,
     const int* __restrict__ offsets,
@@ -1005,9 +1005,10 @@ __global__ __launch_bounds__(256) void k_radix_agg_stats(
     int64_t clamp_cap,
     uint64_t mask,
     int region_bits,
+    int lds_bits,  // LDS staging slots (>= region_bits for coarse segments)
     int* __restrict__ error_flag) {
   extern __shared__ char smem[];
-  int region = 1 << region_bits;
+  int region = 1 << lds_bits;
   uint64_t* lkeys = (uint64_t*)smem;
   long long* lcnt = (long long*)(smem + (size_t)region * 8);
   long long* lsum = (long long*)(smem + (size_t)region * 16);
@@ -1028,6 +1029,7 @@ __global__ __launch_bounds__(256) void k_radix_agg_stats(
   int start = offsets[b];
   for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
     uint64_t packed = ev_packed[start + j];
+    if (packed == EMPTY_SLOT) continue;  // staged-scatter pad
     long long v = ev_vals[start + j];
     uint64_t h64 = mix64(packed);
     int lh = (int)((h64 >> 32) & (region - 1));
@@ -2673,39 +2675,93 @@ void radix_stats_insert(
               "stats radix needs 0 < region_bits <= 10 (40 B/slot LDS)");
   int64_t nb = nslots >> region_bits;
   TORCH_CHECK(nb >= 1 && nb <= 8192, "region count out of range");
-  int64_t cap = ev_packed.numel() / nb;
-  TORCH_CHECK(cap * nb >= 2 * n || cap >= n,
-              "scatter buffers too small (need ~2x batch)");
-  TORCH_CHECK(ev_vals.numel() >= nb * cap, "ev_vals too small");
   if (n == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
   uint64_t mask = (uint64_t)(nslots - 1);
   dim3 block(256);
   dim3 grid(n_blocks(n, 256));
-  gcursors.narrow(0, 0, nb).zero_();
+
+  // Same scatter-variant decision as the window path; the stats agg
+  // stages 40 B/LDS slot, capping coarse segments at lds_bits 11.
+  ScatterKind kind = scatter_kind_env();
+  int coarse = scatter_coarse_bits(kind);
+  int seg_bits = (int)region_bits + coarse;
+  int64_t nseg = nslots >> seg_bits;
+  size_t staged_lds = (size_t)nseg * SC_GRAN * 8 * 2 +
+                      4 * (size_t)nseg * sizeof(int);
+  while (kind == SCAT_STAGED &&
+         (nseg < 1 || staged_lds > 160 * 1024 || seg_bits > 11)) {
+    if (coarse > 0 && seg_bits > 11) {
+      coarse -= 1;
+    } else {
+      kind = SCAT_FIXED;
+      coarse = 0;
+    }
+    seg_bits = (int)region_bits + coarse;
+    nseg = nslots >> seg_bits;
+    staged_lds = (size_t)nseg * SC_GRAN * 8 * 2 +
+                 4 * (size_t)nseg * sizeof(int);
+  }
+  if (seg_bits > 11) {
+    coarse = 0;
+    seg_bits = (int)region_bits;
+    nseg = nb;
+  }
+  int64_t cap = ev_packed.numel() / nseg;
+  if (kind == SCAT_STAGED) {
+    cap &= ~(int64_t)(SC_GRAN - 1);
+    if (cap < SC_GRAN) {
+      kind = SCAT_FIXED;
+      coarse = 0;
+      seg_bits = (int)region_bits;
+      nseg = nb;
+      cap = ev_packed.numel() / nseg;
+    }
+  }
+  TORCH_CHECK(cap * nseg >= 2 * n || cap >= n,
+              "scatter buffers too small (need ~2x batch)");
+  TORCH_CHECK(ev_vals.numel() >= nseg * cap, "ev_vals too small");
+  gcursors.narrow(0, 0, nseg).zero_();
   ov_cursor.zero_();
-  size_t hist_lds = (size_t)nb * sizeof(int);
+  size_t hist_lds = (size_t)nseg * sizeof(int);
   uint64_t win_m2, win_maxfast;
   magic_div_u64(len_ms, &win_m2, &win_maxfast);
-  hipLaunchKernelGGL(
-      k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
-      keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-      vals.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
-      (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
-      (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
-      ov_cursor.data_ptr<int32_t>(),
-      (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
-      ov_packed.numel(),
-      (unsigned long long*)max_ts.data_ptr<int64_t>(),
-      error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+  if (kind == SCAT_STAGED) {
+    unsigned gs = (unsigned)((n + 4095) / 4096);
+    if (gs > 512) gs = 512;
+    if (gs < 1) gs = 1;
+    hipLaunchKernelGGL(
+        (k_radix_scatter_staged<AGG_SUM>), dim3(gs), block, staged_lds,
+        stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
+        vals.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
+        seg_bits, cap, gcursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(),
+        ev_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(),
+        ov_vals.data_ptr<int64_t>(), ov_packed.numel(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+  } else {
+    hipLaunchKernelGGL(
+        k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
+        keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
+        vals.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
+        seg_bits, cap, gcursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
+        ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
+        ov_packed.numel(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+  }
 
   auto offsets = at::arange(
-      nb, at::TensorOptions().dtype(at::kInt).device(keys.device()));
+      nseg, at::TensorOptions().dtype(at::kInt).device(keys.device()));
   offsets = offsets * (int)cap;
-  int region = 1 << region_bits;
-  size_t agg_lds = (size_t)region * 40;
+  size_t agg_lds = (size_t)40 << seg_bits;
+  dim3 agg_block(seg_bits >= 11 ? 1024 : 256);
   hipLaunchKernelGGL(
-      k_radix_agg_stats, dim3((unsigned)nb), block, agg_lds, stream,
+      k_radix_agg_stats, dim3((unsigned)nseg), agg_block, agg_lds, stream,
       (const uint64_t*)ev_packed.data_ptr<int64_t>(),
       ev_vals.data_ptr<int64_t>(), offsets.data_ptr<int32_t>(),
       gcursors.data_ptr<int32_t>(), (uint64_t*)tkeys.data_ptr<int64_t>(),
@@ -2713,7 +2769,7 @@ void radix_stats_insert(
       (long long*)tsum.data_ptr<int64_t>(),
       (long long*)tmin.data_ptr<int64_t>(),
       (long long*)tmax.data_ptr<int64_t>(), cap, mask, (int)region_bits,
-      error_flag.data_ptr<int32_t>());
+      seg_bits, error_flag.data_ptr<int32_t>());
 
   hipLaunchKernelGGL(
       k_overflow_agg_stats, dim3(64), block, 0, stream,
