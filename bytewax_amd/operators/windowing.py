@@ -1362,7 +1362,35 @@ def min_window(
     windower: Windower[Any],
     by=_identity,
 ) -> WindowOut[V, V]:
-    """Find the minimum value for each key per window."""
+    """Find the minimum value for each key per window.
+
+    Example:
+
+    >>> align = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    >>> inp = [
+    ...     (align + timedelta(seconds=s), v)
+    ...     for s, v in [(1, 5), (2, 9), (61, 3)]
+    ... ]
+    >>> flow = Dataflow("min_window_eg")
+    >>> s = op.input("inp", flow, TestingSource(inp))
+    >>> keyed = op.key_on("key", s, lambda x: "ALL")
+    >>> clock = win.EventClock(
+    ...     ts_getter=lambda x: x[0],
+    ...     wait_for_system_duration=timedelta(0),
+    ... )
+    >>> wo = win.min_window(
+    ...     "min",
+    ...     keyed,
+    ...     clock,
+    ...     win.TumblingWindower(align_to=align, length=timedelta(minutes=1)),
+    ...     by=lambda x: x[1],
+    ... )
+    >>> out = []
+    >>> op.output("out", wo.down, TestingSink(out))
+    >>> run_main(flow)
+    >>> [(k, (wid, v[1])) for k, (wid, v) in out]
+    [('ALL', (0, 5)), ('ALL', (1, 3))]
+    """
     return reduce_window(
         "reduce_window", up, clock, windower, partial(min, key=by)
     )
@@ -1490,6 +1518,33 @@ def join_window(
 
     :returns: Window result streams; downstream contains tuples with
         the value from each side in argument order.
+
+    Example:
+
+    >>> align = datetime(2024, 1, 1, tzinfo=timezone.utc)
+    >>> names = [(align + timedelta(seconds=1), ("1", "alice"))]
+    >>> mails = [(align + timedelta(seconds=2), ("1", "a@x.io"))]
+    >>> flow = Dataflow("join_window_eg")
+    >>> n = op.input("n", flow, TestingSource(names))
+    >>> m = op.input("m", flow, TestingSource(mails))
+    >>> clock = win.EventClock(
+    ...     ts_getter=lambda ts_kv: ts_kv[0],
+    ...     wait_for_system_duration=timedelta(0),
+    ... )
+    >>> n_keyed = op.map("nk", n, lambda ts_kv: (ts_kv[1][0], ts_kv))
+    >>> m_keyed = op.map("mk", m, lambda ts_kv: (ts_kv[1][0], ts_kv))
+    >>> wo = win.join_window(
+    ...     "join",
+    ...     clock,
+    ...     win.TumblingWindower(align_to=align, length=timedelta(minutes=1)),
+    ...     n_keyed,
+    ...     m_keyed,
+    ... )
+    >>> out = []
+    >>> op.output("out", wo.down, TestingSink(out))
+    >>> run_main(flow)
+    >>> [(k, (wid, tuple(v[1][1] for v in vals))) for k, (wid, vals) in out]
+    [('1', (0, ('alice', 'a@x.io')))]
     """
     if insert_mode not in typing.get_args(JoinInsertMode):
         msg = f"unknown join insert mode {insert_mode!r}"
