@@ -102,3 +102,20 @@ def test_search_session_example():
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     out = res.stdout.decode()
     assert "('1', 1.0)" in out and "('2', 0.0)" in out
+
+
+def test_events_to_parquet_example(tmp_path):
+    res = _run(
+        [sys.executable, str(REPO / "examples" / "events_to_parquet.py")],
+    )
+    # The example writes ./parquet_demo_out relative to the cwd (repo
+    # root under _run); assert and clean up.
+    out_dir = REPO / "parquet_demo_out"
+    try:
+        assert res.returncode == 0, res.stderr.decode()[-1500:]
+        assert "wrote" in res.stdout.decode()
+        assert list(out_dir.rglob("*.parquet"))
+    finally:
+        import shutil
+
+        shutil.rmtree(out_dir, ignore_errors=True)
