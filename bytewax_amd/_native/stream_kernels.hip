@@ -491,6 +491,16 @@ __global__ __launch_bounds__(256) void k_agg_cached(
 // precomputes a Granlund-Montgomery magic multiplier instead:
 // win = mulhi64(x, m2) is exact for 0 <= x < maxfast (m2 == 0 or
 // x outside the window falls back to the hardware-free division).
+// Oldest window overlapping t for a sliding windower: the window ids
+// are strides of `off`; an event at t lands in ids
+// [floor((t-align-len)/off)+1, (t-align)/off].
+__device__ __forceinline__ int64_t win_lo_of(
+    int64_t t, int64_t align_ms, int64_t len_ms, int64_t off_ms) {
+  int64_t num = t - align_ms - len_ms;
+  int64_t fl = num >= 0 ? num / off_ms : -((-num + off_ms - 1) / off_ms);
+  return fl + 1;
+}
+
 __device__ __forceinline__ int64_t win_of(
     int64_t t, int64_t align_ms, int64_t len_ms, uint64_t m2,
     uint64_t maxfast) {
@@ -535,6 +545,7 @@ __global__ void k_radix_scatter_fixed(
     int64_t n,
     int64_t align_ms,
     int64_t len_ms,
+    int64_t off_ms,  // sliding stride; == len_ms for tumbling
     int64_t ts_base,
     uint64_t mask,
     int region_bits,
@@ -563,13 +574,20 @@ __global__ void k_radix_scatter_fixed(
     int64_t t = (int64_t)ts[i] + ts_base;
     if (t > local_max) local_max = t;
     uint64_t packed;
+    int nwin = 1;
     if (MODE == AGG_TS) {
       packed = (uint64_t)(uint32_t)keys[i];
     } else {
-      int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+      int64_t win = win_of(t, align_ms, off_ms, win_m2, win_maxfast);
       packed = ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+      if (off_ms < len_ms) {
+        nwin = (int)(win - win_lo_of(t, align_ms, len_ms, off_ms)) + 1;
+      }
     }
-    atomicAdd(&lhist[(int)region_of(mix64(packed), mask, region_bits)], 1);
+    for (int w = 0; w < nwin; ++w) {
+      atomicAdd(&lhist[(int)region_of(mix64(packed), mask, region_bits)], 1);
+      packed -= (uint64_t)1 << 32;
+    }
   }
   __syncthreads();
   for (int b = threadIdx.x; b < nb; b += blockDim.x) {
@@ -585,27 +603,34 @@ __global__ void k_radix_scatter_fixed(
   for (int64_t i = start; i < n; i += stride) {
     int64_t t = (int64_t)ts[i] + ts_base;
     uint64_t packed;
+    int nwin = 1;
     if (MODE == AGG_TS) {
       packed = (uint64_t)(uint32_t)keys[i];
     } else {
-      int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+      int64_t win = win_of(t, align_ms, off_ms, win_m2, win_maxfast);
       packed = ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
-    }
-    int b = (int)region_of(mix64(packed), mask, region_bits);
-    int64_t in_bucket = lbase[b] + atomicAdd(&lhist[b], 1);
-    int64_t v = (MODE == AGG_SUM) ? vals[i] : (MODE == AGG_TS ? t : 0);
-    if (in_bucket < cap) {
-      int64_t pos = (int64_t)b * cap + in_bucket;
-      ev_packed[pos] = packed;
-      if (MODE != AGG_COUNT) ev_vals[pos] = v;
-    } else {
-      int opos = atomicAdd(ov_cursor, 1);
-      if (opos < ov_cap) {
-        ov_packed[opos] = packed;
-        if (MODE != AGG_COUNT) ov_vals[opos] = v;
-      } else {
-        atomicExch(error_flag, 1);
+      if (off_ms < len_ms) {
+        nwin = (int)(win - win_lo_of(t, align_ms, len_ms, off_ms)) + 1;
       }
+    }
+    int64_t v = (MODE == AGG_SUM) ? vals[i] : (MODE == AGG_TS ? t : 0);
+    for (int w = 0; w < nwin; ++w) {
+      int b = (int)region_of(mix64(packed), mask, region_bits);
+      int64_t in_bucket = lbase[b] + atomicAdd(&lhist[b], 1);
+      if (in_bucket < cap) {
+        int64_t pos = (int64_t)b * cap + in_bucket;
+        ev_packed[pos] = packed;
+        if (MODE != AGG_COUNT) ev_vals[pos] = v;
+      } else {
+        int opos = atomicAdd(ov_cursor, 1);
+        if (opos < ov_cap) {
+          ov_packed[opos] = packed;
+          if (MODE != AGG_COUNT) ov_vals[opos] = v;
+        } else {
+          atomicExch(error_flag, 1);
+        }
+      }
+      packed -= (uint64_t)1 << 32;
     }
   }
   for (int off = WAVE / 2; off > 0; off >>= 1) {
@@ -632,6 +657,8 @@ __global__ void k_radix_scatter_direct(
     int64_t n,
     int64_t align_ms,
     int64_t len_ms,
+    int64_t off_ms,  // accepted for launch-shape parity; the launcher
+                     // routes sliding (off < len) to the fixed variant
     int64_t ts_base,
     uint64_t mask,
     int region_bits,
@@ -709,6 +736,7 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
     int64_t n,
     int64_t align_ms,
     int64_t len_ms,
+    int64_t off_ms,  // sliding stride; == len_ms for tumbling
     int64_t ts_base,
     uint64_t mask,
     int seg_bits,
@@ -766,6 +794,7 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
     uint64_t pk[U];
     int64_t pv[U];
     int sg[U];
+    int nw[U];
     // P1: load, classify, histogram.
     for (int u = 0; u < U; ++u) {
       int64_t i = t0 + (int64_t)u * blockDim.x + threadIdx.x;
@@ -774,19 +803,35 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
       int64_t t = (int64_t)ts[i] + ts_base;
       if (t > local_max) local_max = t;
       uint64_t packed;
+      int nwin = 1;
       if (MODE == AGG_TS) {
         packed = (uint64_t)(uint32_t)keys[i];
       } else {
-        int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+        // `win_m2` is the magic divisor of the STRIDE (off_ms); for a
+        // sliding windower the event expands into every window id in
+        // [win_lo, win].
+        int64_t win = win_of(t, align_ms, off_ms, win_m2, win_maxfast);
         packed =
             ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+        if (off_ms < len_ms) {
+          nwin =
+              (int)(win - win_lo_of(t, align_ms, len_ms, off_ms)) + 1;
+        }
       }
       pk[u] = packed;
+      nw[u] = nwin;
       if (MODE == AGG_SUM) pv[u] = vals[i];
       else if (MODE == AGG_TS) pv[u] = t;
       int b = (int)region_of(mix64(packed), mask, seg_bits);
       sg[u] = b;
       atomicAdd(&lhist[b], 1);
+      // Expanded entries get their own histogram slots.
+      uint64_t p2 = packed;
+      for (int w = 1; w < nwin; ++w) {
+        p2 -= (uint64_t)1 << 32;
+        atomicAdd(
+            &lhist[(int)region_of(mix64(p2), mask, seg_bits)], 1);
+      }
     }
     __syncthreads();
     // P2: per-segment granule reservation.
@@ -813,18 +858,25 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
       }
     }
     __syncthreads();
-    // P3b: place this tile's events at their virtual positions.
+    // P3b: place this tile's (possibly window-expanded) entries at
+    // their virtual positions.
     for (int u = 0; u < U; ++u) {
-      int b = sg[u];
-      if (b < 0) continue;
-      int vpos = atomicAdd(&lofs[b], 1);
-      int grant = lhist[b];
-      if (vpos < grant) {
-        emit(b, (int64_t)lbase[b] + vpos, pk[u],
-             MODE != AGG_COUNT ? pv[u] : 0);
-      } else {
-        res[(size_t)b * SC_GRAN + (vpos - grant)] = pk[u];
-        if (MODE != AGG_COUNT) res_v[(size_t)b * SC_GRAN + (vpos - grant)] = pv[u];
+      if (sg[u] < 0) continue;
+      uint64_t p2 = pk[u];
+      for (int w = 0; w < nw[u]; ++w) {
+        int b = w == 0 ? sg[u]
+                       : (int)region_of(mix64(p2), mask, seg_bits);
+        int vpos = atomicAdd(&lofs[b], 1);
+        int grant = lhist[b];
+        if (vpos < grant) {
+          emit(b, (int64_t)lbase[b] + vpos, p2,
+               MODE != AGG_COUNT ? pv[u] : 0);
+        } else {
+          res[(size_t)b * SC_GRAN + (vpos - grant)] = p2;
+          if (MODE != AGG_COUNT)
+            res_v[(size_t)b * SC_GRAN + (vpos - grant)] = pv[u];
+        }
+        p2 -= (uint64_t)1 << 32;
       }
     }
     __syncthreads();
@@ -2055,8 +2107,11 @@ void radix_window_insert(
     int64_t mode,
     int64_t ts_base,
     int64_t region_bits,
+    int64_t off_ms = 0,  // sliding stride; 0 or == len_ms for tumbling
     std::vector<int64_t> seg_counts = {},
     std::vector<int64_t> seg_bases = {}) {
+  if (off_ms <= 0) off_ms = len_ms;
+  TORCH_CHECK(off_ms <= len_ms, "window offset must be <= length");
   check_dev(keys, torch::kInt32, "keys");
   // Segmented form: `ts` holds int32 deltas laid out as contiguous
   // per-source-rank segments (the RCCL all-to-allv wire format);
@@ -2093,6 +2148,11 @@ void radix_window_insert(
   dim3 grid(n_blocks(n, 256));
 
   ScatterKind kind = scatter_kind_env();
+  // An event expands into ceil(len/off) windows when sliding; the
+  // direct variant has no expansion loop.
+  int64_t xf = off_ms < len_ms ? (len_ms + off_ms - 1) / off_ms : 1;
+  TORCH_CHECK(xf <= 64, "sliding expansion > 64x; use a larger offset");
+  if (kind == SCAT_DIRECT && xf > 1) kind = SCAT_FIXED;
   int coarse = scatter_coarse_bits(kind);
   int seg_bits = (int)region_bits + coarse;
   int64_t nseg = nslots >> seg_bits;
@@ -2132,8 +2192,8 @@ void radix_window_insert(
       cap = ev_packed.numel() / nseg;
     }
   }
-  TORCH_CHECK(cap * nseg >= 2 * n || cap >= n,
-              "scatter buffers too small (need ~2x batch)");
+  TORCH_CHECK(cap * nseg >= 2 * n * xf || cap >= n * xf,
+              "scatter buffers too small (need ~2x batch x expansion)");
   if (mode == AGG_SUM) {
     TORCH_CHECK(ev_vals.numel() >= nseg * cap, "ev_vals too small");
   }
@@ -2167,14 +2227,14 @@ void radix_window_insert(
   }
 
   uint64_t win_m2, win_maxfast;
-  magic_div_u64(len_ms, &win_m2, &win_maxfast);
+  magic_div_u64(off_ms, &win_m2, &win_maxfast);
   auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx,
                   size_t lds) {
     hipLaunchKernelGGL(
         kern, dim3(gx), block, lds, stream,
         keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
         vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
-        sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
+        off_ms, sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
         ov_cursor.data_ptr<int32_t>(),
@@ -2193,7 +2253,7 @@ void radix_window_insert(
       // chip at the LDS-bounded occupancy, few enough to keep the
       // end-of-kernel residual padding small.
       unsigned cap_gs = env_blocks > 0 ? (unsigned)env_blocks : 512u;
-      unsigned gs = (unsigned)((sg.n + 4095) / 4096);
+      unsigned gs = (unsigned)((sg.n * xf + 4095) / 4096);
       if (gs > cap_gs) gs = cap_gs;
       if (gs < 1) gs = 1;
       if (mode == AGG_COUNT)
@@ -2281,8 +2341,11 @@ void radix_scatter_only(
     int64_t mode,
     int64_t ts_base,
     int64_t region_bits,
+    int64_t off_ms,
     std::vector<int64_t> seg_counts,
     std::vector<int64_t> seg_bases) {
+  if (off_ms <= 0) off_ms = len_ms;
+  TORCH_CHECK(off_ms <= len_ms, "window offset must be <= length");
   check_dev(keys, torch::kInt32, "keys");
   bool seg32 = !seg_counts.empty();
   bool ts32 = ts.scalar_type() == torch::kInt32;
@@ -2305,11 +2368,14 @@ void radix_scatter_only(
   uint64_t mask = (uint64_t)(nslots - 1);
   dim3 block(256);
   uint64_t win_m2, win_maxfast;
-  magic_div_u64(len_ms, &win_m2, &win_maxfast);
+  magic_div_u64(off_ms <= 0 ? len_ms : off_ms, &win_m2, &win_maxfast);
 
   // Same variant/segmentation decision as radix_agg_only (both read
   // the env once per call; one process = one configuration).
   ScatterKind kind = scatter_kind_env();
+  int64_t xf = off_ms < len_ms ? (len_ms + off_ms - 1) / off_ms : 1;
+  TORCH_CHECK(xf <= 64, "sliding expansion > 64x; use a larger offset");
+  if (kind == SCAT_DIRECT && xf > 1) kind = SCAT_FIXED;
   int coarse = scatter_coarse_bits(kind);
   int seg_bits = (int)region_bits + coarse;
   int64_t nseg = nslots >> seg_bits;
@@ -2345,7 +2411,8 @@ void radix_scatter_only(
       cap = ev_packed.numel() / nseg;
     }
   }
-  TORCH_CHECK(cap * nseg >= 2 * n || cap >= n, "scatter buffers too small");
+  TORCH_CHECK(cap * nseg >= 2 * n * xf || cap >= n * xf,
+              "scatter buffers too small");
   size_t hist_lds = (size_t)nseg * sizeof(int);
   HIP_CHECK(hipMemsetAsync(
       gcursors.data_ptr<int32_t>(), 0, (size_t)nseg * sizeof(int), stream));
@@ -2372,7 +2439,7 @@ void radix_scatter_only(
         kern, dim3(gx), block, lds,
         stream, keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
         vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
-        sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
+        off_ms, sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
         mode == AGG_SUM ? ev_vals.data_ptr<int64_t>() : nullptr,
         ov_cursor.data_ptr<int32_t>(),
@@ -2733,7 +2800,7 @@ void radix_stats_insert(
     hipLaunchKernelGGL(
         (k_radix_scatter_staged<AGG_SUM>), dim3(gs), block, staged_lds,
         stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-        vals.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
+        vals.data_ptr<int64_t>(), n, align_ms, len_ms, len_ms, ts_base, mask,
         seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
         ev_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
@@ -2745,7 +2812,7 @@ void radix_stats_insert(
     hipLaunchKernelGGL(
         k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
         keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-        vals.data_ptr<int64_t>(), n, align_ms, len_ms, ts_base, mask,
+        vals.data_ptr<int64_t>(), n, align_ms, len_ms, len_ms, ts_base, mask,
         seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
         ov_cursor.data_ptr<int32_t>(),
@@ -2924,7 +2991,7 @@ void radix_join_insert(
   hipLaunchKernelGGL(
       k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
       keys.data_ptr<int32_t>(), zeros_ts.data_ptr<int64_t>(),
-      vals.data_ptr<int64_t>(), n, 0, (int64_t)1 << 40, 0, mask,
+      vals.data_ptr<int64_t>(), n, 0, (int64_t)1 << 40, (int64_t)1 << 40, 0, mask,
       (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
       (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
       ov_cursor.data_ptr<int32_t>(),
@@ -3668,7 +3735,7 @@ int64_t native_run_window_steps(
                 (k_radix_scatter_staged<AGG_COUNT, TSV>), dim3(gs), block,
                 staged_lds, sc_stream, keys.data_ptr<int32_t>(),
                 tsptr, (const int64_t*)nullptr, n,
-                align_ms, len_ms, base, mask, seg_bits, cap,
+                align_ms, len_ms, len_ms, base, mask, seg_bits, cap,
                 gcur.data_ptr<int32_t>(), (uint64_t*)evp.data_ptr<int64_t>(),
                 (int64_t*)nullptr, ovc.data_ptr<int32_t>(),
                 (uint64_t*)ovp.data_ptr<int64_t>(), (int64_t*)nullptr,
@@ -3679,7 +3746,8 @@ int64_t native_run_window_steps(
             hipLaunchKernelGGL(
                 (k_radix_scatter_direct<AGG_COUNT, TSV>), grid, block, 0,
                 sc_stream, keys.data_ptr<int32_t>(), tsptr,
-                (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
+                (const int64_t*)nullptr, n, align_ms, len_ms, len_ms, base,
+                mask,
                 seg_bits, cap, gcur.data_ptr<int32_t>(),
                 (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
                 ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
@@ -3692,7 +3760,7 @@ int64_t native_run_window_steps(
                 (k_radix_scatter_fixed<AGG_COUNT, TSV>), grid, block,
                 2 * hist_lds,
                 sc_stream, keys.data_ptr<int32_t>(), tsptr,
-                (const int64_t*)nullptr, n, align_ms, len_ms, base, mask,
+                (const int64_t*)nullptr, n, align_ms, len_ms, len_ms, base, mask,
                 seg_bits, cap, gcur.data_ptr<int32_t>(),
                 (uint64_t*)evp.data_ptr<int64_t>(), (int64_t*)nullptr,
                 ovc.data_ptr<int32_t>(), (uint64_t*)ovp.data_ptr<int64_t>(),
@@ -4017,7 +4085,7 @@ void session_radix_insert(
     hipLaunchKernelGGL(
         kern, dim3(gx), dim3(256), lds, stream,
         keys.data_ptr<int32_t>(), tsptr, (const int64_t*)nullptr, n, 0,
-        (int64_t)1 << 40, ts_base, mask, (int)seg_bits, cap,
+        (int64_t)1 << 40, (int64_t)1 << 40, ts_base, mask, (int)seg_bits, cap,
         gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(),
         ev_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),

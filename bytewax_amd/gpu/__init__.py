@@ -291,9 +291,14 @@ class WindowAggState:
         if self.off_ms > len_ms:
             msg = "window offset must be <= length"
             raise ValueError(msg)
-        if self.off_ms < len_ms and (radix or dedup):
-            msg = "sliding windows require the plain single-pass path"
+        if self.off_ms < len_ms and dedup:
+            msg = "sliding windows are incompatible with the dedup path"
             raise ValueError(msg)
+        # Sliding expansion factor: each event lands in this many
+        # windows; radix scatter buffers scale by it.
+        self.expand = (
+            -(-len_ms // self.off_ms) if self.off_ms < len_ms else 1
+        )
         self.mode = mode
         self.dedup = dedup
         self.max_ts_host = 0  # watermark if batches carry max_ts
@@ -397,7 +402,8 @@ class WindowAggState:
         # 8-event reservation granularity always has working capacity
         # even for tiny states/batches.
         per_region = max(
-            32, -(-max_batch * 5 // 2) // self.n_regions + 1
+            32,
+            -(-max_batch * self.expand * 5 // 2) // self.n_regions + 1,
         )
         total = per_region * self.n_regions
         self.rx_max_batch = max_batch
@@ -455,7 +461,8 @@ class WindowAggState:
             return
         if self.radix and len(batch) > self.rx_max_batch:
             # Scatter buffers grow to fit the largest batch seen
-            # (exchange-received batches vary in size).
+            # (exchange-received batches vary in size; sliding
+            # expansion is applied inside _alloc_rx).
             self._alloc_rx(int(len(batch) * 5 // 4))
         import torch
 
@@ -496,6 +503,7 @@ class WindowAggState:
                 self.mode,
                 batch.ts_base,
                 self.region_bits,
+                self.off_ms,
                 [],
                 [],
             )
@@ -614,6 +622,7 @@ class WindowAggState:
                 self.mode,
                 ts_base,
                 self.region_bits,
+                self.off_ms,
                 list(seg_counts),
                 [int(b) for b in seg_bases],
             )
@@ -666,6 +675,7 @@ class WindowAggState:
             self.mode,
             0,
             self.region_bits,
+            self.off_ms,
             lz.seg_counts,
             lz.seg_bases,
         )

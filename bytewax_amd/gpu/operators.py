@@ -609,7 +609,7 @@ def keyed_window_agg(
             slots_pow=slots_pow,
             dedup=dedup and not sliding,
             out_cap=out_cap,
-            radix=radix and dev.type != "cpu" and not sliding,
+            radix=radix and dev.type != "cpu",
             region_bits=region_bits,
             off_ms=off_ms,
         )

@@ -915,7 +915,6 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
 
         dev = batch.keys.device
         on_gpu = dev.type != "cpu"
-        sliding = self.spec.off_ms < self.spec.len_ms
         self.state = WindowAggState(
             dev,
             self.spec.align_ms,
@@ -923,7 +922,7 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
             AGG_COUNT if self.spec.mode == "count" else AGG_SUM,
             slots_pow=22 if on_gpu else 16,
             out_cap=1 << (22 if on_gpu else 16),
-            radix=on_gpu and not sliding,
+            radix=on_gpu,
             off_ms=self.spec.off_ms,
         )
         if self._resume is not None:

@@ -392,3 +392,43 @@ def test_close_due_respects_wait_allowance():
     )
     out = st.close_due(wait_ms=20_000)  # watermark 90s - 20s >= 60s
     assert out is not None and out.vals.tolist() == [2]
+
+
+@pytest.mark.gpu
+def test_gpu_sliding_radix_matches_cpu_twin():
+    """Round-2: sliding windows on the RADIX path (scatter expands
+    each event into its overlapped windows) vs the CPU twin."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    from bytewax_amd.gpu import AGG_SUM
+
+    align_ms = _ms(ALIGN)
+    g = torch.Generator().manual_seed(17)
+    n = 200_000
+    keys = torch.randint(0, 5000, (n,), dtype=torch.int32, generator=g)
+    ts = align_ms + torch.randint(
+        0, 300_000, (n,), dtype=torch.int64, generator=g
+    )
+    vals = torch.randint(0, 50, (n,), dtype=torch.int64, generator=g)
+
+    def run(device, radix):
+        state = WindowAggState(
+            torch.device(device), align_ms, 60_000, AGG_SUM,
+            slots_pow=22, out_cap=1 << 22, radix=radix,
+            off_ms=20_000, max_batch=n,
+        )
+        state.insert(
+            RecordBatch(
+                keys.to(device), ts.to(device), vals.to(device),
+                max_ts=int(ts.max()),
+            )
+        )
+        out = state.close_all()
+        return sorted(
+            zip(
+                out.keys.cpu().tolist(), out.ts.cpu().tolist(),
+                out.vals.cpu().tolist(),
+            )
+        )
+
+    assert run("cuda:0", True) == run("cpu", False)

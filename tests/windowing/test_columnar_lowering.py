@@ -167,3 +167,93 @@ def test_count_window_lowering_gpu():
         assert wid == COLUMNAR_WINDOW_ID
         total += int(rb.vals.sum().item())
     assert total == len(events)
+
+
+def _sliding_run(events, folder, device="cpu", batch=41):
+    from bytewax_amd.operators.windowing import SlidingWindower
+
+    batches = []
+    for i in range(0, len(events), batch):
+        chunk = events[i : i + batch]
+        batches.append(
+            RecordBatch(
+                torch.tensor([k for k, _, _ in chunk], dtype=torch.int32,
+                             device=device),
+                torch.tensor([ms for _, ms, _ in chunk], dtype=torch.int64,
+                             device=device),
+                torch.tensor([v for _, _, v in chunk], dtype=torch.int64,
+                             device=device),
+            )
+        )
+    out = []
+    flow = Dataflow("columnar_sliding")
+    s = op.input("inp", flow, TestingSource(batches))
+    keyed = op.key_on("k", s, lambda b: "shard-0")
+    clock = EventClock(
+        ts_getter=lambda it: it, wait_for_system_duration=timedelta(0)
+    )
+    wo = w.fold_window(
+        "fw", keyed, clock,
+        SlidingWindower(
+            align_to=ALIGN, length=WINDOW, offset=timedelta(seconds=20)
+        ),
+        int, folder, lambda a, b: a + b,
+    )
+    op.output("out", wo.down, TestingSink(out))
+    run_main(flow)
+    res = {}
+    for _key, (wid, rb) in out:
+        assert wid == COLUMNAR_WINDOW_ID
+        for k, ms, v in zip(
+            rb.keys.cpu().tolist(), rb.ts.cpu().tolist(),
+            rb.vals.cpu().tolist(),
+        ):
+            res[(k, ms)] = res.get((k, ms), 0) + v
+    return res
+
+
+def _sliding_host(events, folder):
+    from bytewax_amd.operators.windowing import SlidingWindower
+
+    items = [
+        (datetime.fromtimestamp(ms / 1000, tz=timezone.utc), k, v)
+        for k, ms, v in events
+    ]
+    out = []
+    flow = Dataflow("host_sliding")
+    s = op.input("inp", flow, TestingSource(items))
+    keyed = op.key_on("k", s, lambda it: str(it[1]))
+    clock = EventClock(
+        ts_getter=lambda it: it[0], wait_for_system_duration=timedelta(0)
+    )
+    wo = w.fold_window(
+        "fw", keyed, clock,
+        SlidingWindower(
+            align_to=ALIGN, length=WINDOW, offset=timedelta(seconds=20)
+        ),
+        int, folder, lambda a, b: a + b,
+    )
+    op.output("out", wo.down, TestingSink(out))
+    run_main(flow)
+    off_ms = 20_000
+    return {
+        (int(key), ALIGN_MS + wid * off_ms): acc
+        for key, (wid, acc) in out
+    }
+
+
+def test_sliding_lowering_cpu_twin_matches_host():
+    events = _mk_events(600)
+    folder = w.device_sum(lambda it: it[2])
+    assert _sliding_run(events, folder) == _sliding_host(events, folder)
+
+
+@pytest.mark.gpu
+def test_sliding_lowering_gpu_matches_host():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    events = _mk_events(20_000, vocab=300)
+    folder = w.device_sum(lambda it: it[2])
+    assert _sliding_run(events, folder, "cuda:0") == _sliding_host(
+        events, folder
+    )
