@@ -93,6 +93,8 @@ __all__ = [
     "collect_window",
     "count_window",
     "device_count",
+    "device_max",
+    "device_min",
     "device_sum",
     "fold_window",
     "join_window",
@@ -854,6 +856,24 @@ def device_count() -> DeviceFoldable:
     return COUNT_FOLD
 
 
+def device_min(value_getter: Callable = _identity) -> DeviceFoldable:
+    """A `fold_window` min fold that lowers to the fused stats
+    kernels over RecordBatch streams (tumbling windows); pair with a
+    builder like ``lambda: float("inf")`` for the host path."""
+    return DeviceFoldable(
+        "min", lambda acc, value: min(acc, value_getter(value))
+    )
+
+
+def device_max(value_getter: Callable = _identity) -> DeviceFoldable:
+    """A `fold_window` max fold that lowers to the fused stats
+    kernels over RecordBatch streams (tumbling windows); pair with a
+    builder like ``lambda: -float("inf")`` for the host path."""
+    return DeviceFoldable(
+        "max", lambda acc, value: max(acc, value_getter(value))
+    )
+
+
 @dataclass
 class _ColumnarSpec:
     """Device-lowering parameters resolved from clock + windower."""
@@ -879,6 +899,10 @@ class _ColumnarSpec:
             len_ms = int(windower.length.total_seconds() * 1000)
             off_ms = int(windower.offset.total_seconds() * 1000)
         else:
+            return None
+        if mode in ("min", "max") and off_ms != len_ms:
+            # The stats table is tumbling-only; sliding min/max stays
+            # on the host path.
             return None
         align_ms = int(
             (windower.align_to - _EPOCH_UTC).total_seconds() * 1000
@@ -915,6 +939,20 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
 
         dev = batch.keys.device
         on_gpu = dev.type != "cpu"
+        if self.spec.mode in ("min", "max"):
+            from ..gpu.state import StatsAggState
+
+            self.state = StatsAggState(
+                dev,
+                self.spec.align_ms,
+                self.spec.len_ms,
+                slots_pow=22 if on_gpu else 16,
+                out_cap=1 << (22 if on_gpu else 16),
+            )
+            if self._resume is not None:
+                self.state.restore_from_host(self._resume)
+                self._resume = None
+            return
         self.state = WindowAggState(
             dev,
             self.spec.align_ms,
@@ -929,12 +967,38 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
             self.state.restore_from_host(self._resume)
             self._resume = None
 
+    def _stats_close(self, horizon) -> Optional[Any]:
+        """Extract closed (key, window) stats rows as a RecordBatch of
+        the fold's column (min or max)."""
+        from ..gpu import RecordBatch
+
+        if horizon is not None and horizon <= self.state.closed_horizon:
+            return None
+        cols = self.state.extract(horizon, clear=True)
+        if horizon is not None:
+            self.state.closed_horizon = horizon
+        if cols is None:
+            return None
+        wins = cols["wins"].to(dtype=cols["cnt"].dtype)
+        return RecordBatch(
+            cols["keys"],
+            wins * self.spec.len_ms + self.spec.align_ms,
+            cols[self.spec.mode],
+        )
+
     def on_batch(self, values) -> Tuple[Iterable, bool]:
         events: List = []
         for batch in values:
             self._ensure(batch)
             self.state.insert(batch)
-        closed = self.state.close_due(self.spec.wait_ms)
+        if self.spec.mode in ("min", "max"):
+            wm = self.state.max_ts_host
+            horizon = (
+                wm - self.spec.wait_ms - self.spec.align_ms
+            ) // self.spec.len_ms
+            closed = self._stats_close(horizon)
+        else:
+            closed = self.state.close_due(self.spec.wait_ms)
         if closed is not None:
             events.append((COLUMNAR_WINDOW_ID, "E", closed))
         return (events, False)
@@ -945,7 +1009,10 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
     def on_eof(self) -> Tuple[Iterable, bool]:
         if self.state is None:
             return ([], True)
-        final = self.state.close_all()
+        if self.spec.mode in ("min", "max"):
+            final = self._stats_close(None)
+        else:
+            final = self.state.close_all()
         if final is None:
             return ([], True)
         return ([(COLUMNAR_WINDOW_ID, "E", final)], True)

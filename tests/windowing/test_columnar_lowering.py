@@ -257,3 +257,56 @@ def test_sliding_lowering_gpu_matches_host():
     assert _sliding_run(events, folder, "cuda:0") == _sliding_host(
         events, folder
     )
+
+
+def _minmax_host(events, folder, builder):
+    items = [
+        (datetime.fromtimestamp(ms / 1000, tz=timezone.utc), k, v)
+        for k, ms, v in events
+    ]
+    out = []
+    flow = Dataflow("host_minmax")
+    s = op.input("inp", flow, TestingSource(items))
+    keyed = op.key_on("k", s, lambda it: str(it[1]))
+    clock = EventClock(
+        ts_getter=lambda it: it[0], wait_for_system_duration=timedelta(0)
+    )
+    wo = w.fold_window(
+        "fw", keyed, clock,
+        TumblingWindower(align_to=ALIGN, length=WINDOW),
+        builder, folder, lambda a, b: folder(a, (None, None, b)),
+    )
+    op.output("out", wo.down, TestingSink(out))
+    run_main(flow)
+    win_len = int(WINDOW.total_seconds() * 1000)
+    return {
+        (int(key), ALIGN_MS + wid * win_len): acc
+        for key, (wid, acc) in out
+    }
+
+
+def test_fold_window_min_max_lowering_cpu_twin():
+    events = _mk_events(700)
+    for name, builder in (("min", lambda: 10**9), ("max", lambda: -1)):
+        folder = (
+            w.device_min(lambda it: it[2])
+            if name == "min"
+            else w.device_max(lambda it: it[2])
+        )
+        host = {
+            k: v
+            for k, v in _minmax_host(events, folder, builder).items()
+        }
+        col = _columnar_run(events, folder)
+        assert col == host, name
+
+
+@pytest.mark.gpu
+def test_fold_window_min_max_lowering_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    events = _mk_events(20_000, vocab=400)
+    folder = w.device_max(lambda it: it[2])
+    assert _columnar_run(events, folder, "cuda:0") == _minmax_host(
+        events, folder, lambda: -1
+    )
