@@ -1489,6 +1489,7 @@ __global__ void k_join_region(
   int full = (1 << n_sides) - 1;
   for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
     uint64_t packed = ev_packed[start + j];
+    if (packed == EMPTY_SLOT) continue;  // staged-scatter pad
     uint64_t slot = find_slot_r(tkeys, mask, region_bits, packed);
     if (slot == ~0ULL) {
       atomicExch(error_flag, 1);
@@ -2978,6 +2979,19 @@ void radix_join_insert(
   TORCH_CHECK(region_bits > 0 && region_bits <= 12, "bad region_bits");
   int64_t nb = nslots >> region_bits;
   int64_t cap = ev_packed.numel() / nb;
+  // Staged (line-granule) scatter when the per-region capacity
+  // supports it; the fixed variant otherwise.
+  ScatterKind kind = scatter_kind_env() == SCAT_FIXED ? SCAT_FIXED
+                                                      : SCAT_STAGED;
+  size_t staged_lds =
+      (size_t)nb * SC_GRAN * 8 * 2 + 4 * (size_t)nb * sizeof(int);
+  if (kind == SCAT_STAGED) {
+    cap &= ~(int64_t)(SC_GRAN - 1);
+    if (cap < SC_GRAN || staged_lds > 160 * 1024) {
+      kind = SCAT_FIXED;
+      cap = ev_packed.numel() / nb;
+    }
+  }
   TORCH_CHECK(cap * nb >= 2 * n || cap >= n, "scatter buffers too small");
   if (n == 0) return;
   auto stream = at::hip::getCurrentHIPStream();
@@ -2988,17 +3002,36 @@ void radix_join_insert(
   ov_cursor.zero_();
   size_t hist_lds = (size_t)nb * sizeof(int);
   // Key-only packing: ts == 0, align 0, huge window -> win = 0.
-  hipLaunchKernelGGL(
-      k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
-      keys.data_ptr<int32_t>(), zeros_ts.data_ptr<int64_t>(),
-      vals.data_ptr<int64_t>(), n, 0, (int64_t)1 << 40, (int64_t)1 << 40, 0, mask,
-      (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
-      (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
-      ov_cursor.data_ptr<int32_t>(),
-      (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
-      ov_packed.numel(),
-      (unsigned long long*)max_ts_scratch.data_ptr<int64_t>(),
-      error_flag.data_ptr<int32_t>(), (uint64_t)1 << 24, ~(uint64_t)0);
+  if (kind == SCAT_STAGED) {
+    unsigned gs = (unsigned)((n + 4095) / 4096);
+    if (gs > 512) gs = 512;
+    if (gs < 1) gs = 1;
+    hipLaunchKernelGGL(
+        (k_radix_scatter_staged<AGG_SUM>), dim3(gs), block, staged_lds,
+        stream, keys.data_ptr<int32_t>(), zeros_ts.data_ptr<int64_t>(),
+        vals.data_ptr<int64_t>(), n, 0, (int64_t)1 << 40,
+        (int64_t)1 << 40, 0, mask,
+        (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
+        ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
+        ov_packed.numel(),
+        (unsigned long long*)max_ts_scratch.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(), (uint64_t)1 << 24, ~(uint64_t)0);
+  } else {
+    hipLaunchKernelGGL(
+        k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
+        keys.data_ptr<int32_t>(), zeros_ts.data_ptr<int64_t>(),
+        vals.data_ptr<int64_t>(), n, 0, (int64_t)1 << 40, (int64_t)1 << 40,
+        0, mask,
+        (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
+        ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
+        ov_packed.numel(),
+        (unsigned long long*)max_ts_scratch.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(), (uint64_t)1 << 24, ~(uint64_t)0);
+  }
   hipLaunchKernelGGL(
       k_join_region, dim3((unsigned)nb), block, 0, stream,
       (const uint64_t*)ev_packed.data_ptr<int64_t>(),

@@ -366,11 +366,18 @@ class HashJoinState:
         device,
         slots_pow: int = 20,
         out_cap: int = 1 << 20,
-        radix: bool = False,  # measured slower than direct at 1M keys
+        radix: Optional[bool] = None,
         region_bits: int = 11,
     ):
+        import os
+
         import torch
 
+        if radix is None:
+            # Direct atomics measured faster with the r1 fixed
+            # scatter; the staged scatter changed the economics —
+            # BYTEWAX_JOIN_RADIX A/Bs it (see profiles/).
+            radix = os.environ.get("BYTEWAX_JOIN_RADIX", "0") == "1"
         self.device = device
         self.cpu = device.type == "cpu"
         if self.cpu:
