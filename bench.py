@@ -320,6 +320,16 @@ def main():
                         "radix": args.radix,
                         "radix_v2": args.radix_v2,
                         "p99_step_ms": p99_ms,
+                        # How p99_step_ms was measured: the dataflow
+                        # engine reports steady-state poll-gap deltas
+                        # (latency under load); the native loop runs
+                        # isolated synced single-batch probes after
+                        # the throughput region.
+                        "latency_method": (
+                            "poll-gap under load"
+                            if engine == "python"
+                            else "isolated synced single-step probes"
+                        ),
                         "closed_window_rows": closed_rows,
                     },
                 }
