@@ -119,3 +119,23 @@ def test_events_to_parquet_example(tmp_path):
         import shutil
 
         shutil.rmtree(out_dir, ignore_errors=True)
+
+
+def test_basic_example():
+    res = _run([sys.executable, "examples/basic.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "<dance>" in res.stdout.decode()
+
+
+def test_batch_operator_example():
+    res = _run([sys.executable, "examples/batch_operator.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "avg batch:" in res.stdout.decode()
+
+
+def test_event_time_processing_example():
+    res = _run([sys.executable, "examples/event_time_processing.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    out = res.stdout.decode()
+    assert "('LATE'" in out
+    assert "('s1', (1, 19.0))" in out
