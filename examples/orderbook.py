@@ -33,7 +33,7 @@ class SyntheticL2Partition(StatefulSourcePartition):
     def __init__(self, product: str, resume_i: Optional[int]):
         self.product = product
         self.i = resume_i or 0
-        self.rng = random.Random(hash(product) & 0xFFFF)
+        self.rng = random.Random(sum(product.encode()))
         mid = MID[product]
         self._msgs = []
         bids = [[f"{mid - j - 1:.2f}", f"{self.rng.random():.4f}"] for j in range(5)]
@@ -105,7 +105,9 @@ class OrderBook:
             else:
                 book[price] = size
 
-    def summarize(self) -> Summary:
+    def summarize(self) -> Optional[Summary]:
+        if not self.bids or not self.asks:
+            return None  # one side fully consumed; no quote
         bid = max(self.bids)
         ask = min(self.asks)
         return Summary(bid, self.bids[bid], ask, self.asks[ask], ask - bid)
@@ -124,7 +126,7 @@ books = op.stateful_map("book", feed, maintain)
 
 def wide_spread(prod_summary):
     _product, s = prod_summary
-    return s.spread / s.ask_price > 0.0001
+    return s is not None and s.spread / s.ask_price > 0.0001
 
 
 wide = op.filter("wide_spread", books, wide_spread)
