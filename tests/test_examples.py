@@ -139,3 +139,16 @@ def test_event_time_processing_example():
     out = res.stdout.decode()
     assert "('LATE'" in out
     assert "('s1', (1, 19.0))" in out
+
+
+def test_poll_and_split_example():
+    res = _run([sys.executable, "examples/poll_and_split.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    out = res.stdout.decode()
+    assert "'type': 'story'" in out and "'type': 'comment'" in out
+
+
+def test_wikistream_example():
+    res = _run([sys.executable, "examples/wikistream.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "edits" in res.stdout.decode()
