@@ -859,50 +859,24 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
     }
     __syncthreads();
     // P3b: place this tile's (possibly window-expanded) entries at
-    // their virtual positions.  Fast path (no expansion): issue all
-    // U position atomics back-to-back, THEN consume — the PMC showed
-    // waves waiting ~12x their busy cycles on the per-event
-    // ds_add_rtn round trip; batching the issues pays one LDS
-    // latency for the whole register tile instead of U.
-    bool any_expand = false;
-    for (int u = 0; u < U; ++u) any_expand |= nw[u] > 1;
-    if (!any_expand) {
-      int vpos[U];
-      for (int u = 0; u < U; ++u) {
-        if (sg[u] >= 0) vpos[u] = atomicAdd(&lofs[sg[u]], 1);
-      }
-      for (int u = 0; u < U; ++u) {
-        int b = sg[u];
-        if (b < 0) continue;
+    // their virtual positions.
+    for (int u = 0; u < U; ++u) {
+      if (sg[u] < 0) continue;
+      uint64_t p2 = pk[u];
+      for (int w = 0; w < nw[u]; ++w) {
+        int b = w == 0 ? sg[u]
+                       : (int)region_of(mix64(p2), mask, seg_bits);
+        int vpos = atomicAdd(&lofs[b], 1);
         int grant = lhist[b];
-        if (vpos[u] < grant) {
-          emit(b, (int64_t)lbase[b] + vpos[u], pk[u],
+        if (vpos < grant) {
+          emit(b, (int64_t)lbase[b] + vpos, p2,
                MODE != AGG_COUNT ? pv[u] : 0);
         } else {
-          res[(size_t)b * SC_GRAN + (vpos[u] - grant)] = pk[u];
+          res[(size_t)b * SC_GRAN + (vpos - grant)] = p2;
           if (MODE != AGG_COUNT)
-            res_v[(size_t)b * SC_GRAN + (vpos[u] - grant)] = pv[u];
+            res_v[(size_t)b * SC_GRAN + (vpos - grant)] = pv[u];
         }
-      }
-    } else {
-      for (int u = 0; u < U; ++u) {
-        if (sg[u] < 0) continue;
-        uint64_t p2 = pk[u];
-        for (int w = 0; w < nw[u]; ++w) {
-          int b = w == 0 ? sg[u]
-                         : (int)region_of(mix64(p2), mask, seg_bits);
-          int vpos = atomicAdd(&lofs[b], 1);
-          int grant = lhist[b];
-          if (vpos < grant) {
-            emit(b, (int64_t)lbase[b] + vpos, p2,
-                 MODE != AGG_COUNT ? pv[u] : 0);
-          } else {
-            res[(size_t)b * SC_GRAN + (vpos - grant)] = p2;
-            if (MODE != AGG_COUNT)
-              res_v[(size_t)b * SC_GRAN + (vpos - grant)] = pv[u];
-          }
-          p2 -= (uint64_t)1 << 32;
-        }
+        p2 -= (uint64_t)1 << 32;
       }
     }
     __syncthreads();
@@ -989,14 +963,13 @@ __global__ __launch_bounds__(1024) void k_radix_agg(
   int cnt = counts[b];
   if (clamp_cap > 0 && cnt > (int)clamp_cap) cnt = (int)clamp_cap;
   int start = offsets[b];
-  // Chunked probe: batch 8 events' global loads + first LDS reads,
-  // wait once, then resolve.  The common case (slot already holds the
-  // key) is a fire-and-forget LDS atomicAdd; slot keys only ever
-  // transition EMPTY -> key inside one launch, so a stale first read
-  // is safe (mismatch or EMPTY falls to the fresh-probing slow path).
-  const int CH = 8;
-  int stride = blockDim.x;
-  auto slow = [&](uint64_t packed, unsigned long long inc, int lh) {
+  for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
+    uint64_t packed = ev_packed[start + j];
+    if (packed == EMPTY_SLOT) continue;  // staged-scatter pad
+    unsigned long long inc =
+        (MODE == AGG_SUM) ? (unsigned long long)ev_vals[start + j] : 1ULL;
+    uint64_t h64 = mix64(packed);
+    int lh = (int)((h64 >> 32) & (region - 1));
     bool done = false;
     for (int p = 0; p < region; ++p) {
       uint64_t cur = lkeys[lh];
@@ -1021,39 +994,6 @@ __global__ __launch_bounds__(1024) void k_radix_agg(
       // at least as full): flush straight to the global region.
       if (!hash_add(tkeys, tvals, mask, region_bits, packed, inc)) {
         atomicExch(error_flag, 1);
-      }
-    }
-  };
-  for (int j0 = threadIdx.x; j0 < cnt; j0 += stride * CH) {
-    uint64_t pk[CH];
-    unsigned long long inc[CH];
-    int lh[CH];
-    bool act[CH];
-    for (int c = 0; c < CH; ++c) {
-      int j = j0 + c * stride;
-      act[c] = j < cnt;
-      if (act[c]) {
-        pk[c] = ev_packed[start + j];
-        if (pk[c] == EMPTY_SLOT) {
-          act[c] = false;  // staged-scatter pad
-        } else {
-          inc[c] = (MODE == AGG_SUM)
-                       ? (unsigned long long)ev_vals[start + j]
-                       : 1ULL;
-          lh[c] = (int)((mix64(pk[c]) >> 32) & (region - 1));
-        }
-      }
-    }
-    uint64_t cur[CH];
-    for (int c = 0; c < CH; ++c) {
-      if (act[c]) cur[c] = lkeys[lh[c]];
-    }
-    for (int c = 0; c < CH; ++c) {
-      if (!act[c]) continue;
-      if (cur[c] == pk[c]) {
-        atomicAdd(&lvals[lh[c]], inc[c]);
-      } else {
-        slow(pk[c], inc[c], lh[c]);
       }
     }
   }
