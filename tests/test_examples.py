@@ -152,3 +152,15 @@ def test_wikistream_example():
     res = _run([sys.executable, "examples/wikistream.py"])
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     assert "edits" in res.stdout.decode()
+
+
+def test_custom_metrics_example():
+    res = _run([sys.executable, "examples/custom_metrics.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "next_batch_delay_seconds{" in res.stdout.decode()
+
+
+def test_tracing_otlp_example():
+    res = _run([sys.executable, "examples/tracing_otlp.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "bytes of spans" in res.stdout.decode()
