@@ -164,3 +164,17 @@ def test_tracing_otlp_example():
     res = _run([sys.executable, "examples/tracing_otlp.py"])
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     assert "bytes of spans" in res.stdout.decode()
+
+
+def test_apriori_example():
+    res = _run([sys.executable, "examples/apriori.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    out = res.stdout.decode()
+    assert "('milk', 4)" in out and "('bread,milk', 3)" in out
+
+
+def test_partials_example():
+    res = _run([sys.executable, "examples/partials.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    # range(5) + 5 chained add_one steps -> 5..9
+    assert "partials.insp: 9" in res.stdout.decode()
