@@ -419,6 +419,31 @@ def stateful(
     """Advanced generic stateful operator (per-item variant).
 
     Subclass {py:obj}`StatefulLogic` to define its behavior.
+
+    Example:
+
+    >>> class RunningMax(op.StatefulLogic):
+    ...     def __init__(self, resume):
+    ...         self.mx = resume if resume is not None else 0
+    ...     def on_item(self, v):
+    ...         self.mx = max(self.mx, v)
+    ...         return ((self.mx,), op.StatefulLogic.RETAIN)
+    ...     def on_notify(self):
+    ...         return ((), op.StatefulLogic.RETAIN)
+    ...     def on_eof(self):
+    ...         return ((), op.StatefulLogic.DISCARD)
+    ...     def notify_at(self):
+    ...         return None
+    ...     def snapshot(self):
+    ...         return self.mx
+    >>> flow = Dataflow("stateful_eg")
+    >>> s = op.input("inp", flow, TestingSource([("a", 2), ("a", 1)]))
+    >>> s = op.stateful("max", s, RunningMax)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 2)
+    ('a', 2)
+    
     """
 
     def shim_builder(resume_state: Optional[S]) -> _StatefulLogicShim[V, W, S]:
@@ -1375,6 +1400,19 @@ def join(
     >>> op.output("out", s, StdOutSink())
     >>> run_main(flow)
     ('1', ('alice', 'a@x.io'))
+
+    With ``emit_mode="running"`` every update emits the current
+    (possibly incomplete) tuple:
+
+    >>> flow = Dataflow("join_running_eg")
+    >>> a = op.input("a", flow, TestingSource([("1", "x1"), ("1", "x2")]))
+    >>> b = op.input("b", flow, TestingSource([("1", "y1")]))
+    >>> s = op.join("join", a, b, emit_mode="running")
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('1', ('x1', None))
+    ('1', ('x1', 'y1'))
+    ('1', ('x2', 'y1'))
     """
     if insert_mode not in typing.get_args(JoinInsertMode):
         msg = f"unknown join insert mode {insert_mode!r}"
