@@ -262,6 +262,7 @@ def main():
             device=device,
             exchange=(world > 1),
             radix=args.radix and on_gpu,
+            dedup=args.dedup,
         )
         op.output("out", agg, CollectCountsSink(out))
 

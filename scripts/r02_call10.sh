@@ -32,13 +32,16 @@ vocab = [f"word-{i}" for i in range(200_000)]
 N, B = 4_000_000, 10
 packed = pack_strings([vocab[rng.randrange(len(vocab))] for _ in range(N)])
 
+import numpy as np
+ts_tmpl = (np.arange(N, dtype=np.int64) % 1000)
+
 class Part(StatelessSourcePartition):
     def __init__(self):
         self.i = 0
     def next_batch(self):
         if self.i >= B:
             raise StopIteration()
-        ts = [ALIGN_MS + self.i * 1000 + j % 1000 for j in range(N)]
+        ts = ts_tmpl + (ALIGN_MS + self.i * 1000)
         self.i += 1
         return [(packed, ts)]
 
