@@ -2734,7 +2734,8 @@ void radix_stats_insert(
     int64_t ts_base,
     int64_t region_bits) {
   check_dev(keys, torch::kInt32, "keys");
-  check_dev(ts, torch::kInt64, "ts");
+  bool ts32 = ts.scalar_type() == torch::kInt32;
+  if (!ts32) check_dev(ts, torch::kInt64, "ts");
   check_dev(vals, torch::kInt64, "vals");
   int64_t n = keys.numel();
   int64_t nslots = tkeys.numel();
@@ -2794,34 +2795,43 @@ void radix_stats_insert(
   size_t hist_lds = (size_t)nseg * sizeof(int);
   uint64_t win_m2, win_maxfast;
   magic_div_u64(len_ms, &win_m2, &win_maxfast);
-  if (kind == SCAT_STAGED) {
-    unsigned gs = (unsigned)((n + 4095) / 4096);
-    if (gs > 512) gs = 512;
-    if (gs < 1) gs = 1;
-    hipLaunchKernelGGL(
-        (k_radix_scatter_staged<AGG_SUM>), dim3(gs), block, staged_lds,
-        stream, keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-        vals.data_ptr<int64_t>(), n, align_ms, len_ms, len_ms, ts_base, mask,
-        seg_bits, cap, gcursors.data_ptr<int32_t>(),
-        (uint64_t*)ev_packed.data_ptr<int64_t>(),
-        ev_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
-        (uint64_t*)ov_packed.data_ptr<int64_t>(),
-        ov_vals.data_ptr<int64_t>(), ov_packed.numel(),
-        (unsigned long long*)max_ts.data_ptr<int64_t>(),
-        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
-  } else {
-    hipLaunchKernelGGL(
-        k_radix_scatter_fixed<AGG_SUM>, grid, block, 2 * hist_lds, stream,
-        keys.data_ptr<int32_t>(), ts.data_ptr<int64_t>(),
-        vals.data_ptr<int64_t>(), n, align_ms, len_ms, len_ms, ts_base, mask,
-        seg_bits, cap, gcursors.data_ptr<int32_t>(),
-        (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
-        ov_cursor.data_ptr<int32_t>(),
-        (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
-        ov_packed.numel(),
-        (unsigned long long*)max_ts.data_ptr<int64_t>(),
-        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
-  }
+  auto scat_any = [&](auto tsptr) {
+    using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
+    if (kind == SCAT_STAGED) {
+      unsigned gs = (unsigned)((n + 4095) / 4096);
+      if (gs > 512) gs = 512;
+      if (gs < 1) gs = 1;
+      hipLaunchKernelGGL(
+          (k_radix_scatter_staged<AGG_SUM, TSV>), dim3(gs), block,
+          staged_lds,
+          stream, keys.data_ptr<int32_t>(), tsptr,
+          vals.data_ptr<int64_t>(), n, align_ms, len_ms, len_ms, ts_base,
+          mask,
+          seg_bits, cap, gcursors.data_ptr<int32_t>(),
+          (uint64_t*)ev_packed.data_ptr<int64_t>(),
+          ev_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
+          (uint64_t*)ov_packed.data_ptr<int64_t>(),
+          ov_vals.data_ptr<int64_t>(), ov_packed.numel(),
+          (unsigned long long*)max_ts.data_ptr<int64_t>(),
+          error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+    } else {
+      hipLaunchKernelGGL(
+          (k_radix_scatter_fixed<AGG_SUM, TSV>), grid, block,
+          2 * hist_lds, stream,
+          keys.data_ptr<int32_t>(), tsptr,
+          vals.data_ptr<int64_t>(), n, align_ms, len_ms, len_ms, ts_base,
+          mask,
+          seg_bits, cap, gcursors.data_ptr<int32_t>(),
+          (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
+          ov_cursor.data_ptr<int32_t>(),
+          (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
+          ov_packed.numel(),
+          (unsigned long long*)max_ts.data_ptr<int64_t>(),
+          error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+    }
+  };
+  if (ts32) scat_any(ts.data_ptr<int32_t>());
+  else scat_any(ts.data_ptr<int64_t>());
 
   auto offsets = at::arange(
       nseg, at::TensorOptions().dtype(at::kInt).device(keys.device()));

@@ -96,8 +96,14 @@ class StatsAggState:
         if batch.vals is None:
             msg = "stats aggregation requires a `vals` column"
             raise ValueError(msg)
-        if not self.cpu and batch.ts.dtype != torch.int64:
-            # Stats kernels take absolute int64 timestamps.
+        if (
+            not self.cpu
+            and not self.radix
+            and batch.ts.dtype != torch.int64
+        ):
+            # The plain (non-radix) stats kernel takes absolute int64
+            # timestamps; the radix scatter consumes int32 templates
+            # natively.
             batch = RecordBatch(
                 batch.keys,
                 batch.ts.to(torch.int64),
