@@ -178,3 +178,9 @@ def test_partials_example():
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     # range(5) + 5 chained add_one steps -> 5..9
     assert "partials.insp: 9" in res.stdout.decode()
+
+
+def test_split_demo_example():
+    res = _run([sys.executable, "examples/split_demo.py"])
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
+    assert "_value" in res.stdout.decode()
