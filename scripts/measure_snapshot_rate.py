@@ -17,8 +17,8 @@ def main():
     dev = torch.device("cuda:0")
     n_cells = 16_000_000
     st = WindowAggState(
-        dev, 0, 60_000, AGG_COUNT, slots_pow=26, out_cap=n_cells + 1024,
-        radix=True, max_batch=n_cells,
+        dev, 0, 60_000, AGG_COUNT, slots_pow=25, out_cap=n_cells + 1024,
+        radix=True, region_bits=12, max_batch=n_cells,
     )
     g = torch.Generator(device="cuda").manual_seed(9)
     keys = torch.randint(
@@ -37,8 +37,8 @@ def main():
     nbytes = rows * 16  # key4 + win4 + val8
 
     st2 = WindowAggState(
-        dev, 0, 60_000, AGG_COUNT, slots_pow=26, out_cap=n_cells + 1024,
-        radix=True, max_batch=n_cells,
+        dev, 0, 60_000, AGG_COUNT, slots_pow=25, out_cap=n_cells + 1024,
+        radix=True, region_bits=12, max_batch=n_cells,
     )
     torch.cuda.synchronize()
     t0 = time.perf_counter()
