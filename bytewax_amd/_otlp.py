@@ -127,6 +127,10 @@ class OtlpHttpExporter:
             target=self._loop, args=(flush_interval_s,), daemon=True
         )
         self._thread.start()
+        # Final flush even if the guard is never closed explicitly.
+        import atexit
+
+        atexit.register(self.flush)
 
     def record(self, span: OtlpSpan) -> None:
         with self._lock:
