@@ -184,3 +184,15 @@ def test_split_demo_example():
     res = _run([sys.executable, "examples/split_demo.py"])
     assert res.returncode == 0, res.stderr.decode()[-1500:]
     assert "_value" in res.stdout.decode()
+
+
+def test_simple_kafka_example_builds():
+    pytest.importorskip("confluent_kafka")
+    res = _run(
+        [
+            sys.executable,
+            "-c",
+            "import examples.simple_kafka_in_and_out as m; print(m.flow.flow_id)",
+        ]
+    )
+    assert res.returncode == 0, res.stderr.decode()[-1500:]
