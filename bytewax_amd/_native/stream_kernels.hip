@@ -728,7 +728,7 @@ __global__ void k_radix_scatter_direct(
 // 2^(seg_bits-region_bits) regions per block.
 #define SC_GRAN 8  // events per cursor reservation = one 64 B line
 
-template <int MODE, typename TS = int64_t>
+template <int MODE, typename TS = int64_t, int U = 16>
 __global__ __launch_bounds__(256) void k_radix_scatter_staged(
     const int32_t* __restrict__ keys,
     const TS* __restrict__ ts,
@@ -786,7 +786,6 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
     }
   };
 
-  const int U = 16;
   const int64_t tile = (int64_t)blockDim.x * U;
   int64_t local_max = 0;
   for (int64_t t0 = (int64_t)blockIdx.x * tile; t0 < n;
@@ -2250,19 +2249,33 @@ void radix_window_insert(
     unsigned gx = (unsigned)n_blocks(sg.n, 256);
     if (env_blocks > 0 && (unsigned)env_blocks < gx) gx = (unsigned)env_blocks;
     if (kind == SCAT_STAGED) {
-      // One tile = blockDim * 16 events; enough blocks to fill the
+      // One tile = blockDim * U events; enough blocks to fill the
       // chip at the LDS-bounded occupancy, few enough to keep the
-      // end-of-kernel residual padding small.
+      // end-of-kernel residual padding small.  U (events per thread
+      // per tile) tunable via BYTEWAX_SCATTER_U (8/16/32).
+      int su = 16;
+      if (const char* e = std::getenv("BYTEWAX_SCATTER_U")) {
+        int v = atoi(e);
+        if (v == 8 || v == 16 || v == 32) su = v;
+      }
       unsigned cap_gs = env_blocks > 0 ? (unsigned)env_blocks : 512u;
-      unsigned gs = (unsigned)((sg.n * xf + 4095) / 4096);
+      unsigned gs = (unsigned)((sg.n * xf + 256 * su - 1) / (256 * su));
       if (gs > cap_gs) gs = cap_gs;
       if (gs < 1) gs = 1;
-      if (mode == AGG_COUNT)
-        scat(k_radix_scatter_staged<AGG_COUNT, TSV>, tsptr, sg, gs,
-             staged_lds);
-      else
+      if (mode == AGG_COUNT) {
+        if (su == 8)
+          scat(k_radix_scatter_staged<AGG_COUNT, TSV, 8>, tsptr, sg, gs,
+               staged_lds);
+        else if (su == 32)
+          scat(k_radix_scatter_staged<AGG_COUNT, TSV, 32>, tsptr, sg, gs,
+               staged_lds);
+        else
+          scat(k_radix_scatter_staged<AGG_COUNT, TSV, 16>, tsptr, sg, gs,
+               staged_lds);
+      } else {
         scat(k_radix_scatter_staged<AGG_SUM, TSV>, tsptr, sg, gs,
              staged_lds);
+      }
     } else if (kind == SCAT_DIRECT) {
       if (mode == AGG_COUNT)
         scat(k_radix_scatter_direct<AGG_COUNT, TSV>, tsptr, sg, gx, 0);
@@ -3771,20 +3784,33 @@ int64_t native_run_window_steps(
           using TSV =
               std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
           if (kind == SCAT_STAGED) {
-            unsigned gs = (unsigned)((n + 4095) / 4096);
+            int su = 16;
+            if (const char* e = std::getenv("BYTEWAX_SCATTER_U")) {
+              int v = atoi(e);
+              if (v == 8 || v == 16 || v == 32) su = v;
+            }
+            unsigned gs = (unsigned)((n + 256 * su - 1) / (256 * su));
             if (gs > 512) gs = 512;
             if (gs < 1) gs = 1;
-            hipLaunchKernelGGL(
-                (k_radix_scatter_staged<AGG_COUNT, TSV>), dim3(gs), block,
-                staged_lds, sc_stream, keys.data_ptr<int32_t>(),
-                tsptr, (const int64_t*)nullptr, n,
-                align_ms, len_ms, len_ms, base, mask, seg_bits, cap,
-                gcur.data_ptr<int32_t>(), (uint64_t*)evp.data_ptr<int64_t>(),
-                (int64_t*)nullptr, ovc.data_ptr<int32_t>(),
-                (uint64_t*)ovp.data_ptr<int64_t>(), (int64_t*)nullptr,
-                ovp.numel(),
-                (unsigned long long*)max_ts.data_ptr<int64_t>(),
-                error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+            auto launch_staged = [&](auto kern) {
+              hipLaunchKernelGGL(
+                  kern, dim3(gs), block,
+                  staged_lds, sc_stream, keys.data_ptr<int32_t>(),
+                  tsptr, (const int64_t*)nullptr, n,
+                  align_ms, len_ms, len_ms, base, mask, seg_bits, cap,
+                  gcur.data_ptr<int32_t>(), (uint64_t*)evp.data_ptr<int64_t>(),
+                  (int64_t*)nullptr, ovc.data_ptr<int32_t>(),
+                  (uint64_t*)ovp.data_ptr<int64_t>(), (int64_t*)nullptr,
+                  ovp.numel(),
+                  (unsigned long long*)max_ts.data_ptr<int64_t>(),
+                  error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+            };
+            if (su == 8)
+              launch_staged(k_radix_scatter_staged<AGG_COUNT, TSV, 8>);
+            else if (su == 32)
+              launch_staged(k_radix_scatter_staged<AGG_COUNT, TSV, 32>);
+            else
+              launch_staged(k_radix_scatter_staged<AGG_COUNT, TSV, 16>);
           } else if (kind == SCAT_DIRECT) {
             hipLaunchKernelGGL(
                 (k_radix_scatter_direct<AGG_COUNT, TSV>), grid, block, 0,
