@@ -7,8 +7,9 @@ streams: per GPU-worker, a columnar source emits fixed-size event
 batches; events are exchanged across workers by key hash (RCCL
 all-to-allv over xGMI), folded into HBM-resident keyed window state by
 the fused HIP insert kernel, and emitted when the watermark closes each
-window.  One "step" = one source batch fully processed through the
-engine.
+window.  One "step" = one engine scheduling step = `--batches-per-poll`
+source batches fully processed through the engine (big polls amortize
+per-step engine overhead and keep the default timed region >= ~5 s).
 
 Run: python bench.py [--gpus N] [--steps K] [--warmup W]
 Multi-GPU: python -m torch.distributed.run --nnodes=1 --nproc-per-node N
