@@ -76,3 +76,129 @@ def test_exchange_ts_base_fold_two_ranks(tmp_path: Path):
     for p, (so, se) in zip(procs, outs):
         assert p.returncode == 0, se.decode()[-1500:]
     assert "EXCHANGE OK" in outs[0][0].decode()
+
+
+@pytest.mark.timeout(240)
+def test_sliding_window_exchange_2rank_gloo(tmp_path):
+    """Sliding windows + the wire-format exchange at world 2 (gloo
+    CPU twins): per-key sums across ranks must equal the single-rank
+    brute force (the scatter expands events into overlapped windows
+    AFTER the int32-segment exchange)."""
+    import json
+    import subprocess
+    import sys
+    import textwrap
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    script = tmp_path / "slide2.py"
+    script.write_text(
+        textwrap.dedent(
+            """
+            import os, json, torch
+            import torch.distributed as dist
+            from datetime import datetime, timedelta, timezone
+            import bytewax_amd.operators as op
+            from bytewax_amd.dataflow import Dataflow
+            from bytewax_amd.gpu import RecordBatch
+            from bytewax_amd.gpu.operators import (
+                CollectCountsSink, keyed_window_agg,
+            )
+            from bytewax_amd.inputs import (
+                DynamicSource, StatelessSourcePartition,
+            )
+            from bytewax_amd.testing import run_main
+
+            dist.init_process_group("gloo")
+            rank = dist.get_rank()
+            ALIGN = datetime(2024, 1, 1, tzinfo=timezone.utc)
+            ALIGN_MS = int(ALIGN.timestamp() * 1000)
+
+            class Part(StatelessSourcePartition):
+                def __init__(self):
+                    self.i = 0
+                def next_batch(self):
+                    if self.i >= 4:
+                        raise StopIteration()
+                    g = torch.Generator().manual_seed(100 * rank + self.i)
+                    n = 500
+                    keys = torch.randint(0, 50, (n,), dtype=torch.int32,
+                                         generator=g)
+                    ts = torch.randint(0, 5000, (n,), dtype=torch.int32,
+                                       generator=g)
+                    self.i += 1
+                    return [RecordBatch(
+                        keys, ts, keys.to(torch.int64),
+                        max_ts=ALIGN_MS + self.i * 5000 - 1,
+                        ts_base=ALIGN_MS + (self.i - 1) * 5000,
+                    )]
+
+            class Src(DynamicSource):
+                def build(self, *_a):
+                    return Part()
+
+            out = []
+            flow = Dataflow("slide2")
+            s = op.input("inp", flow, Src())
+            agg = keyed_window_agg(
+                "agg", s, align_to=ALIGN,
+                length=timedelta(seconds=6),
+                offset=timedelta(seconds=2),
+                mode="sum", device="cpu", exchange=True,
+            )
+            op.output("out", agg, CollectCountsSink(out))
+            run_main(flow, epoch_interval=timedelta(days=1))
+            rows = {}
+            for b in out:
+                for k, t, v in zip(b.keys.tolist(), b.ts.tolist(),
+                                   b.vals.tolist()):
+                    rows[(k, t)] = rows.get((k, t), 0) + v
+            gathered = [None, None]
+            dist.all_gather_object(gathered, rows)
+            if rank == 0:
+                merged = {}
+                for g2 in gathered:
+                    for k, v in g2.items():
+                        merged[k] = merged.get(k, 0) + v
+                # Brute force over all ranks' events.
+                expect = {}
+                for r in range(2):
+                    for i in range(4):
+                        g = torch.Generator().manual_seed(100 * r + i)
+                        n = 500
+                        keys = torch.randint(0, 50, (n,),
+                                             dtype=torch.int32, generator=g)
+                        ts = torch.randint(0, 5000, (n,),
+                                           dtype=torch.int32, generator=g)
+                        base = ALIGN_MS + i * 5000
+                        for k, t in zip(keys.tolist(), ts.tolist()):
+                            at = base + t
+                            hi = (at - ALIGN_MS) // 2000
+                            lo = (at - ALIGN_MS - 6000) // 2000 + 1
+                            for w in range(lo, hi + 1):
+                                key = (k, ALIGN_MS + w * 2000)
+                                expect[key] = expect.get(key, 0) + k
+                assert merged == expect, "sliding exchange mismatch"
+                print("SLIDE2_OK")
+            dist.destroy_process_group()
+            """
+        )
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(repo)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["MASTER_PORT"] = str(29870 + os.getpid() % 100)
+    procs = []
+    for rank in range(2):
+        e = dict(env)
+        e.update(RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK=str(rank))
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, str(script)],
+                env=e, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            )
+        )
+    outs = [p.communicate(timeout=200) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se.decode()[-2000:]
+    assert "SLIDE2_OK" in outs[0][0].decode()

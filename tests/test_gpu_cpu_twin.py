@@ -136,6 +136,8 @@ def test_bench_ranks_gloo(tmp_path: Path, world):
                     "1",
                     "--events-per-batch",
                     "20000",
+                    "--batches-per-poll",
+                    "3",
                     "--vocab",
                     "1000",
                     "--device",
