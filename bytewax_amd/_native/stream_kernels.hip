@@ -34,6 +34,7 @@
 #include <type_traits>
 #include <cstdlib>
 #include <cstdint>
+#include <cstring>
 #include <vector>
 
 #define WAVE 64
@@ -66,6 +67,13 @@ static ScatterKind scatter_kind_env() {
     if (atoi(sd) != 0) return SCAT_DIRECT;
   }
   return SCAT_STAGED;
+}
+
+// Experimental coalesced-placement staged variant ("staged2"): only
+// consulted where it is implemented (COUNT, tumbling).
+static bool scatter_staged2_env() {
+  const char* s = std::getenv("BYTEWAX_SCATTER");
+  return s != nullptr && std::strncmp(s, "staged2", 7) == 0;
 }
 
 // Extra segment coarsening: scatter segments cover 2^coarse table
@@ -897,6 +905,199 @@ __global__ __launch_bounds__(256) void k_radix_scatter_staged(
       emit(b, (int64_t)gb + j, p,
            (MODE != AGG_COUNT && j < rc) ? res_v[(size_t)b * SC_GRAN + j]
                                          : 0);
+    }
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    int64_t other = __shfl_down((long long)local_max, off);
+    if (other > local_max) local_max = other;
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0 && local_max > 0) {
+    atomicMax(max_ts, (unsigned long long)local_max);
+  }
+}
+
+
+// Coalesced-placement variant of the staged scatter (COUNT,
+// tumbling): stage the tile's granted entries in LDS together with
+// their precomputed global positions, then write them back with a
+// LINEAR copy loop — consecutive lanes hit consecutive addresses
+// inside each segment's granule run, so the hardware coalesces ~8
+// scattered-by-segment events into one full-line transaction instead
+// of 8.  Probes the hypothesis that the scatter is bound by L2
+// random-store transaction rate (~1 tx/event in the default staged
+// kernel), not bytes.  Selected via BYTEWAX_SCATTER=staged2.
+template <typename TS = int64_t, int U = 16>
+__global__ __launch_bounds__(256) void k_radix_scatter_staged2(
+    const int32_t* __restrict__ keys,
+    const TS* __restrict__ ts,
+    const int64_t* __restrict__ vals,  // unused (COUNT)
+    int64_t n,
+    int64_t align_ms,
+    int64_t len_ms,
+    int64_t off_ms,
+    int64_t ts_base,
+    uint64_t mask,
+    int seg_bits,
+    int64_t cap,
+    int* __restrict__ gcursors,
+    uint64_t* __restrict__ ev_packed,
+    int64_t* __restrict__ ev_vals,  // unused (COUNT)
+    int* __restrict__ ov_cursor,
+    uint64_t* __restrict__ ov_packed,
+    int64_t* __restrict__ ov_vals,  // unused (COUNT)
+    int64_t ov_cap,
+    unsigned long long* __restrict__ max_ts,
+    int* __restrict__ error_flag,
+    uint64_t win_m2,
+    uint64_t win_maxfast) {
+  extern __shared__ char smem[];
+  const int nseg = (int)(((mask + 1) >> seg_bits));
+  const int T = (int)blockDim.x * U;
+  const int MAXSTAGE = T + 7 * 2048;  // upper bound; sized by host LDS
+  uint64_t* stage = (uint64_t*)smem;                       // [maxstage]
+  int* gpos = (int*)(smem + (size_t)(T + 7 * nseg) * 8);   // [maxstage]
+  uint64_t* res = (uint64_t*)(smem + (size_t)(T + 7 * nseg) * 12);
+  int* res_cnt = (int*)((char*)res + (size_t)nseg * SC_GRAN * 8);
+  int* lhist = res_cnt + nseg;
+  int* lbase = lhist + nseg;
+  int* ltoff = lbase + nseg;
+  int* lvirt = ltoff + nseg;
+  int* lgrant = lvirt + nseg;
+  int* ltotal = lgrant + nseg;  // [1]
+  (void)MAXSTAGE;
+  for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+    res_cnt[b] = 0;
+    lhist[b] = 0;
+  }
+  __syncthreads();
+
+  auto spill = [&](uint64_t packed) {
+    int opos = atomicAdd(ov_cursor, 1);
+    if (opos < ov_cap) ov_packed[opos] = packed;
+    else atomicExch(error_flag, 1);
+  };
+
+  const int64_t tile = (int64_t)blockDim.x * U;
+  int64_t local_max = 0;
+  for (int64_t t0 = (int64_t)blockIdx.x * tile; t0 < n;
+       t0 += (int64_t)gridDim.x * tile) {
+    uint64_t pk[U];
+    int sg[U];
+    // P1: classify + histogram.
+    for (int u = 0; u < U; ++u) {
+      int64_t i = t0 + (int64_t)u * blockDim.x + threadIdx.x;
+      sg[u] = -1;
+      if (i >= n) continue;
+      int64_t t = (int64_t)ts[i] + ts_base;
+      if (t > local_max) local_max = t;
+      int64_t win = win_of(t, align_ms, len_ms, win_m2, win_maxfast);
+      uint64_t packed =
+          ((uint64_t)(uint32_t)(int32_t)win << 32) | (uint32_t)keys[i];
+      pk[u] = packed;
+      int b = (int)region_of(mix64(packed), mask, seg_bits);
+      sg[u] = b;
+      atomicAdd(&lhist[b], 1);
+    }
+    __syncthreads();
+    // P2: granule grants + in-tile staging offsets (exclusive scan of
+    // the per-segment grants).
+    for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+      int tot = res_cnt[b] + lhist[b];
+      int grant = tot & ~(SC_GRAN - 1);
+      lgrant[b] = grant;
+      lbase[b] = grant > 0 ? atomicAdd(&gcursors[b], grant) : 0;
+      // Virtual positions count residual + new; ungranted segments
+      // keep their residual in place and append after it.
+      lvirt[b] = res_cnt[b];
+      ltoff[b] = grant;  // scanned in place below
+    }
+    __syncthreads();
+    // Hillis-Steele inclusive scan over ltoff[nseg], then shift to
+    // exclusive via (incl - grant).
+    for (int d = 1; d < nseg; d <<= 1) {
+      int v[8];
+      int nb = 0;
+      for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+        v[nb++] = b >= d ? ltoff[b - d] : 0;
+      }
+      __syncthreads();
+      nb = 0;
+      for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+        ltoff[b] += v[nb++];
+      }
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) ltotal[0] = ltoff[nseg - 1];
+    __syncthreads();
+    // P3a: drain old residual into staging (virtual slots 0..rc).
+    for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+      int grant = lgrant[b];
+      if (grant > 0) {
+        int rc = res_cnt[b];
+        int off = ltoff[b] - grant;
+        int gb = lbase[b];
+        for (int j = 0; j < rc; ++j) {
+          int64_t gp = (int64_t)gb + j;
+          if (gp < cap) {
+            stage[off + j] = res[(size_t)b * SC_GRAN + j];
+            gpos[off + j] = (int)((int64_t)b * cap + gp);
+          } else {
+            stage[off + j] = 0;
+            gpos[off + j] = -1;
+            spill(res[(size_t)b * SC_GRAN + j]);
+          }
+        }
+      }
+    }
+    __syncthreads();
+    // P3b: place new events (staged if granted, else residual).
+    for (int u = 0; u < U; ++u) {
+      int b = sg[u];
+      if (b < 0) continue;
+      int virt = atomicAdd(&lvirt[b], 1);
+      int grant = lgrant[b];
+      if (virt < grant) {
+        int off = ltoff[b] - grant + virt;
+        int64_t gp = (int64_t)lbase[b] + virt;
+        if (gp < cap) {
+          stage[off] = pk[u];
+          gpos[off] = (int)((int64_t)b * cap + gp);
+        } else {
+          stage[off] = 0;
+          gpos[off] = -1;
+          spill(pk[u]);
+        }
+      } else {
+        res[(size_t)b * SC_GRAN + (virt - grant)] = pk[u];
+      }
+    }
+    __syncthreads();
+    // P4: coalesced copy-out — consecutive threads write consecutive
+    // staged slots; slots are segment-grouped and granule-aligned, so
+    // adjacent lanes hit adjacent addresses of the same 64 B lines.
+    int total = ltotal[0];
+    for (int j = threadIdx.x; j < total; j += blockDim.x) {
+      int gp = gpos[j];
+      if (gp >= 0) ev_packed[gp] = stage[j];
+    }
+    // P5: carry leftovers.
+    for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+      res_cnt[b] = lvirt[b] - lgrant[b];
+      if (res_cnt[b] < 0) res_cnt[b] = 0;
+      lhist[b] = 0;
+    }
+    __syncthreads();
+  }
+  // Final flush: sentinel-padded granules, like the default staged.
+  for (int b = threadIdx.x; b < nseg; b += blockDim.x) {
+    int rc = res_cnt[b];
+    if (rc == 0) continue;
+    int gb = atomicAdd(&gcursors[b], SC_GRAN);
+    for (int j = 0; j < SC_GRAN; ++j) {
+      uint64_t p = j < rc ? res[(size_t)b * SC_GRAN + j] : EMPTY_SLOT;
+      int64_t gp = (int64_t)gb + j;
+      if (gp < cap) ev_packed[(int64_t)b * cap + gp] = p;
+      else if (p != EMPTY_SLOT) spill(p);
     }
   }
   for (int off = WAVE / 2; off > 0; off >>= 1) {
@@ -2262,7 +2463,13 @@ void radix_window_insert(
       unsigned gs = (unsigned)((sg.n * xf + 256 * su - 1) / (256 * su));
       if (gs > cap_gs) gs = cap_gs;
       if (gs < 1) gs = 1;
-      if (mode == AGG_COUNT) {
+      size_t s2_lds = (size_t)(256 * 16 + 7 * nseg) * 12 +
+                      (size_t)nseg * SC_GRAN * 8 +
+                      7 * (size_t)nseg * sizeof(int) + 16;
+      if (mode == AGG_COUNT && xf == 1 && scatter_staged2_env() &&
+          s2_lds <= 160 * 1024) {
+        scat(k_radix_scatter_staged2<TSV>, tsptr, sg, gs, s2_lds);
+      } else if (mode == AGG_COUNT) {
         if (su == 8)
           scat(k_radix_scatter_staged<AGG_COUNT, TSV, 8>, tsptr, sg, gs,
                staged_lds);
