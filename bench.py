@@ -52,10 +52,13 @@ def parse_args():
     )
     p.add_argument("--region-bits", type=int, default=11)
     p.add_argument(
-        "--no-pipeline",
-        action="store_true",
-        help="disable the two-stream scatter/agg pipeline (native "
-        "radix engine)",
+        "--pipeline",
+        action=argparse.BooleanOptionalAction,
+        default=False,
+        help="two-stream scatter/agg overlap in the native radix "
+        "engine (default off: with the line-staged scatter the "
+        "overlap contends for the same LDS/CUs and measures ~4% "
+        "slower than serial execution; see profiles/)",
     )
     p.add_argument(
         "--dedup",
@@ -220,7 +223,7 @@ def main():
                 ), None
             return state.native_run(
                 part.key_pool, ts_pool, start, count,
-                args.sim_ms_per_batch, pipelined=not args.no_pipeline,
+                args.sim_ms_per_batch, pipelined=args.pipeline,
             )
 
         # Warmup.  One timed step = B batches (matching the dataflow
