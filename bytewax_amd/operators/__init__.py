@@ -219,6 +219,16 @@ def inspect_debug(
 
     :arg inspector: Called with the step ID, item, epoch, and worker
         index for each item.
+
+    Example:
+
+    >>> flow = Dataflow("dbg")
+    >>> s = op.input("inp", flow, TestingSource([1]))
+    >>> s = op.inspect_debug("d", s)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    dbg.d W0 @1: 1
+    1
     """
     return "down"  # type: ignore[return-value]
 
@@ -292,6 +302,24 @@ def stateful_batch(
 
     :arg builder: Called whenever a new key is encountered with the
         resume state returned from snapshotting, if any.
+
+    Example:
+
+    >>> class BatchSum(op.StatefulBatchLogic):
+    ...     def __init__(self, resume):
+    ...         self.total = resume or 0
+    ...     def on_batch(self, values):
+    ...         self.total += sum(values)
+    ...         return ([self.total], op.StatefulBatchLogic.RETAIN)
+    ...     def snapshot(self):
+    ...         return self.total
+    >>> flow = Dataflow("sb")
+    >>> s = op.input("inp", flow, TestingSource([("a", 1), ("a", 2)]))
+    >>> s = op.stateful_batch("sum", s, BatchSum)
+    >>> op.output("out", s, StdOutSink())
+    >>> run_main(flow)
+    ('a', 1)
+    ('a', 3)
     """
     if not callable(builder):
         msg = "`builder` must be callable"
