@@ -2459,7 +2459,7 @@ void radix_window_insert(
         int v = atoi(e);
         if (v == 8 || v == 16 || v == 32) su = v;
       }
-      unsigned cap_gs = env_blocks > 0 ? (unsigned)env_blocks : 512u;
+      unsigned cap_gs = env_blocks > 0 ? (unsigned)env_blocks : 1024u;
       unsigned gs = (unsigned)((sg.n * xf + 256 * su - 1) / (256 * su));
       if (gs > cap_gs) gs = cap_gs;
       if (gs < 1) gs = 1;
@@ -2674,7 +2674,7 @@ void radix_scatter_only(
     using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
     if (kind == SCAT_STAGED) {
       unsigned gs = (unsigned)((sg.n + 4095) / 4096);
-      if (gs > 512) gs = 512;
+      if (gs > 1024) gs = 1024;
       if (gs < 1) gs = 1;
       if (mode == AGG_COUNT)
         scat(k_radix_scatter_staged<AGG_COUNT, TSV>, tsptr, sg, gs,
@@ -3019,7 +3019,7 @@ void radix_stats_insert(
     using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
     if (kind == SCAT_STAGED) {
       unsigned gs = (unsigned)((n + 4095) / 4096);
-      if (gs > 512) gs = 512;
+      if (gs > 1024) gs = 1024;
       if (gs < 1) gs = 1;
       hipLaunchKernelGGL(
           (k_radix_scatter_staged<AGG_SUM, TSV>), dim3(gs), block,
@@ -3234,7 +3234,7 @@ void radix_join_insert(
   // Key-only packing: ts == 0, align 0, huge window -> win = 0.
   if (kind == SCAT_STAGED) {
     unsigned gs = (unsigned)((n + 4095) / 4096);
-    if (gs > 512) gs = 512;
+    if (gs > 1024) gs = 1024;
     if (gs < 1) gs = 1;
     hipLaunchKernelGGL(
         (k_radix_scatter_staged<AGG_SUM>), dim3(gs), block, staged_lds,
@@ -3997,7 +3997,7 @@ int64_t native_run_window_steps(
               if (v == 8 || v == 16 || v == 32) su = v;
             }
             unsigned gs = (unsigned)((n + 256 * su - 1) / (256 * su));
-            if (gs > 512) gs = 512;
+            if (gs > 1024) gs = 1024;
             if (gs < 1) gs = 1;
             auto launch_staged = [&](auto kern) {
               hipLaunchKernelGGL(
@@ -4374,7 +4374,7 @@ void session_radix_insert(
     using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
     if (staged && staged_lds <= 160 * 1024) {
       unsigned gs = (unsigned)((n + 4095) / 4096);
-      if (gs > 512) gs = 512;
+      if (gs > 1024) gs = 1024;
       if (gs < 1) gs = 1;
       scat(k_radix_scatter_staged<AGG_TS, TSV>, tsptr, gs, staged_lds);
     } else {
