@@ -2429,10 +2429,11 @@ void radix_window_insert(
 
   uint64_t win_m2, win_maxfast;
   magic_div_u64(off_ms, &win_m2, &win_maxfast);
+  unsigned scat_threads = 256;
   auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx,
                   size_t lds) {
     hipLaunchKernelGGL(
-        kern, dim3(gx), block, lds, stream,
+        kern, dim3(gx), dim3(scat_threads), lds, stream,
         keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
         vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
         off_ms, sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
@@ -2459,8 +2460,15 @@ void radix_window_insert(
         int v = atoi(e);
         if (v == 8 || v == 16 || v == 32) su = v;
       }
+      int sthreads = 256;
+      if (const char* e = std::getenv("BYTEWAX_SCATTER_THREADS")) {
+        int v = atoi(e);
+        if (v == 256 || v == 512 || v == 1024) sthreads = v;
+      }
+      scat_threads = (unsigned)sthreads;
       unsigned cap_gs = env_blocks > 0 ? (unsigned)env_blocks : 1024u;
-      unsigned gs = (unsigned)((sg.n * xf + 256 * su - 1) / (256 * su));
+      unsigned gs =
+          (unsigned)((sg.n * xf + sthreads * su - 1) / (sthreads * su));
       if (gs > cap_gs) gs = cap_gs;
       if (gs < 1) gs = 1;
       size_t s2_lds = (size_t)(256 * 16 + 7 * nseg) * 12 +
@@ -2484,11 +2492,13 @@ void radix_window_insert(
              staged_lds);
       }
     } else if (kind == SCAT_DIRECT) {
+      scat_threads = 256;
       if (mode == AGG_COUNT)
         scat(k_radix_scatter_direct<AGG_COUNT, TSV>, tsptr, sg, gx, 0);
       else
         scat(k_radix_scatter_direct<AGG_SUM, TSV>, tsptr, sg, gx, 0);
     } else {
+      scat_threads = 256;
       if (mode == AGG_COUNT)
         scat(k_radix_scatter_fixed<AGG_COUNT, TSV>, tsptr, sg, gx,
              2 * hist_lds);
