@@ -736,8 +736,8 @@ __global__ void k_radix_scatter_direct(
 // 2^(seg_bits-region_bits) regions per block.
 #define SC_GRAN 8  // events per cursor reservation = one 64 B line
 
-template <int MODE, typename TS = int64_t, int U = 16>
-__global__ __launch_bounds__(256) void k_radix_scatter_staged(
+template <int MODE, typename TS = int64_t, int U = 16, int BLK = 256>
+__global__ __launch_bounds__(BLK) void k_radix_scatter_staged(
     const int32_t* __restrict__ keys,
     const TS* __restrict__ ts,
     const int64_t* __restrict__ vals,  // nullptr for COUNT
@@ -2476,9 +2476,16 @@ void radix_window_insert(
                       7 * (size_t)nseg * sizeof(int) + 16;
       if (mode == AGG_COUNT && xf == 1 && scatter_staged2_env() &&
           s2_lds <= 160 * 1024) {
+        scat_threads = 256;
         scat(k_radix_scatter_staged2<TSV>, tsptr, sg, gs, s2_lds);
       } else if (mode == AGG_COUNT) {
-        if (su == 8)
+        if (sthreads == 512)
+          scat((k_radix_scatter_staged<AGG_COUNT, TSV, 16, 512>), tsptr,
+               sg, gs, staged_lds);
+        else if (sthreads == 1024)
+          scat((k_radix_scatter_staged<AGG_COUNT, TSV, 16, 1024>), tsptr,
+               sg, gs, staged_lds);
+        else if (su == 8)
           scat(k_radix_scatter_staged<AGG_COUNT, TSV, 8>, tsptr, sg, gs,
                staged_lds);
         else if (su == 32)
@@ -2488,6 +2495,7 @@ void radix_window_insert(
           scat(k_radix_scatter_staged<AGG_COUNT, TSV, 16>, tsptr, sg, gs,
                staged_lds);
       } else {
+        scat_threads = 256;
         scat(k_radix_scatter_staged<AGG_SUM, TSV>, tsptr, sg, gs,
              staged_lds);
       }
