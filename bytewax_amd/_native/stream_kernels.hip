@@ -2460,7 +2460,9 @@ void radix_window_insert(
         int v = atoi(e);
         if (v == 8 || v == 16 || v == 32) su = v;
       }
-      int sthreads = 256;
+      // 512-thread workgroups measured fastest for the COUNT staged
+      // scatter (60.1 vs 56.6e9 at 256, 57.2 at 1024 — see profiles/).
+      int sthreads = mode == AGG_COUNT ? 512 : 256;
       if (const char* e = std::getenv("BYTEWAX_SCATTER_THREADS")) {
         int v = atoi(e);
         if (v == 256 || v == 512 || v == 1024) sthreads = v;
@@ -2672,10 +2674,11 @@ void radix_scatter_only(
   } else {
     segs.push_back({0, n, ts_base});
   }
+  unsigned scat_threads = 256;
   auto scat = [&](auto kern, auto tsptr, const Seg& sg, unsigned gx,
                   size_t lds) {
     hipLaunchKernelGGL(
-        kern, dim3(gx), block, lds,
+        kern, dim3(gx), dim3(scat_threads), lds,
         stream, keys.data_ptr<int32_t>() + sg.off, tsptr + sg.off,
         vptr != nullptr ? vptr + sg.off : nullptr, sg.n, align_ms, len_ms,
         off_ms, sg.base, mask, seg_bits, cap, gcursors.data_ptr<int32_t>(),
@@ -2691,15 +2694,22 @@ void radix_scatter_only(
   auto scat_any = [&](auto tsptr, const Seg& sg) {
     using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
     if (kind == SCAT_STAGED) {
-      unsigned gs = (unsigned)((sg.n + 4095) / 4096);
-      if (gs > 1024) gs = 1024;
-      if (gs < 1) gs = 1;
-      if (mode == AGG_COUNT)
-        scat(k_radix_scatter_staged<AGG_COUNT, TSV>, tsptr, sg, gs,
-             staged_lds);
-      else
+      if (mode == AGG_COUNT) {
+        // 512-thread workgroups (see radix_window_insert).
+        unsigned gs = (unsigned)((sg.n + 8191) / 8192);
+        if (gs > 1024) gs = 1024;
+        if (gs < 1) gs = 1;
+        scat_threads = 512;
+        scat((k_radix_scatter_staged<AGG_COUNT, TSV, 16, 512>), tsptr,
+             sg, gs, staged_lds);
+        scat_threads = 256;
+      } else {
+        unsigned gs = (unsigned)((sg.n + 4095) / 4096);
+        if (gs > 1024) gs = 1024;
+        if (gs < 1) gs = 1;
         scat(k_radix_scatter_staged<AGG_SUM, TSV>, tsptr, sg, gs,
              staged_lds);
+      }
     } else if (kind == SCAT_DIRECT) {
       unsigned gx = (unsigned)n_blocks(sg.n, 256);
       if (mode == AGG_COUNT)
@@ -4036,6 +4046,7 @@ int64_t native_run_window_steps(
               launch_staged(k_radix_scatter_staged<AGG_COUNT, TSV, 32>);
             else
               launch_staged(k_radix_scatter_staged<AGG_COUNT, TSV, 16>);
+            // (pipe path keeps 256-thread blocks; serial is default)
           } else if (kind == SCAT_DIRECT) {
             hipLaunchKernelGGL(
                 (k_radix_scatter_direct<AGG_COUNT, TSV>), grid, block, 0,
