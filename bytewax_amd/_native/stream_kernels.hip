@@ -3046,12 +3046,12 @@ void radix_stats_insert(
   auto scat_any = [&](auto tsptr) {
     using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
     if (kind == SCAT_STAGED) {
-      unsigned gs = (unsigned)((n + 4095) / 4096);
+      unsigned gs = (unsigned)((n + 8191) / 8192);
       if (gs > 1024) gs = 1024;
       if (gs < 1) gs = 1;
       hipLaunchKernelGGL(
-          (k_radix_scatter_staged<AGG_SUM, TSV>), dim3(gs), block,
-          staged_lds,
+          (k_radix_scatter_staged<AGG_SUM, TSV, 16, 512>), dim3(gs),
+          dim3(512), staged_lds,
           stream, keys.data_ptr<int32_t>(), tsptr,
           vals.data_ptr<int64_t>(), n, align_ms, len_ms, len_ms, ts_base,
           mask,
@@ -4386,6 +4386,20 @@ void session_radix_insert(
   magic_div_u64((int64_t)1 << 40, &win_m2, &win_maxfast);
   size_t staged_lds = (size_t)nseg * SC_GRAN * 8 * 2 +
                       4 * (size_t)nseg * sizeof(int);
+  auto scat512 = [&](auto kern, auto tsptr, unsigned gx, size_t lds) {
+    hipLaunchKernelGGL(
+        kern, dim3(gx), dim3(512), lds, stream,
+        keys.data_ptr<int32_t>(), tsptr, (const int64_t*)nullptr, n, 0,
+        (int64_t)1 << 40, (int64_t)1 << 40, ts_base, mask, (int)seg_bits,
+        cap,
+        gcursors.data_ptr<int32_t>(),
+        (uint64_t*)ev_packed.data_ptr<int64_t>(),
+        ev_vals.data_ptr<int64_t>(), ov_cursor.data_ptr<int32_t>(),
+        (uint64_t*)ov_packed.data_ptr<int64_t>(),
+        ov_vals.data_ptr<int64_t>(), ov_packed.numel(),
+        (unsigned long long*)max_ts.data_ptr<int64_t>(),
+        error_flag.data_ptr<int32_t>(), win_m2, win_maxfast);
+  };
   auto scat = [&](auto kern, auto tsptr, unsigned gx, size_t lds) {
     hipLaunchKernelGGL(
         kern, dim3(gx), dim3(256), lds, stream,
@@ -4402,10 +4416,11 @@ void session_radix_insert(
   auto scat_any = [&](auto tsptr) {
     using TSV = std::remove_const_t<std::remove_pointer_t<decltype(tsptr)>>;
     if (staged && staged_lds <= 160 * 1024) {
-      unsigned gs = (unsigned)((n + 4095) / 4096);
+      unsigned gs = (unsigned)((n + 8191) / 8192);
       if (gs > 1024) gs = 1024;
       if (gs < 1) gs = 1;
-      scat(k_radix_scatter_staged<AGG_TS, TSV>, tsptr, gs, staged_lds);
+      scat512(k_radix_scatter_staged<AGG_TS, TSV, 16, 512>, tsptr, gs,
+              staged_lds);
     } else {
       size_t hist_lds = (size_t)nseg * sizeof(int);
       scat(k_radix_scatter_fixed<AGG_TS, TSV>, tsptr,
