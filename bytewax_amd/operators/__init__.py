@@ -874,7 +874,23 @@ def key_rm(step_id: str, up: KeyedStream[X]) -> Stream[X]:
 
 @operator
 def raises(step_id: str, up: Stream[Any]) -> None:
-    """Raise an exception and crash the dataflow on any item."""
+    """Raise an exception and crash the dataflow on any item.
+
+    Example:
+
+    Mark a branch that must stay empty; any item that reaches it
+    crashes the execution:
+
+    >>> flow = Dataflow("raises_eg")
+    >>> s = op.input("inp", flow, TestingSource([1, -2]))
+    >>> b = op.branch("valid", s, lambda x: x > 0)
+    >>> op.raises("no_negatives", b.falses)
+    >>> op.output("out", b.trues, StdOutSink())
+    >>> run_main(flow)
+    Traceback (most recent call last):
+        ...
+    RuntimeError: `raises` step 'raises_eg.no_negatives' got an item: -2
+    """
 
     def shim_mapper(x: Any) -> Iterable[Any]:
         msg = f"`raises` step {step_id!r} got an item: {x!r}"
