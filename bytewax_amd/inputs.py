@@ -179,6 +179,21 @@ class SimplePollingSource(FixedPartitionedSource[X, None]):
     """Calls a user-defined function at a regular interval.
 
     Subclass and override {py:obj}`next_item`.  Only one worker polls.
+
+    Example:
+        >>> from datetime import timedelta
+        >>> from bytewax_amd.inputs import SimplePollingSource
+        >>> class Counter(SimplePollingSource):
+        ...     n = 0
+        ...     def next_item(self):
+        ...         Counter.n += 1
+        ...         return Counter.n
+        >>> src = Counter(timedelta(seconds=10))
+        >>> src.list_parts()
+        ['singleton']
+        >>> part = src.build_part("input", "singleton", None)
+        >>> part.next_batch()
+        [1]
     """
 
     class Retry(Exception):
@@ -214,7 +229,17 @@ class SimplePollingSource(FixedPartitionedSource[X, None]):
 
 
 def batch(ib: Iterable[X], batch_size: int) -> Iterator[List[X]]:
-    """Batch an iterable into fixed-size lists."""
+    """Batch an iterable into fixed-size lists.
+
+    Use this to build `next_batch` for a partition over a plain
+    iterator (reference documents the same helper on
+    pysrc/bytewax/inputs.py).
+
+    Example:
+        >>> from bytewax_amd.inputs import batch
+        >>> list(batch(range(5), 2))
+        [[0, 1], [2, 3], [4]]
+    """
     it = iter(ib)
     while True:
         out: List[X] = []
@@ -235,6 +260,15 @@ def batch_getter(
 
     The resulting iterator is infinite: it yields (possibly empty)
     batches forever; callers should stop polling it externally.
+
+    Example:
+        >>> from bytewax_amd.inputs import batch_getter
+        >>> q = [1, 2, 3]
+        >>> def poll():
+        ...     return q.pop(0) if q else None
+        >>> it = batch_getter(poll, 2)
+        >>> next(it), next(it), next(it)
+        ([1, 2], [3], [])
     """
     while True:
         out: List[X] = []
@@ -249,7 +283,15 @@ def batch_getter(
 def batch_getter_ex(
     getter: Callable[[], X], batch_size: int, yield_ex: type = IndexError
 ) -> Iterator[List[X]]:
-    """Batch from a getter function that raises when empty."""
+    """Batch from a getter function that raises when empty.
+
+    Example:
+        >>> from bytewax_amd.inputs import batch_getter_ex
+        >>> q = [1, 2, 3]
+        >>> it = batch_getter_ex(lambda: q.pop(0), 2)
+        >>> next(it), next(it), next(it)
+        ([1, 2], [3], [])
+    """
     while True:
         out: List[X] = []
         for _ in range(batch_size):
@@ -271,6 +313,15 @@ def batch_async(
     Each advance collects up to `batch_size` items, waiting at most
     `timeout` total; yields possibly-empty batches until the async
     iterator is exhausted.
+
+    Example:
+        >>> from datetime import timedelta
+        >>> from bytewax_amd.inputs import batch_async
+        >>> async def gen():
+        ...     for i in range(3):
+        ...         yield i
+        >>> list(batch_async(gen(), timedelta(seconds=1), 2))
+        [[0, 1], [2]]
     """
     loop = loop if loop is not None else asyncio.new_event_loop()
 

@@ -68,6 +68,11 @@ class FixedPartitionedSink(Sink[Any], Generic[X, S]):
         The default hash is `zlib.adler32`, which is consistent across
         processes (unlike Python's builtin `hash`).  The return value
         is wrapped modulo the partition count.
+
+        Example:
+            >>> from bytewax_amd.outputs import FixedPartitionedSink
+            >>> FixedPartitionedSink.part_fn(None, "user-1") % 3
+            1
         """
         return zlib.adler32(item_key.encode())
 

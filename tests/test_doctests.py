@@ -51,3 +51,15 @@ def test_operator_docstring_examples():
 
 def test_windowing_docstring_examples():
     _run(win, 40)
+
+
+def test_inputs_sdk_docstring_examples():
+    import bytewax_amd.inputs as inputs
+
+    _run(inputs, 20)
+
+
+def test_outputs_sdk_docstring_examples():
+    import bytewax_amd.outputs as outputs
+
+    _run(outputs, 2)
