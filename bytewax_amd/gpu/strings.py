@@ -61,6 +61,11 @@ class StringDict:
         self.new_idx = torch.empty(new_cap, dtype=torch.int32, device=device)
         self.new_n = torch.zeros(1, dtype=torch.int32, device=device)
         self.error_flag = torch.zeros(1, dtype=torch.int32, device=device)
+        # Pinned staging for the per-batch H2D of string bytes: the
+        # encode path is H2D-bound (see profiles/r02_kernel_stats.md),
+        # and pageable copies stage through a driver bounce buffer.
+        self._pin_bytes = None
+        self._pin_offs = None
 
     def __len__(self) -> int:
         return len(self.id2str)
@@ -100,8 +105,7 @@ class StringDict:
                     self.id2str.append(s)
                 ids[i] = j
             return torch.from_numpy(ids)
-        d_bytes = torch.from_numpy(data).to(self.device)
-        d_offs = torch.from_numpy(offs).to(self.device)
+        d_bytes, d_offs = self._stage(data, offs)
         out_ids = torch.empty(n, dtype=torch.int32, device=self.device)
         self.k.dict_encode(
             d_bytes, d_offs, self.dlo, self.dhi, self.dids, self.counter,
@@ -130,6 +134,31 @@ class StringDict:
                 self.id2str[i] = self._strings_of(data, offs, bi, strings)
         return out_ids
 
+    def _stage(self, data: np.ndarray, offs: np.ndarray):
+        """Copy packed bytes/offsets through grow-on-demand pinned
+        host buffers and ship them with async DMA copies (both land
+        on the current stream, so the following kernel launch orders
+        after them)."""
+        import torch
+
+        nb, no = data.shape[0], offs.shape[0]
+        if self._pin_bytes is None or self._pin_bytes.numel() < nb:
+            cap = max(1 << 16, 1 << max(0, nb - 1).bit_length())
+            self._pin_bytes = torch.empty(
+                cap, dtype=torch.uint8, pin_memory=True
+            )
+        if self._pin_offs is None or self._pin_offs.numel() < no:
+            cap = max(1 << 10, 1 << max(0, no - 1).bit_length())
+            self._pin_offs = torch.empty(
+                cap, dtype=torch.int64, pin_memory=True
+            )
+        if nb:
+            self._pin_bytes[:nb].copy_(torch.from_numpy(data))
+        self._pin_offs[:no].copy_(torch.from_numpy(offs))
+        d_bytes = self._pin_bytes[:nb].to(self.device, non_blocking=True)
+        d_offs = self._pin_offs[:no].to(self.device, non_blocking=True)
+        return d_bytes, d_offs
+
     def decode(self, ids) -> List[str]:
         """Map an id tensor (any device) back to strings."""
         arr = ids.cpu().numpy() if hasattr(ids, "cpu") else np.asarray(ids)
@@ -153,9 +182,10 @@ class StringDict:
             return
         data, offs = pack_strings(strings)
         ids = np.arange(len(strings), dtype=np.int32)
+        d_bytes, d_offs = self._stage(data, offs)
         self.k.dict_restore(
-            torch.from_numpy(data).to(self.device),
-            torch.from_numpy(offs).to(self.device),
+            d_bytes,
+            d_offs,
             torch.from_numpy(ids).to(self.device),
             self.dlo, self.dhi, self.dids, self.error_flag,
         )
