@@ -94,6 +94,7 @@ __all__ = [
     "count_window",
     "device_count",
     "device_max",
+    "device_mean",
     "device_min",
     "device_sum",
     "fold_window",
@@ -823,9 +824,18 @@ class DeviceFoldable:
     (`WindowAggState`), chosen by stream type at runtime.
     """
 
-    def __init__(self, mode: str, host_fn: Callable):
+    def __init__(
+        self,
+        mode: str,
+        host_fn: Callable,
+        finish: Optional[Callable] = None,
+    ):
         self.mode = mode
         self._host_fn = host_fn
+        #: Optional host-path finisher applied to the accumulator at
+        #: window close (e.g. mean = sum / count); the columnar path
+        #: computes the finished value in the extract kernel instead.
+        self.finish = finish
 
     def __call__(self, acc, value):
         return self._host_fn(acc, value)
@@ -871,6 +881,22 @@ def device_max(value_getter: Callable = _identity) -> DeviceFoldable:
     builder like ``lambda: -float("inf")`` for the host path."""
     return DeviceFoldable(
         "max", lambda acc, value: max(acc, value_getter(value))
+    )
+
+
+def device_mean(value_getter: Callable = _identity) -> DeviceFoldable:
+    """A `fold_window` mean fold that lowers to the fused stats
+    kernels over RecordBatch streams (tumbling windows): the kernels
+    keep (count, sum) per cell and the close emits sum / count.
+
+    Host path: the accumulator is a ``(sum, count)`` pair — pair with
+    a builder of ``lambda: (0, 0)``; the finisher divides at close so
+    both paths emit the mean itself.
+    """
+    return DeviceFoldable(
+        "mean",
+        lambda acc, value: (acc[0] + value_getter(value), acc[1] + 1),
+        finish=lambda acc: acc[0] / acc[1],
     )
 
 
@@ -939,7 +965,7 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
 
         dev = batch.keys.device
         on_gpu = dev.type != "cpu"
-        if self.spec.mode in ("min", "max"):
+        if self.spec.mode in ("min", "max", "mean"):
             from ..gpu.state import StatsAggState
 
             self.state = StatsAggState(
@@ -980,10 +1006,14 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
         if cols is None:
             return None
         wins = cols["wins"].to(dtype=cols["cnt"].dtype)
+        if self.spec.mode == "mean":
+            vals = cols["sum"].double() / cols["cnt"].double()
+        else:
+            vals = cols[self.spec.mode]
         return RecordBatch(
             cols["keys"],
             wins * self.spec.len_ms + self.spec.align_ms,
-            cols[self.spec.mode],
+            vals,
         )
 
     def on_batch(self, values) -> Tuple[Iterable, bool]:
@@ -991,7 +1021,7 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
         for batch in values:
             self._ensure(batch)
             self.state.insert(batch)
-        if self.spec.mode in ("min", "max"):
+        if self.spec.mode in ("min", "max", "mean"):
             wm = self.state.max_ts_host
             horizon = (
                 wm - self.spec.wait_ms - self.spec.align_ms
@@ -1009,7 +1039,7 @@ class _ColumnarWindowLogic(StatefulBatchLogic):
     def on_eof(self) -> Tuple[Iterable, bool]:
         if self.state is None:
             return ([], True)
-        if self.spec.mode in ("min", "max"):
+        if self.spec.mode in ("min", "max", "mean"):
             final = self._stats_close(None)
         else:
             final = self.state.close_all()
@@ -1239,10 +1269,23 @@ def fold_window(
         return _FoldWindowLogic(folder, merger, state)
 
     mode = folder.mode if isinstance(folder, DeviceFoldable) else None
-    return window(
+    out = window(
         "window", up, clock, windower, shim_builder, ordered=ordered,
         device_mode=mode,
     )
+    fin = folder.finish if isinstance(folder, DeviceFoldable) else None
+    if fin is not None:
+        # Host-path accumulators need the finisher at close; columnar
+        # closes (COLUMNAR_WINDOW_ID) already emit finished values.
+        def _finish_value(wid_acc):
+            wid, acc = wid_acc
+            if wid == COLUMNAR_WINDOW_ID:
+                return wid_acc
+            return (wid, fin(acc))
+
+        downs = op.map_value("finish", out.down, _finish_value)
+        return WindowOut(downs, out.late, out.meta)
+    return out
 
 
 def _collect_list_folder(s: List[V], v: V) -> List[V]:

@@ -310,3 +310,51 @@ def test_fold_window_min_max_lowering_gpu():
     assert _columnar_run(events, folder, "cuda:0") == _minmax_host(
         events, folder, lambda: -1
     )
+
+
+def _mean_host(events, folder):
+    """Host path for device_mean: (sum, count) accumulator + the
+    finisher wired by fold_window (tests the finish map too)."""
+    items = [
+        (datetime.fromtimestamp(ms / 1000, tz=timezone.utc), k, v)
+        for k, ms, v in events
+    ]
+    out = []
+    flow = Dataflow("host_mean")
+    s = op.input("inp", flow, TestingSource(items))
+    keyed = op.key_on("k", s, lambda it: str(it[1]))
+    clock = EventClock(
+        ts_getter=lambda it: it[0], wait_for_system_duration=timedelta(0)
+    )
+    wo = w.fold_window(
+        "fw", keyed, clock,
+        TumblingWindower(align_to=ALIGN, length=WINDOW),
+        lambda: (0, 0), folder,
+        lambda a, b: (a[0] + b[0], a[1] + b[1]),
+    )
+    op.output("out", wo.down, TestingSink(out))
+    run_main(flow)
+    win_len = int(WINDOW.total_seconds() * 1000)
+    return {
+        (int(key), ALIGN_MS + wid * win_len): acc
+        for key, (wid, acc) in out
+    }
+
+
+def test_fold_window_mean_lowering_cpu_twin():
+    events = _mk_events(700)
+    folder = w.device_mean(lambda it: it[2])
+    host = _mean_host(events, folder)
+    assert all(isinstance(v, float) for v in host.values())
+    assert _columnar_run(events, folder) == host
+
+
+@pytest.mark.gpu
+def test_fold_window_mean_lowering_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    events = _mk_events(20_000, vocab=400)
+    folder = w.device_mean(lambda it: it[2])
+    assert _columnar_run(events, folder, "cuda:0") == _mean_host(
+        events, folder
+    )
