@@ -61,9 +61,14 @@ class StringDict:
         self.new_idx = torch.empty(new_cap, dtype=torch.int32, device=device)
         self.new_n = torch.zeros(1, dtype=torch.int32, device=device)
         self.error_flag = torch.zeros(1, dtype=torch.int32, device=device)
-        # Pinned staging for the per-batch H2D of string bytes: the
-        # encode path is H2D-bound (see profiles/r02_kernel_stats.md),
-        # and pageable copies stage through a driver bounce buffer.
+        # Optional pinned staging for the per-batch H2D of string
+        # bytes (BYTEWAX_STR_PINNED=1).  Measured SLOWER than the
+        # direct pageable copy on MI355X (the extra serial host
+        # memcpy into the pinned buffer outweighs the DMA gain; see
+        # profiles/r02_kernel_stats.md), so off by default.
+        import os
+
+        self._pinned = os.environ.get("BYTEWAX_STR_PINNED", "0") == "1"
         self._pin_bytes = None
         self._pin_offs = None
 
@@ -135,12 +140,17 @@ class StringDict:
         return out_ids
 
     def _stage(self, data: np.ndarray, offs: np.ndarray):
-        """Copy packed bytes/offsets through grow-on-demand pinned
-        host buffers and ship them with async DMA copies (both land
-        on the current stream, so the following kernel launch orders
-        after them)."""
+        """Ship packed bytes/offsets to the device: direct pageable
+        copies by default; grow-on-demand pinned staging under
+        BYTEWAX_STR_PINNED=1 (async DMA on the current stream, so the
+        following kernel launch orders after it)."""
         import torch
 
+        if not self._pinned:
+            return (
+                torch.from_numpy(data).to(self.device),
+                torch.from_numpy(offs).to(self.device),
+            )
         nb, no = data.shape[0], offs.shape[0]
         if self._pin_bytes is None or self._pin_bytes.numel() < nb:
             cap = max(1 << 16, 1 << max(0, nb - 1).bit_length())
