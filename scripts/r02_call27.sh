@@ -9,6 +9,8 @@ mkdir -p $REPO/gpurun_out
 : > $L
 cd $REPO
 
+timeout 300 python -m pytest tests/test_gpu_strings.py -m gpu -q >> $L 2>&1
+echo "canary pytest rc=$?" >> $L
 for MODE in 0 1 0 1; do
   echo "=== BYTEWAX_STR_PINNED=$MODE ===" >> $L
   BYTEWAX_STR_PINNED=$MODE timeout 200 python - >> $L 2>&1 <<'PYEOF'
