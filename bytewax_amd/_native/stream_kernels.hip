@@ -76,6 +76,13 @@ static bool scatter_staged2_env() {
   return s != nullptr && std::strncmp(s, "staged2", 7) == 0;
 }
 
+// LDS-deduped region join (BYTEWAX_JOIN_LDS=0 falls back to the
+// per-event region kernel for A/B).
+static bool join_lds_env() {
+  const char* s = std::getenv("BYTEWAX_JOIN_LDS");
+  return s == nullptr || s[0] != '0';
+}
+
 // Extra segment coarsening: scatter segments cover 2^coarse table
 // regions each (longer runs per segment; the agg kernel stages
 // 2^(region_bits+coarse) LDS slots per block).
@@ -1606,11 +1613,53 @@ __global__ void k_stats_extract(
   }
 }
 
-// Stream-stream hash join, "last" insert / "complete" emit semantics
-// (reference operators/__init__.py _JoinLogic): insert this side's
-// value; the insert that makes all sides present emits the joined
-// row and resets the presence flags, so the next emission again
-// requires a fresh value from every side.
+// Per-key global join-cell update, "last" insert / "complete" emit
+// semantics (reference operators/__init__.py _JoinLogic): store this
+// side's value; the update that makes all sides present emits the
+// joined row and resets the presence flags, so the next emission
+// again requires a fresh value from every side.  Shared by the
+// direct insert, the region kernels, and the overflow spill.
+__device__ __forceinline__ void join_apply(
+    uint64_t packed,
+    long long v,
+    int side,
+    int full,
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tval0,
+    long long* __restrict__ tval1,
+    int* __restrict__ tflags,
+    uint64_t mask,
+    int region_bits,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_v0,
+    int64_t* __restrict__ out_v1,
+    int* __restrict__ out_n,
+    int64_t out_cap,
+    int* __restrict__ error_flag) {
+  uint64_t slot = find_slot_r(tkeys, mask, region_bits, packed);
+  if (slot == ~0ULL) {
+    atomicExch(error_flag, 1);
+    return;
+  }
+  if (side == 0) {
+    atomicExch((unsigned long long*)&tval0[slot], (unsigned long long)v);
+  } else {
+    atomicExch((unsigned long long*)&tval1[slot], (unsigned long long)v);
+  }
+  int old = atomicOr(&tflags[slot], 1 << side);
+  if ((old | (1 << side)) == full && old != full) {
+    int idx = atomicAdd(out_n, 1);
+    if (idx < out_cap) {
+      out_keys[idx] = (int32_t)(uint32_t)(packed & 0xFFFFFFFFULL);
+      out_v0[idx] = tval0[slot];
+      out_v1[idx] = tval1[slot];
+    }
+    atomicAnd(&tflags[slot], 0);
+  }
+}
+
+// Direct (unpartitioned) join insert: one global-table update per
+// event.
 __global__ void k_join_insert(
     const int32_t* __restrict__ keys,
     const int64_t* __restrict__ vals,
@@ -1632,30 +1681,9 @@ __global__ void k_join_insert(
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   int full = (1 << n_sides) - 1;
   for (; i < n; i += stride) {
-    uint64_t packed = (uint64_t)(uint32_t)keys[i];
-    uint64_t slot = find_slot(tkeys, mask, packed);
-    if (slot == ~0ULL) {
-      atomicExch(error_flag, 1);
-      continue;
-    }
-    long long v = vals[i];
-    if (side == 0) {
-      atomicExch((unsigned long long*)&tval0[slot],
-                 (unsigned long long)v);
-    } else {
-      atomicExch((unsigned long long*)&tval1[slot],
-                 (unsigned long long)v);
-    }
-    int old = atomicOr(&tflags[slot], 1 << side);
-    if ((old | (1 << side)) == full && old != full) {
-      int idx = atomicAdd(out_n, 1);
-      if (idx < cap) {
-        out_keys[idx] = keys[i];
-        out_v0[idx] = tval0[slot];
-        out_v1[idx] = tval1[slot];
-      }
-      atomicAnd(&tflags[slot], 0);
-    }
+    join_apply((uint64_t)(uint32_t)keys[i], vals[i], side, full, tkeys,
+               tval0, tval1, tflags, mask, 0, out_keys, out_v0, out_v1,
+               out_n, cap, error_flag);
   }
 }
 
@@ -1690,28 +1718,93 @@ __global__ void k_join_region(
   for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
     uint64_t packed = ev_packed[start + j];
     if (packed == EMPTY_SLOT) continue;  // staged-scatter pad
-    uint64_t slot = find_slot_r(tkeys, mask, region_bits, packed);
-    if (slot == ~0ULL) {
-      atomicExch(error_flag, 1);
-      continue;
-    }
-    long long v = ev_vals[start + j];
-    if (side == 0) {
-      atomicExch((unsigned long long*)&tval0[slot],
-                 (unsigned long long)v);
-    } else {
-      atomicExch((unsigned long long*)&tval1[slot],
-                 (unsigned long long)v);
-    }
-    int old = atomicOr(&tflags[slot], 1 << side);
-    if ((old | (1 << side)) == full && old != full) {
-      int idx = atomicAdd(out_n, 1);
-      if (idx < out_cap) {
-        out_keys[idx] = (int32_t)(uint32_t)(packed & 0xFFFFFFFFULL);
-        out_v0[idx] = tval0[slot];
-        out_v1[idx] = tval1[slot];
+    join_apply(packed, ev_vals[start + j], side, full, tkeys, tval0,
+               tval1, tflags, mask, region_bits, out_keys, out_v0,
+               out_v1, out_n, out_cap, error_flag);
+  }
+}
+
+// LDS-deduped region join: one workgroup per region; the region's
+// batch segment is first collapsed in LDS to (key -> last value)
+// for this side, then merged — ONE global-table update per distinct
+// key instead of per event (same staging shape as k_radix_agg).
+// Since each launch carries a single side, a per-batch dedup cannot
+// change emission counts: at most one not-full -> full transition
+// per key per single-side batch either way.
+__global__ __launch_bounds__(256) void k_join_region_lds(
+    const uint64_t* __restrict__ ev_packed,
+    const int64_t* __restrict__ ev_vals,
+    const int* __restrict__ counts,
+    int64_t cap,
+    int side,
+    int n_sides,
+    uint64_t* __restrict__ tkeys,
+    long long* __restrict__ tval0,
+    long long* __restrict__ tval1,
+    int* __restrict__ tflags,
+    uint64_t mask,
+    int region_bits,
+    int lds_bits,
+    int32_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_v0,
+    int64_t* __restrict__ out_v1,
+    int* __restrict__ out_n,
+    int64_t out_cap,
+    int* __restrict__ error_flag) {
+  extern __shared__ char smem[];
+  int R = 1 << lds_bits;
+  uint64_t* lkeys = (uint64_t*)smem;
+  unsigned long long* lvals =
+      (unsigned long long*)(smem + (size_t)R * sizeof(uint64_t));
+  for (int s = threadIdx.x; s < R; s += blockDim.x) {
+    lkeys[s] = EMPTY_SLOT;
+  }
+  __syncthreads();
+  int b = blockIdx.x;
+  int cnt = counts[b];
+  if (cnt > (int)cap) cnt = (int)cap;
+  int64_t start = (int64_t)b * cap;
+  int full = (1 << n_sides) - 1;
+  for (int j = threadIdx.x; j < cnt; j += blockDim.x) {
+    uint64_t packed = ev_packed[start + j];
+    if (packed == EMPTY_SLOT) continue;  // staged-scatter pad
+    unsigned long long v = (unsigned long long)ev_vals[start + j];
+    uint64_t h64 = mix64(packed);
+    int lh = (int)((h64 >> 32) & (R - 1));
+    bool done = false;
+    for (int p = 0; p < R; ++p) {
+      uint64_t cur = lkeys[lh];
+      if (cur == packed) {
+        atomicExch(&lvals[lh], v);
+        done = true;
+        break;
       }
-      atomicAnd(&tflags[slot], 0);
+      if (cur == EMPTY_SLOT) {
+        uint64_t prev = atomicCAS(
+            (unsigned long long*)&lkeys[lh], EMPTY_SLOT, packed);
+        if (prev == EMPTY_SLOT || prev == packed) {
+          atomicExch(&lvals[lh], v);
+          done = true;
+          break;
+        }
+      }
+      lh = (lh + 1) & (R - 1);
+    }
+    if (!done) {
+      // LDS staging full: apply straight to the global region.
+      join_apply(packed, (long long)v, side, full, tkeys, tval0, tval1,
+                 tflags, mask, region_bits, out_keys, out_v0, out_v1,
+                 out_n, out_cap, error_flag);
+    }
+  }
+  __syncthreads();
+  // Merge distinct keys once into the global table (the segment's
+  // region is contiguous in HBM, so the probes stay local).
+  for (int s = threadIdx.x; s < R; s += blockDim.x) {
+    if (lkeys[s] != EMPTY_SLOT) {
+      join_apply(lkeys[s], (long long)lvals[s], side, full, tkeys,
+                 tval0, tval1, tflags, mask, region_bits, out_keys,
+                 out_v0, out_v1, out_n, out_cap, error_flag);
     }
   }
 }
@@ -1742,30 +1835,9 @@ __global__ void k_join_overflow(
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; i < n; i += stride) {
-    uint64_t packed = ov_packed[i];
-    uint64_t slot = find_slot_r(tkeys, mask, region_bits, packed);
-    if (slot == ~0ULL) {
-      atomicExch(error_flag, 1);
-      continue;
-    }
-    long long v = ov_vals[i];
-    if (side == 0) {
-      atomicExch((unsigned long long*)&tval0[slot],
-                 (unsigned long long)v);
-    } else {
-      atomicExch((unsigned long long*)&tval1[slot],
-                 (unsigned long long)v);
-    }
-    int old = atomicOr(&tflags[slot], 1 << side);
-    if ((old | (1 << side)) == full && old != full) {
-      int idx = atomicAdd(out_n, 1);
-      if (idx < out_cap) {
-        out_keys[idx] = (int32_t)(uint32_t)(packed & 0xFFFFFFFFULL);
-        out_v0[idx] = tval0[slot];
-        out_v1[idx] = tval1[slot];
-      }
-      atomicAnd(&tflags[slot], 0);
-    }
+    join_apply(ov_packed[i], ov_vals[i], side, full, tkeys, tval0,
+               tval1, tflags, mask, region_bits, out_keys, out_v0,
+               out_v1, out_n, out_cap, error_flag);
   }
 }
 
@@ -3290,17 +3362,39 @@ void radix_join_insert(
         (unsigned long long*)max_ts_scratch.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>(), (uint64_t)1 << 24, ~(uint64_t)0);
   }
-  hipLaunchKernelGGL(
-      k_join_region, dim3((unsigned)nb), block, 0, stream,
-      (const uint64_t*)ev_packed.data_ptr<int64_t>(),
-      ev_vals.data_ptr<int64_t>(), gcursors.data_ptr<int32_t>(), cap,
-      (int)side, (int)n_sides, (uint64_t*)tkeys.data_ptr<int64_t>(),
-      (long long*)tval0.data_ptr<int64_t>(),
-      (long long*)tval1.data_ptr<int64_t>(), tflags.data_ptr<int32_t>(),
-      mask, (int)region_bits, out_keys.data_ptr<int32_t>(),
-      out_v0.data_ptr<int64_t>(), out_v1.data_ptr<int64_t>(),
-      out_n.data_ptr<int32_t>(), out_keys.numel(),
-      error_flag.data_ptr<int32_t>());
+  // LDS staging sized one bit above the region span (headroom keeps
+  // the in-LDS probe chains short at high per-region cardinality).
+  int slots_bits = 0;
+  while (((int64_t)1 << slots_bits) < nslots) ++slots_bits;
+  int lds_bits = slots_bits - (int)region_bits + 1;
+  if (lds_bits < 6) lds_bits = 6;
+  if (lds_bits > 13) lds_bits = 13;
+  size_t join_lds = (size_t)16 << lds_bits;
+  if (join_lds_env() && join_lds <= 144 * 1024) {
+    hipLaunchKernelGGL(
+        k_join_region_lds, dim3((unsigned)nb), block, join_lds, stream,
+        (const uint64_t*)ev_packed.data_ptr<int64_t>(),
+        ev_vals.data_ptr<int64_t>(), gcursors.data_ptr<int32_t>(), cap,
+        (int)side, (int)n_sides, (uint64_t*)tkeys.data_ptr<int64_t>(),
+        (long long*)tval0.data_ptr<int64_t>(),
+        (long long*)tval1.data_ptr<int64_t>(), tflags.data_ptr<int32_t>(),
+        mask, (int)region_bits, lds_bits, out_keys.data_ptr<int32_t>(),
+        out_v0.data_ptr<int64_t>(), out_v1.data_ptr<int64_t>(),
+        out_n.data_ptr<int32_t>(), out_keys.numel(),
+        error_flag.data_ptr<int32_t>());
+  } else {
+    hipLaunchKernelGGL(
+        k_join_region, dim3((unsigned)nb), block, 0, stream,
+        (const uint64_t*)ev_packed.data_ptr<int64_t>(),
+        ev_vals.data_ptr<int64_t>(), gcursors.data_ptr<int32_t>(), cap,
+        (int)side, (int)n_sides, (uint64_t*)tkeys.data_ptr<int64_t>(),
+        (long long*)tval0.data_ptr<int64_t>(),
+        (long long*)tval1.data_ptr<int64_t>(), tflags.data_ptr<int32_t>(),
+        mask, (int)region_bits, out_keys.data_ptr<int32_t>(),
+        out_v0.data_ptr<int64_t>(), out_v1.data_ptr<int64_t>(),
+        out_n.data_ptr<int32_t>(), out_keys.numel(),
+        error_flag.data_ptr<int32_t>());
+  }
   hipLaunchKernelGGL(
       k_join_overflow, dim3(64), block, 0, stream,
       (const uint64_t*)ov_packed.data_ptr<int64_t>(),
