@@ -63,7 +63,19 @@ class KafkaSourceMessage(Generic[K, V]):
     timestamp: Optional[Tuple[int, int]] = field(default=None)
 
     def to_sink(self) -> "KafkaSinkMessage[K, V]":
-        """Convert to a sink message, keeping key, value, headers."""
+        """Convert to a sink message, keeping key, value, headers.
+
+        Consumed metadata (topic, offset, partition, timestamp) is
+        dropped: the sink assigns its own.
+
+        Example:
+            >>> from bytewax_amd.connectors.kafka import KafkaSourceMessage
+            >>> msg = KafkaSourceMessage(
+            ...     key=b"k", value=b"v", topic="in", offset=17
+            ... )
+            >>> msg.to_sink()
+            KafkaSinkMessage(key=b'k', value=b'v', topic=None, headers=[], partition=None, timestamp=0)
+        """
         return KafkaSinkMessage(
             key=self.key, value=self.value, headers=self.headers
         )

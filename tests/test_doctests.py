@@ -63,3 +63,9 @@ def test_outputs_sdk_docstring_examples():
     import bytewax_amd.outputs as outputs
 
     _run(outputs, 2)
+
+
+def test_kafka_message_docstring_examples():
+    import bytewax_amd.connectors.kafka as kafka
+
+    _run(kafka, 2)
