@@ -1635,7 +1635,12 @@ __device__ __forceinline__ void join_apply(
     int64_t* __restrict__ out_v1,
     int* __restrict__ out_n,
     int64_t out_cap,
-    int* __restrict__ error_flag) {
+    int* __restrict__ error_flag,
+    int rearm = 0) {  // the key had >= 2 events this batch: per-item
+                      // semantics leave the side flag RE-SET after an
+                      // emission (a duplicate lands after the
+                      // completing event clears the flags), so a
+                      // deduped caller must restore it
   uint64_t slot = find_slot_r(tkeys, mask, region_bits, packed);
   if (slot == ~0ULL) {
     atomicExch(error_flag, 1);
@@ -1655,6 +1660,7 @@ __device__ __forceinline__ void join_apply(
       out_v1[idx] = tval1[slot];
     }
     atomicAnd(&tflags[slot], 0);
+    if (rearm) atomicOr(&tflags[slot], 1 << side);
   }
 }
 
@@ -1756,8 +1762,10 @@ __global__ __launch_bounds__(256) void k_join_region_lds(
   uint64_t* lkeys = (uint64_t*)smem;
   unsigned long long* lvals =
       (unsigned long long*)(smem + (size_t)R * sizeof(uint64_t));
+  int* lcnt = (int*)(smem + (size_t)R * 2 * sizeof(uint64_t));
   for (int s = threadIdx.x; s < R; s += blockDim.x) {
     lkeys[s] = EMPTY_SLOT;
+    lcnt[s] = 0;
   }
   __syncthreads();
   int b = blockIdx.x;
@@ -1776,6 +1784,7 @@ __global__ __launch_bounds__(256) void k_join_region_lds(
       uint64_t cur = lkeys[lh];
       if (cur == packed) {
         atomicExch(&lvals[lh], v);
+        atomicAdd(&lcnt[lh], 1);
         done = true;
         break;
       }
@@ -1784,6 +1793,7 @@ __global__ __launch_bounds__(256) void k_join_region_lds(
             (unsigned long long*)&lkeys[lh], EMPTY_SLOT, packed);
         if (prev == EMPTY_SLOT || prev == packed) {
           atomicExch(&lvals[lh], v);
+          atomicAdd(&lcnt[lh], 1);
           done = true;
           break;
         }
@@ -1799,12 +1809,14 @@ __global__ __launch_bounds__(256) void k_join_region_lds(
   }
   __syncthreads();
   // Merge distinct keys once into the global table (the segment's
-  // region is contiguous in HBM, so the probes stay local).
+  // region is contiguous in HBM, so the probes stay local); keys
+  // with >= 2 events keep per-item re-arm semantics via join_apply.
   for (int s = threadIdx.x; s < R; s += blockDim.x) {
     if (lkeys[s] != EMPTY_SLOT) {
       join_apply(lkeys[s], (long long)lvals[s], side, full, tkeys,
                  tval0, tval1, tflags, mask, region_bits, out_keys,
-                 out_v0, out_v1, out_n, out_cap, error_flag);
+                 out_v0, out_v1, out_n, out_cap, error_flag,
+                 lcnt[s] >= 2);
     }
   }
 }
@@ -3369,7 +3381,8 @@ void radix_join_insert(
   int lds_bits = slots_bits - (int)region_bits + 1;
   if (lds_bits < 6) lds_bits = 6;
   if (lds_bits > 13) lds_bits = 13;
-  size_t join_lds = (size_t)16 << lds_bits;
+  // 8 B key + 8 B value + 4 B event count per LDS slot.
+  size_t join_lds = (size_t)20 << lds_bits;
   if (join_lds_env() && join_lds <= 144 * 1024) {
     hipLaunchKernelGGL(
         k_join_region_lds, dim3((unsigned)nb), block, join_lds, stream,

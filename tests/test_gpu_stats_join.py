@@ -183,6 +183,30 @@ def test_join_gpu_matches_reference(radix):
 
 
 @pytest.mark.gpu
+@pytest.mark.parametrize("radix", [False, True])
+def test_join_gpu_hot_key_rearm(radix):
+    """Per-item join semantics under massive key duplication: a batch
+    with >= 2 events of a key leaves that side's flag RE-SET after an
+    emission, so the key joins again on the next opposite-side batch.
+    The LDS-deduped kernel must reproduce this (lcnt re-arm)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    n = 1_000_000
+    k = torch.full((n,), 7, dtype=torch.int32).cuda()
+    v = torch.arange(n, dtype=torch.int64).cuda()
+    st = HashJoinState(
+        torch.device("cuda:0"), slots_pow=17, out_cap=1 << 16, radix=radix
+    )
+    st.insert(0, k, v)  # side 0: flag set, no emit
+    st.insert(1, k, v)  # completes once; dups re-arm side 1
+    st.insert(0, k, v)  # completes again off the re-armed side 1
+    out = st.take_joined()
+    assert out is not None
+    keys, v0, v1 = (t.cpu() for t in out)
+    assert keys.tolist() == [7, 7]
+
+
+@pytest.mark.gpu
 def test_stats_merge_rows_into_occupied_cells_gpu():
     """merge_rows (rescale path) must compose with live cells and
     duplicate rows on device exactly like the host reference."""
