@@ -3319,7 +3319,21 @@ void radix_join_insert(
   TORCH_CHECK((nslots & (nslots - 1)) == 0, "table size must be 2^k");
   TORCH_CHECK(n_sides == 2, "device join currently supports 2 sides");
   TORCH_CHECK(region_bits > 0 && region_bits <= 12, "bad region_bits");
-  int64_t nb = nslots >> region_bits;
+  // Scatter segments one bit coarser than table regions
+  // (BYTEWAX_JOIN_COARSE): halves the scatter's cursor LDS (better
+  // occupancy, longer runs); the dedup/merge kernel probes the
+  // global table by key, so segments need not match regions.
+  int64_t coarse = 1;
+  if (const char* c = std::getenv("BYTEWAX_JOIN_COARSE")) {
+    if (c[0] >= '0' && c[0] <= '3' && c[1] == '\0') coarse = c[0] - '0';
+  }
+  int64_t seg_bits = region_bits + coarse;
+  int64_t nb = nslots >> seg_bits;
+  if (nb < 1) {
+    nb = 1;
+    seg_bits = 0;
+    while (((int64_t)1 << seg_bits) < nslots) ++seg_bits;
+  }
   int64_t cap = ev_packed.numel() / nb;
   // Staged (line-granule) scatter when the per-region capacity
   // supports it; the fixed variant otherwise.
@@ -3344,16 +3358,19 @@ void radix_join_insert(
   ov_cursor.zero_();
   size_t hist_lds = (size_t)nb * sizeof(int);
   // Key-only packing: ts == 0, align 0, huge window -> win = 0.
+  // 512-thread workgroups + grid 1024: same tuning as the stats SUM
+  // scatter (profiles/r02 call 29: the scatter is 63% of join time).
   if (kind == SCAT_STAGED) {
-    unsigned gs = (unsigned)((n + 4095) / 4096);
+    unsigned gs = (unsigned)((n + 8191) / 8192);
     if (gs > 1024) gs = 1024;
     if (gs < 1) gs = 1;
     hipLaunchKernelGGL(
-        (k_radix_scatter_staged<AGG_SUM>), dim3(gs), block, staged_lds,
+        (k_radix_scatter_staged<AGG_SUM, int64_t, 16, 512>), dim3(gs),
+        dim3(512), staged_lds,
         stream, keys.data_ptr<int32_t>(), zeros_ts.data_ptr<int64_t>(),
         vals.data_ptr<int64_t>(), n, 0, (int64_t)1 << 40,
         (int64_t)1 << 40, 0, mask,
-        (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        (int)seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
         ov_cursor.data_ptr<int32_t>(),
         (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
@@ -3366,7 +3383,7 @@ void radix_join_insert(
         keys.data_ptr<int32_t>(), zeros_ts.data_ptr<int64_t>(),
         vals.data_ptr<int64_t>(), n, 0, (int64_t)1 << 40, (int64_t)1 << 40,
         0, mask,
-        (int)region_bits, cap, gcursors.data_ptr<int32_t>(),
+        (int)seg_bits, cap, gcursors.data_ptr<int32_t>(),
         (uint64_t*)ev_packed.data_ptr<int64_t>(), ev_vals.data_ptr<int64_t>(),
         ov_cursor.data_ptr<int32_t>(),
         (uint64_t*)ov_packed.data_ptr<int64_t>(), ov_vals.data_ptr<int64_t>(),
@@ -3374,13 +3391,12 @@ void radix_join_insert(
         (unsigned long long*)max_ts_scratch.data_ptr<int64_t>(),
         error_flag.data_ptr<int32_t>(), (uint64_t)1 << 24, ~(uint64_t)0);
   }
-  // LDS staging sized one bit above the region span (headroom keeps
-  // the in-LDS probe chains short at high per-region cardinality).
-  int slots_bits = 0;
-  while (((int64_t)1 << slots_bits) < nslots) ++slots_bits;
-  int lds_bits = slots_bits - (int)region_bits + 1;
+  // LDS staging sized one bit above the segment's table span
+  // (2^seg_bits cells per segment); headroom keeps the in-LDS probe
+  // chains short at high per-segment cardinality.
+  int lds_bits = (int)seg_bits + 1;
   if (lds_bits < 6) lds_bits = 6;
-  if (lds_bits > 13) lds_bits = 13;
+  if (lds_bits > 12) lds_bits = 12;
   // 8 B key + 8 B value + 4 B event count per LDS slot.
   size_t join_lds = (size_t)20 << lds_bits;
   if (join_lds_env() && join_lds <= 144 * 1024) {

@@ -46,6 +46,68 @@ def _kafka_error_split(step_id: str, up: Stream) -> KafkaOpOut:
     return KafkaOpOut(oks=b.falses, errs=b.trues)
 
 
+def _de_fn(deserializer, field: str):
+    """Adapt a deserializer to `fn(msg, payload) -> obj`.
+
+    Accepts our :class:`SchemaDeserializer` or a confluent-kafka
+    style callable invoked as ``deserializer(payload, ctx)`` with a
+    `SerializationContext` (the reference's serde operators call
+    confluent (de)serializers that way; kafka/operators.py:225-429).
+    """
+    if isinstance(deserializer, SchemaDeserializer):
+        return lambda _msg, payload: deserializer.de(payload)
+    if callable(deserializer):
+
+        def fn(msg, payload):
+            try:
+                from confluent_kafka.serialization import (
+                    MessageField,
+                    SerializationContext,
+                )
+
+                ctx = SerializationContext(
+                    msg.topic, getattr(MessageField, field)
+                )
+            except ImportError:
+                ctx = None
+            return deserializer(payload, ctx)
+
+        return fn
+    msg = (
+        "expected a SchemaDeserializer or confluent-kafka style "
+        f"callable; got {type(deserializer)!r}"
+    )
+    raise TypeError(msg)
+
+
+def _ser_fn(serializer, field: str):
+    """Adapt a serializer to `fn(msg, obj) -> payload` (see _de_fn)."""
+    if isinstance(serializer, SchemaSerializer):
+        return lambda _msg, obj: serializer.ser(obj)
+    if callable(serializer):
+
+        def fn(msg, obj):
+            try:
+                from confluent_kafka.serialization import (
+                    MessageField,
+                    SerializationContext,
+                )
+
+                ctx = SerializationContext(
+                    msg.topic, getattr(MessageField, field)
+                )
+            except ImportError:
+                ctx = None
+            return serializer(obj, ctx)
+
+        return fn
+    msg = (
+        "expected a SchemaSerializer or confluent-kafka style "
+        f"callable; got {type(serializer)!r}"
+    )
+    raise TypeError(msg)
+
+
 @operator
 def input(  # noqa: A001
     step_id: str,
@@ -98,9 +160,11 @@ def deserialize_key(
 ) -> KafkaOpOut:
     """Deserialize message keys; failures go to the `errs` stream."""
 
+    de = _de_fn(deserializer, "KEY")
+
     def shim_mapper(msg: KafkaSourceMessage):
         try:
-            return msg._with_key(deserializer.de(msg.key))
+            return msg._with_key(de(msg, msg.key))
         except Exception as ex:  # noqa: BLE001
             return KafkaError(ex, msg)
 
@@ -116,9 +180,11 @@ def deserialize_value(
 ) -> KafkaOpOut:
     """Deserialize message values; failures go to the `errs` stream."""
 
+    de = _de_fn(deserializer, "VALUE")
+
     def shim_mapper(msg: KafkaSourceMessage):
         try:
-            return msg._with_value(deserializer.de(msg.value))
+            return msg._with_value(de(msg, msg.value))
         except Exception as ex:  # noqa: BLE001
             return KafkaError(ex, msg)
 
@@ -136,10 +202,13 @@ def deserialize(
 ) -> KafkaOpOut:
     """Deserialize both keys and values."""
 
+    kde = _de_fn(key_deserializer, "KEY")
+    vde = _de_fn(val_deserializer, "VALUE")
+
     def shim_mapper(msg: KafkaSourceMessage):
         try:
-            key = key_deserializer.de(msg.key)
-            value = val_deserializer.de(msg.value)
+            key = kde(msg, msg.key)
+            value = vde(msg, msg.value)
             return msg._with_key_and_value(key, value)
         except Exception as ex:  # noqa: BLE001
             return KafkaError(ex, msg)
@@ -156,8 +225,10 @@ def serialize_key(
 ) -> Stream:
     """Serialize message keys; crashes on failure."""
 
+    ser = _ser_fn(serializer, "KEY")
+
     def shim_mapper(msg):
-        return msg._with_key(serializer.ser(msg.key))
+        return msg._with_key(ser(msg, msg.key))
 
     return op.map("map", up, shim_mapper)
 
@@ -170,8 +241,10 @@ def serialize_value(
 ) -> Stream:
     """Serialize message values; crashes on failure."""
 
+    ser = _ser_fn(serializer, "VALUE")
+
     def shim_mapper(msg):
-        return msg._with_value(serializer.ser(msg.value))
+        return msg._with_value(ser(msg, msg.value))
 
     return op.map("map", up, shim_mapper)
 
@@ -186,9 +259,12 @@ def serialize(
 ) -> Stream:
     """Serialize both keys and values."""
 
+    kser = _ser_fn(key_serializer, "KEY")
+    vser = _ser_fn(val_serializer, "VALUE")
+
     def shim_mapper(msg):
         return msg._with_key_and_value(
-            key_serializer.ser(msg.key), val_serializer.ser(msg.value)
+            kser(msg, msg.key), vser(msg, msg.value)
         )
 
     return op.map("map", up, shim_mapper)

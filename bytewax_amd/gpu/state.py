@@ -380,10 +380,10 @@ class HashJoinState:
         import torch
 
         if radix is None:
-            # Direct atomics measured faster with the r1 fixed
-            # scatter; the staged scatter changed the economics —
-            # BYTEWAX_JOIN_RADIX A/Bs it (see profiles/).
-            radix = os.environ.get("BYTEWAX_JOIN_RADIX", "0") == "1"
+            # Radix default: the LDS-deduped region join measures
+            # 1.6x the direct atomics at 1M keys (r02 call 29/30,
+            # profiles/); BYTEWAX_JOIN_RADIX=0 restores direct.
+            radix = os.environ.get("BYTEWAX_JOIN_RADIX", "1") == "1"
         self.device = device
         self.cpu = device.type == "cpu"
         if self.cpu:
