@@ -103,3 +103,46 @@ def test_kafka_source_requires_confluent():
 
     with pytest.raises(ImportError):
         KafkaSource(["localhost:9092"], ["topic"])
+
+
+def test_serde_accepts_confluent_style_callables():
+    """The (de)serialize operators accept confluent-kafka style
+    callables — invoked as fn(payload, ctx) — in addition to our
+    SchemaSerializer/SchemaDeserializer ABCs (reference
+    kafka/operators.py:225-429 calls confluent serializers that
+    way).  Without confluent-kafka installed ctx is None."""
+    msgs = [
+        KafkaSourceMessage(key=b"7", value=b"70", topic="t"),
+        KafkaSourceMessage(key=b"8", value=b"80", topic="t"),
+    ]
+    seen_ctx = []
+
+    def de_int(payload, ctx):
+        seen_ctx.append(ctx)
+        return int(payload)
+
+    out = []
+    flow = Dataflow("confluent_style")
+    s = op.input("inp", flow, TestingSource(msgs))
+    des = kop.deserialize(
+        "de", s, key_deserializer=de_int, val_deserializer=de_int
+    )
+    ser = kop.serialize_value(
+        "ser", des.oks, lambda obj, ctx: str(obj * 2).encode()
+    )
+    op.output("out", ser, TestingSink(out))
+    run_main(flow)
+    assert [(m.key, m.value) for m in out] == [(7, b"140"), (8, b"160")]
+    try:
+        from confluent_kafka.serialization import SerializationContext
+
+        assert all(isinstance(c, SerializationContext) for c in seen_ctx)
+    except ImportError:
+        assert seen_ctx == [None] * 4
+
+
+def test_serde_rejects_non_callables():
+    flow = Dataflow("bad_serde")
+    s = op.input("inp", flow, TestingSource([]))
+    with pytest.raises(TypeError, match="SchemaDeserializer"):
+        kop.deserialize_value("de", s, 42)
