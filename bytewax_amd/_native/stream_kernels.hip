@@ -1737,7 +1737,8 @@ __global__ void k_join_region(
 // Since each launch carries a single side, a per-batch dedup cannot
 // change emission counts: at most one not-full -> full transition
 // per key per single-side batch either way.
-__global__ __launch_bounds__(256) void k_join_region_lds(
+template <int BLK = 256>
+__global__ __launch_bounds__(BLK) void k_join_region_lds(
     const uint64_t* __restrict__ ev_packed,
     const int64_t* __restrict__ ev_vals,
     const int* __restrict__ counts,
@@ -3400,17 +3401,30 @@ void radix_join_insert(
   // 8 B key + 8 B value + 4 B event count per LDS slot.
   size_t join_lds = (size_t)20 << lds_bits;
   if (join_lds_env() && join_lds <= 144 * 1024) {
-    hipLaunchKernelGGL(
-        k_join_region_lds, dim3((unsigned)nb), block, join_lds, stream,
-        (const uint64_t*)ev_packed.data_ptr<int64_t>(),
-        ev_vals.data_ptr<int64_t>(), gcursors.data_ptr<int32_t>(), cap,
-        (int)side, (int)n_sides, (uint64_t*)tkeys.data_ptr<int64_t>(),
-        (long long*)tval0.data_ptr<int64_t>(),
-        (long long*)tval1.data_ptr<int64_t>(), tflags.data_ptr<int32_t>(),
-        mask, (int)region_bits, lds_bits, out_keys.data_ptr<int32_t>(),
-        out_v0.data_ptr<int64_t>(), out_v1.data_ptr<int64_t>(),
-        out_n.data_ptr<int32_t>(), out_keys.numel(),
-        error_flag.data_ptr<int32_t>());
+    // 512 threads/workgroup: the merge kernel is LDS-atomic latency
+    // bound, so more waves in flight per block hide it
+    // (BYTEWAX_JOIN_THREADS sweeps 256/512/1024).
+    int jthreads = 512;
+    if (const char* t = std::getenv("BYTEWAX_JOIN_THREADS")) {
+      int v = std::atoi(t);
+      if (v == 256 || v == 512 || v == 1024) jthreads = v;
+    }
+    auto launch_lds = [&](auto kern, int thr) {
+      hipLaunchKernelGGL(
+          kern, dim3((unsigned)nb), dim3(thr), join_lds, stream,
+          (const uint64_t*)ev_packed.data_ptr<int64_t>(),
+          ev_vals.data_ptr<int64_t>(), gcursors.data_ptr<int32_t>(), cap,
+          (int)side, (int)n_sides, (uint64_t*)tkeys.data_ptr<int64_t>(),
+          (long long*)tval0.data_ptr<int64_t>(),
+          (long long*)tval1.data_ptr<int64_t>(), tflags.data_ptr<int32_t>(),
+          mask, (int)region_bits, lds_bits, out_keys.data_ptr<int32_t>(),
+          out_v0.data_ptr<int64_t>(), out_v1.data_ptr<int64_t>(),
+          out_n.data_ptr<int32_t>(), out_keys.numel(),
+          error_flag.data_ptr<int32_t>());
+    };
+    if (jthreads == 1024) launch_lds(k_join_region_lds<1024>, 1024);
+    else if (jthreads == 512) launch_lds(k_join_region_lds<512>, 512);
+    else launch_lds(k_join_region_lds<256>, 256);
   } else {
     hipLaunchKernelGGL(
         k_join_region, dim3((unsigned)nb), block, 0, stream,
