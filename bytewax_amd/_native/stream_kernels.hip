@@ -2296,6 +2296,83 @@ __global__ void k_dict_insert_pinned(
 }
 
 // ---------------------------------------------------------------------------
+// String exchange: route raw string bytes to their owning rank by
+// CONTENT hash, so multi-GPU str-keyed streams encode at the owner
+// and dictionary ids never cross ranks (bytewax_amd/gpu/strings.py).
+// Wire order is bucket-major: lengths/ts/vals in per-destination
+// contiguous segments, bytes packed in the same relative order
+// (receivers rebuild offsets with a cumsum over lengths).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int str_owner(
+    const uint8_t* __restrict__ bytes, const int64_t* __restrict__ offs,
+    int64_t i, int world) {
+  uint64_t lo, hi;
+  str_hash128(bytes + offs[i], (int)(offs[i + 1] - offs[i]), &lo, &hi);
+  return (int)(hi % (uint64_t)world);
+}
+
+__global__ void k_str_bucket_hist(
+    const uint8_t* __restrict__ bytes,
+    const int64_t* __restrict__ offs,
+    int64_t n,
+    int world,
+    int* __restrict__ counts,            // [world] strings
+    long long* __restrict__ byte_counts  // [world] bytes
+) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    int dst = str_owner(bytes, offs, i, world);
+    atomicAdd(&counts[dst], 1);
+    atomicAdd((unsigned long long*)&byte_counts[dst],
+              (unsigned long long)(offs[i + 1] - offs[i]));
+  }
+}
+
+__global__ void k_str_meta_scatter(
+    const uint8_t* __restrict__ bytes,
+    const int64_t* __restrict__ offs,
+    const int64_t* __restrict__ ts,
+    const int64_t* __restrict__ vals,  // may be nullptr
+    int64_t n,
+    int world,
+    int* __restrict__ cursors,  // [world] exclusive prefix of counts
+    int32_t* __restrict__ send_lens,
+    int64_t* __restrict__ send_ts,
+    int64_t* __restrict__ send_vals,
+    int32_t* __restrict__ src_idx) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    int dst = str_owner(bytes, offs, i, world);
+    int pos = atomicAdd(&cursors[dst], 1);
+    send_lens[pos] = (int32_t)(offs[i + 1] - offs[i]);
+    send_ts[pos] = ts[i];
+    if (vals != nullptr) send_vals[pos] = vals[i];
+    src_idx[pos] = (int32_t)i;
+  }
+}
+
+__global__ void k_str_byte_gather(
+    const uint8_t* __restrict__ bytes,
+    const int64_t* __restrict__ offs,
+    const int32_t* __restrict__ src_idx,
+    const int64_t* __restrict__ send_offs,  // [n] exclusive cumsum of lens
+    int64_t n,
+    uint8_t* __restrict__ send_bytes) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; j < n; j += stride) {
+    int64_t i = src_idx[j];
+    int64_t src = offs[i];
+    int len = (int)(offs[i + 1] - src);
+    int64_t dst = send_offs[j];
+    for (int b = 0; b < len; ++b) send_bytes[dst + b] = bytes[src + b];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Host-side wrappers (torch extension API)
 // ---------------------------------------------------------------------------
 
@@ -3401,10 +3478,11 @@ void radix_join_insert(
   // 8 B key + 8 B value + 4 B event count per LDS slot.
   size_t join_lds = (size_t)20 << lds_bits;
   if (join_lds_env() && join_lds <= 144 * 1024) {
-    // 512 threads/workgroup: the merge kernel is LDS-atomic latency
-    // bound, so more waves in flight per block hide it
-    // (BYTEWAX_JOIN_THREADS sweeps 256/512/1024).
-    int jthreads = 512;
+    // 1024 threads/workgroup: the merge kernel is LDS-atomic latency
+    // bound, so more waves in flight per block hide it (sweep r02
+    // call 31: 256 -> 17.3, 512 -> 20.2, 1024 -> 20.5 Ge/s;
+    // BYTEWAX_JOIN_THREADS overrides).
+    int jthreads = 1024;
     if (const char* t = std::getenv("BYTEWAX_JOIN_THREADS")) {
       int v = std::atoi(t);
       if (v == 256 || v == 512 || v == 1024) jthreads = v;
@@ -4673,7 +4751,74 @@ void dict_restore(
       (uint64_t)(nslots - 1), error_flag.data_ptr<int32_t>());
 }
 
+// Pack a str batch into per-destination-rank wire segments for the
+// all-to-allv string exchange (owner = content hash % world).
+// Returns nothing; fills `counts` (strings/rank), `byte_counts`
+// (bytes/rank), `send_lens`/`send_ts`/`send_vals` (bucket-major wire
+// order) and `send_bytes`.  All device work stays on the current
+// stream (cumsums included); callers read counts to host for the
+// collective split sizes.
+void str_exchange_pack(
+    torch::Tensor bytes,   // uint8 [total]
+    torch::Tensor offs,    // int64 [n+1]
+    torch::Tensor ts,      // int64 [n]
+    c10::optional<torch::Tensor> vals,  // int64 [n]
+    int64_t world,
+    torch::Tensor counts,       // int32 [world] out (zeroed here)
+    torch::Tensor byte_counts,  // int64 [world] out (zeroed here)
+    torch::Tensor send_lens,    // int32 [n] out
+    torch::Tensor send_ts,      // int64 [n] out
+    torch::Tensor send_vals,    // int64 [n or 0] out
+    torch::Tensor send_bytes    // uint8 [total] out
+) {
+  check_dev(bytes, torch::kUInt8, "bytes");
+  check_dev(offs, torch::kInt64, "offs");
+  check_dev(ts, torch::kInt64, "ts");
+  check_dev(send_lens, torch::kInt32, "send_lens");
+  int64_t n = offs.numel() - 1;
+  TORCH_CHECK(world >= 1, "world must be >= 1");
+  TORCH_CHECK(send_lens.numel() >= n, "send_lens too small");
+  TORCH_CHECK(send_bytes.numel() >= bytes.numel(), "send_bytes too small");
+  counts.zero_();
+  byte_counts.zero_();
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 block(256);
+  dim3 grid(n_blocks(n, 256));
+  hipLaunchKernelGGL(
+      k_str_bucket_hist, grid, block, 0, stream,
+      bytes.data_ptr<uint8_t>(), offs.data_ptr<int64_t>(), n, (int)world,
+      counts.data_ptr<int32_t>(),
+      (long long*)byte_counts.data_ptr<int64_t>());
+  auto cursors =
+      (torch::cumsum(counts, 0, torch::kInt32) - counts).contiguous();
+  const int64_t* vptr =
+      vals.has_value() ? vals->data_ptr<int64_t>() : nullptr;
+  auto src_idx = torch::empty(
+      {n}, torch::TensorOptions()
+               .dtype(torch::kInt32)
+               .device(bytes.device()));
+  hipLaunchKernelGGL(
+      k_str_meta_scatter, grid, block, 0, stream,
+      bytes.data_ptr<uint8_t>(), offs.data_ptr<int64_t>(),
+      ts.data_ptr<int64_t>(), vptr, n, (int)world,
+      cursors.data_ptr<int32_t>(), send_lens.data_ptr<int32_t>(),
+      send_ts.data_ptr<int64_t>(),
+      vptr != nullptr ? send_vals.data_ptr<int64_t>() : nullptr,
+      src_idx.data_ptr<int32_t>());
+  auto lens64 = send_lens.narrow(0, 0, n).to(torch::kInt64);
+  auto send_offs = (torch::cumsum(lens64, 0) - lens64).contiguous();
+  hipLaunchKernelGGL(
+      k_str_byte_gather, grid, block, 0, stream,
+      bytes.data_ptr<uint8_t>(), offs.data_ptr<int64_t>(),
+      src_idx.data_ptr<int32_t>(), send_offs.data_ptr<int64_t>(), n,
+      send_bytes.data_ptr<uint8_t>());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("str_exchange_pack", &str_exchange_pack,
+        "pack a str batch into per-rank wire segments (content-hash "
+        "owner) for the all-to-allv string exchange");
   m.def("session_radix_insert", &session_radix_insert,
         "fused radix session insert: AGG_TS scatter + per-segment LDS "
         "session aggregation + sequential overflow walk");

@@ -32,6 +32,171 @@ def pack_strings(strings: Sequence[str]) -> Tuple[np.ndarray, np.ndarray]:
     return data, offs
 
 
+def str_owner_cpu(b: bytes, world: int) -> int:
+    """CPU twin of the device content-hash owner (same FNV/mix
+    pipeline as `str_hash128`, so CPU tests predict device routing)."""
+    h = 0xCBF29CE484222325
+    for c in b:
+        h = ((h ^ c) * 0x100000001B3) & 0xFFFFFFFFFFFFFFFF
+    hi = _mix64((h + 0x2545F4914F6CDD1D) & 0xFFFFFFFFFFFFFFFF)
+    return hi % world
+
+
+def _mix64(x: int) -> int:
+    m = 0xFFFFFFFFFFFFFFFF
+    x = (x ^ (x >> 33)) & m
+    x = (x * 0xFF51AFD7ED558CCD) & m
+    x = (x ^ (x >> 33)) & m
+    x = (x * 0xC4CEB9FE1A85EC53) & m
+    return (x ^ (x >> 33)) & m
+
+
+def exchange_str_by_key(
+    data, offs, ts, vals=None, group=None, force: bool = False
+):
+    """Exchange a str-keyed batch so every string's events land on
+    the rank that owns the string by CONTENT hash: the multi-GPU
+    str-keyed design (dictionary ids are per-rank, so raw bytes are
+    exchanged and encoded at the owner — never ids).
+
+    Inputs are device tensors (uint8 bytes, int64 offs[n+1], int64
+    ts[n], optional int64 vals[n]) or numpy/CPU tensors for the gloo
+    twin.  Returns `(data, offs, ts, vals)` in the same form.
+    Collective: every rank must call this once per scheduling step.
+    """
+    import torch
+    import torch.distributed as dist
+
+    from ._ext import ext as _ext_mod
+
+    world = dist.get_world_size(group)
+    if world == 1 and not force:
+        return data, offs, ts, vals
+    dev = (
+        data.device
+        if isinstance(data, torch.Tensor)
+        else torch.device("cpu")
+    )
+    if dev.type == "cpu":
+        # CPU twin: python bucketing + gloo all-to-all.
+        data_np = np.asarray(data, dtype=np.uint8)
+        offs_np = np.asarray(offs, dtype=np.int64)
+        ts_np = np.asarray(ts.numpy() if hasattr(ts, "numpy") else ts)
+        vals_np = None
+        if vals is not None:
+            vals_np = np.asarray(
+                vals.numpy() if hasattr(vals, "numpy") else vals
+            )
+        n = len(offs_np) - 1
+        buckets = [[] for _ in range(world)]
+        for i in range(n):
+            b = bytes(data_np[offs_np[i] : offs_np[i + 1]])
+            buckets[str_owner_cpu(b, world)].append(i)
+        send_lens, send_ts, send_vals, send_bytes = [], [], [], []
+        counts = []
+        for idxs in buckets:
+            counts.append(len(idxs))
+            for i in idxs:
+                send_lens.append(int(offs_np[i + 1] - offs_np[i]))
+                send_ts.append(int(ts_np[i]))
+                if vals_np is not None:
+                    send_vals.append(int(vals_np[i]))
+                send_bytes.append(
+                    data_np[offs_np[i] : offs_np[i + 1]].tobytes()
+                )
+        byte_counts = []
+        k = 0
+        for c in counts:
+            byte_counts.append(sum(send_lens[k : k + c]))
+            k += c
+        t_counts = torch.tensor(counts, dtype=torch.int32)
+        t_bcounts = torch.tensor(byte_counts, dtype=torch.int64)
+        t_lens = torch.tensor(send_lens, dtype=torch.int32)
+        t_ts = torch.tensor(send_ts, dtype=torch.int64)
+        t_vals = (
+            torch.tensor(send_vals, dtype=torch.int64)
+            if vals_np is not None
+            else None
+        )
+        t_bytes = torch.frombuffer(
+            bytearray(b"".join(send_bytes)), dtype=torch.uint8
+        ) if send_bytes else torch.empty(0, dtype=torch.uint8)
+    else:
+        k = _ext_mod()
+        n = int(offs.numel()) - 1
+        t_counts = torch.zeros(world, dtype=torch.int32, device=dev)
+        t_bcounts = torch.zeros(world, dtype=torch.int64, device=dev)
+        t_lens = torch.empty(max(n, 1), dtype=torch.int32, device=dev)
+        t_ts = torch.empty(max(n, 1), dtype=torch.int64, device=dev)
+        t_vals = (
+            torch.empty(max(n, 1), dtype=torch.int64, device=dev)
+            if vals is not None
+            else None
+        )
+        t_bytes = torch.empty(
+            max(int(data.numel()), 1), dtype=torch.uint8, device=dev
+        )
+        k.str_exchange_pack(
+            data, offs, ts, vals, world, t_counts, t_bcounts,
+            t_lens,
+            t_ts,
+            t_vals if t_vals is not None
+            else torch.empty(0, dtype=torch.int64, device=dev),
+            t_bytes,
+        )
+        t_lens = t_lens[:n]
+        t_ts = t_ts[:n]
+        if t_vals is not None:
+            t_vals = t_vals[:n]
+        t_bytes = t_bytes[: int(data.numel())]
+
+    # Control plane: exchange split sizes (strings and bytes).
+    in_splits = t_counts.tolist()
+    in_bsplits = (
+        t_bcounts.tolist()
+        if dev.type == "cpu"
+        else t_bcounts.cpu().tolist()
+    )
+    recv_counts = torch.empty_like(t_counts)
+    dist.all_to_all_single(recv_counts, t_counts.contiguous(), group=group)
+    bc = t_bcounts.contiguous()
+    recv_bcounts = torch.empty_like(bc)
+    dist.all_to_all_single(recv_bcounts, bc, group=group)
+    out_splits = recv_counts.tolist()
+    out_bsplits = (
+        recv_bcounts.tolist()
+        if dev.type == "cpu"
+        else recv_bcounts.cpu().tolist()
+    )
+    m = int(sum(out_splits))
+    mb = int(sum(out_bsplits))
+
+    recv_lens = torch.empty(m, dtype=torch.int32, device=dev)
+    recv_ts = torch.empty(m, dtype=torch.int64, device=dev)
+    recv_bytes = torch.empty(mb, dtype=torch.uint8, device=dev)
+    dist.all_to_all_single(
+        recv_lens, t_lens.contiguous(), out_splits, in_splits, group=group
+    )
+    dist.all_to_all_single(
+        recv_ts, t_ts.contiguous(), out_splits, in_splits, group=group
+    )
+    dist.all_to_all_single(
+        recv_bytes, t_bytes.contiguous(), out_bsplits, in_bsplits,
+        group=group,
+    )
+    recv_vals = None
+    if t_vals is not None:
+        recv_vals = torch.empty(m, dtype=torch.int64, device=dev)
+        dist.all_to_all_single(
+            recv_vals, t_vals.contiguous(), out_splits, in_splits,
+            group=group,
+        )
+    lens64 = recv_lens.to(torch.int64)
+    recv_offs = torch.zeros(m + 1, dtype=torch.int64, device=dev)
+    recv_offs[1:] = torch.cumsum(lens64, 0)
+    return recv_bytes, recv_offs, recv_ts, recv_vals
+
+
 class StringDict:
     """Keyed-stream string dictionary with a device-resident table.
 

@@ -202,3 +202,81 @@ def test_sliding_window_exchange_2rank_gloo(tmp_path):
     for p, (so, se) in zip(procs, outs):
         assert p.returncode == 0, se.decode()[-2000:]
     assert "SLIDE2_OK" in outs[0][0].decode()
+
+
+_STR_PROG = """
+import os
+
+import torch
+import torch.distributed as dist
+
+from bytewax_amd.gpu.strings import (
+    exchange_str_by_key,
+    pack_strings,
+    str_owner_cpu,
+)
+
+dist.init_process_group("gloo")
+rank = dist.get_rank()
+
+import random
+
+rng = random.Random(31 + rank)
+words = [f"word-{rng.randrange(400)}" for _ in range(3000)]
+data, offs = pack_strings(words)
+ts = torch.arange(3000, dtype=torch.int64) + rank * 1_000_000
+vals = torch.arange(3000, dtype=torch.int64) * (rank + 1)
+
+rb, ro, rt, rv = exchange_str_by_key(
+    data, offs, ts, vals
+)
+got = []
+rb_np = rb.numpy()
+ro_np = ro.numpy()
+for i in range(len(ro_np) - 1):
+    s = bytes(rb_np[ro_np[i] : ro_np[i + 1]]).decode()
+    got.append((s, int(rt[i]), int(rv[i])))
+# Routing: every received string owned by this rank.
+for s, _t, _v in got:
+    assert str_owner_cpu(s.encode(), 2) == rank, s
+# Conservation: the union of received triples == union of sent.
+sent = [
+    (w, int(t), int(v)) for w, t, v in zip(words, ts.tolist(), vals.tolist())
+]
+gathered = [None, None]
+dist.all_gather_object(gathered, (sent, got))
+if rank == 0:
+    all_sent = sorted(gathered[0][0] + gathered[1][0])
+    all_got = sorted(gathered[0][1] + gathered[1][1])
+    assert all_sent == all_got, (len(all_sent), len(all_got))
+    print("STR EXCHANGE OK")
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.timeout(180)
+def test_str_exchange_two_ranks_gloo(tmp_path: Path):
+    """String-key exchange at world 2 (CPU twin): raw bytes routed by
+    content hash, events conserved, every string wholly owned by one
+    rank (the multi-GPU str-keyed design in gpu/strings.py)."""
+    prog = tmp_path / "strex.py"
+    prog.write_text(textwrap.dedent(_STR_PROG))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["MASTER_PORT"] = str(29700 + os.getpid() % 60)
+    procs = []
+    for rank in range(2):
+        e = dict(env, RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK=str(rank))
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, str(prog)],
+                env=e,
+                stdout=subprocess.PIPE,
+                stderr=subprocess.PIPE,
+            )
+        )
+    outs = [p.communicate(timeout=150) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se.decode()[-1500:]
+    assert "STR EXCHANGE OK" in outs[0][0].decode()

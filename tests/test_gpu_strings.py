@@ -7,6 +7,8 @@ exactly.
 """
 
 import random
+
+import numpy as np
 from datetime import datetime, timedelta, timezone
 
 import pytest
@@ -155,3 +157,65 @@ def test_string_dict_gpu_large_random():
     assert len(d) == 200_000
     assert sorted(ids.cpu().tolist()) == list(range(200_000))
     assert d.decode(ids[:100]) == vocab[:100]
+
+
+@pytest.mark.gpu
+def test_str_exchange_pack_matches_cpu_twin():
+    """The device string-exchange pack (hash-bucket + meta scatter +
+    byte gather) must agree with the CPU twin: same per-rank string
+    and byte counts, and each wire segment's (string, ts, val)
+    multiset equals the twin's bucket."""
+    import torch
+
+    from bytewax_amd.gpu import ext
+    from bytewax_amd.gpu.strings import pack_strings, str_owner_cpu
+
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    rng = random.Random(13)
+    world = 4
+    words = [f"w{rng.randrange(500)}-{rng.randrange(7)}" for _ in range(20_000)]
+    data, offs = pack_strings(words)
+    n = len(words)
+    ts = list(range(n))
+    vals = [i * 3 for i in range(n)]
+
+    dev = torch.device("cuda:0")
+    d_bytes = torch.from_numpy(data).to(dev)
+    d_offs = torch.from_numpy(offs).to(dev)
+    d_ts = torch.tensor(ts, dtype=torch.int64, device=dev)
+    d_vals = torch.tensor(vals, dtype=torch.int64, device=dev)
+    counts = torch.zeros(world, dtype=torch.int32, device=dev)
+    bcounts = torch.zeros(world, dtype=torch.int64, device=dev)
+    send_lens = torch.empty(n, dtype=torch.int32, device=dev)
+    send_ts = torch.empty(n, dtype=torch.int64, device=dev)
+    send_vals = torch.empty(n, dtype=torch.int64, device=dev)
+    send_bytes = torch.empty(len(data), dtype=torch.uint8, device=dev)
+    ext().str_exchange_pack(
+        d_bytes, d_offs, d_ts, d_vals, world, counts, bcounts,
+        send_lens, send_ts, send_vals, send_bytes,
+    )
+
+    twin = [[] for _ in range(world)]
+    for i, w in enumerate(words):
+        twin[str_owner_cpu(w.encode(), world)].append(
+            (w, ts[i], vals[i])
+        )
+    assert counts.cpu().tolist() == [len(b) for b in twin]
+    assert bcounts.cpu().tolist() == [
+        sum(len(w.encode()) for w, _t, _v in b) for b in twin
+    ]
+    # Reconstruct wire strings per segment and compare multisets.
+    lens = send_lens.cpu().numpy()
+    sb = send_bytes.cpu().numpy()
+    sts = send_ts.cpu().tolist()
+    svs = send_vals.cpu().tolist()
+    boffs = np.concatenate([[0], np.cumsum(lens)])
+    pos = 0
+    for r, bucket in enumerate(twin):
+        seg = []
+        for j in range(pos, pos + len(bucket)):
+            s = bytes(sb[boffs[j] : boffs[j + 1]]).decode()
+            seg.append((s, sts[j], svs[j]))
+        assert sorted(seg) == sorted(bucket), f"rank {r}"
+        pos += len(bucket)
