@@ -640,10 +640,13 @@ class _StrWindowLogic(StatefulBatchLogic):
     fold_window semantics under watermark close).
     """
 
-    def __init__(self, sdict, state, wait_ms: int, resume):
+    def __init__(
+        self, sdict, state, wait_ms: int, resume, exchange: bool = False
+    ):
         self.sdict = sdict
         self.state = state
         self.wait_ms = wait_ms
+        self.exchange = exchange
         if resume is not None:
             self.sdict.restore(resume["dict"])
             self.state.restore_from_host(resume["win"])
@@ -663,7 +666,6 @@ class _StrWindowLogic(StatefulBatchLogic):
         for item in batches:
             strings, ts = item[0], item[1]
             vals = item[2] if len(item) > 2 else None
-            ids = self.sdict.encode(strings)
             dev = self.state.device
             ts_t = torch.as_tensor(ts, dtype=torch.int64).to(dev)
             vals_t = (
@@ -671,6 +673,38 @@ class _StrWindowLogic(StatefulBatchLogic):
                 if vals is not None
                 else None
             )
+            if self.exchange:
+                # Multi-GPU: route raw bytes by CONTENT hash so every
+                # string is encoded at its owning rank (per-rank
+                # dictionary ids never cross ranks).  Collective:
+                # every rank feeds one batch per scheduling step.
+                import numpy as np
+
+                from .strings import exchange_str_by_key, pack_strings
+
+                if isinstance(strings, tuple):
+                    data, offs = strings
+                else:
+                    data, offs = pack_strings(strings)
+                if not isinstance(data, torch.Tensor):
+                    data = torch.from_numpy(
+                        np.asarray(data, dtype=np.uint8)
+                    )
+                    offs = torch.from_numpy(
+                        np.asarray(offs, dtype=np.int64)
+                    )
+                if dev.type != "cpu":
+                    data, offs = data.to(dev), offs.to(dev)
+                data, offs, ts_t, vals_t = exchange_str_by_key(
+                    data, offs, ts_t, vals_t
+                )
+                ids = self.sdict.encode(
+                    (data, offs)
+                    if dev.type != "cpu"
+                    else (data.numpy(), offs.numpy())
+                )
+            else:
+                ids = self.sdict.encode(strings)
             self.state.insert(
                 RecordBatch(ids.to(dev), ts_t, vals_t)
             )
@@ -703,6 +737,7 @@ def keyed_window_agg_str(
     dict_slots_pow: int = 21,
     out_cap: int = 1 << 20,
     device: str = "cuda",
+    exchange: bool = False,
 ) -> Stream:
     """Str-keyed tumbling-window aggregation on GPU.
 
@@ -716,6 +751,12 @@ def keyed_window_agg_str(
     ``(key_str, win_start_ms, value)`` tuples at watermark close,
     matching the host windowing path's rows exactly (gpu test
     `tests/test_gpu_strings.py`).
+
+    With ``exchange=True`` (multi-GPU) raw string bytes are routed to
+    their owning rank by content hash before encoding
+    (:func:`bytewax_amd.gpu.strings.exchange_str_by_key`), so each
+    rank's dictionary only ever holds the strings it owns.
+    Collective: every rank must feed one batch per scheduling step.
     """
     import torch
 
@@ -734,7 +775,9 @@ def keyed_window_agg_str(
             slots_pow=slots_pow, out_cap=out_cap,
             radix=dev.type != "cpu",
         )
-        return _StrWindowLogic(sdict, state, wait_ms, resume_state)
+        return _StrWindowLogic(
+            sdict, state, wait_ms, resume_state, exchange=exchange
+        )
 
     keyed = op.map("wrap", up, lambda b: ("shard-0", b))
     agg = op.stateful_batch("agg", keyed, shim_builder)

@@ -259,9 +259,23 @@ class StringDict:
 
     def encode(self, batch: StrBatch):
         """Encode a batch of strings -> int32 id tensor on the device
-        (CPU twin returns a CPU tensor)."""
+        (CPU twin returns a CPU tensor).
+
+        Also accepts a ``(uint8 bytes, int64 offs)`` pair of DEVICE
+        tensors (e.g. straight from :func:`exchange_str_by_key`): the
+        kernels consume them in place and host bytes are fetched only
+        when the batch introduced new ids.
+        """
         import torch
 
+        if (
+            not self.cpu
+            and isinstance(batch, tuple)
+            and len(batch) == 2
+            and isinstance(batch[0], torch.Tensor)
+            and batch[0].device.type != "cpu"
+        ):
+            return self._encode_dev(batch[0], batch[1])
         data, offs, strings = self._as_packed(batch)
         n = len(offs) - 1
         if self.cpu:
@@ -276,7 +290,18 @@ class StringDict:
                 ids[i] = j
             return torch.from_numpy(ids)
         d_bytes, d_offs = self._stage(data, offs)
+        return self._run_encode(d_bytes, d_offs, n, data, offs, strings)
+
+    def _encode_dev(self, d_bytes, d_offs):
+        n = int(d_offs.numel()) - 1
+        return self._run_encode(d_bytes, d_offs, n, None, None, None)
+
+    def _run_encode(self, d_bytes, d_offs, n, data, offs, strings):
+        import torch
+
         out_ids = torch.empty(n, dtype=torch.int32, device=self.device)
+        if n == 0:
+            return out_ids
         self.k.dict_encode(
             d_bytes, d_offs, self.dlo, self.dhi, self.dids, self.counter,
             self.new_ids, self.new_idx, self.new_n, out_ids,
@@ -293,6 +318,11 @@ class StringDict:
             }.get(code, f"string dictionary error {code}")
             raise RuntimeError(msg)
         if n_new:
+            if data is None:
+                # Device-tensor input: fetch host bytes only when the
+                # batch introduced new ids.
+                data = d_bytes.cpu().numpy()
+                offs = d_offs.cpu().numpy()
             ids_h = self.new_ids[:n_new].cpu().numpy()
             idx_h = self.new_idx[:n_new].cpu().numpy()
             need = len(self.id2str) + n_new

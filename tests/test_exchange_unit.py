@@ -280,3 +280,84 @@ def test_str_exchange_two_ranks_gloo(tmp_path: Path):
     for p, (so, se) in zip(procs, outs):
         assert p.returncode == 0, se.decode()[-1500:]
     assert "STR EXCHANGE OK" in outs[0][0].decode()
+
+
+_STR_E2E_PROG = """
+import os
+import random
+
+import torch
+import torch.distributed as dist
+
+from bytewax_amd.gpu import AGG_COUNT, WindowAggState
+from bytewax_amd.gpu.operators import _StrWindowLogic
+from bytewax_amd.gpu.strings import StringDict
+
+dist.init_process_group("gloo")
+rank = dist.get_rank()
+
+dev = torch.device("cpu")
+sdict = StringDict(dev)
+state = WindowAggState(
+    dev, 0, 60_000, AGG_COUNT, slots_pow=16, out_cap=1 << 16, radix=False
+)
+logic = _StrWindowLogic(sdict, state, 0, None, exchange=True)
+
+rng = random.Random(7 + rank)
+B, N = 4, 800
+sent = []
+out = []
+for b in range(B):
+    words = [f"user-{rng.randrange(50)}" for _ in range(N)]
+    ts = [b * 30_000 + i % 1000 for i in range(N)]
+    sent.extend(zip(words, ts))
+    rows, _ = logic.on_batch([(words, ts)])
+    out.extend(rows)
+rows, _ = logic.on_eof()
+out.extend(rows)
+
+gathered = [None, None]
+dist.all_gather_object(gathered, (sent, out))
+if rank == 0:
+    from collections import Counter
+
+    truth = Counter()
+    for sent_r, _o in gathered:
+        for w, t in sent_r:
+            truth[(w, (t // 60_000) * 60_000)] += 1
+    got = Counter()
+    for _s, out_r in gathered:
+        for key, win_ms, v in out_r:
+            got[(key, win_ms)] += v
+    assert got == truth, (len(got), len(truth))
+    print("STR E2E OK")
+dist.destroy_process_group()
+"""
+
+
+@pytest.mark.timeout(180)
+def test_str_windowing_exchange_two_ranks_gloo(tmp_path: Path):
+    """keyed_window_agg_str's exchange path at world 2 (CPU twins):
+    windowed counts over the union of both ranks' str events equal
+    the brute force, with each string counted at its owning rank."""
+    prog = tmp_path / "stre2e.py"
+    prog.write_text(textwrap.dedent(_STR_E2E_PROG))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    env["MASTER_PORT"] = str(29640 + os.getpid() % 60)
+    procs = []
+    for rank in range(2):
+        e = dict(env, RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK=str(rank))
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, str(prog)],
+                env=e,
+                stdout=subprocess.PIPE,
+                stderr=subprocess.PIPE,
+            )
+        )
+    outs = [p.communicate(timeout=150) for p in procs]
+    for p, (so, se) in zip(procs, outs):
+        assert p.returncode == 0, se.decode()[-1500:]
+    assert "STR E2E OK" in outs[0][0].decode()
