@@ -1457,9 +1457,17 @@ def max_window(
     >>> run_main(flow)
     >>> [(k, (wid, v[1])) for k, (wid, v) in out]
     [('ALL', (0, 9)), ('ALL', (1, 3))]
+
+    Over :class:`bytewax_amd.gpu.RecordBatch` streams this lowers
+    onto the fused stats kernels (the batch's `vals` column is the
+    compared value; `by` applies to host items only).
     """
     return reduce_window(
-        "reduce_window", up, clock, windower, partial(max, key=by)
+        "reduce_window",
+        up,
+        clock,
+        windower,
+        DeviceFoldable("max", partial(max, key=by)),
     )
 
 
@@ -1499,9 +1507,17 @@ def min_window(
     >>> run_main(flow)
     >>> [(k, (wid, v[1])) for k, (wid, v) in out]
     [('ALL', (0, 5)), ('ALL', (1, 3))]
+
+    Over :class:`bytewax_amd.gpu.RecordBatch` streams this lowers
+    onto the fused stats kernels (the batch's `vals` column is the
+    compared value; `by` applies to host items only).
     """
     return reduce_window(
-        "reduce_window", up, clock, windower, partial(min, key=by)
+        "reduce_window",
+        up,
+        clock,
+        windower,
+        DeviceFoldable("min", partial(min, key=by)),
     )
 
 
@@ -1547,6 +1563,15 @@ def reduce_window(
 
     def shim_folder(s, v):
         return v if s is None else reducer(s, v)
+
+    if isinstance(reducer, DeviceFoldable):
+        # Keep the device-lowering mode through the reduce shim
+        # (max_window/min_window pass device folds so RecordBatch
+        # streams hit the stats kernels; host items still reduce
+        # with the first value as the initial accumulator).
+        shim_folder = DeviceFoldable(
+            reducer.mode, shim_folder, finish=reducer.finish
+        )
 
     def none_builder():
         return None

@@ -358,3 +358,66 @@ def test_fold_window_mean_lowering_gpu():
     assert _columnar_run(events, folder, "cuda:0") == _mean_host(
         events, folder
     )
+
+
+def _minmax_columnar_run(events, which, device="cpu", batch=37):
+    """max_window/min_window over RecordBatch streams (auto
+    lowering through the reduce shim's DeviceFoldable)."""
+    batches = []
+    for i in range(0, len(events), batch):
+        chunk = events[i : i + batch]
+        batches.append(
+            RecordBatch(
+                torch.tensor([k for k, _, _ in chunk], dtype=torch.int32,
+                             device=device),
+                torch.tensor([ms for _, ms, _ in chunk], dtype=torch.int64,
+                             device=device),
+                torch.tensor([v for _, _, v in chunk], dtype=torch.int64,
+                             device=device),
+            )
+        )
+    out = []
+    flow = Dataflow("columnar_minmax_pub")
+    s = op.input("inp", flow, TestingSource(batches))
+    keyed = op.key_on("k", s, lambda b: "shard-0")
+    clock = EventClock(
+        ts_getter=lambda it: it, wait_for_system_duration=timedelta(0)
+    )
+    fn = w.max_window if which == "max" else w.min_window
+    wo = fn("mw", keyed, clock,
+            TumblingWindower(align_to=ALIGN, length=WINDOW))
+    op.output("out", wo.down, TestingSink(out))
+    run_main(flow)
+    res = {}
+    for _key, (wid, rb) in out:
+        assert wid == COLUMNAR_WINDOW_ID
+        for k, ms, v in zip(
+            rb.keys.cpu().tolist(), rb.ts.cpu().tolist(),
+            rb.vals.cpu().tolist(),
+        ):
+            res[(k, ms)] = v
+    return res
+
+
+def test_public_max_min_window_lower_on_columnar_cpu_twin():
+    events = _mk_events(600)
+    win_len = int(WINDOW.total_seconds() * 1000)
+    for which, red in (("max", max), ("min", min)):
+        brute = {}
+        for k, ms, v in events:
+            cell = (k, ALIGN_MS + ((ms - ALIGN_MS) // win_len) * win_len)
+            brute[cell] = red(brute.get(cell, v), v)
+        assert _minmax_columnar_run(events, which) == brute, which
+
+
+@pytest.mark.gpu
+def test_public_max_window_lowers_on_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    events = _mk_events(20_000, vocab=300)
+    win_len = int(WINDOW.total_seconds() * 1000)
+    brute = {}
+    for k, ms, v in events:
+        cell = (k, ALIGN_MS + ((ms - ALIGN_MS) // win_len) * win_len)
+        brute[cell] = max(brute.get(cell, v), v)
+    assert _minmax_columnar_run(events, "max", "cuda:0") == brute
