@@ -757,6 +757,9 @@ def keyed_window_agg_str(
     (:func:`bytewax_amd.gpu.strings.exchange_str_by_key`), so each
     rank's dictionary only ever holds the strings it owns.
     Collective: every rank must feed one batch per scheduling step.
+    Snapshots restore at the SAME world size (ownership is
+    ``hash % world``; resuming at a different world would strand
+    state at its old owner — rescale through the host path).
     """
     import torch
 
@@ -1054,6 +1057,11 @@ def stream_join(
     is unspecified, mirroring the reference's unspecified cross-worker
     arrival order.  Pre-deduplicate per batch if you need the serial
     "last" semantics exactly.
+
+    Engine: by default each side's batch is radix-partitioned into
+    table-region segments, collapsed to distinct keys in LDS, and
+    merged once per key (20.5e9 events/s at 1M keys; see profiles/).
+    ``BYTEWAX_JOIN_RADIX=0`` restores the direct-atomic path.
     """
     import torch
 
