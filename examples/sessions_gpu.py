@@ -56,12 +56,15 @@ def main() -> None:
     dev = torch.device("cuda:0")
     scratch = torch.empty(8 * (1 << 30), dtype=torch.int8, device=dev)
     del scratch  # pre-touch the allocator (stays cached for reuse)
-    n, vocab, gap_ms, batches = 32_000_000, 1_000_000, 30_000, 8
+    n, vocab, gap_ms, batches = 32_000_000, 1_000_000, 30_000, 32
 
     # Pre-generate batches before the timed region (the measurement
-    # is the session kernels, not torch.randint).
+    # is the session kernels, not torch.randint).  4 batches per
+    # engine poll amortize the per-step close sweep + host sync,
+    # matching bench.py's batches-per-poll methodology (sessions
+    # then close with up to 4 batches of extra latency).
     part = _SyntheticPartition(
-        dev, n, batches, vocab, 10_000, _ms(ALIGN), seed=1
+        dev, n, batches, vocab, 10_000, _ms(ALIGN), seed=1, per_poll=4
     )
 
     class Prebuilt(DynamicSource):
