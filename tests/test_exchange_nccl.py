@@ -35,3 +35,52 @@ def test_exchange_2rank_rccl_one_gpu():
     out = res.stdout.decode()
     assert res.returncode == 0, (out + res.stderr.decode())[-2000:]
     assert "PROBE_2RANK_RESULT: PASS" in out
+
+
+def test_str_exchange_1rank_rccl_self_roundtrip():
+    """The string-byte exchange over an actual NCCL(RCCL) group:
+    1-rank force self-exchange — device pack kernels + the four
+    all-to-allv collectives + offset rebuild return the same string
+    multiset with its ts/vals attached."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    prog = r"""
+import os, random, sys
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+os.environ.setdefault("MASTER_PORT", "29733")
+import torch
+import torch.distributed as dist
+from bytewax_amd.gpu.strings import exchange_str_by_key, pack_strings
+
+dist.init_process_group("nccl", rank=0, world_size=1)
+rng = random.Random(3)
+words = [f"word-{rng.randrange(2000)}" for _ in range(50_000)]
+data, offs = pack_strings(words)
+dev = torch.device("cuda:0")
+d = torch.from_numpy(data).to(dev)
+o = torch.from_numpy(offs).to(dev)
+ts = torch.arange(len(words), dtype=torch.int64, device=dev)
+vals = ts * 7
+rb, ro, rt, rv = exchange_str_by_key(d, o, ts, vals, force=True)
+rb_h, ro_h = rb.cpu().numpy(), ro.cpu().numpy()
+got = sorted(
+    (bytes(rb_h[ro_h[i]:ro_h[i+1]]).decode(), int(t), int(v))
+    for i, (t, v) in enumerate(zip(rt.cpu().tolist(), rv.cpu().tolist()))
+)
+sent = sorted((w, i, i * 7) for i, w in enumerate(words))
+assert got == sent, (len(got), len(sent))
+print("STR_NCCL_SELF OK", flush=True)
+dist.destroy_process_group()
+"""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    res = subprocess.run(
+        [sys.executable, "-c", prog],
+        capture_output=True,
+        timeout=300,
+        cwd=str(REPO),
+        env=env,
+    )
+    out = res.stdout.decode()
+    assert res.returncode == 0, (out + res.stderr.decode())[-2000:]
+    assert "STR_NCCL_SELF OK" in out
