@@ -579,14 +579,16 @@ class HashJoinState:
                 torch.tensor([v0 for _k, v0, _v1 in out], dtype=torch.int64),
                 torch.tensor([v1 for *_kv, v1 in out], dtype=torch.int64),
             )
-        n = int(self.out_n.item())
+        # One D2H fetch for (count, error) — two .item() calls would
+        # sync the stream twice per drained batch.
+        n, err = torch.cat((self.out_n, self.error_flag)).cpu().tolist()
+        if err != 0:
+            msg = "join table overflowed; increase slots_pow"
+            raise RuntimeError(msg)
         if n == 0:
             return None
         if n > self.out_cap:
             msg = f"join produced {n} rows > out_cap"
-            raise RuntimeError(msg)
-        if int(self.error_flag.item()) != 0:
-            msg = "join table overflowed; increase slots_pow"
             raise RuntimeError(msg)
         out = (
             self.out_keys[:n].clone(),
