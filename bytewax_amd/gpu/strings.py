@@ -118,9 +118,12 @@ def exchange_str_by_key(
             if vals_np is not None
             else None
         )
-        t_bytes = torch.frombuffer(
-            bytearray(b"".join(send_bytes)), dtype=torch.uint8
-        ) if send_bytes else torch.empty(0, dtype=torch.uint8)
+        joined = b"".join(send_bytes)
+        t_bytes = (
+            torch.frombuffer(bytearray(joined), dtype=torch.uint8)
+            if joined
+            else torch.empty(0, dtype=torch.uint8)
+        )
     else:
         k = _ext_mod()
         n = int(offs.numel()) - 1
