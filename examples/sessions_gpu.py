@@ -1,6 +1,6 @@
 """Device session windows: per-user activity sessions at GPU scale.
 
-256M events over 1M users; sessions close after a 30s idle gap.
+1B events over 1M users; sessions close after a 30s idle gap.
 Runs through the dataflow engine (`run_main`): synthetic columnar
 source -> `keyed_session_agg` (fused radix session kernels) -> sink.
 """
