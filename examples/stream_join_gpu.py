@@ -39,10 +39,14 @@ def main():
     device = "cuda" if torch.cuda.is_available() else "cpu"
 
     events = 10_000_000
-    n_batches = 20
+    n_batches = 40
 
     # Pre-generate both sides on device before the timed region (the
     # join measures the hash-state kernels, not torch.randint).
+    # 4 batches per poll amortize the per-step engine overhead +
+    # drain sync (bench.py's batches-per-poll methodology); each
+    # poll hands one side's 4 batches, so sides interleave in
+    # 4-batch runs — pair counts are deterministic for the schedule.
     from bytewax_amd.gpu import _ms
     from bytewax_amd.gpu.operators import _SyntheticPartition
     from bytewax_amd.inputs import DynamicSource
@@ -50,7 +54,7 @@ def main():
     parts = [
         _SyntheticPartition(
             torch.device(device), events, n_batches, 1_000_000, 1000,
-            _ms(ALIGN), seed, vals=True,
+            _ms(ALIGN), seed, vals=True, per_poll=4,
         )
         for seed in (1, 2)
     ]
