@@ -87,6 +87,9 @@ def main():
         _ms(ALIGN),
         11 + rank,
         vals=True,  # temperatures (int fixed-point)
+        # 4 batches per engine poll amortize per-step overhead
+        # (bench.py's batches-per-poll methodology).
+        per_poll=4,
     )
 
     class PrebuiltSource(DynamicSource):
