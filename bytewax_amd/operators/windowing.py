@@ -926,9 +926,9 @@ class _ColumnarSpec:
             off_ms = int(windower.offset.total_seconds() * 1000)
         else:
             return None
-        if mode in ("min", "max") and off_ms != len_ms:
-            # The stats table is tumbling-only; sliding min/max stays
-            # on the host path.
+        if mode in ("min", "max", "mean") and off_ms != len_ms:
+            # The stats table is tumbling-only; sliding min/max/mean
+            # stays on the host path.
             return None
         align_ms = int(
             (windower.align_to - _EPOCH_UTC).total_seconds() * 1000

@@ -421,3 +421,27 @@ def test_public_max_window_lowers_on_gpu():
         cell = (k, ALIGN_MS + ((ms - ALIGN_MS) // win_len) * win_len)
         brute[cell] = max(brute.get(cell, v), v)
     assert _minmax_columnar_run(events, "max", "cuda:0") == brute
+
+
+def test_sliding_mean_stays_on_host():
+    """Sliding + device_mean must NOT lower (the stats backend is
+    tumbling-only); the host path computes the overlapped windows."""
+    from bytewax_amd.operators.windowing import SlidingWindower, _ColumnarSpec
+
+    spec = _ColumnarSpec.resolve(
+        "mean",
+        EventClock(ts_getter=lambda it: it,
+                   wait_for_system_duration=timedelta(0)),
+        SlidingWindower(
+            align_to=ALIGN, length=WINDOW, offset=timedelta(seconds=20)
+        ),
+    )
+    assert spec is None
+    # Tumbling mean does lower.
+    spec = _ColumnarSpec.resolve(
+        "mean",
+        EventClock(ts_getter=lambda it: it,
+                   wait_for_system_duration=timedelta(0)),
+        TumblingWindower(align_to=ALIGN, length=WINDOW),
+    )
+    assert spec is not None
