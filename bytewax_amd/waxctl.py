@@ -195,6 +195,26 @@ def build_manifests(
         ]
 
     pod_spec = {"containers": [container], "volumes": volumes}
+    if recovery:
+        # First boot initializes the recovery DB partitions
+        # (idempotent: `python -m bytewax_amd.recovery` skips an
+        # already-initialized directory).
+        pod_spec["initContainers"] = [
+            {
+                "name": "init-recovery",
+                "image": f"{image}:{tag}",
+                "imagePullPolicy": image_pull_policy,
+                "command": [
+                    "sh",
+                    "-c",
+                    "python -m bytewax_amd.recovery /var/recovery "
+                    f"{recovery_parts} || true",
+                ],
+                "volumeMounts": [
+                    {"name": "recovery", "mountPath": "/var/recovery"}
+                ],
+            }
+        ]
     template = {
         "metadata": {"labels": _labels(name)},
         "spec": pod_spec,
