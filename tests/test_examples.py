@@ -196,3 +196,14 @@ def test_simple_kafka_example_builds():
         ]
     )
     assert res.returncode == 0, res.stderr.decode()[-1500:]
+
+
+def test_serde_examples_compile():
+    """The serde examples need a live broker + registry to IMPORT
+    (module-level clients), so CI checks they at least compile."""
+    import py_compile
+
+    for name in ("confluent_serde.py", "redpanda_serde.py"):
+        py_compile.compile(
+            str(REPO / "examples" / name), doraise=True
+        )
