@@ -463,13 +463,32 @@ class _StatefulBatchExec(_Exec):
             else:
                 lst.append(v)
         out: List[Tuple[str, Any]] = []
+        # Hot loop: `_handle`/`_refresh_notify` inlined (two Python
+        # frames per (key, batch) otherwise), and the sched pop is
+        # skipped while no key schedules notifications at all.
+        out_append = out.append
+        logics = self.logics
+        sched = self.sched
+        awoken_add = self.awoken.add
+        builder = self.builder
         for k in sorted(grouped):
-            logic = self.logics.get(k)
+            logic = logics.get(k)
             if logic is None:
-                logic = self.builder(None)
-                self.logics[k] = logic
+                logic = builder(None)
+                logics[k] = logic
             ws, discard = logic.on_batch(grouped[k])
-            self._handle(k, ws, discard, out)
+            for w in ws:
+                out_append((k, w))
+            awoken_add(k)
+            if discard:
+                logics.pop(k, None)
+                sched.pop(k, None)
+            else:
+                at = logic.notify_at()
+                if at is not None:
+                    sched[k] = at
+                elif sched:
+                    sched.pop(k, None)
         if out:
             return [(self.out, out)]
         return []
