@@ -993,18 +993,32 @@ def collect(
     return stateful("stateful", up, shim_builder)
 
 
-class _FoldFinalLogic(StatefulLogic[V, S, S]):
+class _FoldFinalLogic(StatefulBatchLogic[V, S, S]):
+    """Batch-level fold (the engine's hottest host-path logic:
+    count/max/min/reduce_final all route here) — a tight fold loop
+    instead of the per-item StatefulLogic shim."""
+
     def __init__(self, step_id: str, folder: Callable[[S, V], S], state: S):
         self.step_id = step_id
         self.folder = folder
         self.state = state
 
-    def on_item(self, value: V) -> Tuple[Iterable[S], bool]:
-        self.state = self.folder(self.state, value)
-        return ((), StatefulLogic.RETAIN)
+    def on_batch(self, values: List[V]) -> Tuple[Iterable[S], bool]:
+        folder = self.folder
+        state = self.state
+        for v in values:
+            state = folder(state, v)
+        self.state = state
+        return ((), StatefulBatchLogic.RETAIN)
+
+    def on_notify(self) -> Tuple[Iterable[S], bool]:
+        return ((), StatefulBatchLogic.RETAIN)
 
     def on_eof(self) -> Tuple[Iterable[S], bool]:
-        return ((self.state,), StatefulLogic.DISCARD)
+        return ((self.state,), StatefulBatchLogic.DISCARD)
+
+    def notify_at(self) -> Optional[datetime]:
+        return None
 
     def snapshot(self) -> S:
         return copy.deepcopy(self.state)
@@ -1040,7 +1054,7 @@ def fold_final(
         state = resume_state if resume_state is not None else builder()
         return _FoldFinalLogic(step_id, folder, state)
 
-    return stateful("stateful", up, shim_builder)
+    return stateful_batch("stateful_batch", up, shim_builder)
 
 
 @operator
