@@ -75,8 +75,17 @@ class _IterSourcePartition(StatefulSourcePartition[X, int]):
         self._next_awake = None
 
         batch: List[X] = []
+        batch_append = batch.append
+        batch_size = self._batch_size
+        sentinels = _SENTINELS
         for item in self._it:
-            if isinstance(item, TestingSource.EOF):
+            # Hot path: one combined isinstance for the three
+            # sentinel classes instead of three checks per item.
+            if not isinstance(item, sentinels):
+                batch_append(item)
+                if len(batch) >= batch_size:
+                    break
+            elif isinstance(item, TestingSource.EOF):
                 self._raise = StopIteration()
                 # Skip over this on continuation.
                 self._start_idx += 1
@@ -87,14 +96,10 @@ class _IterSourcePartition(StatefulSourcePartition[X, int]):
                     # Only trigger once; skipped on resume executions.
                     item._triggered = True
                     break
-            elif isinstance(item, TestingSource.PAUSE):
+            else:  # PAUSE
                 self._next_awake = datetime.now(tz=timezone.utc) + item.for_duration
                 self._start_idx += 1
                 break
-            else:
-                batch.append(item)
-                if len(batch) >= self._batch_size:
-                    break
 
         if len(batch) > 0 or self._raise is not None or self._next_awake is not None:
             self._start_idx += len(batch)
@@ -150,6 +155,10 @@ class TestingSource(FixedPartitionedSource[X, int]):
         self, step_id: str, for_part: str, resume_state: Optional[int]
     ) -> _IterSourcePartition[X]:
         return _IterSourcePartition(self._ib, self._batch_size, resume_state)
+
+
+#: Sentinel classes checked once per item on the hot read path.
+_SENTINELS = (TestingSource.EOF, TestingSource.ABORT, TestingSource.PAUSE)
 
 
 class _ListSinkPartition(StatelessSinkPartition[X]):
