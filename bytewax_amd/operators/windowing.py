@@ -415,6 +415,9 @@ class _SlidingWindowerLogic(WindowerLogic[_SlidingWindowerState]):
 
     def intersects(self, timestamp: datetime) -> List[int]:
         since = timestamp - self.align_to
+        if self.length == self.offset:
+            # Tumbling fast path: exactly one window, one floordiv.
+            return [since // self.offset]
         lo = (since - self.length) // self.offset + 1
         hi = since // self.offset + 1
         return list(range(lo, hi))
@@ -424,12 +427,13 @@ class _SlidingWindowerLogic(WindowerLogic[_SlidingWindowerState]):
         return WindowMetadata(open_time, open_time + self.length)
 
     def open_for(self, timestamp: datetime) -> List[int]:
-        found = []
-        for window_id in self.intersects(timestamp):
-            self.state.opened.setdefault(
-                window_id, self._metadata_for(window_id)
-            )
-            found.append(window_id)
+        found = self.intersects(timestamp)
+        opened = self.state.opened
+        for window_id in found:
+            # Metadata built lazily: `setdefault` would construct it
+            # per ITEM even for an already-open window.
+            if window_id not in opened:
+                opened[window_id] = self._metadata_for(window_id)
         return found
 
     def late_for(self, timestamp: datetime) -> List[int]:
