@@ -413,16 +413,33 @@ class StatefulLogic(ABC, Generic[V, W, S]):
 
 
 class _StatefulLogicShim(StatefulBatchLogic[V, W, S]):
-    def __init__(self, inner: StatefulLogic[V, W, S]):
-        self.inner = inner
+    def __init__(
+        self,
+        inner: StatefulLogic[V, W, S],
+        builder: Optional[Callable[[Optional[S]], StatefulLogic[V, W, S]]] = None,
+    ):
+        self.inner: Optional[StatefulLogic[V, W, S]] = inner
+        self.builder = builder
 
     def on_batch(self, values: List[V]) -> Tuple[Iterable[W], bool]:
+        # A mid-batch complete DISCARDS the per-item logic but must
+        # NOT drop the remaining values: a fresh logic is built for
+        # them, like the reference's shim (which rebuilds and
+        # continues the loop; pysrc _StatefulLogic.on_batch).
         out: List[W] = []
+        inner = self.inner
         for v in values:
-            ws, is_complete = self.inner.on_item(v)
+            if inner is None:
+                inner = self.builder(None) if self.builder else None
+                if inner is None:
+                    break
+            ws, is_complete = inner.on_item(v)
             out.extend(ws)
             if is_complete:
-                return (out, StatefulBatchLogic.DISCARD)
+                inner = None
+        self.inner = inner
+        if inner is None:
+            return (out, StatefulBatchLogic.DISCARD)
         return (out, StatefulBatchLogic.RETAIN)
 
     def on_notify(self) -> Tuple[Iterable[W], bool]:
@@ -475,7 +492,7 @@ def stateful(
     """
 
     def shim_builder(resume_state: Optional[S]) -> _StatefulLogicShim[V, W, S]:
-        return _StatefulLogicShim(builder(resume_state))
+        return _StatefulLogicShim(builder(resume_state), builder)
 
     return stateful_batch("stateful_batch", up, shim_builder)
 

@@ -242,3 +242,26 @@ def test_stateful_flat_map_expands_and_discards(entry_point):
     # k=1 (seen 1 -> [1]), k=2 (seen 2 -> [2,2], discard),
     # k=3 (fresh, seen 1 -> [3]).
     assert sorted(out) == [("k", 1), ("k", 2), ("k", 2), ("k", 3)]
+
+
+def test_stateful_midbatch_discard_keeps_remaining_values():
+    """A per-item logic that completes mid-batch must NOT drop the
+    batch's remaining values: the shim rebuilds a fresh logic for
+    them (reference pysrc _StatefulLogic.on_batch rebuilds and
+    continues).  Regression: join complete-mode with two right-side
+    values in one batch previously lost the second."""
+    out = []
+    flow = Dataflow("midbatch_discard")
+    l = op.input(
+        "l", flow, TestingSource([("k", "a1"), ("k", "a3")], batch_size=1)
+    )
+    r = op.input(
+        "r", flow, TestingSource([("k", "b1"), ("k", "b2")], batch_size=10)
+    )
+    j = op.join("j", l, r)
+    op.output("out", j, TestingSink(out))
+    run_main(flow)
+    # b1 completes the first pair (discard); b2 must persist in a
+    # fresh logic and join the later a3.
+    assert ("k", ("a1", "b1")) in out
+    assert ("k", ("a3", "b2")) in out
