@@ -260,6 +260,8 @@ def test_stateful_midbatch_discard_keeps_remaining_values():
     )
     j = op.join("j", l, r)
     op.output("out", j, TestingSink(out))
+    from bytewax_amd.testing import run_main
+
     run_main(flow)
     # b1 completes the first pair (discard); b2 must persist in a
     # fresh logic and join the later a3.
