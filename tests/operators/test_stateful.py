@@ -267,3 +267,25 @@ def test_stateful_midbatch_discard_keeps_remaining_values():
     # fresh logic and join the later a3.
     assert ("k", ("a1", "b1")) in out
     assert ("k", ("a3", "b2")) in out
+
+
+def test_collect_multiple_max_size_completions_in_one_batch():
+    """collect(max_size=3) fed 7 same-key items in one source batch
+    must emit [0,1,2], [3,4,5] and (at EOF) [6] — each max_size
+    completion discards the logic mid-batch and a fresh one collects
+    the rest (same shim path as the join regression above)."""
+    out = []
+    flow = Dataflow("collect_midbatch")
+    s = op.input(
+        "inp",
+        flow,
+        TestingSource([("k", i) for i in range(7)], batch_size=10),
+    )
+    c = op.collect(
+        "collect", s, timeout=timedelta(seconds=10), max_size=3
+    )
+    op.output("out", c, TestingSink(out))
+    from bytewax_amd.testing import run_main
+
+    run_main(flow)
+    assert [v for _k, v in out] == [[0, 1, 2], [3, 4, 5], [6]]
