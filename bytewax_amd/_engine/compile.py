@@ -128,7 +128,10 @@ def compile_graph(flow: Dataflow) -> ExecGraph:
     for s in steps:
         if s.op_name == "input":
             n_inputs += 1
-        if s.op_name == "output":
+        if s.op_name in ("output", "inspect_debug"):
+            # `inspect` counts as a terminal step like the reference
+            # (its inspect-only pytest flows run without an
+            # `op.output`).
             n_outputs += 1
         for inp_idx, sid in enumerate(s.inp_streams):
             if sid not in producers:
@@ -139,12 +142,17 @@ def compile_graph(flow: Dataflow) -> ExecGraph:
                 raise AssertionError(msg)
             graph.consumers.setdefault(sid, []).append((s.idx, inp_idx))
 
+    # Missing input/output surface as RuntimeError like the
+    # reference (its panic bridging; pytests assert RuntimeError and
+    # NOT ValueError).
+    from ..errors import BytewaxRuntimeError
+
     if n_inputs < 1:
         msg = "Dataflow needs to contain at least one input step"
-        raise ValueError(msg)
+        raise BytewaxRuntimeError(msg)
     if n_outputs < 1:
         msg = "Dataflow needs to contain at least one output step"
-        raise ValueError(msg)
+        raise BytewaxRuntimeError(msg)
 
     # Topological order + exchange depth (number of exchange edges on
     # the longest path ending at each step).

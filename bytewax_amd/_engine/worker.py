@@ -288,7 +288,17 @@ class _BranchExec(_Exec):
         ts: List[Any] = []
         fs: List[Any] = []
         for x in items:
-            (ts if self.predicate(x) else fs).append(x)
+            keep = self.predicate(x)
+            if keep.__class__ is not bool:
+                from ..errors import BytewaxTypeError
+
+                msg = (
+                    "return value of `predicate` in step "
+                    f"{self.step.step_id!r} must be a `bool`; "
+                    f"got a {type(keep)!r} instead"
+                )
+                raise BytewaxTypeError(msg)
+            (ts if keep else fs).append(x)
         out = []
         if ts:
             out.append((self.trues, ts))

@@ -251,9 +251,9 @@ class DirSink(FixedPartitionedSink[str, int]):
             raise ValueError(msg)
         self._dir_path = dir_path
         self._file_count = file_count
-        self._file_namer = file_namer or (
-            lambda i, n: f"part_{i}.out"
-        )
+        # Default name matches the reference ("part_{i}", no
+        # extension; reference files.py:372).
+        self._file_namer = file_namer or (lambda i, n: f"part_{i}")
         self._assign_file = assign_file
         self._end = end
 

@@ -7,6 +7,7 @@ cost nothing.
 """
 
 import asyncio
+import queue
 from abc import ABC, abstractmethod
 from datetime import datetime, timedelta, timezone
 from typing import (
@@ -139,43 +140,52 @@ class DynamicSource(Source[X]):
         ...
 
 
-class _SimplePollingPartition(StatefulSourcePartition[X, None]):
+class _SimplePollingPartition(StatefulSourcePartition[X, S]):
+    """Drift-free polling partition (reference inputs.py:285-330
+    contract: `now` injected for deterministic tests; awakes advance
+    from the SCHEDULED time, not from the current clock; an exactly
+    on-mark `align_to` activates immediately)."""
+
     def __init__(
         self,
+        now: datetime,
         interval: timedelta,
         align_to: Optional[datetime],
         getter: Callable[[], X],
+        snapshot: Callable[[], Any] = lambda: None,
     ):
         self._interval = interval
         self._getter = getter
-        now = datetime.now(timezone.utc)
-        if align_to is not None and align_to > now:
-            self._next_awake = align_to
-        elif align_to is not None:
-            since = (now - align_to) // interval
-            self._next_awake = align_to + interval * (since + 1)
+        self._snapshot = snapshot
+        if align_to is not None:
+            since_last_awake = (now - align_to) % interval
+            if since_last_awake > timedelta(seconds=0):
+                until_next_awake = interval - since_last_awake
+            else:
+                until_next_awake = timedelta(seconds=0)
+            self._next_awake = now + until_next_awake
         else:
             self._next_awake = now
 
     def next_batch(self) -> List[X]:
-        self._next_awake = datetime.now(timezone.utc) + self._interval
         try:
             item = self._getter()
+            self._next_awake += self._interval
+            if item is None:
+                return []
+            return [item]
         except SimplePollingSource.Retry as ex:
-            self._next_awake = datetime.now(timezone.utc) + ex.timeout
+            self._next_awake += ex.timeout
             return []
-        if item is None:
-            return []
-        return [item]
 
     def next_awake(self) -> Optional[datetime]:
         return self._next_awake
 
-    def snapshot(self) -> None:
-        return None
+    def snapshot(self) -> Any:
+        return self._snapshot()
 
 
-class SimplePollingSource(FixedPartitionedSource[X, None]):
+class SimplePollingSource(FixedPartitionedSource[X, S]):
     """Calls a user-defined function at a regular interval.
 
     Subclass and override {py:obj}`next_item`.  Only one worker polls.
@@ -215,9 +225,12 @@ class SimplePollingSource(FixedPartitionedSource[X, None]):
 
     def build_part(
         self, step_id: str, for_part: str, resume_state: Optional[None]
-    ) -> _SimplePollingPartition[X]:
+    ) -> "_SimplePollingPartition[X, None]":
         return _SimplePollingPartition(
-            self._interval, self._align_to, self.next_item
+            datetime.now(timezone.utc),
+            self._interval,
+            self._align_to,
+            self.next_item,
         )
 
     @abstractmethod
@@ -256,10 +269,10 @@ def batch(ib: Iterable[X], batch_size: int) -> Iterator[List[X]]:
 def batch_getter(
     getter: Callable[[], X], batch_size: int, yield_on: Any = None
 ) -> Iterator[List[X]]:
-    """Batch from a getter function that returns a sentinel when empty.
-
-    The resulting iterator is infinite: it yields (possibly empty)
-    batches forever; callers should stop polling it externally.
+    """Batch from a getter function that returns a sentinel when
+    there are no items YET; the getter raises `StopIteration` at
+    EOF, which yields the final partial batch and ends the iterator
+    (reference inputs.py:477-510 contract).
 
     Example:
         >>> from bytewax_amd.inputs import batch_getter
@@ -272,8 +285,12 @@ def batch_getter(
     """
     while True:
         out: List[X] = []
-        for _ in range(batch_size):
-            item = getter()
+        while len(out) < batch_size:
+            try:
+                item = getter()
+            except StopIteration:
+                yield out
+                return
             if item == yield_on:
                 break
             out.append(item)
@@ -281,24 +298,35 @@ def batch_getter(
 
 
 def batch_getter_ex(
-    getter: Callable[[], X], batch_size: int, yield_ex: type = IndexError
+    getter: Callable[[], X], batch_size: int, yield_ex: type = queue.Empty
 ) -> Iterator[List[X]]:
-    """Batch from a getter function that raises when empty.
+    """Batch from a getter that raises `yield_ex` (default
+    `queue.Empty`) when there are no items YET and `StopIteration`
+    at EOF — the final partial batch is yielded, then the iterator
+    ends (reference inputs.py:513-547 contract).
 
     Example:
+        >>> import queue
         >>> from bytewax_amd.inputs import batch_getter_ex
         >>> q = [1, 2, 3]
-        >>> it = batch_getter_ex(lambda: q.pop(0), 2)
+        >>> def poll():
+        ...     if not q:
+        ...         raise queue.Empty()
+        ...     return q.pop(0)
+        >>> it = batch_getter_ex(poll, 2)
         >>> next(it), next(it), next(it)
         ([1, 2], [3], [])
     """
     while True:
         out: List[X] = []
-        for _ in range(batch_size):
+        while len(out) < batch_size:
             try:
                 out.append(getter())
             except yield_ex:
                 break
+            except StopIteration:
+                yield out
+                return
         yield out
 
 

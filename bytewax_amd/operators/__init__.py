@@ -28,6 +28,7 @@ from typing import (
     TypeVar,
 )
 
+from ..errors import BytewaxTypeError
 from ..dataflow import (
     Dataflow,
     KeyedStream,
@@ -589,7 +590,7 @@ def flatten(
                 f"step {step_id!r} requires iterable items; "
                 f"got a {type(x)!r} instead"
             )
-            raise TypeError(msg)
+            raise BytewaxTypeError(msg)
         return x
 
     return flat_map("flat_map", up, shim_mapper)
@@ -628,7 +629,7 @@ def filter(  # noqa: A001
                 f"in step {step_id!r} must be a `bool`; "
                 f"got a {type(keep)!r} instead"
             )
-            raise TypeError(msg)
+            raise BytewaxTypeError(msg)
         if keep:
             return (x,)
         return ()
@@ -662,7 +663,7 @@ def filter_value(
                 f"in step {step_id!r} must be a `bool`; "
                 f"got a {type(keep)!r} instead"
             )
-            raise TypeError(msg)
+            raise BytewaxTypeError(msg)
         if keep:
             return (v,)
         return ()
@@ -861,7 +862,7 @@ def key_on(step_id: str, up: Stream[X], key: Callable[[X], str]) -> KeyedStream[
                 f"in step {step_id!r} must be a `str`; "
                 f"got a {type(k)!r} instead"
             )
-            raise TypeError(msg)
+            raise BytewaxTypeError(msg)
         return (k, x)
 
     return map("map", up, shim_mapper)
@@ -1026,6 +1027,12 @@ class _FoldFinalLogic(StatefulBatchLogic[V, S, S]):
         for v in values:
             state = folder(state, v)
         self.state = state
+        return ((), StatefulBatchLogic.RETAIN)
+
+    def on_item(self, value: V) -> Tuple[Iterable[S], bool]:
+        """Per-item compatibility entry (the engine uses on_batch;
+        the reference's internals tests drive on_item directly)."""
+        self.state = self.folder(self.state, value)
         return ((), StatefulBatchLogic.RETAIN)
 
     def on_notify(self) -> Tuple[Iterable[S], bool]:
@@ -1196,13 +1203,13 @@ class _StatefulMapLogic(StatefulLogic[V, W, S]):
         res = self.mapper(self.state, value)
         try:
             self.state, out = res
-        except TypeError as ex:
+        except (TypeError, ValueError) as ex:
             msg = (
                 f"return value of `mapper` {f_repr(self.mapper)} "
                 f"in step {self.step_id!r} must be a 2-tuple of "
                 f"`(updated_state, emit)`; got a {type(res)!r} instead"
             )
-            raise TypeError(msg) from ex
+            raise BytewaxTypeError(msg) from ex
         if self.single:
             emitted: Iterable[W] = (out,)  # type: ignore[assignment]
         else:

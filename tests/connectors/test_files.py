@@ -106,7 +106,7 @@ def test_dir_sink(tmp_path: Path):
     run_main(flow)
     written = sorted(
         line
-        for p in outdir.glob("part_*.out")
+        for p in outdir.glob("part_*")
         for line in p.read_text().splitlines()
     )
     assert written == ["1", "2"]

@@ -45,7 +45,7 @@ def test_batch_getter_ex():
             return stack.pop()
         raise IndexError()
 
-    b = batch_getter_ex(getter, 10)
+    b = batch_getter_ex(getter, 10, yield_ex=IndexError)
     assert next(b) == [1, 2, 3]
     assert next(b) == []
 
