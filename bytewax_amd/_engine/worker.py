@@ -448,25 +448,25 @@ class _StatefulBatchExec(_Exec):
         for kv in items:
             # Hot loop: exact-type checks first (`type(x) is` is much
             # cheaper than isinstance), subclass fallbacks second.
-            if type(kv) is not tuple and not isinstance(kv, tuple):
+            if (
+                type(kv) is not tuple and not isinstance(kv, tuple)
+            ) or len(kv) != 2:
+                from ..errors import BytewaxTypeError
+
                 msg = (
                     f"step {self.step.step_id!r} requires `(key, value)` "
                     f"2-tuples from upstream; got a {type(kv)!r} instead"
                 )
-                raise TypeError(msg)
-            if len(kv) != 2:
-                msg = (
-                    f"step {self.step.step_id!r} requires `(key, value)` "
-                    f"2-tuples from upstream; got a {type(kv)!r} instead"
-                )
-                raise TypeError(msg)
+                raise BytewaxTypeError(msg)
             k, v = kv
             if type(k) is not str and not isinstance(k, str):
+                from ..errors import BytewaxTypeError
+
                 msg = (
                     f"step {self.step.step_id!r} requires keys to be `str`; "
                     f"got a {type(k)!r} instead"
                 )
-                raise TypeError(msg)
+                raise BytewaxTypeError(msg)
             lst = get(k)
             if lst is None:
                 grouped[k] = [v]
@@ -665,17 +665,21 @@ class _Worker:
                 try:
                     k, _v = kv
                 except (TypeError, ValueError) as ex:
+                    from ..errors import BytewaxTypeError
+
                     msg = (
                         f"step {step.step_id!r} requires `(key, value)` "
                         f"2-tuples from upstream; got a {type(kv)!r} instead"
                     )
-                    raise TypeError(msg) from ex
+                    raise BytewaxTypeError(msg) from ex
                 if not isinstance(k, str):
+                    from ..errors import BytewaxTypeError
+
                     msg = (
                         f"step {step.step_id!r} requires keys to be `str`; "
                         f"got a {type(k)!r} instead"
                     )
-                    raise TypeError(msg)
+                    raise BytewaxTypeError(msg)
                 buckets.setdefault(_route_key(k, n), []).append(kv)
         elif step.exchange == "random":
             for kv in items:
@@ -968,7 +972,21 @@ def _run_cluster(
     if cluster.abort is not None:
         if isinstance(cluster.abort, AbortExecution):
             return  # aborted execution ends cleanly; resume replays
-        raise cluster.abort
+        if isinstance(cluster.abort, KeyboardInterrupt):
+            raise cluster.abort
+        # User/step errors surface WRAPPED as BytewaxRuntimeError
+        # (the original chained as __cause__), matching the
+        # reference's panic bridging: its pytests assert the custom
+        # exception does NOT escape raw (test_execution.py
+        # test_reraises_custom_exception's nested raises).
+        from ..errors import BytewaxRuntimeError, BytewaxTypeError
+
+        if isinstance(cluster.abort, (BytewaxRuntimeError, BytewaxTypeError)):
+            # Engine-origin contract errors are already RuntimeError
+            # (or the dual TypeError/RuntimeError) — no double wrap.
+            raise cluster.abort
+        msg = "error while executing dataflow; see the cause above"
+        raise BytewaxRuntimeError(msg) from cluster.abort
 
 
 def run_main(

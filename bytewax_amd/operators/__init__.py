@@ -907,12 +907,13 @@ def raises(step_id: str, up: Stream[Any]) -> None:
     >>> run_main(flow)
     Traceback (most recent call last):
         ...
-    RuntimeError: `raises` step 'raises_eg.no_negatives' got an item: -2
+    bytewax_amd.errors.BytewaxRuntimeError: `raises` step 'raises_eg.no_negatives' got an item: -2
     """
+    from ..errors import BytewaxRuntimeError
 
     def shim_mapper(x: Any) -> Iterable[Any]:
         msg = f"`raises` step {step_id!r} got an item: {x!r}"
-        raise RuntimeError(msg)
+        raise BytewaxRuntimeError(msg)
 
     from ..connectors.stdio import StdOutSink
 

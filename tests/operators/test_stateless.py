@@ -224,8 +224,15 @@ def test_user_exception_propagates(entry_point):
 
     s = op.map("boom", s, boom)
     op.output("out", s, TestingSink(out))
-    with pytest.raises(ValueError, match="boom"):
+    # User exceptions surface WRAPPED as BytewaxRuntimeError with the
+    # original chained as __cause__ (the reference's convention; its
+    # pytests assert the custom exception does not escape raw).
+    from bytewax_amd.errors import BytewaxRuntimeError
+
+    with pytest.raises(BytewaxRuntimeError) as exc_info:
         entry_point(flow)
+    assert isinstance(exc_info.value.__cause__, ValueError)
+    assert "boom" in str(exc_info.value.__cause__)
 
 
 def test_mid_flow_merge_of_keyed_streams(entry_point):
