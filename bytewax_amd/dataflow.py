@@ -185,7 +185,7 @@ def _validate_step_id(step_id: str) -> None:
         msg = f"step ID must be a `str`; got {step_id!r}"
         raise TypeError(msg)
     if not _STEP_ID_RE.match(step_id):
-        msg = f"step ID {step_id!r} must not contain periods or whitespace"
+        msg = f"step ID {step_id!r} can't contain any periods or whitespace"
         raise ValueError(msg)
 
 
@@ -251,8 +251,9 @@ def operator(builder: Optional[Callable] = None, *, _core: bool = False):
             scope = _find_scope(args) or _find_scope(kwargs.values())
             if scope is None:
                 msg = (
-                    f"operator {op_name!r} requires at least one `Stream` "
-                    "or `Dataflow` argument to anchor it to a flow"
+                    f"operator {op_name!r} requires at least one "
+                    "upstream argument that must be a `Stream` (or a "
+                    "`Dataflow`) to anchor it to a flow"
                 )
                 raise TypeError(msg)
             full_id = f"{scope.prefix}.{step_id}"
