@@ -92,17 +92,21 @@ def _locate_dataflow(module_name: str, dataflow_name: str) -> Dataflow:
     raise TypeError(msg)
 
 
-def _prepare_import(import_str: str) -> Dataflow:
-    """Resolve `"module:flow"` or `"path/to/file.py:flow"`."""
-    import_str = import_str.replace(":", ";", 0)  # noop; keep str
+def _prepare_import(import_str: str):
+    """Split `"module:flow"` / `"path/to/file.py:flow"` into a
+    `(module_str, attrs_str)` pair (reference run.py:153-190
+    contract: file paths become dotted module paths)."""
     module_str, _, flow_str = import_str.partition(":")
     if not flow_str:
         flow_str = "flow"
     path = Path(module_str)
     if path.suffix == ".py":
-        sys.path.insert(0, str(path.parent.resolve()))
-        module_str = path.stem
-    return _locate_dataflow(module_str, flow_str)
+        if path.is_absolute():
+            sys.path.insert(0, str(path.parent.resolve()))
+            module_str = path.stem
+        else:
+            module_str = ".".join(path.with_suffix("").parts)
+    return module_str, flow_str
 
 
 class _EnvDefault(argparse.Action):
@@ -173,7 +177,7 @@ def _parse_args(args=None):
     recovery.add_argument(
         "-s",
         "--snapshot-interval",
-        type=float,
+        type=lambda v: timedelta(seconds=float(v)),
         action=_EnvDefault,
         envvar="BYTEWAX_SNAPSHOT_INTERVAL",
         required=False,
@@ -182,7 +186,7 @@ def _parse_args(args=None):
     recovery.add_argument(
         "-b",
         "--backup-interval",
-        type=float,
+        type=lambda v: timedelta(seconds=float(v)),
         action=_EnvDefault,
         envvar="BYTEWAX_RECOVERY_BACKUP_INTERVAL",
         required=False,
@@ -258,19 +262,11 @@ def cli_main(
 
 def _main() -> None:
     args = _parse_args()
-    flow = _prepare_import(args.import_str)
-    epoch_interval = (
-        timedelta(seconds=float(args.snapshot_interval))
-        if args.snapshot_interval is not None
-        else None
-    )
+    flow = _locate_dataflow(*_prepare_import(args.import_str))
+    epoch_interval = args.snapshot_interval
     recovery_config = None
     if args.recovery_directory is not None:
-        backup = (
-            timedelta(seconds=float(args.backup_interval))
-            if args.backup_interval is not None
-            else None
-        )
+        backup = args.backup_interval
         recovery_config = RecoveryConfig(
             args.recovery_directory, backup_interval=backup
         )

@@ -113,7 +113,7 @@ def to_mermaid(flow: Dataflow) -> str:
 def _main() -> None:
     import argparse
 
-    from .run import _prepare_import
+    from .run import _locate_dataflow, _prepare_import
 
     parser = argparse.ArgumentParser(
         prog="python -m bytewax_amd.visualize",
@@ -124,7 +124,7 @@ def _main() -> None:
         "--format", choices=["json", "mermaid"], default="mermaid"
     )
     args = parser.parse_args()
-    flow = _prepare_import(args.import_str)
+    flow = _locate_dataflow(*_prepare_import(args.import_str))
     if args.format == "json":
         print(to_json(flow))
     else:

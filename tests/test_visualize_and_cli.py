@@ -62,7 +62,7 @@ def test_prepare_import_py_path(tmp_path: Path):
             """
         )
     )
-    flow = _prepare_import(f"{f}:flow")
+    flow = _locate_dataflow(*_prepare_import(f"{f}:flow"))
     assert flow.flow_id == "fromfile"
 
 
@@ -81,7 +81,7 @@ def test_prepare_import_default_name(tmp_path: Path):
             """
         )
     )
-    flow = _prepare_import(str(f))
+    flow = _locate_dataflow(*_prepare_import(str(f)))
     assert flow.flow_id == "deffy"
 
 
