@@ -222,6 +222,43 @@ def _ports_of(obj: Any, step_id: str, kind: str) -> List[Any]:
     return ports
 
 
+def _named_ports(named_values, step_id: str) -> List[Any]:
+    """Port descriptors named like the reference: input ports carry
+    the builder's parameter names ("up", "lefts", ...), output ports
+    the returned port names ("down") or dataclass field names
+    ("trues"/"falses"/...)."""
+    ports: List[Any] = []
+    for name, o in named_values:
+        if isinstance(o, Stream):
+            ports.append(SinglePort(f"{step_id}.{name}", o.stream_id))
+        elif isinstance(o, (list, tuple)):
+            if all(isinstance(el, Stream) for el in o) and o:
+                ports.append(
+                    MultiPort(
+                        f"{step_id}.{name}",
+                        {str(i): el.stream_id for i, el in enumerate(o)},
+                    )
+                )
+            else:
+                for i, el in enumerate(o):
+                    ports.extend(
+                        _named_ports([(f"{name}_{i}", el)], step_id)
+                    )
+        elif isinstance(o, Dataflow) or o is None:
+            pass
+        elif hasattr(o, "__dataclass_fields__"):
+            ports.extend(
+                _named_ports(
+                    [
+                        (f, getattr(o, f))
+                        for f in o.__dataclass_fields__
+                    ],
+                    step_id,
+                )
+            )
+    return ports
+
+
 def operator(builder: Optional[Callable] = None, *, _core: bool = False):
     """Decorate a function to make it a dataflow operator.
 
@@ -268,7 +305,15 @@ def operator(builder: Optional[Callable] = None, *, _core: bool = False):
                 op_name=op_name,
                 is_core=_core,
             )
-            step.inp_ports = _ports_of((args, kwargs), full_id, "inp")
+            _b = sig.bind(full_id, *args, **kwargs)
+            step.inp_ports = _named_ports(
+                [
+                    (n, v)
+                    for n, v in _b.arguments.items()
+                    if n != "step_id"
+                ],
+                full_id,
+            )
             scope.substeps.append(step)
             # Builders see the fully-qualified step id (matching the
             # reference: `op.inspect("help", s)` prints "flow.help").
@@ -289,7 +334,10 @@ def operator(builder: Optional[Callable] = None, *, _core: bool = False):
                     out = fn(full_id, *args, **kwargs)
                 finally:
                     scope.pop()
-            step.out_ports = _ports_of(out, full_id, "out")
+            if isinstance(out, Stream):
+                step.out_ports = _named_ports([("down", out)], full_id)
+            else:
+                step.out_ports = _named_ports([("out", out)], full_id)
             return out
 
         wrapper.__name__ = op_name

@@ -1,28 +1,48 @@
 """Visualize dataflow structure.
 
 Parity target: ``bytewax.visualize`` (reference pysrc/bytewax/
-visualize.py:52-405): render a `Dataflow` to a JSON-able structure or
-Mermaid diagram; ``python -m bytewax_amd.visualize`` CLI.
+visualize.py:30-405): render a `Dataflow` to the "rendered" data
+model (ports resolved to globally-unique IDs, `from_port_ids`
+pointing at each stream's originating port), encode it as JSON, or
+draw a Mermaid flowchart; ``python -m bytewax_amd.visualize`` CLI.
+The JSON schema and Mermaid text match the reference byte-for-byte
+(its pytests/test_visualize.py runs green against this module).
 """
 
 import json
+from collections import ChainMap
 from dataclasses import dataclass
 from typing import Any, Dict, List
 
 from .dataflow import Dataflow, Operator
 
-__all__ = ["RenderedDataflow", "RenderedOperator", "render_dataflow", "to_json", "to_mermaid"]
+__all__ = [
+    "RenderedDataflow",
+    "RenderedOperator",
+    "RenderedPort",
+    "render_dataflow",
+    "to_rendered",
+    "to_json",
+    "to_mermaid",
+]
 
 
 @dataclass(frozen=True)
 class RenderedPort:
+    """Port with stream links resolved to originating ports."""
+
+    port_name: str
     port_id: str
-    stream_ids: List[str]
+    from_port_ids: List[str]
+    from_stream_ids: List[str]
 
 
 @dataclass(frozen=True)
 class RenderedOperator:
-    op_name: str
+    """Operator with all ports resolved to globally-unique IDs."""
+
+    op_type: str
+    step_name: str
     step_id: str
     inp_ports: List[RenderedPort]
     out_ports: List[RenderedPort]
@@ -31,81 +51,133 @@ class RenderedOperator:
 
 @dataclass(frozen=True)
 class RenderedDataflow:
+    """Dataflow with streams, ports resolved to globally-unique IDs."""
+
     flow_id: str
     substeps: List[RenderedOperator]
 
 
-def _render_op(op: Operator) -> RenderedOperator:
-    def ports(ps):
-        out = []
-        for p in ps:
-            if hasattr(p, "stream_id"):
-                out.append(RenderedPort(p.port_id, [p.stream_id]))
-            elif hasattr(p, "stream_ids"):
-                out.append(RenderedPort(p.port_id, list(p.stream_ids.values())))
-        return out
+def _port_name(port) -> str:
+    return port.port_id.rsplit(".", 1)[1]
 
+
+def _stream_ids(port) -> List[str]:
+    if hasattr(port, "stream_id"):
+        return [port.stream_id]
+    return list(port.stream_ids.values())
+
+
+def _to_rendered(step: Operator, stream_to_orig: ChainMap) -> RenderedOperator:
+    inp_rports = [
+        RenderedPort(
+            _port_name(p),
+            p.port_id,
+            [stream_to_orig[s] for s in _stream_ids(p)],
+            list(_stream_ids(p)),
+        )
+        for p in step.inp_ports
+    ]
+    stream_to_orig.update(
+        {s: p.port_id for p in step.out_ports for s in _stream_ids(p)}
+    )
+    # Inner scope: a stream entering this step originates (for the
+    # substeps) at this step's fake input port.
+    inner = stream_to_orig.new_child(
+        {s: p.port_id for p in step.inp_ports for s in _stream_ids(p)}
+    )
+    substeps = [_to_rendered(sub, inner) for sub in step.substeps]
+    out_rports = [
+        RenderedPort(
+            _port_name(p),
+            p.port_id,
+            [inner[s] for s in _stream_ids(p)] if substeps else [],
+            list(_stream_ids(p)) if substeps else [],
+        )
+        for p in step.out_ports
+    ]
     return RenderedOperator(
-        op.op_name,
-        op.step_id,
-        ports(op.inp_ports),
-        ports(op.out_ports),
-        [_render_op(s) for s in op.substeps],
+        step.op_name,
+        step.step_name,
+        step.step_id,
+        inp_rports,
+        out_rports,
+        substeps,
     )
 
 
-def render_dataflow(flow: Dataflow) -> RenderedDataflow:
-    """Convert a dataflow into a renderable structure."""
+def to_rendered(flow: Dataflow) -> RenderedDataflow:
+    """Convert a dataflow into the "rendered" data model: all port
+    links resolved, so a renderer only connects each
+    `RenderedPort.port_id` to its `from_port_ids`."""
+    stream_to_orig: ChainMap = ChainMap()
     return RenderedDataflow(
-        flow.flow_id, [_render_op(s) for s in flow.substeps]
+        flow.flow_id,
+        [_to_rendered(s, stream_to_orig) for s in flow.substeps],
     )
+
+
+#: Back-compat alias (pre-round-2 name).
+render_dataflow = to_rendered
 
 
 def _to_plain(obj: Any) -> Any:
-    if hasattr(obj, "__dataclass_fields__"):
-        return {
-            f: _to_plain(getattr(obj, f)) for f in obj.__dataclass_fields__
-        }
+    if isinstance(obj, (RenderedDataflow, RenderedOperator, RenderedPort)):
+        return dict(
+            typ=type(obj).__name__,
+            **{
+                f: _to_plain(getattr(obj, f))
+                for f in obj.__dataclass_fields__
+            },
+        )
     if isinstance(obj, list):
         return [_to_plain(o) for o in obj]
     return obj
 
 
 def to_json(flow: Dataflow) -> str:
-    """Encode this dataflow into JSON."""
-    return json.dumps(_to_plain(render_dataflow(flow)), indent=2)
+    """Encode this dataflow into JSON (reference schema: every node
+    carries a `typ` discriminator)."""
+    return json.dumps(_to_plain(to_rendered(flow)), indent=2)
 
 
 def to_mermaid(flow: Dataflow) -> str:
-    """Render a dataflow as a Mermaid flowchart (top-level steps)."""
-    lines = ["flowchart TD", f'subgraph "{flow.flow_id} (Dataflow)"']
-    stream_producers: Dict[str, str] = {}
+    """Render a dataflow as a Mermaid flowchart of the top-level
+    steps, edges labeled `producer_port → consumer_port`."""
+    rflow = to_rendered(flow)
+    lines = ["flowchart TD", f'subgraph "{rflow.flow_id} (Dataflow)"']
+    # Map every port id (at any depth) to its top-level step.
+    port_step: Dict[str, RenderedOperator] = {}
 
-    def walk_core(op: Operator):
-        if op.is_core:
-            yield op
-        for s in op.substeps:
-            yield from walk_core(s)
+    def index(top: RenderedOperator, step: RenderedOperator) -> None:
+        for p in step.inp_ports + step.out_ports:
+            port_step[p.port_id] = top
+        for sub in step.substeps:
+            index(top, sub)
 
-    top = list(flow.substeps)
-    for op in top:
-        lines.append(f'{op.step_id}["{op.step_name} ({op.op_name})"]')
-        for core in walk_core(op):
-            for sid in [f"{core.step_id}.{p}" for p in ("down", "trues", "falses")]:
-                stream_producers[sid] = op.step_id
-    for op in top:
-        seen = set()
-        for core in walk_core(op):
-            for p in core.inp_ports:
-                for sid in (
-                    [p.stream_id]
-                    if hasattr(p, "stream_id")
-                    else list(p.stream_ids.values())
-                ):
-                    prod = stream_producers.get(sid)
-                    if prod and prod != op.step_id and (prod, op.step_id) not in seen:
-                        seen.add((prod, op.step_id))
-                        lines.append(f"{prod} --> {op.step_id}")
+    for step in rflow.substeps:
+        index(step, step)
+    port_names = {}
+
+    def names(step: RenderedOperator) -> None:
+        for p in step.inp_ports + step.out_ports:
+            port_names[p.port_id] = p.port_name
+        for sub in step.substeps:
+            names(sub)
+
+    for step in rflow.substeps:
+        names(step)
+
+    for step in rflow.substeps:
+        lines.append(f'{step.step_id}["{step.step_name} ({step.op_type})"]')
+        for p in step.inp_ports:
+            for from_pid in p.from_port_ids:
+                prod = port_step.get(from_pid)
+                if prod is None or prod.step_id == step.step_id:
+                    continue
+                label = f"{port_names[from_pid]} → {p.port_name}"
+                lines.append(
+                    f'{prod.step_id} -- "{label}" --> {step.step_id}'
+                )
     lines.append("end")
     return "\n".join(lines)
 

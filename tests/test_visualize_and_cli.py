@@ -31,7 +31,7 @@ def test_render_dataflow():
     assert names == ["viz.inp", "viz.double", "viz.out"]
     # Composite steps have substeps.
     double = r.substeps[1]
-    assert double.substeps[0].op_name == "flat_map_batch"
+    assert double.substeps[0].op_type == "flat_map_batch"
 
 
 def test_to_json_roundtrips():
@@ -43,8 +43,8 @@ def test_to_json_roundtrips():
 def test_to_mermaid():
     m = to_mermaid(_flow())
     assert "flowchart TD" in m
-    assert "viz.inp --> viz.double" in m
-    assert "viz.double --> viz.out" in m
+    assert 'viz.inp -- "down → up" --> viz.double' in m
+    assert 'viz.double -- "down → up" --> viz.out' in m
 
 
 def test_prepare_import_py_path(tmp_path: Path):
